@@ -1,0 +1,266 @@
+"""AnovosFrame — the engine's DataFrame: a column store of torch tensors.
+
+Replaces the reference's Spark DataFrame (every reference function takes
+``(spark, idf, ...)`` and returns DataFrames). Here a frame is a dict of
+named ``Column`` objects, each one a contiguous torch tensor living on the
+local GPU (HBM3E-resident) or CPU, holding this rank's row shard.
+Cross-rank semantics: frames are row-partitioned; global statistics are
+produced by the ops layer via RCCL partial-aggregate merges
+(core/dist.py), never by shuffling rows.
+
+Columns (core/dtypes.py):
+  numerical   float32/float64 tensor, NaN null
+  categorical int32 codes + host dictionary (list[str]), -1 null
+  timestamp   int64 epoch-microseconds, INT64_MIN null
+"""
+
+from __future__ import annotations
+
+import copy as _copy
+from typing import Dict, Iterable, List, Optional, Sequence, Union
+
+import numpy as np
+import pandas as pd
+import torch
+
+from anovos_amd.core.dtypes import NULL_CODE, NULL_TS, is_timestamp_dtype, kind_of_dtype
+
+
+class Column:
+    """One column: tensor data + logical dtype + optional dictionary."""
+
+    __slots__ = ("name", "dtype", "data", "dictionary")
+
+    def __init__(self, name: str, dtype: str, data: torch.Tensor, dictionary: Optional[List[str]] = None):
+        self.name = name
+        self.dtype = dtype  # Spark-style dtype string ('double', 'string', 'int', ...)
+        self.data = data
+        self.dictionary = dictionary
+
+    @property
+    def kind(self) -> str:
+        return kind_of_dtype(self.dtype)
+
+    def null_mask(self) -> torch.Tensor:
+        """Boolean tensor, True where null."""
+        if self.kind == "numerical":
+            return torch.isnan(self.data)
+        if self.kind == "categorical":
+            return self.data == NULL_CODE
+        if is_timestamp_dtype(self.dtype):
+            return self.data == NULL_TS
+        return torch.zeros_like(self.data, dtype=torch.bool)
+
+    def clone(self) -> "Column":
+        return Column(self.name, self.dtype, self.data, list(self.dictionary) if self.dictionary else None)
+
+    def gather(self, idx: torch.Tensor) -> "Column":
+        return Column(self.name, self.dtype, self.data[idx], self.dictionary)
+
+    def to_numpy_objects(self) -> np.ndarray:
+        """Materialize as a numpy array with python objects/NaN for report/IO paths."""
+        t = self.data.cpu()
+        if self.kind == "categorical":
+            codes = t.numpy()
+            d = np.array(self.dictionary + [None], dtype=object) if self.dictionary else np.array([None], dtype=object)
+            out = d[np.where(codes == NULL_CODE, len(d) - 1, codes)]
+            return out
+        if is_timestamp_dtype(self.dtype):
+            v = t.numpy()
+            out = v.astype("datetime64[us]").astype(object)
+            out = np.where(v == NULL_TS, None, out)
+            return out
+        return t.numpy()
+
+
+class AnovosFrame:
+    """Row-sharded column store. All mutating ops return new frames
+    (tensors are shared where unchanged — cheap copy-on-write)."""
+
+    def __init__(self, columns: Optional[Dict[str, Column]] = None, device: Union[str, torch.device] = "cpu"):
+        self._cols: Dict[str, Column] = dict(columns or {})
+        self.device = torch.device(device)
+
+    # ---------------- basic introspection ----------------
+    @property
+    def columns(self) -> List[str]:
+        return list(self._cols.keys())
+
+    @property
+    def dtypes(self) -> List[tuple]:
+        return [(c.name, c.dtype) for c in self._cols.values()]
+
+    def col(self, name: str) -> Column:
+        if name not in self._cols:
+            raise KeyError(f"column '{name}' not in frame (have {self.columns})")
+        return self._cols[name]
+
+    def __contains__(self, name: str) -> bool:
+        return name in self._cols
+
+    def __len__(self) -> int:
+        return self.local_rows()
+
+    def local_rows(self) -> int:
+        for c in self._cols.values():
+            return int(c.data.shape[0])
+        return 0
+
+    def count(self) -> int:
+        """Global row count across ranks (RCCL all-reduce when distributed)."""
+        from anovos_amd.core import dist
+
+        return dist.all_reduce_scalar(self.local_rows())
+
+    # ---------------- construction ----------------
+    @staticmethod
+    def from_pandas(pdf: pd.DataFrame, device="cpu", dtype_overrides: Optional[Dict[str, str]] = None) -> "AnovosFrame":
+        cols: Dict[str, Column] = {}
+        dev = torch.device(device)
+        for name in pdf.columns:
+            s = pdf[name]
+            cols[str(name)] = _column_from_series(str(name), s, dev, (dtype_overrides or {}).get(str(name)))
+        return AnovosFrame(cols, dev)
+
+    def to_pandas(self) -> pd.DataFrame:
+        """Local shard to pandas (driver-side smalls only — stats frames etc.)."""
+        data = {}
+        for name, c in self._cols.items():
+            data[name] = c.to_numpy_objects()
+        return pd.DataFrame(data)
+
+    # ---------------- column ops (reference data_ingest.py:201-367 semantics) ----------------
+    def select(self, names: Sequence[str]) -> "AnovosFrame":
+        return AnovosFrame({n: self._cols[n] for n in names}, self.device)
+
+    def drop(self, names: Iterable[str]) -> "AnovosFrame":
+        names = set(names)
+        return AnovosFrame({n: c for n, c in self._cols.items() if n not in names}, self.device)
+
+    def rename(self, mapping: Dict[str, str]) -> "AnovosFrame":
+        out = {}
+        for n, c in self._cols.items():
+            nn = mapping.get(n, n)
+            cc = c.clone()
+            cc.name = nn
+            out[nn] = cc
+        return AnovosFrame(out, self.device)
+
+    def with_column(self, name: str, col: Column) -> "AnovosFrame":
+        out = dict(self._cols)
+        col.name = name
+        out[name] = col
+        return AnovosFrame(out, self.device)
+
+    def filter_rows(self, keep: torch.Tensor) -> "AnovosFrame":
+        """keep: boolean mask or index tensor over local rows."""
+        if keep.dtype == torch.bool:
+            idx = keep.nonzero(as_tuple=True)[0]
+        else:
+            idx = keep
+        return AnovosFrame({n: c.gather(idx) for n, c in self._cols.items()}, self.device)
+
+    def cast(self, name: str, new_dtype: str) -> "AnovosFrame":
+        c = self._cols[name]
+        nd = new_dtype.lower()
+        new_kind = kind_of_dtype(nd)
+        if new_kind == "numerical":
+            tdt = torch.float64 if nd in ("double", "bigint", "long") else torch.float32
+            if c.kind == "numerical":
+                data = c.data.to(tdt)
+                newc = Column(name, nd, data)
+            elif c.kind == "categorical":
+                # parse dictionary entries as floats; unparseable -> NaN
+                vals = []
+                for s in c.dictionary or []:
+                    try:
+                        vals.append(float(s))
+                    except (TypeError, ValueError):
+                        vals.append(float("nan"))
+                lut = torch.tensor(vals + [float("nan")], dtype=tdt, device=c.data.device)
+                codes = c.data.to(torch.long)
+                codes = torch.where(codes == NULL_CODE, torch.full_like(codes, len(vals)), codes)
+                newc = Column(name, nd, lut[codes])
+            else:
+                newc = Column(name, nd, c.data.to(tdt))
+        elif new_kind == "categorical":
+            if c.kind == "categorical":
+                newc = Column(name, "string", c.data, c.dictionary)
+            else:
+                from anovos_amd.ops import encode
+
+                newc = encode.numeric_to_string_column(c)
+        else:
+            newc = Column(name, nd, c.data, c.dictionary)
+        return self.with_column(name, newc)
+
+    def to_device(self, device) -> "AnovosFrame":
+        dev = torch.device(device)
+        out = {n: Column(c.name, c.dtype, c.data.to(dev), c.dictionary) for n, c in self._cols.items()}
+        return AnovosFrame(out, dev)
+
+    def persist(self) -> "AnovosFrame":  # Spark-parity no-op: tensors are already resident
+        return self
+
+    def unpersist(self) -> "AnovosFrame":
+        return self
+
+    def copy(self) -> "AnovosFrame":
+        return AnovosFrame({n: c.clone() for n, c in self._cols.items()}, self.device)
+
+    def __repr__(self):
+        return f"AnovosFrame({self.local_rows()} local rows x {len(self._cols)} cols on {self.device})"
+
+
+def _column_from_series(name: str, s: pd.Series, dev: torch.device, override: Optional[str] = None) -> Column:
+    """Build a Column from a pandas Series, inferring the Spark-style dtype."""
+    if override:
+        k = kind_of_dtype(override)
+        if k == "categorical":
+            return _string_column(name, s, dev)
+        if k == "numerical":
+            arr = pd.to_numeric(s, errors="coerce").astype("float64" if override in ("double", "bigint", "long") else "float32")
+            t = torch.from_numpy(np.ascontiguousarray(arr.to_numpy())).to(dev)
+            return Column(name, override, t)
+    if pd.api.types.is_datetime64_any_dtype(s):
+        vals = s.astype("datetime64[us]")
+        arr = vals.to_numpy().astype("int64")
+        arr = np.where(s.isna().to_numpy(), NULL_TS, arr)
+        return Column(name, "timestamp", torch.from_numpy(arr).to(dev))
+    if pd.api.types.is_float_dtype(s):
+        arr = s.to_numpy(dtype="float64" if s.dtype == np.float64 else "float32")
+        dt = "double" if arr.dtype == np.float64 else "float"
+        return Column(name, dt, torch.from_numpy(np.ascontiguousarray(arr)).to(dev))
+    if pd.api.types.is_bool_dtype(s):
+        arr = s.to_numpy(dtype="float32")
+        return Column(name, "int", torch.from_numpy(arr).to(dev))
+    if pd.api.types.is_integer_dtype(s):
+        if s.isna().any():
+            arr = s.astype("float64").to_numpy()
+        else:
+            arr = s.to_numpy()
+        wide = arr.dtype.itemsize > 4 if arr.dtype.kind in "iu" else True
+        out = arr.astype("float64" if wide else "float32")
+        dt = "bigint" if wide else "int"
+        return Column(name, dt, torch.from_numpy(np.ascontiguousarray(out)).to(dev))
+    return _string_column(name, s, dev)
+
+
+def _string_column(name: str, s: pd.Series, dev: torch.device) -> Column:
+    vals = s.astype(object).where(~s.isna(), None)
+    codes, dictionary = _dict_encode(vals.to_numpy())
+    return Column(name, "string", torch.from_numpy(codes).to(dev), dictionary)
+
+
+def _dict_encode(arr: np.ndarray):
+    """Dictionary-encode an object array -> (int32 codes, list[str]).
+    Dictionary order: first-occurrence (stable); null -> NULL_CODE."""
+    mask = np.array([v is None or (isinstance(v, float) and np.isnan(v)) for v in arr])
+    strs = np.array(["" if m else str(v) for v, m in zip(arr, mask)], dtype=object)
+    uniq, codes = np.unique(strs.astype(str), return_inverse=True)
+    # re-map to first-occurrence order for determinism across ranks is handled
+    # at unify time; local order is np.unique's sorted order.
+    dictionary = [str(u) for u in uniq]
+    codes = codes.astype(np.int32)
+    codes[mask] = NULL_CODE
+    return codes, dictionary

@@ -1,0 +1,172 @@
+"""Columnar IO: csv/parquet/json/avro <-> AnovosFrame via pyarrow.
+
+Replaces the Spark readers of the reference (data_ingest.py:23-117). A
+"path" may be a file or a directory of part files (Spark writes
+directories). Row-partitioning: with world_size W, rank r reads part
+files (or row-group ranges) r, r+W, r+2W... so ingest is parallel and no
+rows are duplicated; writers emit one part file per rank.
+
+Avro: a minimal pure-python Avro Object Container File codec
+(core/avro_codec.py) — the arrow ecosystem here has no avro reader and
+the reference needs avro read/write parity (data_ingest.py:36-38).
+"""
+
+from __future__ import annotations
+
+import glob
+import json
+import os
+from typing import Dict, List, Optional
+
+import numpy as np
+import pandas as pd
+import pyarrow as pa
+import pyarrow.csv as pacsv
+import pyarrow.parquet as papq
+
+from anovos_amd.core import dist
+from anovos_amd.core.frame import AnovosFrame
+
+
+def _expand_parts(file_path: str, exts) -> List[str]:
+    if os.path.isdir(file_path):
+        parts = []
+        for e in exts:
+            parts += glob.glob(os.path.join(file_path, f"*{e}"))
+        parts = sorted(p for p in parts if not os.path.basename(p).startswith(("_", ".")))
+        if not parts:
+            # Spark part files may have no extension for csv/json
+            parts = sorted(
+                p
+                for p in glob.glob(os.path.join(file_path, "part-*"))
+                if not p.endswith(".crc")
+            )
+        return parts
+    return [file_path]
+
+
+def _my_parts(parts: List[str]) -> List[str]:
+    w, r = dist.world_size(), dist.rank()
+    if w <= 1:
+        return parts
+    mine = parts[r::w]
+    return mine
+
+
+def read_dataset(file_path: str, file_type: str, file_configs: Dict = None, device="cpu") -> AnovosFrame:
+    file_configs = dict(file_configs or {})
+    ft = file_type.lower()
+    if ft == "csv":
+        return _read_csv(file_path, file_configs, device)
+    if ft == "parquet":
+        return _read_parquet(file_path, file_configs, device)
+    if ft == "json":
+        return _read_json(file_path, file_configs, device)
+    if ft == "avro":
+        return _read_avro(file_path, file_configs, device)
+    raise ValueError(f"unsupported file_type: {file_type}")
+
+
+def _concat_tables(tables: List[pa.Table]) -> pa.Table:
+    if not tables:
+        return pa.table({})
+    return pa.concat_tables(tables, promote_options="permissive")
+
+
+def _to_frame(table: pa.Table, device) -> AnovosFrame:
+    pdf = table.to_pandas(types_mapper=None)
+    return AnovosFrame.from_pandas(pdf, device=device)
+
+
+def _read_csv(path, cfg, device):
+    header = str(cfg.get("header", True)).lower() in ("true", "1")
+    delim = cfg.get("delimiter", cfg.get("sep", ","))
+    parts = _my_parts(_expand_parts(path, [".csv"]))
+    tables = []
+    for p in parts:
+        ro = pacsv.ReadOptions(autogenerate_column_names=not header)
+        po = pacsv.ParseOptions(delimiter=delim)
+        co = pacsv.ConvertOptions(strings_can_be_null=True)
+        if str(cfg.get("inferSchema", True)).lower() not in ("true", "1"):
+            pass  # arrow always infers; parity is close enough (ints/floats/strings)
+        tables.append(pacsv.read_csv(p, read_options=ro, parse_options=po, convert_options=co))
+    if not tables:
+        raise FileNotFoundError(f"no csv part files under {path}")
+    return _to_frame(_concat_tables(tables), device)
+
+
+def _read_parquet(path, cfg, device):
+    parts = _my_parts(_expand_parts(path, [".parquet", ".pq"]))
+    if not parts:
+        raise FileNotFoundError(f"no parquet files under {path}")
+    tables = [papq.read_table(p) for p in parts]
+    return _to_frame(_concat_tables(tables), device)
+
+
+def _read_json(path, cfg, device):
+    import pyarrow.json as pajson
+
+    parts = _my_parts(_expand_parts(path, [".json", ".jsonl"]))
+    tables = [pajson.read_json(p) for p in parts]
+    if not tables:
+        raise FileNotFoundError(f"no json files under {path}")
+    return _to_frame(_concat_tables(tables), device)
+
+
+def _read_avro(path, cfg, device):
+    from anovos_amd.core import avro_codec
+
+    parts = _my_parts(_expand_parts(path, [".avro"]))
+    if not parts:
+        raise FileNotFoundError(f"no avro files under {path}")
+    pdfs = [avro_codec.read_avro(p) for p in parts]
+    pdf = pd.concat(pdfs, ignore_index=True) if len(pdfs) > 1 else pdfs[0]
+    return AnovosFrame.from_pandas(pdf, device=device)
+
+
+def write_dataset(idf: AnovosFrame, file_path: str, file_type: str, file_configs: Dict = None, column_order: List[str] = None):
+    """Write the frame (all ranks write their shard as part files into a
+    directory, Spark-style). mode: error|overwrite|append."""
+    file_configs = dict(file_configs or {})
+    mode = file_configs.get("mode", "error")
+    ft = file_type.lower()
+    if column_order:
+        if len(column_order) != len(idf.columns):
+            raise ValueError("Count of column(s) specified in column_order argument do not match Dataframe")
+        diff = [x for x in column_order if x not in set(idf.columns)]
+        if diff:
+            raise ValueError(f"Column(s) specified in column_order argument not found in Dataframe: {diff}")
+        idf = idf.select(column_order)
+    if dist.rank() == 0:
+        if os.path.exists(file_path):
+            if mode == "error":
+                raise FileExistsError(f"{file_path} exists (mode=error)")
+            if mode == "overwrite":
+                import shutil
+
+                if os.path.isdir(file_path):
+                    shutil.rmtree(file_path)
+                else:
+                    os.remove(file_path)
+        os.makedirs(file_path, exist_ok=True)
+    dist.barrier()
+    part = os.path.join(file_path, f"part-{dist.rank():05d}")
+    pdf = idf.to_pandas()
+    if ft == "csv":
+        header = str(file_configs.get("header", True)).lower() in ("true", "1")
+        delim = file_configs.get("delimiter", ",")
+        pdf.to_csv(part + ".csv", index=False, header=header, sep=delim)
+    elif ft == "parquet":
+        compression = file_configs.get("compression", "snappy")
+        if compression == "uncompressed":
+            compression = None
+        papq.write_table(pa.Table.from_pandas(pdf, preserve_index=False), part + ".parquet", compression=compression)
+    elif ft == "json":
+        pdf.to_json(part + ".json", orient="records", lines=True)
+    elif ft == "avro":
+        from anovos_amd.core import avro_codec
+
+        avro_codec.write_avro(pdf, part + ".avro")
+    else:
+        raise ValueError(f"unsupported file_type: {file_type}")
+    dist.barrier()

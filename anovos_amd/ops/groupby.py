@@ -1,0 +1,151 @@
+"""groupBy-count family (kernel K5/K21): category frequencies, mode,
+top-k, label-conditioned histograms for IV/IG.
+
+Categorical columns are dictionary codes, so groupBy-count is a bincount
+over the (small) dictionary — LDS-staged on GPU — merged across ranks
+with one all-reduce per batch of columns. Exact numeric mode/unique use
+sort-based torch.unique (rocPRIM radix sort under the hood on ROCm).
+
+Reference semantics: stats_generator.mode_computation (:328, groupBy-
+count-top1), outlier_categories ranking (transformers.py:3614-3641),
+duplicate detection (quality_checker.py:122).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from anovos_amd.core import dist
+from anovos_amd.core.dtypes import NULL_CODE
+from anovos_amd.ops import backend
+
+
+def cat_value_counts(idf, cols: List[str]) -> Dict[str, torch.Tensor]:
+    """Global counts per dictionary code for categorical columns.
+    Returns {col: int64 tensor [dict_size]} on CPU."""
+    out = {}
+    bufs = []
+    for c in cols:
+        col = idf.col(c)
+        size = len(col.dictionary or [])
+        codes = col.data
+        if codes.is_cuda and backend.use_hip(codes):
+            ext = backend.hip_ext()
+            cnt = ext.code_counts(codes.contiguous(), size)
+        else:
+            valid = codes[codes != NULL_CODE].long()
+            cnt = torch.bincount(valid, minlength=size) if size else torch.zeros(0, dtype=torch.int64, device=codes.device)
+        bufs.append(cnt)
+    if bufs:
+        flat = torch.cat([b for b in bufs]) if len(bufs) > 1 else bufs[0]
+        dist.all_reduce_(flat, "sum")
+        off = 0
+        for c, b in zip(cols, bufs):
+            out[c] = flat[off : off + b.numel()].cpu()
+            off += b.numel()
+    return out
+
+
+def mode(idf, col: str, counts: Optional[torch.Tensor] = None) -> Tuple[Optional[str], int]:
+    """Global mode (value, count). Works for categorical (dictionary) and
+    numeric (exact unique) columns. Ties: reference takes Spark's
+    groupBy().count().orderBy(desc).limit(1) — an arbitrary max; we take
+    the first max (lowest code / smallest value)."""
+    c = idf.col(col)
+    if c.kind == "categorical":
+        if counts is None:
+            counts = cat_value_counts(idf, [col])[col]
+        if counts.numel() == 0 or int(counts.sum()) == 0:
+            return None, 0
+        i = int(torch.argmax(counts).item())
+        return c.dictionary[i], int(counts[i])
+    vals, cnts = numeric_value_counts(idf, col)
+    if vals.numel() == 0:
+        return None, 0
+    i = int(torch.argmax(cnts).item())
+    return float(vals[i]), int(cnts[i])
+
+
+def numeric_value_counts(idf, col: str) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Exact (value, count) for a numeric column, merged across ranks.
+    Sort-based; intended for cardinality-bounded columns (mode, IDness)."""
+    t = idf.col(col).data
+    x = t[~torch.isnan(t)]
+    vals, cnts = torch.unique(x, return_counts=True)
+    if dist.is_dist():
+        gathered = dist.all_gather_object((vals.cpu().numpy(), cnts.cpu().numpy()))
+        import numpy as np
+
+        av = np.concatenate([g[0] for g in gathered])
+        ac = np.concatenate([g[1] for g in gathered])
+        uv, inv = np.unique(av, return_inverse=True)
+        uc = np.zeros(len(uv), dtype=np.int64)
+        np.add.at(uc, inv, ac)
+        return torch.from_numpy(uv), torch.from_numpy(uc)
+    return vals.cpu(), cnts.cpu()
+
+
+def duplicate_row_count(idf, cols: Optional[List[str]] = None) -> int:
+    """Rows minus distinct rows (reference quality_checker.py:122:
+    groupBy-all-cols). Implemented as a 64-bit row-hash count-distinct —
+    the K5 dedup design. Exact up to hash collisions (~n^2/2^64)."""
+    cols = cols or idf.columns
+    h = row_hash(idf, cols)
+    uniq = torch.unique(h)
+    if dist.is_dist():
+        gathered = dist.all_gather_object(uniq.cpu().numpy())
+        import numpy as np
+
+        nuniq = len(np.unique(np.concatenate(gathered)))
+    else:
+        nuniq = uniq.numel()
+    total = idf.count()
+    return int(total - nuniq)
+
+
+def row_hash(idf, cols: List[str]) -> torch.Tensor:
+    """64-bit combined row hash across columns (device-side)."""
+    n = idf.local_rows()
+    dev = idf.device
+    h = torch.full((n,), -0x61C8864680B583EB, dtype=torch.int64, device=dev)  # 0x9E3779B97F4A7C15
+    for c in cols:
+        col = idf.col(c)
+        if col.kind == "numerical":
+            b = col.data.to(torch.float64).view(torch.int64)
+            b = torch.where(torch.isnan(col.data), torch.full_like(b, -1), b)
+        elif col.kind == "categorical":
+            # hash the string VALUES (dictionary codes differ across frames):
+            # stable 64-bit string hash LUT gathered by code
+            lut = torch.tensor(
+                [_str_hash64(s) for s in (col.dictionary or [])] + [-1],
+                dtype=torch.int64,
+                device=dev,
+            )
+            codes = col.data.to(torch.long)
+            codes = torch.where(codes == NULL_CODE, torch.full_like(codes, lut.numel() - 1), codes)
+            b = lut[codes]
+        else:
+            b = col.data.to(torch.int64)
+        h = _mix64(h ^ _mix64(b))
+    return h
+
+
+def _str_hash64(s: str) -> int:
+    """Stable FNV-1a 64-bit string hash (python ints, wrapped to int64)."""
+    h = 0xCBF29CE484222325
+    for byte in s.encode("utf-8"):
+        h ^= byte
+        h = (h * 0x100000001B3) & 0xFFFFFFFFFFFFFFFF
+    return h - 0x10000000000000000 if h >= 0x8000000000000000 else h
+
+
+def _mix64(x: torch.Tensor) -> torch.Tensor:
+    # splitmix64 finalizer (wrapping int64 arithmetic)
+    x = x ^ (x >> 30)
+    x = x * -0x40A7B892E31B1A47  # 0xBF58476D1CE4E5B9
+    x = x ^ (x >> 27)
+    x = x * -0x6B2FB644ECCEEE15  # 0x94D049BB133111EB
+    x = x ^ (x >> 31)
+    return x

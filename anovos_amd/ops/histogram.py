@@ -1,0 +1,158 @@
+"""Histogram build + approximate quantiles (kernels K3/K6).
+
+Replaces Spark's Greenwald-Khanna `approxQuantile` (rel-err 0.01,
+reference stats_generator.py:906-908, quality_checker.py:843-847,
+transformers.py:210-215). Algorithm: fused equal-width histogram over all
+columns in one kernel (LDS-staged bins on GPU), all-reduced across ranks,
+then per-quantile bracket refinement passes — two passes give rank
+resolution nbins^2 (≈4e6), far inside Spark's 1% error band.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Sequence
+
+import torch
+
+from anovos_amd.core import dist
+from anovos_amd.ops import backend
+
+DEFAULT_BINS = 2048
+
+
+def column_histograms(
+    tensors: Sequence[torch.Tensor],
+    lo: torch.Tensor,
+    hi: torch.Tensor,
+    nbins: int = DEFAULT_BINS,
+) -> torch.Tensor:
+    """Per-column equal-width histograms over [lo_i, hi_i].
+
+    Values outside the range are clamped into the edge bins (the range is
+    normally the column's global min/max so nothing clips). NaN skipped.
+    Returns int64 [ncols, nbins] LOCAL counts on the device.
+    """
+    ncols = len(tensors)
+    dev = tensors[0].device if ncols else torch.device("cpu")
+    if ncols and dev.type == "cuda" and backend.use_hip(tensors[0]):
+        ext = backend.hip_ext()
+        return ext.column_histograms([t.contiguous() for t in tensors], lo.to(dev), hi.to(dev), nbins)
+    out = torch.zeros(ncols, nbins, dtype=torch.int64, device=dev)
+    for i, t in enumerate(tensors):
+        x = t[~torch.isnan(t)].to(torch.float64)
+        l, h = float(lo[i]), float(hi[i])
+        if x.numel() == 0 or not (h > l):
+            if x.numel() and h == l:
+                out[i, 0] = x.numel()
+            continue
+        idx = ((x - l) * (nbins / (h - l))).long().clamp_(0, nbins - 1)
+        out[i] = torch.bincount(idx, minlength=nbins)
+    return out
+
+
+def global_histograms(tensors, lo, hi, nbins=DEFAULT_BINS) -> torch.Tensor:
+    h = column_histograms(tensors, lo, hi, nbins)
+    dist.all_reduce_(h, "sum")
+    return h
+
+
+def approx_quantiles(
+    idf,
+    cols: List[str],
+    probs: Sequence[float],
+    nbins: int = DEFAULT_BINS,
+    refine: int = 1,
+    moments: Optional[dict] = None,
+) -> Dict[str, List[float]]:
+    """Approximate quantiles for each column at the given probabilities.
+
+    Semantics follow Spark approxQuantile: returns a value whose rank is
+    within rel-err of prob*n; after one refinement pass the bracket is
+    ~(range/nbins^2) wide and we return its midpoint interpolation.
+    """
+    from anovos_amd.ops import stats as stats_ops
+
+    if moments is None:
+        moments = stats_ops.frame_moments(idf, cols)
+    tensors = [idf.col(c).data for c in cols]
+    dev = idf.device
+    lo = torch.tensor([moments[c].min for c in cols], dtype=torch.float64)
+    hi = torch.tensor([moments[c].max for c in cols], dtype=torch.float64)
+    n = torch.tensor([moments[c].n for c in cols], dtype=torch.float64)
+    probs = list(probs)
+
+    hist = global_histograms(tensors, lo, hi, nbins).cpu().to(torch.float64)
+    cdf = torch.cumsum(hist, dim=1)
+
+    result = {c: [float("nan")] * len(probs) for c in cols}
+    # bracket per (col, prob): bin containing target rank
+    brackets = {}
+    for i, c in enumerate(cols):
+        if moments[c].n == 0 or moments[c].min != moments[c].min:
+            continue
+        l, h = float(lo[i]), float(hi[i])
+        if h <= l:
+            result[c] = [l] * len(probs)
+            continue
+        w = (h - l) / nbins
+        for j, p in enumerate(probs):
+            target = p * (moments[c].n - 1)
+            b = int(torch.searchsorted(cdf[i], torch.tensor(target + 0.5)).item())
+            b = min(b, nbins - 1)
+            below = float(cdf[i, b - 1]) if b > 0 else 0.0
+            brackets[(i, j)] = (l + b * w, l + (b + 1) * w, target - below, float(hist[i, b]))
+    for _ in range(refine):
+        brackets = _refine_pass(tensors, cols, brackets, nbins)
+    for (i, j), (bl, bh, off, cnt) in brackets.items():
+        c = cols[i]
+        if cnt <= 1 or bh - bl < 1e-12 * max(1.0, abs(bl)):
+            result[c][j] = bl
+        else:
+            # interpolate inside the (now tiny) bracket by rank fraction
+            frac = min(max(off / max(cnt - 1, 1e-9), 0.0), 1.0) if cnt > 1 else 0.0
+            result[c][j] = bl + frac * (bh - bl)
+    return result
+
+
+def _refine_pass(tensors, cols, brackets, nbins):
+    """One narrowing pass: histogram each active bracket, all columns and
+    brackets in one fused launch (HIP) / vectorized torch loop (CPU)."""
+    if not brackets:
+        return brackets
+    keys = list(brackets.keys())
+    dev = tensors[0].device
+    blo = torch.tensor([brackets[k][0] for k in keys], dtype=torch.float64)
+    bhi = torch.tensor([brackets[k][1] for k in keys], dtype=torch.float64)
+    colidx = torch.tensor([k[0] for k in keys], dtype=torch.int64)
+    if dev.type == "cuda" and backend.use_hip(tensors[0]):
+        ext = backend.hip_ext()
+        h = ext.bracket_histograms([t.contiguous() for t in tensors], colidx.to(dev), blo.to(dev), bhi.to(dev), nbins)
+    else:
+        h = torch.zeros(len(keys), nbins, dtype=torch.int64, device=dev)
+        for kk, (i, j) in enumerate(keys):
+            t = tensors[i]
+            x = t[~torch.isnan(t)].to(torch.float64)
+            l, hh = float(blo[kk]), float(bhi[kk])
+            if hh <= l:
+                continue
+            m = (x >= l) & (x < hh)
+            xv = x[m]
+            if xv.numel() == 0:
+                continue
+            idx = ((xv - l) * (nbins / (hh - l))).long().clamp_(0, nbins - 1)
+            h[kk] = torch.bincount(idx, minlength=nbins)
+    dist.all_reduce_(h, "sum")
+    h = h.cpu().to(torch.float64)
+    cdf = torch.cumsum(h, dim=1)
+    out = {}
+    for kk, k in enumerate(keys):
+        bl, bh, off, _ = brackets[k]
+        if bh <= bl:
+            out[k] = brackets[k]
+            continue
+        w = (bh - bl) / nbins
+        b = int(torch.searchsorted(cdf[kk], torch.tensor(off + 0.5)).item())
+        b = min(b, nbins - 1)
+        below = float(cdf[kk, b - 1]) if b > 0 else 0.0
+        out[k] = (bl + b * w, bl + (b + 1) * w, off - below, float(h[kk, b]))
+    return out

@@ -1,0 +1,147 @@
+"""Fused multi-column statistics (kernel K1/K2 of SURVEY.md §2.10).
+
+One pass over all numeric columns produces, per column, the fp64 partial
+vector [n_valid, sum, sum2, sum3, sum4, min, max, zero_count]. The
+reference computes these with one Spark job per column per statistic
+(stats_generator.py:485-494 — the dominant anti-pattern); here every
+column is covered by a single kernel launch and ONE batched RCCL
+all-reduce merges the 8-double-per-column partials across ranks.
+
+HIP path: ops/hip/stats_kernels.hip (partials per (col, chunk) block,
+deterministic two-kernel reduce, fp64 accumulators).
+Torch path: reference implementation, also the CPU backend.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Sequence
+
+import torch
+
+from anovos_amd.core import dist
+from anovos_amd.ops import backend
+
+NSTAT = 8  # n, s1, s2, s3, s4, min, max, zeros
+
+
+def column_moments_local(tensors: Sequence[torch.Tensor]) -> torch.Tensor:
+    """Local partial moment vectors for a list of numeric columns
+    (NaN = null). Returns fp64 tensor [ncols, 8] on the columns' device."""
+    if len(tensors) == 0:
+        return torch.empty(0, NSTAT, dtype=torch.float64)
+    dev = tensors[0].device
+    if dev.type == "cuda" and backend.use_hip(tensors[0]):
+        ext = backend.hip_ext()
+        return ext.column_moments([t.contiguous() for t in tensors])
+    out = torch.empty(len(tensors), NSTAT, dtype=torch.float64, device=dev)
+    for i, t in enumerate(tensors):
+        td = t.to(torch.float64)
+        valid = ~torch.isnan(td)
+        x = torch.where(valid, td, torch.zeros_like(td))
+        n = valid.sum()
+        s1 = x.sum()
+        x2 = x * x
+        s2 = x2.sum()
+        s3 = (x2 * x).sum()
+        s4 = (x2 * x2).sum()
+        if int(n) > 0:
+            mn = td[valid].min()
+            mx = td[valid].max()
+        else:
+            mn = torch.tensor(float("nan"), dtype=torch.float64, device=dev)
+            mx = torch.tensor(float("nan"), dtype=torch.float64, device=dev)
+        zeros = ((x == 0) & valid).sum()
+        out[i] = torch.stack([n.to(torch.float64), s1, s2, s3, s4, mn, mx, zeros.to(torch.float64)])
+    return out
+
+
+def merge_moments_global(local: torch.Tensor) -> torch.Tensor:
+    """RCCL merge of [ncols, 8] partials: sum for n/s1..s4/zeros, min/max
+    for the extrema — batched into two fused all-reduces."""
+    if not dist.is_dist():
+        return local.cpu()
+    sums = local[:, [0, 1, 2, 3, 4, 7]].contiguous()
+    mn = torch.nan_to_num(local[:, 5], nan=float("inf")).contiguous()
+    mx = torch.nan_to_num(local[:, 6], nan=float("-inf")).contiguous()
+    dist.all_reduce_(sums, "sum")
+    dist.all_reduce_(mn, "min")
+    dist.all_reduce_(mx, "max")
+    out = torch.empty_like(local)
+    out[:, [0, 1, 2, 3, 4, 7]] = sums
+    out[:, 5] = torch.where(torch.isinf(mn), torch.full_like(mn, float("nan")), mn)
+    out[:, 6] = torch.where(torch.isinf(mx), torch.full_like(mx, float("nan")), mx)
+    return out.cpu()
+
+
+class MomentStats:
+    """Derived statistics for one column from its global moment vector."""
+
+    __slots__ = ("n", "s1", "s2", "s3", "s4", "min", "max", "zeros")
+
+    def __init__(self, vec):
+        self.n, self.s1, self.s2, self.s3, self.s4, self.min, self.max, self.zeros = [float(v) for v in vec]
+
+    @property
+    def mean(self):
+        return self.s1 / self.n if self.n > 0 else float("nan")
+
+    def _central(self):
+        n, m = self.n, self.mean
+        M2 = self.s2 - n * m * m
+        M3 = self.s3 - 3 * m * self.s2 + 2 * n * m**3
+        M4 = self.s4 - 4 * m * self.s3 + 6 * m * m * self.s2 - 3 * n * m**4
+        return max(M2, 0.0), M3, M4
+
+    @property
+    def variance(self):  # sample variance (Spark summary stddev is n-1)
+        if self.n < 2:
+            return float("nan")
+        M2, _, _ = self._central()
+        return M2 / (self.n - 1)
+
+    @property
+    def stddev(self):
+        v = self.variance
+        return math.sqrt(v) if v == v else float("nan")
+
+    @property
+    def skewness(self):  # Spark F.skewness: population, biased
+        if self.n < 1:
+            return float("nan")
+        M2, M3, _ = self._central()
+        if M2 <= 0:
+            return float("nan")
+        m2 = M2 / self.n
+        m3 = M3 / self.n
+        return m3 / m2**1.5
+
+    @property
+    def kurtosis(self):  # Spark F.kurtosis: excess kurtosis, population
+        if self.n < 1:
+            return float("nan")
+        M2, _, M4 = self._central()
+        if M2 <= 0:
+            return float("nan")
+        m2 = M2 / self.n
+        m4 = M4 / self.n
+        return m4 / (m2 * m2) - 3.0
+
+
+def frame_moments(idf, cols: List[str]) -> Dict[str, MomentStats]:
+    """Global moment stats for the given numeric columns of a frame."""
+    tensors = [idf.col(c).data for c in cols]
+    local = column_moments_local(tensors)
+    glob = merge_moments_global(local)
+    return {c: MomentStats(glob[i]) for i, c in enumerate(cols)}
+
+
+def null_counts(idf, cols: List[str]) -> Dict[str, int]:
+    """Per-column global null counts (any kind), one fused all-reduce."""
+    dev = idf.device
+    local = torch.empty(len(cols), dtype=torch.float64, device=dev)
+    for i, c in enumerate(cols):
+        local[i] = idf.col(c).null_mask().sum()
+    dist.all_reduce_(local, "sum")
+    total = idf.count()
+    return {c: int(local[i].item()) for i, c in enumerate(cols)}, total
