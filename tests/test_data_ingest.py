@@ -81,7 +81,7 @@ def test_delete_select_rename_recast(ctx, small_pdf):
 
 
 def test_recommend_type(ctx):
-    n = 300
+    n = 500  # dynamic threshold 0.01*500 = 5 > 3 distinct values
     rng = np.random.default_rng(0)
     pdf = pd.DataFrame(
         {
