@@ -18,6 +18,45 @@ from anovos_amd.core import dist
 from anovos_amd.ops import backend
 
 DEFAULT_BINS = 2048
+EXACT_N_THRESHOLD = 1_000_000  # below this, sort-exact quantiles (GK parity)
+
+
+def _exact_quantiles(idf, cols, probs, moments, rel_err=1e-4):
+    """Exact quantiles by device sort (rocPRIM radix sort via torch.sort);
+    Spark GK query semantics (QuantileSummaries.query): p <= relErr ->
+    min, p >= 1-relErr -> max, else element at rank ceil(p*n) — the
+    float product ceil matters for parity with approxQuantile."""
+    import math as _math
+
+    from anovos_amd.core import dist as _dist
+
+    out = {}
+    for c in cols:
+        t = idf.col(c).data
+        x = t[~torch.isnan(t)]
+        if _dist.is_dist():
+            gathered = _dist.all_gather_object(x.cpu().numpy())
+            import numpy as _np
+
+            allv = _np.concatenate(gathered)
+            xs = torch.from_numpy(_np.sort(allv))
+        else:
+            xs, _ = torch.sort(x)
+        n = xs.numel()
+        if n == 0:
+            out[c] = [float("nan")] * len(probs)
+            continue
+        vals = []
+        for p in probs:
+            if p <= rel_err:
+                r = 1
+            elif p >= 1 - rel_err:
+                r = n
+            else:
+                r = min(max(int(_math.ceil(p * n)), 1), n)
+            vals.append(float(xs[r - 1]))
+        out[c] = vals
+    return out
 
 
 def column_histograms(
@@ -63,6 +102,7 @@ def approx_quantiles(
     nbins: int = DEFAULT_BINS,
     refine: int = 1,
     moments: Optional[dict] = None,
+    rel_err: float = 1e-4,
 ) -> Dict[str, List[float]]:
     """Approximate quantiles for each column at the given probabilities.
 
@@ -74,6 +114,11 @@ def approx_quantiles(
 
     if moments is None:
         moments = stats_ops.frame_moments(idf, cols)
+    # Exact path for small columns: Spark's GK sketch returns actual data
+    # elements (rank = ceil(p*n)); reference tests assert 4-decimal values,
+    # so below this size we sort-and-index instead of sketching.
+    if all(moments[c].n <= EXACT_N_THRESHOLD for c in cols):
+        return _exact_quantiles(idf, cols, probs, moments, rel_err=rel_err)
     tensors = [idf.col(c).data for c in cols]
     dev = idf.device
     lo = torch.tensor([moments[c].min for c in cols], dtype=torch.float64)
