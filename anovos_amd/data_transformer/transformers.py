@@ -932,3 +932,14 @@ def _eval_expr(idf: AnovosFrame, expr: str) -> torch.Tensor:
     if not torch.is_tensor(out):
         out = torch.full((idf.local_rows(),), float(out), dtype=torch.float32, device=idf.device)
     return out.to(torch.float32)
+
+
+# Advanced imputers / latent features live in transformers_advanced but are
+# part of this module's API surface in the reference (transformers.py:1677-3168).
+from anovos_amd.data_transformer.transformers_advanced import (  # noqa: E402,F401
+    PCA_latentFeatures,
+    auto_imputation,
+    autoencoder_latentFeatures,
+    imputation_matrixFactorization,
+    imputation_sklearn,
+)

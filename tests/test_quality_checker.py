@@ -1,0 +1,118 @@
+import numpy as np
+import pandas as pd
+import pytest
+import torch
+
+from anovos_amd.core.frame import AnovosFrame
+from anovos_amd.data_analyzer import quality_checker as qc
+
+
+@pytest.fixture()
+def dup_frame():
+    pdf = pd.DataFrame(
+        {
+            "a": [1.0, 1.0, 2.0, 3.0, 3.0, 3.0],
+            "b": ["x", "x", "y", "z", "z", "z"],
+        }
+    )
+    return AnovosFrame.from_pandas(pdf)
+
+
+def test_duplicate_detection(ctx, dup_frame):
+    odf, stats = qc.duplicate_detection(ctx, dup_frame, treatment=True, print_impact=True)
+    d = dict(zip(stats["metric"], stats["value"]))
+    assert d["rows_count"] == 6.0
+    assert d["unique_rows_count"] == 3.0
+    assert d["duplicate_rows"] == 3.0
+    assert odf.count() == 3
+
+
+def test_null_rows_detection(ctx):
+    pdf = pd.DataFrame(
+        {
+            "a": [1.0, np.nan, np.nan, 4.0],
+            "b": ["x", None, None, "w"],
+            "c": [1.0, 2.0, np.nan, 4.0],
+        }
+    )
+    f = AnovosFrame.from_pandas(pdf)
+    odf, stats = qc.nullRows_detection(ctx, f, treatment=True, treatment_threshold=0.5)
+    # row 1 has 2/3 null (flagged), row 2 has 3/3 null (flagged)
+    assert odf.count() == 2
+    assert stats[stats["null_cols_count"] == 2]["treated"].iloc[0] == 1
+
+
+def test_null_columns_detection_row_removal(ctx, income_frame):
+    odf, stats = qc.nullColumns_detection(ctx, income_frame, treatment=True, treatment_method="row_removal")
+    assert odf.count() == 400 - int(
+        (income_frame.col("age").null_mask() | income_frame.col("workclass").null_mask()).sum()
+    )
+    assert set(stats["attribute"]) == {"age", "workclass"}
+
+
+def test_null_columns_detection_MMM(ctx, income_frame):
+    odf, stats = qc.nullColumns_detection(ctx, income_frame, treatment=True, treatment_method="MMM")
+    assert int(odf.col("age").null_mask().sum()) == 0
+    assert int(odf.col("workclass").null_mask().sum()) == 0
+
+
+def test_outlier_detection_both_sides(ctx):
+    rng = np.random.default_rng(0)
+    x = rng.normal(0, 1, 5000)
+    x[:5] = 100.0  # upper outliers
+    x[5:10] = -100.0
+    f = AnovosFrame.from_pandas(pd.DataFrame({"x": x}))
+    odf, stats = qc.outlier_detection(ctx, f, ["x"], detection_side="both", print_impact=True)
+    s = stats.set_index("attribute")
+    assert s.loc["x", "upper_outliers"] >= 5
+    assert s.loc["x", "lower_outliers"] >= 5
+    treated = odf.col("x").data
+    assert float(treated.max()) < 100.0
+    assert float(treated.min()) > -100.0
+
+
+def test_outlier_model_roundtrip(ctx, tmp_path):
+    rng = np.random.default_rng(1)
+    f = AnovosFrame.from_pandas(pd.DataFrame({"x": rng.normal(0, 1, 2000)}))
+    mp = str(tmp_path)
+    odf1 = qc.outlier_detection(ctx, f, ["x"], detection_side="both", model_path=mp)
+    odf2 = qc.outlier_detection(ctx, f, ["x"], detection_side="both", pre_existing_model=True, model_path=mp)
+    assert torch.allclose(odf1.col("x").data, odf2.col("x").data, equal_nan=True)
+
+
+def test_idness_detection(ctx, income_frame):
+    odf, stats = qc.IDness_detection(ctx, income_frame, treatment=True, treatment_threshold=0.9)
+    assert "ifa" not in odf.columns  # unique id column removed
+    assert "education" in odf.columns
+
+
+def test_biasedness_detection(ctx):
+    pdf = pd.DataFrame(
+        {
+            "biased": ["a"] * 95 + ["b"] * 5,
+            "ok": ["x", "y"] * 50,
+        }
+    )
+    f = AnovosFrame.from_pandas(pdf)
+    odf, stats = qc.biasedness_detection(ctx, f, treatment=True, treatment_threshold=0.8)
+    assert "biased" not in odf.columns
+    assert "ok" in odf.columns
+    s = stats.set_index("attribute")
+    assert s.loc["biased", "mode_pct"] == 0.95
+
+
+def test_invalid_entries_detection(ctx):
+    pdf = pd.DataFrame(
+        {
+            "cat": ["good", "n/a", "aaa", "abc", "fine", ":", "xyz1"],
+            "num": [1.0, 2.0, 111.0, 4.0, 5.0, 6.0, 7.0],
+        }
+    )
+    f = AnovosFrame.from_pandas(pdf)
+    odf, stats = qc.invalidEntries_detection(ctx, f, treatment=True, treatment_method="null_replacement")
+    s = stats.set_index("attribute")
+    # n/a (null vocab), aaa (repeated), abc (consecutive), : (special)
+    assert s.loc["cat", "invalid_count"] == 4
+    assert s.loc["num", "invalid_count"] == 1  # "111.0" repeated-char match
+    assert int(odf.col("cat").null_mask().sum()) == 4
+    assert int(odf.col("num").null_mask().sum()) == 1
