@@ -1,0 +1,222 @@
+"""Attribute associations — parity with reference
+data_analyzer/association_evaluator.py (586 LoC).
+
+- correlation_matrix (:38): bf16 MFMA Gram kernel / rocBLAS path (K8),
+- variable_clustering (:142): one fused covariance pass + driver-side
+  VarClusHi (association_eval_varclus.py),
+- IV_calculation (:253): per-bin label histograms (K9) + WOE/IV sums,
+- IG_calculation (:427): entropy before/after split.
+"""
+
+from __future__ import annotations
+
+import math
+import warnings
+from typing import Dict, List
+
+import numpy as np
+import pandas as pd
+import torch
+
+from anovos_amd.core import dist
+from anovos_amd.core.dtypes import NULL_CODE
+from anovos_amd.data_analyzer.association_eval_varclus import VarClusHi
+from anovos_amd.data_analyzer.stats_generator import uniqueCount_computation
+from anovos_amd.data_transformer.transformers import (
+    attribute_binning,
+    cat_to_num_unsupervised,
+    imputation_MMM,
+    monotonic_binning,
+    _event_indicator,
+)
+from anovos_amd.ops import corr as corr_ops
+from anovos_amd.shared.utils import attributeType_segregation, normalize_columns
+
+
+def correlation_matrix(ctx, idf, list_of_cols="all", drop_cols=[], use_sampling=False, sample_size=1000000, print_impact=False):
+    """[attribute, <cols...>] — reference association_evaluator.py:38-139."""
+    num_cols = attributeType_segregation(idf)[0]
+    if list_of_cols == "all":
+        list_of_cols = num_cols
+    cols = normalize_columns(idf, list_of_cols, drop_cols, restrict_to=num_cols)
+    if any(x not in num_cols for x in cols) or len(cols) == 0:
+        raise TypeError("Invalid input for Column(s)")
+    if use_sampling and idf.count() > sample_size:
+        warnings.warn("Using sampling. Only " + str(sample_size) + " random sampled rows are considered.")
+        from anovos_amd.data_ingest.data_sampling import data_sample
+
+        idf = data_sample(idf, fraction=float(sample_size) / idf.count(), method_type="random")
+    corr = corr_ops.pearson_matrix(idf, cols)
+    odf = pd.DataFrame(corr, columns=cols, index=cols)
+    odf["attribute"] = odf.index
+    sorted_cols = sorted(cols)
+    odf = odf[["attribute"] + sorted_cols].sort_values("attribute").reset_index(drop=True)
+    if print_impact:
+        print(odf.to_string(index=False))
+    return odf
+
+
+def variable_clustering(ctx, idf, list_of_cols="all", drop_cols=[], stats_mode={}, persist=True, print_impact=False):
+    """[Cluster, Attribute, RS_Ratio] — reference association_evaluator.py:142-250."""
+    if list_of_cols == "all":
+        num_cols, cat_cols, _ = attributeType_segregation(idf)
+        list_of_cols = num_cols + cat_cols
+    cols = normalize_columns(idf, list_of_cols, drop_cols)
+    if len(cols) == 0:
+        raise TypeError("Invalid input for Column(s)")
+    uc = uniqueCount_computation(ctx, idf, cols)
+    remove_cols = uc[uc["unique_values"] < 2]["attribute"].tolist()
+    cols = [e for e in cols if e not in remove_cols]
+    sub = idf.select(cols)
+    cat_cols = attributeType_segregation(sub)[1]
+    idf_encoded = cat_to_num_unsupervised(ctx, sub, list_of_cols=cat_cols, method_type="label_encoding")
+    num_cols = attributeType_segregation(idf_encoded)[0]
+    idf_encoded = idf_encoded.select(num_cols)
+    idf_imputed = imputation_MMM(ctx, idf_encoded, stats_mode=stats_mode)
+    # single fused pass: correlation matrix of standardized features
+    corr = corr_ops.pearson_matrix(idf_imputed, num_cols)
+    corr_df = pd.DataFrame(corr, columns=num_cols, index=num_cols)
+    vc = VarClusHi(corr_df, maxeigval2=1, maxclus=None)
+    vc.varclus()
+    rs = vc.rsquare()
+    odf = rs[["Cluster", "Variable", "RS_Ratio"]].rename(columns={"Variable": "Attribute"})
+    odf["RS_Ratio"] = odf["RS_Ratio"].round(4)
+    if print_impact:
+        print(odf.to_string(index=False))
+    return odf.reset_index(drop=True)
+
+
+def _binned_label_counts(idf, col: str, label: torch.Tensor):
+    """Per-group (label_0, label_1) counts for a column (null = own group),
+    merged across ranks. Returns dict group_key -> (n0, n1)."""
+    c = idf.col(col)
+    if c.kind == "categorical":
+        codes = c.data.to(torch.long)
+        size = len(c.dictionary or []) + 1
+        codes = torch.where(codes == NULL_CODE, torch.full_like(codes, size - 1), codes)
+    else:
+        vals = torch.nan_to_num(c.data, nan=-1.0).to(torch.long) + 1  # bins 1..N -> 2..N+1; null -> 0
+        codes = vals.clamp(min=0)
+        size = int(dist.all_reduce_scalar(int(codes.max().item()) if codes.numel() else 0, "max")) + 1
+    n1 = torch.zeros(size, dtype=torch.float64, device=codes.device)
+    n0 = torch.zeros(size, dtype=torch.float64, device=codes.device)
+    lab = label.to(torch.float64)
+    n1.scatter_add_(0, codes, lab)
+    n0.scatter_add_(0, codes, 1.0 - lab)
+    dist.all_reduce_(n0, "sum")
+    dist.all_reduce_(n1, "sum")
+    return n0.cpu().numpy(), n1.cpu().numpy()
+
+
+def IV_calculation(
+    ctx,
+    idf,
+    list_of_cols="all",
+    drop_cols=[],
+    label_col="label",
+    event_label=1,
+    encoding_configs={"bin_method": "equal_frequency", "bin_size": 10, "monotonicity_check": 0},
+    print_impact=False,
+):
+    """[attribute, iv] — reference association_evaluator.py:253-424."""
+    if list_of_cols == "all":
+        num_cols, cat_cols, _ = attributeType_segregation(idf)
+        list_of_cols = num_cols + cat_cols
+    if isinstance(list_of_cols, str):
+        list_of_cols = [x.strip() for x in list_of_cols.split("|")]
+    if isinstance(drop_cols, str):
+        drop_cols = [x.strip() for x in drop_cols.split("|")]
+    cols = [e for e in dict.fromkeys(list_of_cols) if e not in (list(drop_cols) + [label_col])]
+    if any(x not in idf.columns for x in cols) or len(cols) == 0:
+        raise TypeError("Invalid input for Column(s)")
+    label = _event_indicator(idf, label_col, event_label)
+    if int(dist.all_reduce_scalar(int(label.sum()))) == 0:
+        raise TypeError("Invalid input for Event Label Value")
+    num_cols = attributeType_segregation(idf.select(cols))[0]
+    if len(num_cols) > 0 and bool(encoding_configs):
+        bin_size = encoding_configs["bin_size"]
+        bin_method = encoding_configs["bin_method"]
+        if encoding_configs.get("monotonicity_check", 0) == 1:
+            idf_encoded = monotonic_binning(ctx, idf, num_cols, [], label_col, event_label, bin_method, bin_size)
+        else:
+            idf_encoded = attribute_binning(ctx, idf, num_cols, [], bin_method, bin_size)
+    else:
+        idf_encoded = idf
+    rows = []
+    for col in cols:
+        n0, n1 = _binned_label_counts(idf_encoded, col, label)
+        keep = (n0 + n1) > 0
+        n0, n1 = n0[keep], n1[keep]
+        t0, t1 = n0.sum(), n1.sum()
+        event_pcr = n1 / t1
+        nonevent_pcr = n0 / t0
+        diff = nonevent_pcr - event_pcr
+        with np.errstate(divide="ignore", invalid="ignore"):
+            woe = np.where(
+                (nonevent_pcr != 0) & (event_pcr != 0),
+                np.log(nonevent_pcr / event_pcr),
+                np.log(((n0 + 0.5) / t0) / ((n1 + 0.5) / t1)),
+            )
+        iv = float(np.sum(woe * diff))
+        rows.append([col, iv])
+    odf = pd.DataFrame(rows, columns=["attribute", "iv"])
+    if print_impact:
+        print(odf.to_string(index=False))
+    return odf
+
+
+def IG_calculation(
+    ctx,
+    idf,
+    list_of_cols="all",
+    drop_cols=[],
+    label_col="label",
+    event_label=1,
+    encoding_configs={"bin_method": "equal_frequency", "bin_size": 10, "monotonicity_check": 0},
+    print_impact=False,
+):
+    """[attribute, ig] — reference association_evaluator.py:427-586."""
+    if list_of_cols == "all":
+        num_cols, cat_cols, _ = attributeType_segregation(idf)
+        list_of_cols = num_cols + cat_cols
+    if isinstance(list_of_cols, str):
+        list_of_cols = [x.strip() for x in list_of_cols.split("|")]
+    if isinstance(drop_cols, str):
+        drop_cols = [x.strip() for x in drop_cols.split("|")]
+    cols = [e for e in dict.fromkeys(list_of_cols) if e not in (list(drop_cols) + [label_col])]
+    if any(x not in idf.columns for x in cols) or len(cols) == 0:
+        raise TypeError("Invalid input for Column(s)")
+    label = _event_indicator(idf, label_col, event_label)
+    total_rows = idf.count()
+    total_events = int(dist.all_reduce_scalar(int(label.sum())))
+    if total_events == 0:
+        raise TypeError("Invalid input for Event Label Value")
+    num_cols = attributeType_segregation(idf.select(cols))[0]
+    if len(num_cols) > 0 and bool(encoding_configs):
+        bin_size = encoding_configs["bin_size"]
+        bin_method = encoding_configs["bin_method"]
+        if encoding_configs.get("monotonicity_check", 0) == 1:
+            idf_encoded = monotonic_binning(ctx, idf, num_cols, [], label_col, event_label, bin_method, bin_size)
+        else:
+            idf_encoded = attribute_binning(ctx, idf, num_cols, [], bin_method, bin_size)
+    else:
+        idf_encoded = idf
+    total_event = total_events / total_rows
+    total_entropy = -(total_event * math.log2(total_event) + (1 - total_event) * math.log2(1 - total_event))
+    rows = []
+    for col in cols:
+        n0, n1 = _binned_label_counts(idf_encoded, col, label)
+        tot = n0 + n1
+        keep = tot > 0
+        n0, n1, tot = n0[keep], n1[keep], tot[keep]
+        event_pct = n1 / tot
+        segment_pct = tot / tot.sum()
+        with np.errstate(divide="ignore", invalid="ignore"):
+            ent = -segment_pct * (event_pct * np.log2(event_pct) + (1 - event_pct) * np.log2(1 - event_pct))
+        # Spark's sum skips null (0*log0) entries — emulate by nan-skipping
+        entropy_sum = float(np.nansum(np.where(np.isfinite(ent), ent, np.nan)))
+        rows.append([col, float(total_entropy - entropy_sum)])
+    odf = pd.DataFrame(rows, columns=["attribute", "ig"])
+    if print_impact:
+        print(odf.to_string(index=False))
+    return odf
