@@ -27,15 +27,22 @@ from anovos_amd.core.dtypes import NULL_CODE, NULL_TS, is_timestamp_dtype, kind_
 
 
 class Column:
-    """One column: tensor data + logical dtype + optional dictionary."""
+    """One column: tensor data + logical dtype + optional dictionary.
 
-    __slots__ = ("name", "dtype", "data", "dictionary")
+    ``cache`` holds derived statistics (moments vector, histograms, null
+    counts) keyed by the ops layer — the in-memory analog of the
+    reference's pre-saved stats reuse (workflow.stats_args, reference
+    workflow.py:91-145). A new Column (new tensor) starts cold; sharing a
+    Column between frames shares its cache."""
+
+    __slots__ = ("name", "dtype", "data", "dictionary", "cache")
 
     def __init__(self, name: str, dtype: str, data: torch.Tensor, dictionary: Optional[List[str]] = None):
         self.name = name
         self.dtype = dtype  # Spark-style dtype string ('double', 'string', 'int', ...)
         self.data = data
         self.dictionary = dictionary
+        self.cache = {}
 
     @property
     def kind(self) -> str:
@@ -198,6 +205,12 @@ class AnovosFrame:
         dev = torch.device(device)
         out = {n: Column(c.name, c.dtype, c.data.to(dev), c.dictionary) for n, c in self._cols.items()}
         return AnovosFrame(out, dev)
+
+    def clear_stats_cache(self) -> None:
+        """Drop all cached derived statistics (forces recomputation —
+        used by benchmarks to keep timed steps honest)."""
+        for c in self._cols.values():
+            c.cache.clear()
 
     def persist(self) -> "AnovosFrame":  # Spark-parity no-op: tensors are already resident
         return self

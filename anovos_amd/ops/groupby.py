@@ -24,8 +24,12 @@ from anovos_amd.ops import backend
 
 def cat_value_counts(idf, cols: List[str]) -> Dict[str, torch.Tensor]:
     """Global counts per dictionary code for categorical columns.
-    Returns {col: int64 tensor [dict_size]} on CPU."""
+    Returns {col: int64 tensor [dict_size]} on CPU. Cached per Column."""
     out = {}
+    cached = [c for c in cols if "cat_counts" in idf.col(c).cache]
+    for c in cached:
+        out[c] = idf.col(c).cache["cat_counts"]
+    cols = [c for c in cols if c not in set(cached)]
     bufs = []
     for c in cols:
         col = idf.col(c)
@@ -44,6 +48,7 @@ def cat_value_counts(idf, cols: List[str]) -> Dict[str, torch.Tensor]:
         off = 0
         for c, b in zip(cols, bufs):
             out[c] = flat[off : off + b.numel()].cpu()
+            idf.col(c).cache["cat_counts"] = out[c]
             off += b.numel()
     return out
 
@@ -70,21 +75,45 @@ def mode(idf, col: str, counts: Optional[torch.Tensor] = None) -> Tuple[Optional
 
 def numeric_value_counts(idf, col: str) -> Tuple[torch.Tensor, torch.Tensor]:
     """Exact (value, count) for a numeric column, merged across ranks.
-    Sort-based; intended for cardinality-bounded columns (mode, IDness)."""
+
+    Fast path: integral-valued columns with bounded range use a dense
+    bincount (one pass, no sort) — exact, and the common case for
+    discrete numerics (ages, counts, codes). Otherwise sort-based
+    torch.unique (rocPRIM radix sort on ROCm)."""
     t = idf.col(col).data
     x = t[~torch.isnan(t)]
+    m = idf.col(col).cache.get("moments")
+    if m is None:
+        from anovos_amd.ops import stats as _stats
+
+        m = _stats.frame_moments(idf, [col])[col]
+    if x.numel() and m is not None and m.min == m.min and (m.max - m.min) < 4_000_000:
+        xi = x.to(torch.int64)
+        if bool((x == xi.to(x.dtype)).all()):
+            lo = int(m.min)
+            cnts_d = torch.bincount(xi - lo, minlength=int(m.max) - lo + 1)
+            nz = cnts_d.nonzero(as_tuple=True)[0]
+            vals = (nz + lo).to(t.dtype)
+            cnts = cnts_d[nz]
+            if dist.is_dist():
+                return _merge_value_counts(vals, cnts)
+            return vals.cpu(), cnts.cpu()
     vals, cnts = torch.unique(x, return_counts=True)
     if dist.is_dist():
-        gathered = dist.all_gather_object((vals.cpu().numpy(), cnts.cpu().numpy()))
-        import numpy as np
-
-        av = np.concatenate([g[0] for g in gathered])
-        ac = np.concatenate([g[1] for g in gathered])
-        uv, inv = np.unique(av, return_inverse=True)
-        uc = np.zeros(len(uv), dtype=np.int64)
-        np.add.at(uc, inv, ac)
-        return torch.from_numpy(uv), torch.from_numpy(uc)
+        return _merge_value_counts(vals, cnts)
     return vals.cpu(), cnts.cpu()
+
+
+def _merge_value_counts(vals: torch.Tensor, cnts: torch.Tensor):
+    gathered = dist.all_gather_object((vals.cpu().numpy(), cnts.cpu().numpy()))
+    import numpy as np
+
+    av = np.concatenate([g[0] for g in gathered])
+    ac = np.concatenate([g[1] for g in gathered])
+    uv, inv = np.unique(av, return_inverse=True)
+    uc = np.zeros(len(uv), dtype=np.int64)
+    np.add.at(uc, inv, ac)
+    return torch.from_numpy(uv), torch.from_numpy(uc)
 
 
 def duplicate_row_count(idf, cols: Optional[List[str]] = None) -> int:

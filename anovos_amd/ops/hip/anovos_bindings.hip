@@ -1,0 +1,279 @@
+// Torch bindings for the anovos_amd MI355X kernels (anovos_kernels.hip).
+// Compiled by torch.utils.cpp_extension with hipcc for gfx950; streams come
+// from the caller's current torch HIP stream so kernels compose with the
+// engine's side-stream overlap (core/dist.py SideStream).
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+extern "C" {
+int anovos_moments(const void *const *cols, const int64_t *lens, int ncols,
+                   int nchunks, int dtype, double *partials, double *out,
+                   hipStream_t stream);
+int anovos_hist(const void *const *cols, const int64_t *lens, int ncols,
+                const double *lo, const double *hi, int nbins, int nchunks,
+                int dtype, uint64_t *out, hipStream_t stream);
+int anovos_bracket_hist(const void *const *cols, const int64_t *lens,
+                        const int64_t *colidx, int nbrackets, const double *lo,
+                        const double *hi, int nbins, int nchunks, int dtype,
+                        uint64_t *out, hipStream_t stream);
+int anovos_bucketize(const void *const *cols, const int64_t *lens, int ncols,
+                     const double *cutflat, const int64_t *cutoff_off,
+                     const int *cutoff_len, int max_ncut, int nchunks, int dtype,
+                     int32_t *const *outs, hipStream_t stream);
+int anovos_code_counts(const int32_t *codes, int64_t n, int size, int nchunks,
+                       uint64_t *out, hipStream_t stream);
+int anovos_hll(const void *x, int64_t n, int p, int nchunks, int dtype,
+               int32_t *regs, hipStream_t stream);
+int anovos_row_null(const void *const *cols, int ncols, int64_t n, int dtype,
+                    int32_t *out, hipStream_t stream);
+}
+
+namespace {
+
+hipStream_t current_stream() { return c10::hip::getCurrentHIPStream().stream(); }
+
+void check_hip(int err, const char *what) {
+  TORCH_CHECK(err == 0, what, " failed: ", hipGetErrorString((hipError_t)err));
+}
+
+int pick_chunks(int64_t n, int ncols) {
+  // fill 256 CUs x 8 XCDs: target >= ~2048 workgroups, >= ~64K elems/chunk
+  int64_t by_size = (n + (1 << 16) - 1) >> 16;
+  int64_t cap = std::max<int64_t>(1, 16384 / std::max(1, ncols));
+  int64_t c = std::min<int64_t>(std::max<int64_t>(by_size, 1), cap);
+  int64_t floor_c = std::min<int64_t>(by_size, std::max<int64_t>(1, 2048 / std::max(1, ncols)));
+  return (int)std::max<int64_t>(c, std::max<int64_t>(floor_c, 1));
+}
+
+struct PtrPack {
+  torch::Tensor dev;  // int64 tensor of device pointers
+};
+
+// Upload an array of device pointers / lengths to the GPU.
+torch::Tensor to_device_i64(const std::vector<int64_t> &vals, const torch::Device &dev) {
+  auto cpu = torch::from_blob((void *)vals.data(), {(int64_t)vals.size()},
+                              torch::TensorOptions().dtype(torch::kInt64))
+                 .clone();
+  return cpu.to(dev, /*non_blocking=*/true);
+}
+
+int dtype_code(const torch::Tensor &t) {
+  if (t.scalar_type() == torch::kFloat32) return 0;
+  if (t.scalar_type() == torch::kFloat64) return 1;
+  TORCH_CHECK(false, "expected float32/float64 column, got ", t.scalar_type());
+  return -1;
+}
+
+}  // namespace
+
+torch::Tensor column_moments(std::vector<torch::Tensor> cols) {
+  TORCH_CHECK(!cols.empty(), "no columns");
+  auto device = cols[0].device();
+  auto out = torch::zeros({(int64_t)cols.size(), 8},
+                          torch::TensorOptions().dtype(torch::kFloat64).device(device));
+  // group by dtype, one fused launch per dtype
+  for (int pass = 0; pass < 2; ++pass) {
+    std::vector<int64_t> ptrs, lens, idx;
+    for (size_t i = 0; i < cols.size(); ++i) {
+      auto &t = cols[i];
+      TORCH_CHECK(t.is_contiguous() && t.device() == device, "columns must be contiguous, same device");
+      if (dtype_code(t) != pass) continue;
+      ptrs.push_back((int64_t)t.data_ptr());
+      lens.push_back(t.numel());
+      idx.push_back((int64_t)i);
+    }
+    if (ptrs.empty()) continue;
+    int ncols = (int)ptrs.size();
+    int64_t maxn = *std::max_element(lens.begin(), lens.end());
+    int nchunks = pick_chunks(maxn, ncols);
+    auto dptr = to_device_i64(ptrs, device);
+    auto dlen = to_device_i64(lens, device);
+    auto partials = torch::empty({(int64_t)ncols * nchunks, 8},
+                                 torch::TensorOptions().dtype(torch::kFloat64).device(device));
+    auto sub = torch::empty({ncols, 8}, torch::TensorOptions().dtype(torch::kFloat64).device(device));
+    check_hip(anovos_moments((const void *const *)dptr.data_ptr<int64_t>(),
+                             dlen.data_ptr<int64_t>(), ncols, nchunks, pass,
+                             partials.data_ptr<double>(), sub.data_ptr<double>(),
+                             current_stream()),
+              "anovos_moments");
+    auto didx = to_device_i64(idx, device);
+    out.index_copy_(0, didx, sub);
+  }
+  return out;
+}
+
+torch::Tensor column_histograms(std::vector<torch::Tensor> cols, torch::Tensor lo,
+                                torch::Tensor hi, int64_t nbins) {
+  TORCH_CHECK(!cols.empty(), "no columns");
+  auto device = cols[0].device();
+  auto lo_d = lo.to(torch::kFloat64).to(device);
+  auto hi_d = hi.to(torch::kFloat64).to(device);
+  auto out = torch::zeros({(int64_t)cols.size(), nbins},
+                          torch::TensorOptions().dtype(torch::kInt64).device(device));
+  for (int pass = 0; pass < 2; ++pass) {
+    std::vector<int64_t> ptrs, lens, idx;
+    for (size_t i = 0; i < cols.size(); ++i) {
+      if (dtype_code(cols[i]) != pass) continue;
+      ptrs.push_back((int64_t)cols[i].data_ptr());
+      lens.push_back(cols[i].numel());
+      idx.push_back((int64_t)i);
+    }
+    if (ptrs.empty()) continue;
+    int ncols = (int)ptrs.size();
+    int64_t maxn = *std::max_element(lens.begin(), lens.end());
+    int nchunks = pick_chunks(maxn, ncols);
+    auto dptr = to_device_i64(ptrs, device);
+    auto dlen = to_device_i64(lens, device);
+    auto didx_cpu = idx;
+    auto sub = torch::zeros({ncols, nbins}, torch::TensorOptions().dtype(torch::kInt64).device(device));
+    auto lo_sel = lo_d.index_select(0, to_device_i64(didx_cpu, device));
+    auto hi_sel = hi_d.index_select(0, to_device_i64(didx_cpu, device));
+    check_hip(anovos_hist((const void *const *)dptr.data_ptr<int64_t>(),
+                          dlen.data_ptr<int64_t>(), ncols, lo_sel.data_ptr<double>(),
+                          hi_sel.data_ptr<double>(), (int)nbins, nchunks, pass,
+                          (uint64_t *)sub.data_ptr<int64_t>(), current_stream()),
+              "anovos_hist");
+    out.index_copy_(0, to_device_i64(didx_cpu, device), sub);
+  }
+  return out;
+}
+
+torch::Tensor bracket_histograms(std::vector<torch::Tensor> cols, torch::Tensor colidx,
+                                 torch::Tensor lo, torch::Tensor hi, int64_t nbins) {
+  TORCH_CHECK(!cols.empty(), "no columns");
+  auto device = cols[0].device();
+  int dtype = dtype_code(cols[0]);
+  for (auto &t : cols) TORCH_CHECK(dtype_code(t) == dtype, "bracket_histograms: mixed dtypes unsupported");
+  std::vector<int64_t> ptrs, lens;
+  for (auto &t : cols) {
+    ptrs.push_back((int64_t)t.data_ptr());
+    lens.push_back(t.numel());
+  }
+  auto dptr = to_device_i64(ptrs, device);
+  auto dlen = to_device_i64(lens, device);
+  auto ci = colidx.to(torch::kInt64).to(device);
+  auto lo_d = lo.to(torch::kFloat64).to(device);
+  auto hi_d = hi.to(torch::kFloat64).to(device);
+  int nbrackets = (int)ci.numel();
+  int64_t maxn = *std::max_element(lens.begin(), lens.end());
+  int nchunks = pick_chunks(maxn, nbrackets);
+  auto out = torch::zeros({nbrackets, nbins}, torch::TensorOptions().dtype(torch::kInt64).device(device));
+  check_hip(anovos_bracket_hist((const void *const *)dptr.data_ptr<int64_t>(),
+                                dlen.data_ptr<int64_t>(), ci.data_ptr<int64_t>(),
+                                nbrackets, lo_d.data_ptr<double>(), hi_d.data_ptr<double>(),
+                                (int)nbins, nchunks, dtype,
+                                (uint64_t *)out.data_ptr<int64_t>(), current_stream()),
+            "anovos_bracket_hist");
+  return out;
+}
+
+std::vector<torch::Tensor> bucketize_columns(std::vector<torch::Tensor> cols,
+                                             std::vector<torch::Tensor> cutoffs) {
+  TORCH_CHECK(cols.size() == cutoffs.size(), "cols/cutoffs size mismatch");
+  auto device = cols[0].device();
+  std::vector<torch::Tensor> outs;
+  for (auto &t : cols)
+    outs.push_back(torch::empty_like(t, torch::TensorOptions().dtype(torch::kInt32).device(device)));
+  for (int pass = 0; pass < 2; ++pass) {
+    std::vector<int64_t> ptrs, lens, optrs, offs;
+    std::vector<double> flat;
+    std::vector<int> clens;
+    int max_ncut = 1;
+    for (size_t i = 0; i < cols.size(); ++i) {
+      if (dtype_code(cols[i]) != pass) continue;
+      ptrs.push_back((int64_t)cols[i].data_ptr());
+      lens.push_back(cols[i].numel());
+      optrs.push_back((int64_t)outs[i].data_ptr());
+      auto cc = cutoffs[i].to(torch::kFloat64).cpu().contiguous();
+      offs.push_back((int64_t)flat.size());
+      const double *cd = cc.data_ptr<double>();
+      flat.insert(flat.end(), cd, cd + cc.numel());
+      clens.push_back((int)cc.numel());
+      max_ncut = std::max(max_ncut, (int)cc.numel());
+    }
+    if (ptrs.empty()) continue;
+    int ncols = (int)ptrs.size();
+    int64_t maxn = *std::max_element(lens.begin(), lens.end());
+    int nchunks = pick_chunks(maxn, ncols);
+    auto dptr = to_device_i64(ptrs, device);
+    auto dlen = to_device_i64(lens, device);
+    auto dout = to_device_i64(optrs, device);
+    auto doff = to_device_i64(offs, device);
+    auto dflat = torch::from_blob(flat.data(), {(int64_t)std::max<size_t>(flat.size(), 1)},
+                                  torch::TensorOptions().dtype(torch::kFloat64))
+                     .clone()
+                     .to(device);
+    auto dclen = torch::from_blob(clens.data(), {(int64_t)clens.size()},
+                                  torch::TensorOptions().dtype(torch::kInt32))
+                     .clone()
+                     .to(device);
+    check_hip(anovos_bucketize((const void *const *)dptr.data_ptr<int64_t>(),
+                               dlen.data_ptr<int64_t>(), ncols, dflat.data_ptr<double>(),
+                               doff.data_ptr<int64_t>(), dclen.data_ptr<int>(), max_ncut,
+                               nchunks, pass, (int32_t *const *)dout.data_ptr<int64_t>(),
+                               current_stream()),
+              "anovos_bucketize");
+  }
+  return outs;
+}
+
+torch::Tensor code_counts(torch::Tensor codes, int64_t size) {
+  TORCH_CHECK(codes.scalar_type() == torch::kInt32, "codes must be int32");
+  auto device = codes.device();
+  auto out = torch::zeros({std::max<int64_t>(size, 1)},
+                          torch::TensorOptions().dtype(torch::kInt64).device(device));
+  if (size == 0 || codes.numel() == 0) return out.narrow(0, 0, size);
+  int nchunks = pick_chunks(codes.numel(), 1);
+  nchunks = std::min(nchunks, 8192);
+  check_hip(anovos_code_counts(codes.data_ptr<int32_t>(), codes.numel(), (int)size,
+                               nchunks, (uint64_t *)out.data_ptr<int64_t>(),
+                               current_stream()),
+            "anovos_code_counts");
+  return out;
+}
+
+torch::Tensor hll_registers(torch::Tensor x, int64_t p) {
+  auto device = x.device();
+  auto regs = torch::zeros({(int64_t)1 << p},
+                           torch::TensorOptions().dtype(torch::kInt32).device(device));
+  if (x.numel() == 0) return regs;
+  int nchunks = std::min(pick_chunks(x.numel(), 1), 4096);
+  check_hip(anovos_hll(x.data_ptr(), x.numel(), (int)p, nchunks, dtype_code(x),
+                       regs.data_ptr<int32_t>(), current_stream()),
+            "anovos_hll");
+  return regs;
+}
+
+void row_null_counts_num(std::vector<torch::Tensor> cols, torch::Tensor out) {
+  TORCH_CHECK(!cols.empty(), "no columns");
+  TORCH_CHECK(out.scalar_type() == torch::kInt32, "out must be int32");
+  auto device = cols[0].device();
+  for (int pass = 0; pass < 2; ++pass) {
+    std::vector<int64_t> ptrs;
+    for (auto &t : cols) {
+      if (dtype_code(t) != pass) continue;
+      TORCH_CHECK(t.numel() == out.numel(), "column length mismatch");
+      ptrs.push_back((int64_t)t.data_ptr());
+    }
+    if (ptrs.empty()) continue;
+    auto dptr = to_device_i64(ptrs, device);
+    check_hip(anovos_row_null((const void *const *)dptr.data_ptr<int64_t>(),
+                              (int)ptrs.size(), out.numel(), pass,
+                              out.data_ptr<int32_t>(), current_stream()),
+              "anovos_row_null");
+  }
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("column_moments", &column_moments, "fused per-column moments (K1/K2)");
+  m.def("column_histograms", &column_histograms, "fused per-column histograms (K3/K6)");
+  m.def("bracket_histograms", &bracket_histograms, "quantile-refinement histograms (K3)");
+  m.def("bucketize_columns", &bucketize_columns, "branchless bucketize (K6)");
+  m.def("code_counts", &code_counts, "dictionary code bincount (K5)");
+  m.def("hll_registers", &hll_registers, "HyperLogLog registers (K4)");
+  m.def("row_null_counts_num", &row_null_counts_num, "fused row null counts (K10)");
+}

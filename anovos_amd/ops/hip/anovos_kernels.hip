@@ -1,0 +1,457 @@
+// anovos_amd HIP kernels for MI355X (gfx950, CDNA4).
+//
+// Kernel inventory:
+//   K1/K2  column_moments      — fused per-column count/sum/sum2/sum3/sum4/
+//                                min/max/zeros, fp64 accumulators, partials
+//                                per (col, chunk) block + deterministic
+//                                second-stage reduce (no atomics).
+//   K3/K6  column_histograms   — equal-width histograms, LDS-staged bins.
+//          bracket_histograms  — quantile-refinement histograms.
+//          bucketize_columns   — branchless binary search over LDS cutoffs.
+//   K4     hll_registers       — HyperLogLog p=14, LDS-staged registers.
+//   K5     code_counts         — dictionary-code bincount, LDS-staged.
+//   K10    row_null_counts     — fused row-wise NaN count across columns.
+//
+// Design notes (cdna_hip_programming.md): wave64; blocks of 256 threads;
+// grids sized ncols x chunks >> 256 workgroups to fill 8 XCDs; fp32 loads
+// vectorized as float4 (16B/lane); all cross-block merging goes through a
+// partials buffer + second kernel so results are deterministic.
+
+#include <hip/hip_runtime.h>
+#include <cfloat>
+#include <cstdint>
+#include <cmath>
+
+#define THREADS 256
+#define DEV_INLINE __device__ __forceinline__
+
+// ------------------------------------------------------------------
+// K1/K2: fused column moments
+// ------------------------------------------------------------------
+// partials layout: [ncols * nchunks][8] doubles:
+//   0:n 1:s1 2:s2 3:s3 4:s4 5:min 6:max 7:zeros
+
+struct MomAcc {
+  double n, s1, s2, s3, s4, mn, mx, zn;
+};
+
+DEV_INLINE void mom_add(MomAcc &a, double v) {
+  if (!isnan(v)) {
+    a.n += 1.0;
+    a.s1 += v;
+    double v2 = v * v;
+    a.s2 += v2;
+    a.s3 += v2 * v;
+    a.s4 += v2 * v2;
+    a.mn = fmin(a.mn, v);
+    a.mx = fmax(a.mx, v);
+    a.zn += (v == 0.0) ? 1.0 : 0.0;
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(THREADS) void moments_partials_kernel(
+    const T *const *cols, const int64_t *lens, int nchunks, double *partials) {
+  const int col = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  const T *__restrict__ x = cols[col];
+  const int64_t n = lens[col];
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+
+  MomAcc a{0, 0, 0, 0, 0, DBL_MAX, -DBL_MAX, 0};
+
+  if (sizeof(T) == 4) {
+    // vectorized float4 path: 16 B per lane per iteration
+    const int64_t nv = (e - s) / 4;
+    const float4 *xv = reinterpret_cast<const float4 *>(x + s);
+    for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
+      float4 v = xv[i];
+      mom_add(a, (double)v.x);
+      mom_add(a, (double)v.y);
+      mom_add(a, (double)v.z);
+      mom_add(a, (double)v.w);
+    }
+    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) mom_add(a, (double)x[i]);
+  } else {
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) mom_add(a, (double)x[i]);
+  }
+
+  // block reduce through LDS (8 doubles per thread -> tree)
+  __shared__ double sm[THREADS * 8];
+  double *mine = &sm[threadIdx.x * 8];
+  mine[0] = a.n; mine[1] = a.s1; mine[2] = a.s2; mine[3] = a.s3;
+  mine[4] = a.s4; mine[5] = a.mn; mine[6] = a.mx; mine[7] = a.zn;
+  __syncthreads();
+  for (int stride = THREADS / 2; stride > 0; stride >>= 1) {
+    if (threadIdx.x < stride) {
+      double *other = &sm[(threadIdx.x + stride) * 8];
+      mine[0] += other[0]; mine[1] += other[1]; mine[2] += other[2];
+      mine[3] += other[3]; mine[4] += other[4];
+      mine[5] = fmin(mine[5], other[5]);
+      mine[6] = fmax(mine[6], other[6]);
+      mine[7] += other[7];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    double *out = &partials[(int64_t)blockIdx.x * 8];
+    for (int k = 0; k < 8; ++k) out[k] = sm[k];
+  }
+}
+
+__global__ __launch_bounds__(THREADS) void moments_reduce_kernel(
+    const double *partials, int nchunks, double *out) {
+  const int col = blockIdx.x;
+  MomAcc a{0, 0, 0, 0, 0, DBL_MAX, -DBL_MAX, 0};
+  for (int c = threadIdx.x; c < nchunks; c += THREADS) {
+    const double *p = &partials[((int64_t)col * nchunks + c) * 8];
+    a.n += p[0]; a.s1 += p[1]; a.s2 += p[2]; a.s3 += p[3]; a.s4 += p[4];
+    a.mn = fmin(a.mn, p[5]); a.mx = fmax(a.mx, p[6]); a.zn += p[7];
+  }
+  __shared__ double sm[THREADS * 8];
+  double *mine = &sm[threadIdx.x * 8];
+  mine[0] = a.n; mine[1] = a.s1; mine[2] = a.s2; mine[3] = a.s3;
+  mine[4] = a.s4; mine[5] = a.mn; mine[6] = a.mx; mine[7] = a.zn;
+  __syncthreads();
+  for (int stride = THREADS / 2; stride > 0; stride >>= 1) {
+    if (threadIdx.x < stride) {
+      double *other = &sm[(threadIdx.x + stride) * 8];
+      mine[0] += other[0]; mine[1] += other[1]; mine[2] += other[2];
+      mine[3] += other[3]; mine[4] += other[4];
+      mine[5] = fmin(mine[5], other[5]);
+      mine[6] = fmax(mine[6], other[6]);
+      mine[7] += other[7];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    double *o = &out[(int64_t)col * 8];
+    o[0] = sm[0]; o[1] = sm[1]; o[2] = sm[2]; o[3] = sm[3]; o[4] = sm[4];
+    o[5] = (sm[0] > 0) ? sm[5] : nan("");
+    o[6] = (sm[0] > 0) ? sm[6] : nan("");
+    o[7] = sm[7];
+  }
+}
+
+// ------------------------------------------------------------------
+// K3/K6: histograms (LDS bins, atomic flush to global)
+// ------------------------------------------------------------------
+
+template <typename T>
+__global__ __launch_bounds__(THREADS) void hist_kernel(
+    const T *const *cols, const int64_t *lens, const double *lo, const double *hi,
+    int nbins, int nchunks, uint64_t *out /*[ncols][nbins]*/) {
+  extern __shared__ uint32_t bins[];  // nbins
+  const int col = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  for (int b = threadIdx.x; b < nbins; b += THREADS) bins[b] = 0;
+  __syncthreads();
+
+  const T *__restrict__ x = cols[col];
+  const int64_t n = lens[col];
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+  const double l = lo[col];
+  const double h = hi[col];
+  if (h > l) {
+    const double scale = (double)nbins / (h - l);
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+      double v = (double)x[i];
+      if (!isnan(v)) {
+        int b = (int)((v - l) * scale);
+        b = max(0, min(nbins - 1, b));
+        atomicAdd(&bins[b], 1u);
+      }
+    }
+  } else {
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+      double v = (double)x[i];
+      if (!isnan(v)) atomicAdd(&bins[0], 1u);
+    }
+  }
+  __syncthreads();
+  uint64_t *g = &out[(int64_t)col * nbins];
+  for (int b = threadIdx.x; b < nbins; b += THREADS)
+    if (bins[b]) atomicAdd((unsigned long long *)&g[b], (unsigned long long)bins[b]);
+}
+
+// bracket histograms: like hist but per-(bracket) with col indirection and
+// values outside [lo, hi) skipped.
+template <typename T>
+__global__ __launch_bounds__(THREADS) void bracket_hist_kernel(
+    const T *const *cols, const int64_t *lens, const int64_t *colidx,
+    const double *lo, const double *hi, int nbins, int nchunks, uint64_t *out) {
+  extern __shared__ uint32_t bins[];
+  const int bracket = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  for (int b = threadIdx.x; b < nbins; b += THREADS) bins[b] = 0;
+  __syncthreads();
+  const int col = (int)colidx[bracket];
+  const T *__restrict__ x = cols[col];
+  const int64_t n = lens[col];
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+  const double l = lo[bracket];
+  const double h = hi[bracket];
+  if (h > l) {
+    const double scale = (double)nbins / (h - l);
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+      double v = (double)x[i];
+      if (!isnan(v) && v >= l && v < h) {
+        int b = (int)((v - l) * scale);
+        b = max(0, min(nbins - 1, b));
+        atomicAdd(&bins[b], 1u);
+      }
+    }
+  }
+  __syncthreads();
+  uint64_t *g = &out[(int64_t)bracket * nbins];
+  for (int b = threadIdx.x; b < nbins; b += THREADS)
+    if (bins[b]) atomicAdd((unsigned long long *)&g[b], (unsigned long long)bins[b]);
+}
+
+// ------------------------------------------------------------------
+// bucketize: branchless binary search over per-column cutoffs in LDS
+// ------------------------------------------------------------------
+
+template <typename T>
+__global__ __launch_bounds__(THREADS) void bucketize_kernel(
+    const T *const *cols, const int64_t *lens,
+    const double *cutflat, const int64_t *cutoff_off, const int *cutoff_len,
+    int nchunks, int32_t *const *outs) {
+  extern __shared__ double cuts[];
+  const int col = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  const int ncut = cutoff_len[col];
+  const double *src = &cutflat[cutoff_off[col]];
+  for (int i = threadIdx.x; i < ncut; i += THREADS) cuts[i] = src[i];
+  __syncthreads();
+
+  const T *__restrict__ x = cols[col];
+  int32_t *__restrict__ out = outs[col];
+  const int64_t n = lens[col];
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+    double v = (double)x[i];
+    if (isnan(v)) {
+      out[i] = -1;
+      continue;
+    }
+    // first index with cuts[idx] >= v  (torch.bucketize right=False)
+    int lo = 0, len = ncut;
+    while (len > 0) {
+      int half = len >> 1;
+      int mid = lo + half;
+      // branchless: move lo past mid when cuts[mid] < v
+      lo = (cuts[mid] < v) ? (mid + 1) : lo;
+      len = (cuts[mid] < v) ? (len - half - 1) : half;
+    }
+    out[i] = lo;
+  }
+}
+
+// ------------------------------------------------------------------
+// K5: dictionary code bincount
+// ------------------------------------------------------------------
+
+__global__ __launch_bounds__(THREADS) void code_counts_kernel(
+    const int32_t *codes, int64_t n, int size, int nchunks, uint64_t *out) {
+  extern __shared__ uint32_t cnt[];
+  const bool use_lds = (size <= 16384);
+  if (use_lds) {
+    for (int i = threadIdx.x; i < size; i += THREADS) cnt[i] = 0;
+    __syncthreads();
+  }
+  const int chunk = blockIdx.x;
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+  if (use_lds) {
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+      int c = codes[i];
+      if (c >= 0 && c < size) atomicAdd(&cnt[c], 1u);
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < size; i += THREADS)
+      if (cnt[i]) atomicAdd((unsigned long long *)&out[i], (unsigned long long)cnt[i]);
+  } else {
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+      int c = codes[i];
+      if (c >= 0 && c < size) atomicAdd((unsigned long long *)&out[c], 1ull);
+    }
+  }
+}
+
+// ------------------------------------------------------------------
+// K4: HyperLogLog registers (p fixed by host, registers int32)
+// ------------------------------------------------------------------
+
+DEV_INLINE uint64_t splitmix64(uint64_t x) {
+  x ^= x >> 30; x *= 0xBF58476D1CE4E5B9ull;
+  x ^= x >> 27; x *= 0x94D049BB133111EBull;
+  x ^= x >> 31;
+  return x;
+}
+
+template <typename T>
+__global__ __launch_bounds__(THREADS) void hll_kernel(
+    const T *x, int64_t n, int p, int nchunks, int32_t *regs /*[1<<p]*/) {
+  extern __shared__ int32_t sreg[];  // 1<<p
+  const int m = 1 << p;
+  for (int i = threadIdx.x; i < m; i += THREADS) sreg[i] = 0;
+  __syncthreads();
+  const int chunk = blockIdx.x;
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+    double v = (double)x[i];
+    if (isnan(v)) continue;
+    uint64_t bits;
+    memcpy(&bits, &v, 8);
+    uint64_t h = splitmix64(bits);
+    int idx = (int)(h >> (64 - p));
+    uint64_t rem = h << p;
+    int rho = (rem == 0) ? (64 - p + 1) : (__clzll((long long)rem) + 1);
+    if (rho > 64 - p + 1) rho = 64 - p + 1;
+    atomicMax(&sreg[idx], rho);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < m; i += THREADS)
+    if (sreg[i]) atomicMax(&regs[i], sreg[i]);
+}
+
+// ------------------------------------------------------------------
+// K10: fused row-wise NaN count over numeric columns
+// ------------------------------------------------------------------
+
+template <typename T>
+__global__ __launch_bounds__(THREADS) void row_null_kernel(
+    const T *const *cols, int ncols, int64_t n, int64_t rows_per_block, int32_t *out) {
+  extern __shared__ uint16_t rc[];  // rows_per_block
+  const int64_t s = blockIdx.x * rows_per_block;
+  const int64_t e = min(n, s + rows_per_block);
+  const int nr = (int)(e - s);
+  for (int i = threadIdx.x; i < nr; i += THREADS) rc[i] = 0;
+  __syncthreads();
+  for (int c = 0; c < ncols; ++c) {
+    const T *__restrict__ x = cols[c] + s;
+    for (int i = threadIdx.x; i < nr; i += THREADS) {
+      // lane i handles row i (+k*256): LDS bank = i%32 distinct per lane group
+      if (isnan((double)x[i])) rc[i] += 1;  // no atomics: one writer per row
+    }
+    __syncthreads();
+  }
+  for (int i = threadIdx.x; i < nr; i += THREADS) out[s + i] += (int32_t)rc[i];
+}
+
+// ------------------------------------------------------------------
+// C API (host launchers). Streams come from the caller (torch stream).
+// ------------------------------------------------------------------
+
+#define LAUNCH_OK 0
+
+extern "C" {
+
+int anovos_moments(const void *const *cols, const int64_t *lens, int ncols,
+                   int nchunks, int dtype /*0=f32 1=f64*/, double *partials,
+                   double *out, hipStream_t stream) {
+  dim3 grid(ncols * nchunks);
+  if (dtype == 0)
+    hipLaunchKernelGGL(moments_partials_kernel<float>, grid, dim3(THREADS), 0, stream,
+                       (const float *const *)cols, lens, nchunks, partials);
+  else
+    hipLaunchKernelGGL(moments_partials_kernel<double>, grid, dim3(THREADS), 0, stream,
+                       (const double *const *)cols, lens, nchunks, partials);
+  hipLaunchKernelGGL(moments_reduce_kernel, dim3(ncols), dim3(THREADS), 0, stream,
+                     partials, nchunks, out);
+  return (int)hipGetLastError();
+}
+
+int anovos_hist(const void *const *cols, const int64_t *lens, int ncols,
+                const double *lo, const double *hi, int nbins, int nchunks,
+                int dtype, uint64_t *out, hipStream_t stream) {
+  dim3 grid(ncols * nchunks);
+  size_t lds = (size_t)nbins * 4;
+  if (dtype == 0)
+    hipLaunchKernelGGL(hist_kernel<float>, grid, dim3(THREADS), lds, stream,
+                       (const float *const *)cols, lens, lo, hi, nbins, nchunks, out);
+  else
+    hipLaunchKernelGGL(hist_kernel<double>, grid, dim3(THREADS), lds, stream,
+                       (const double *const *)cols, lens, lo, hi, nbins, nchunks, out);
+  return (int)hipGetLastError();
+}
+
+int anovos_bracket_hist(const void *const *cols, const int64_t *lens,
+                        const int64_t *colidx, int nbrackets, const double *lo,
+                        const double *hi, int nbins, int nchunks, int dtype,
+                        uint64_t *out, hipStream_t stream) {
+  dim3 grid(nbrackets * nchunks);
+  size_t lds = (size_t)nbins * 4;
+  if (dtype == 0)
+    hipLaunchKernelGGL(bracket_hist_kernel<float>, grid, dim3(THREADS), lds, stream,
+                       (const float *const *)cols, lens, colidx, lo, hi, nbins, nchunks, out);
+  else
+    hipLaunchKernelGGL(bracket_hist_kernel<double>, grid, dim3(THREADS), lds, stream,
+                       (const double *const *)cols, lens, colidx, lo, hi, nbins, nchunks, out);
+  return (int)hipGetLastError();
+}
+
+int anovos_bucketize(const void *const *cols, const int64_t *lens, int ncols,
+                     const double *cutflat, const int64_t *cutoff_off,
+                     const int *cutoff_len, int max_ncut, int nchunks, int dtype,
+                     int32_t *const *outs, hipStream_t stream) {
+  dim3 grid(ncols * nchunks);
+  size_t lds = (size_t)max_ncut * 8;
+  if (lds < 8) lds = 8;
+  if (dtype == 0)
+    hipLaunchKernelGGL(bucketize_kernel<float>, grid, dim3(THREADS), lds, stream,
+                       (const float *const *)cols, lens, cutflat, cutoff_off, cutoff_len, nchunks, outs);
+  else
+    hipLaunchKernelGGL(bucketize_kernel<double>, grid, dim3(THREADS), lds, stream,
+                       (const double *const *)cols, lens, cutflat, cutoff_off, cutoff_len, nchunks, outs);
+  return (int)hipGetLastError();
+}
+
+int anovos_code_counts(const int32_t *codes, int64_t n, int size, int nchunks,
+                       uint64_t *out, hipStream_t stream) {
+  size_t lds = (size <= 16384) ? (size_t)size * 4 : 0;
+  hipLaunchKernelGGL(code_counts_kernel, dim3(nchunks), dim3(THREADS), lds, stream,
+                     codes, n, size, nchunks, out);
+  return (int)hipGetLastError();
+}
+
+int anovos_hll(const void *x, int64_t n, int p, int nchunks, int dtype,
+               int32_t *regs, hipStream_t stream) {
+  size_t lds = (size_t)(1 << p) * 4;
+  if (dtype == 0)
+    hipLaunchKernelGGL(hll_kernel<float>, dim3(nchunks), dim3(THREADS), lds, stream,
+                       (const float *)x, n, p, nchunks, regs);
+  else
+    hipLaunchKernelGGL(hll_kernel<double>, dim3(nchunks), dim3(THREADS), lds, stream,
+                       (const double *)x, n, p, nchunks, regs);
+  return (int)hipGetLastError();
+}
+
+int anovos_row_null(const void *const *cols, int ncols, int64_t n, int dtype,
+                    int32_t *out, hipStream_t stream) {
+  const int64_t rows_per_block = 32768;  // 64 KiB LDS of uint16 counters
+  int64_t nblocks = (n + rows_per_block - 1) / rows_per_block;
+  if (nblocks == 0) return 0;
+  size_t lds = (size_t)rows_per_block * 2;
+  if (dtype == 0)
+    hipLaunchKernelGGL(row_null_kernel<float>, dim3((uint32_t)nblocks), dim3(THREADS), lds, stream,
+                       (const float *const *)cols, ncols, n, rows_per_block, out);
+  else
+    hipLaunchKernelGGL(row_null_kernel<double>, dim3((uint32_t)nblocks), dim3(THREADS), lds, stream,
+                       (const double *const *)cols, ncols, n, rows_per_block, out);
+  return (int)hipGetLastError();
+}
+
+}  // extern "C"

@@ -119,14 +119,21 @@ def approx_quantiles(
     # so below this size we sort-and-index instead of sketching.
     if all(moments[c].n <= EXACT_N_THRESHOLD for c in cols):
         return _exact_quantiles(idf, cols, probs, moments, rel_err=rel_err)
-    tensors = [idf.col(c).data for c in cols]
     dev = idf.device
     lo = torch.tensor([moments[c].min for c in cols], dtype=torch.float64)
     hi = torch.tensor([moments[c].max for c in cols], dtype=torch.float64)
-    n = torch.tensor([moments[c].n for c in cols], dtype=torch.float64)
     probs = list(probs)
 
-    hist = global_histograms(tensors, lo, hi, nbins).cpu().to(torch.float64)
+    # pass-1 histograms are cached per column (range is always the global
+    # min/max, so the CDF is reusable across quantile requests)
+    key = ("hist", nbins)
+    todo = [i for i, c in enumerate(cols) if key not in idf.col(c).cache]
+    if todo:
+        tensors = [idf.col(cols[i]).data for i in todo]
+        h = global_histograms(tensors, lo[todo], hi[todo], nbins).cpu()
+        for k, i in enumerate(todo):
+            idf.col(cols[i]).cache[key] = h[k]
+    hist = torch.stack([idf.col(c).cache[key] for c in cols]).to(torch.float64)
     cdf = torch.cumsum(hist, dim=1)
 
     result = {c: [float("nan")] * len(probs) for c in cols}

@@ -129,19 +129,48 @@ class MomentStats:
 
 
 def frame_moments(idf, cols: List[str]) -> Dict[str, MomentStats]:
-    """Global moment stats for the given numeric columns of a frame."""
-    tensors = [idf.col(c).data for c in cols]
-    local = column_moments_local(tensors)
-    glob = merge_moments_global(local)
-    return {c: MomentStats(glob[i]) for i, c in enumerate(cols)}
+    """Global moment stats for the given numeric columns of a frame.
+    Cached per Column (the reference's stats-reuse wiring, workflow.py:91-145);
+    benchmarks clear the cache per step to keep timing honest."""
+    out: Dict[str, MomentStats] = {}
+    todo = []
+    for c in cols:
+        m = idf.col(c).cache.get("moments")
+        if m is not None:
+            out[c] = m
+        else:
+            todo.append(c)
+    if todo:
+        tensors = [idf.col(c).data for c in todo]
+        local = column_moments_local(tensors)
+        glob = merge_moments_global(local)
+        for i, c in enumerate(todo):
+            m = MomentStats(glob[i])
+            idf.col(c).cache["moments"] = m
+            out[c] = m
+    return out
 
 
 def null_counts(idf, cols: List[str]) -> Dict[str, int]:
-    """Per-column global null counts (any kind), one fused all-reduce."""
-    dev = idf.device
-    local = torch.empty(len(cols), dtype=torch.float64, device=dev)
-    for i, c in enumerate(cols):
-        local[i] = idf.col(c).null_mask().sum()
-    dist.all_reduce_(local, "sum")
+    """Per-column global null counts (any kind). Numeric columns derive
+    the count from the fused moments pass (total - n); categorical and
+    other columns use one small reduction each, all-reduced in a batch."""
     total = idf.count()
-    return {c: int(local[i].item()) for i, c in enumerate(cols)}, total
+    res: Dict[str, int] = {}
+    num_cols = [c for c in cols if idf.col(c).kind == "numerical" and "nulls" not in idf.col(c).cache]
+    if num_cols:
+        moments = frame_moments(idf, num_cols)
+        for c in num_cols:
+            idf.col(c).cache["nulls"] = int(total - moments[c].n)
+    other = [c for c in cols if "nulls" not in idf.col(c).cache]
+    if other:
+        dev = idf.device
+        local = torch.empty(len(other), dtype=torch.float64, device=dev)
+        for i, c in enumerate(other):
+            local[i] = idf.col(c).null_mask().sum()
+        dist.all_reduce_(local, "sum")
+        for i, c in enumerate(other):
+            idf.col(c).cache["nulls"] = int(local[i].item())
+    for c in cols:
+        res[c] = idf.col(c).cache["nulls"]
+    return res, total
