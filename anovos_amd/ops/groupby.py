@@ -170,11 +170,17 @@ def _str_hash64(s: str) -> int:
     return h - 0x10000000000000000 if h >= 0x8000000000000000 else h
 
 
+def _lshr(x: torch.Tensor, n: int) -> torch.Tensor:
+    """Logical (zero-fill) right shift on int64 (torch >> is arithmetic)."""
+    return (x >> n) & ((1 << (64 - n)) - 1)
+
+
 def _mix64(x: torch.Tensor) -> torch.Tensor:
-    # splitmix64 finalizer (wrapping int64 arithmetic)
-    x = x ^ (x >> 30)
+    # splitmix64 finalizer (wrapping int64 arithmetic, LOGICAL right shifts
+    # to match the uint64 device kernel bit-for-bit)
+    x = x ^ _lshr(x, 30)
     x = x * -0x40A7B892E31B1A47  # 0xBF58476D1CE4E5B9
-    x = x ^ (x >> 27)
+    x = x ^ _lshr(x, 27)
     x = x * -0x6B2FB644ECCEEE15  # 0x94D049BB133111EB
-    x = x ^ (x >> 31)
+    x = x ^ _lshr(x, 31)
     return x

@@ -135,6 +135,7 @@ def approx_quantiles(
             idf.col(cols[i]).cache[key] = h[k]
     hist = torch.stack([idf.col(c).cache[key] for c in cols]).to(torch.float64)
     cdf = torch.cumsum(hist, dim=1)
+    tensors = [idf.col(c).data for c in cols]
 
     result = {c: [float("nan")] * len(probs) for c in cols}
     # bracket per (col, prob): bin containing target rank
