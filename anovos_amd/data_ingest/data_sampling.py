@@ -112,8 +112,8 @@ def data_sample(
         counts = local_counts
     smallest = float(counts.min())
     frac_per = (fraction * smallest / counts).clamp(max=1.0)
-    g = torch.Generator(device="cpu")
+    g = torch.Generator(device=base.device)
     g.manual_seed(int(seed_value) * 1000003 + dist.rank())
-    u = torch.rand(base.local_rows(), generator=g).to(base.device)
+    u = torch.rand(base.local_rows(), generator=g, device=base.device)
     keep = u < frac_per[inv]
     return base.filter_rows(keep)

@@ -414,10 +414,16 @@ def z_standardization(
         )
     odf = idf
     applied = [c for c in list_of_cols if c not in excluded]
-    for c in applied:
-        mu, sd = params[c]
-        data = (idf.col(c).data.to(torch.float32) - mu) / sd
-        odf = odf.with_column(c + "_scaled", Column(c + "_scaled", "double", data))
+    if applied:
+        from anovos_amd.ops import elementwise
+
+        scaled = elementwise.scale_columns(
+            [idf.col(c).data for c in applied],
+            [params[c][0] for c in applied],
+            [1.0 / params[c][1] for c in applied],
+        )
+        for c, data in zip(applied, scaled):
+            odf = odf.with_column(c + "_scaled", Column(c + "_scaled", "double", data))
     odf = _finish_output(idf, odf, applied, "_scaled", output_mode)
     if print_impact:
         print(odf.columns)
@@ -463,10 +469,16 @@ def IQR_standardization(
         warnings.warn("The following column(s) are excluded from standardization because IQR is zero:" + str(excluded))
     odf = idf
     applied = [c for c in list_of_cols if c not in excluded]
-    for c in applied:
-        p25, p50, p75 = params[c]
-        data = (idf.col(c).data.to(torch.float32) - p50) / (p75 - p25)
-        odf = odf.with_column(c + "_scaled", Column(c + "_scaled", "double", data))
+    if applied:
+        from anovos_amd.ops import elementwise
+
+        scaled = elementwise.scale_columns(
+            [idf.col(c).data for c in applied],
+            [params[c][1] for c in applied],
+            [1.0 / (params[c][2] - params[c][0]) for c in applied],
+        )
+        for c, data in zip(applied, scaled):
+            odf = odf.with_column(c + "_scaled", Column(c + "_scaled", "double", data))
     odf = _finish_output(idf, odf, applied, "_scaled", output_mode)
     if print_impact:
         print(odf.columns)
@@ -507,10 +519,16 @@ def normalization(
         warnings.warn("The following column(s) are excluded from normalization (constant or empty):" + str(excluded))
     odf = idf
     applied = [c for c in list_of_cols if c not in excluded]
-    for c in applied:
-        mn, mx = params[c]
-        data = (idf.col(c).data.to(torch.float32) - mn) / (mx - mn)
-        odf = odf.with_column(c + "_scaled", Column(c + "_scaled", "double", data))
+    if applied:
+        from anovos_amd.ops import elementwise
+
+        scaled = elementwise.scale_columns(
+            [idf.col(c).data for c in applied],
+            [params[c][0] for c in applied],
+            [1.0 / (params[c][1] - params[c][0]) for c in applied],
+        )
+        for c, data in zip(applied, scaled):
+            odf = odf.with_column(c + "_scaled", Column(c + "_scaled", "double", data))
     odf = _finish_output(idf, odf, applied, "_scaled", output_mode)
     if print_impact:
         print(odf.columns)
@@ -621,14 +639,20 @@ def imputation_MMM(
                 )
 
     odf = idf
+    fillable = [c for c in num_cols if fill_num.get(c) is not None and fill_num[c] == fill_num[c]]
+    if fillable:
+        from anovos_amd.ops import elementwise
+
+        filled = elementwise.fill_nan_columns(
+            [idf.col(c).data for c in fillable], [float(fill_num[c]) for c in fillable]
+        )
+        for c, data in zip(fillable, filled):
+            odf = odf.with_column(c + "_imputed", Column(c + "_imputed", idf.col(c).dtype, data))
     for c in num_cols:
-        v = fill_num.get(c)
+        if c in fillable:
+            continue
         col = idf.col(c)
-        if v is None or v != v:
-            data = col.data.clone()
-        else:
-            data = torch.nan_to_num(col.data, nan=float(v))
-        odf = odf.with_column(c + "_imputed", Column(c + "_imputed", col.dtype, data))
+        odf = odf.with_column(c + "_imputed", Column(c + "_imputed", col.dtype, col.data.clone()))
     for c in cat_cols:
         v = fill_cat.get(c)
         col = idf.col(c)

@@ -98,6 +98,11 @@ def statistics(
         bin_size=bin_size, pre_existing_model=True, model_path=model_path,
     )
 
+    # batched frequency pass: one fused histogram launch covers every
+    # numeric (binned) column of each frame
+    q_freqs = batched_bin_frequencies(target_bin, list_of_cols, count_target)
+    if not pre_existing_source:
+        p_freqs = batched_bin_frequencies(source_bin, list_of_cols, count_source)
     rows = []
     for i in list_of_cols:
         if pre_existing_source:
@@ -105,12 +110,12 @@ def statistics(
             p_keys = [str(k) for k in x[i].tolist()]
             p_vals = x["p"].tolist()
         else:
-            p_keys, p_vals = _bin_frequencies(source_bin, i, count_source)
+            p_keys, p_vals = p_freqs[i]
             if source_save and dist.rank() == 0:
                 d = os.path.join(model_path, "frequency_counts", i)
                 os.makedirs(d, exist_ok=True)
                 pd.DataFrame({i: p_keys, "p": p_vals}).to_csv(os.path.join(d, "part-00000.csv"), index=False)
-        q_keys, q_vals = _bin_frequencies(target_bin, i, count_target)
+        q_keys, q_vals = q_freqs[i]
 
         # full-outer join on bin key, fill 0.0001, order by key
         pmap = dict(zip(p_keys, p_vals))
@@ -150,6 +155,43 @@ def _key_order(k):
         return (0, float(k), "")
     except (TypeError, ValueError):
         return (1, 0.0, str(k))
+
+
+def batched_bin_frequencies(binned_idf, cols, total: int):
+    """Per-bin frequencies for many columns at once: numeric (binned)
+    columns go through ONE fused histogram kernel + one all-reduce;
+    categorical columns through the fused dictionary bincount."""
+    from anovos_amd.ops import histogram as hist_ops
+    from anovos_amd.ops import stats as stats_ops
+
+    out = {}
+    num_cols = [c for c in cols if binned_idf.col(c).kind == "numerical"]
+    cat_cols = [c for c in cols if c not in num_cols]
+    if num_cols:
+        moments = stats_ops.frame_moments(binned_idf, num_cols)
+        M = 1
+        for c in num_cols:
+            if moments[c].max == moments[c].max:
+                M = max(M, int(moments[c].max))
+        tensors = [binned_idf.col(c).data for c in num_cols]
+        lo = torch.full((len(num_cols),), 1.0, dtype=torch.float64)
+        hi = torch.full((len(num_cols),), float(M + 1), dtype=torch.float64)
+        hist = hist_ops.global_histograms(tensors, lo, hi, M).cpu()
+        for i, c in enumerate(num_cols):
+            keys, vals = [], []
+            nnull = int(total - moments[c].n)
+            if nnull:
+                keys.append("-1")
+                vals.append(nnull / total)
+            h = hist[i]
+            for b in range(M):
+                if h[b] > 0:
+                    keys.append(str(b + 1))
+                    vals.append(float(h[b]) / total)
+            out[c] = (keys, vals)
+    for c in cat_cols:
+        out[c] = _bin_frequencies(binned_idf, c, total)
+    return out
 
 
 def _bin_frequencies(binned_idf, col: str, total: int):

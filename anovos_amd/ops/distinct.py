@@ -64,16 +64,21 @@ def hll_estimate(regs: torch.Tensor) -> float:
 
 
 def approx_distinct(idf, cols: List[str]) -> Dict[str, int]:
-    """HLL distinct per numeric column; registers merged via all-reduce(max)."""
+    """HLL distinct per numeric column; one fused multi-column kernel on
+    GPU, registers merged via a single all-reduce(max)."""
     out = {}
-    regs_all = []
-    for c in cols:
-        regs_all.append(hll_registers(idf.col(c).data))
-    if regs_all:
-        flat = torch.stack(regs_all).to(torch.int32)
-        dist.all_reduce_(flat, "max")
-        for i, c in enumerate(cols):
-            out[c] = int(round(hll_estimate(flat[i])))
+    if not cols:
+        return out
+    first = idf.col(cols[0]).data
+    if first.is_cuda and backend.use_hip(first):
+        ext = backend.hip_ext()
+        flat = ext.hll_registers_multi([idf.col(c).data.contiguous() for c in cols], HLL_P)
+    else:
+        flat = torch.stack([hll_registers(idf.col(c).data) for c in cols]).to(torch.int32)
+    dist.all_reduce_(flat, "max")
+    flat = flat.cpu()
+    for i, c in enumerate(cols):
+        out[c] = int(round(hll_estimate(flat[i])))
     return out
 
 

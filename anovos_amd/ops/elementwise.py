@@ -1,0 +1,33 @@
+"""Fused elementwise column ops (kernel K11): (x-a)*b scaling and NaN
+fill across many columns in one launch — the apply halves of
+z/IQR/min-max standardization and MMM imputation."""
+
+from __future__ import annotations
+
+from typing import List, Sequence
+
+import torch
+
+from anovos_amd.ops import backend
+
+
+def scale_columns(tensors: Sequence[torch.Tensor], a: Sequence[float], b: Sequence[float]) -> List[torch.Tensor]:
+    """out_i = (x_i - a_i) * b_i as float32; NaN propagates."""
+    if tensors and tensors[0].is_cuda and backend.use_hip(tensors[0]):
+        ext = backend.hip_ext()
+        return ext.scale_columns(
+            [t.contiguous() for t in tensors],
+            torch.tensor(list(a), dtype=torch.float64),
+            torch.tensor(list(b), dtype=torch.float64),
+        )
+    return [((t.to(torch.float32) - float(ai)) * float(bi)) for t, ai, bi in zip(tensors, a, b)]
+
+
+def fill_nan_columns(tensors: Sequence[torch.Tensor], fills: Sequence[float]) -> List[torch.Tensor]:
+    """out_i = isnan(x_i) ? fill_i : x_i (dtype preserved)."""
+    if tensors and tensors[0].is_cuda and backend.use_hip(tensors[0]):
+        ext = backend.hip_ext()
+        return ext.fill_nan_columns(
+            [t.contiguous() for t in tensors], torch.tensor(list(fills), dtype=torch.float64)
+        )
+    return [torch.nan_to_num(t, nan=float(f)) for t, f in zip(tensors, fills)]

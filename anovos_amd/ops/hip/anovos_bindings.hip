@@ -30,6 +30,15 @@ int anovos_hll(const void *x, int64_t n, int p, int nchunks, int dtype,
                int32_t *regs, hipStream_t stream);
 int anovos_row_null(const void *const *cols, int ncols, int64_t n, int dtype,
                     int32_t *out, hipStream_t stream);
+int anovos_hll_multi(const void *const *cols, const int64_t *lens, int ncols,
+                     int p, int nchunks, int dtype, int32_t *regs,
+                     hipStream_t stream);
+int anovos_axpb(const void *const *cols, const int64_t *lens, int ncols,
+                const double *a, const double *b, int nchunks, int dtype,
+                float *const *outs, hipStream_t stream);
+int anovos_fillnan(const void *const *cols, const int64_t *lens, int ncols,
+                   const double *fill, int nchunks, int dtype,
+                   void *const *outs, hipStream_t stream);
 }
 
 namespace {
@@ -268,7 +277,109 @@ void row_null_counts_num(std::vector<torch::Tensor> cols, torch::Tensor out) {
   }
 }
 
+torch::Tensor hll_registers_multi(std::vector<torch::Tensor> cols, int64_t p) {
+  TORCH_CHECK(!cols.empty(), "no columns");
+  auto device = cols[0].device();
+  auto regs = torch::zeros({(int64_t)cols.size(), (int64_t)1 << p},
+                           torch::TensorOptions().dtype(torch::kInt32).device(device));
+  for (int pass = 0; pass < 2; ++pass) {
+    std::vector<int64_t> ptrs, lens, rptrs, idx;
+    for (size_t i = 0; i < cols.size(); ++i) {
+      if (dtype_code(cols[i]) != pass) continue;
+      ptrs.push_back((int64_t)cols[i].data_ptr());
+      lens.push_back(cols[i].numel());
+      idx.push_back((int64_t)i);
+    }
+    if (ptrs.empty()) continue;
+    // contiguous sub-block trick: launch on a per-pass register buffer
+    auto sub = torch::zeros({(int64_t)ptrs.size(), (int64_t)1 << p},
+                            torch::TensorOptions().dtype(torch::kInt32).device(device));
+    auto dptr = to_device_i64(ptrs, device);
+    auto dlen = to_device_i64(lens, device);
+    int64_t maxn = *std::max_element(lens.begin(), lens.end());
+    int nchunks = std::max(1, std::min((int)((maxn + (1 << 20) - 1) >> 20), (int)(4096 / std::max<size_t>(ptrs.size(), 1) + 1)));
+    check_hip(anovos_hll_multi((const void *const *)dptr.data_ptr<int64_t>(),
+                               dlen.data_ptr<int64_t>(), (int)ptrs.size(), (int)p,
+                               nchunks, pass, sub.data_ptr<int32_t>(), current_stream()),
+              "anovos_hll_multi");
+    regs.index_copy_(0, to_device_i64(idx, device), sub);
+  }
+  return regs;
+}
+
+std::vector<torch::Tensor> scale_columns(std::vector<torch::Tensor> cols,
+                                         torch::Tensor a, torch::Tensor b) {
+  TORCH_CHECK(!cols.empty(), "no columns");
+  auto device = cols[0].device();
+  auto a_c = a.to(torch::kFloat64).cpu().contiguous();
+  auto b_c = b.to(torch::kFloat64).cpu().contiguous();
+  std::vector<torch::Tensor> outs;
+  for (auto &t : cols)
+    outs.push_back(torch::empty(t.sizes(), torch::TensorOptions().dtype(torch::kFloat32).device(device)));
+  for (int pass = 0; pass < 2; ++pass) {
+    std::vector<int64_t> ptrs, lens, optrs;
+    std::vector<double> av, bv;
+    for (size_t i = 0; i < cols.size(); ++i) {
+      if (dtype_code(cols[i]) != pass) continue;
+      ptrs.push_back((int64_t)cols[i].data_ptr());
+      lens.push_back(cols[i].numel());
+      optrs.push_back((int64_t)outs[i].data_ptr());
+      av.push_back(a_c.data_ptr<double>()[i]);
+      bv.push_back(b_c.data_ptr<double>()[i]);
+    }
+    if (ptrs.empty()) continue;
+    int64_t maxn = *std::max_element(lens.begin(), lens.end());
+    int nchunks = pick_chunks(maxn, (int)ptrs.size());
+    auto dptr = to_device_i64(ptrs, device);
+    auto dlen = to_device_i64(lens, device);
+    auto dout = to_device_i64(optrs, device);
+    auto da = torch::from_blob(av.data(), {(int64_t)av.size()}, torch::TensorOptions().dtype(torch::kFloat64)).clone().to(device);
+    auto db = torch::from_blob(bv.data(), {(int64_t)bv.size()}, torch::TensorOptions().dtype(torch::kFloat64)).clone().to(device);
+    check_hip(anovos_axpb((const void *const *)dptr.data_ptr<int64_t>(),
+                          dlen.data_ptr<int64_t>(), (int)ptrs.size(),
+                          da.data_ptr<double>(), db.data_ptr<double>(), nchunks, pass,
+                          (float *const *)dout.data_ptr<int64_t>(), current_stream()),
+              "anovos_axpb");
+  }
+  return outs;
+}
+
+std::vector<torch::Tensor> fill_nan_columns(std::vector<torch::Tensor> cols, torch::Tensor fill) {
+  TORCH_CHECK(!cols.empty(), "no columns");
+  auto device = cols[0].device();
+  auto f_c = fill.to(torch::kFloat64).cpu().contiguous();
+  std::vector<torch::Tensor> outs;
+  for (auto &t : cols) outs.push_back(torch::empty_like(t));
+  for (int pass = 0; pass < 2; ++pass) {
+    std::vector<int64_t> ptrs, lens, optrs;
+    std::vector<double> fv;
+    for (size_t i = 0; i < cols.size(); ++i) {
+      if (dtype_code(cols[i]) != pass) continue;
+      ptrs.push_back((int64_t)cols[i].data_ptr());
+      lens.push_back(cols[i].numel());
+      optrs.push_back((int64_t)outs[i].data_ptr());
+      fv.push_back(f_c.data_ptr<double>()[i]);
+    }
+    if (ptrs.empty()) continue;
+    int64_t maxn = *std::max_element(lens.begin(), lens.end());
+    int nchunks = pick_chunks(maxn, (int)ptrs.size());
+    auto dptr = to_device_i64(ptrs, device);
+    auto dlen = to_device_i64(lens, device);
+    auto dout = to_device_i64(optrs, device);
+    auto df = torch::from_blob(fv.data(), {(int64_t)fv.size()}, torch::TensorOptions().dtype(torch::kFloat64)).clone().to(device);
+    check_hip(anovos_fillnan((const void *const *)dptr.data_ptr<int64_t>(),
+                             dlen.data_ptr<int64_t>(), (int)ptrs.size(),
+                             df.data_ptr<double>(), nchunks, pass,
+                             (void *const *)dout.data_ptr<int64_t>(), current_stream()),
+              "anovos_fillnan");
+  }
+  return outs;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("hll_registers_multi", &hll_registers_multi, "fused multi-column HLL (K4)");
+  m.def("scale_columns", &scale_columns, "fused (x-a)*b scaling (K11)");
+  m.def("fill_nan_columns", &fill_nan_columns, "fused NaN fill (K11)");
   m.def("column_moments", &column_moments, "fused per-column moments (K1/K2)");
   m.def("column_histograms", &column_histograms, "fused per-column histograms (K3/K6)");
   m.def("bracket_histograms", &bracket_histograms, "quantile-refinement histograms (K3)");

@@ -330,25 +330,149 @@ __global__ __launch_bounds__(THREADS) void hll_kernel(
 // ------------------------------------------------------------------
 // K10: fused row-wise NaN count over numeric columns
 // ------------------------------------------------------------------
+// Grid: (row_chunk, col_chunk). Each block owns ROWS_PB rows x <=COLS_PB
+// columns; per-row counters live in LDS as uint16, loads are float4, and
+// the block adds its partial counts to global with one atomic per row
+// only when col chunking splits a row.
+
+#define ROWS_PB 8192
+#define COLS_PB 64
 
 template <typename T>
 __global__ __launch_bounds__(THREADS) void row_null_kernel(
-    const T *const *cols, int ncols, int64_t n, int64_t rows_per_block, int32_t *out) {
-  extern __shared__ uint16_t rc[];  // rows_per_block
-  const int64_t s = blockIdx.x * rows_per_block;
-  const int64_t e = min(n, s + rows_per_block);
+    const T *const *cols, int ncols, int64_t n, int32_t *out) {
+  __shared__ uint16_t rc[ROWS_PB];
+  const int ncolchunks = (ncols + COLS_PB - 1) / COLS_PB;
+  const int rowchunk = blockIdx.x / ncolchunks;
+  const int colchunk = blockIdx.x % ncolchunks;
+  const int64_t s = (int64_t)rowchunk * ROWS_PB;
+  const int64_t e = min(n, s + ROWS_PB);
   const int nr = (int)(e - s);
-  for (int i = threadIdx.x; i < nr; i += THREADS) rc[i] = 0;
+  const int c0 = colchunk * COLS_PB;
+  const int c1 = min(ncols, c0 + COLS_PB);
+  for (int i = threadIdx.x; i < ROWS_PB / 4; i += THREADS)
+    ((uint64_t *)rc)[i] = 0;  // 4 counters per store
   __syncthreads();
-  for (int c = 0; c < ncols; ++c) {
+  const bool vec_ok = (sizeof(T) == 4) && (nr % 4 == 0) && ((s & 3) == 0);
+  for (int c = c0; c < c1; ++c) {
     const T *__restrict__ x = cols[c] + s;
-    for (int i = threadIdx.x; i < nr; i += THREADS) {
-      // lane i handles row i (+k*256): LDS bank = i%32 distinct per lane group
-      if (isnan((double)x[i])) rc[i] += 1;  // no atomics: one writer per row
+    if (vec_ok) {
+      const float4 *xv = reinterpret_cast<const float4 *>(x);
+      for (int i = threadIdx.x; i < nr / 4; i += THREADS) {
+        float4 v = xv[i];
+        // 4 consecutive rows per lane: one 8-B LDS read+write
+        ushort4 cnt = ((ushort4 *)rc)[i];
+        cnt.x += isnan(v.x);
+        cnt.y += isnan(v.y);
+        cnt.z += isnan(v.z);
+        cnt.w += isnan(v.w);
+        ((ushort4 *)rc)[i] = cnt;
+      }
+    } else {
+      for (int i = threadIdx.x; i < nr; i += THREADS)
+        if (isnan((double)x[i])) rc[i] += 1;
     }
     __syncthreads();
   }
-  for (int i = threadIdx.x; i < nr; i += THREADS) out[s + i] += (int32_t)rc[i];
+  if (ncolchunks == 1) {
+    for (int i = threadIdx.x; i < nr; i += THREADS) out[s + i] += (int32_t)rc[i];
+  } else {
+    for (int i = threadIdx.x; i < nr; i += THREADS)
+      if (rc[i]) atomicAdd(&out[s + i], (int32_t)rc[i]);
+  }
+}
+
+// ------------------------------------------------------------------
+// K4 fused: multi-column HyperLogLog
+// ------------------------------------------------------------------
+
+template <typename T>
+__global__ __launch_bounds__(THREADS) void hll_multi_kernel(
+    const T *const *cols, const int64_t *lens, int p, int nchunks,
+    int32_t *regs /*[ncols][1<<p]*/) {
+  extern __shared__ int32_t sreg[];
+  const int m = 1 << p;
+  const int col = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  for (int i = threadIdx.x; i < m; i += THREADS) sreg[i] = 0;
+  __syncthreads();
+  const T *__restrict__ x = cols[col];
+  const int64_t n = lens[col];
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+    double v = (double)x[i];
+    if (isnan(v)) continue;
+    uint64_t bits;
+    memcpy(&bits, &v, 8);
+    uint64_t h = splitmix64(bits);
+    int idx = (int)(h >> (64 - p));
+    uint64_t rem = h << p;
+    int rho = (rem == 0) ? (64 - p + 1) : (__clzll((long long)rem) + 1);
+    if (rho > 64 - p + 1) rho = 64 - p + 1;
+    atomicMax(&sreg[idx], rho);
+  }
+  __syncthreads();
+  int32_t *g = &regs[(int64_t)col * m];
+  for (int i = threadIdx.x; i < m; i += THREADS)
+    if (sreg[i]) atomicMax(&g[i], sreg[i]);
+}
+
+// ------------------------------------------------------------------
+// K11 fused: multi-column scale/shift  out = (x - a) * b   (NaN passes)
+//            and multi-column NaN fill out = isnan(x) ? fill : x
+// ------------------------------------------------------------------
+
+template <typename T>
+__global__ __launch_bounds__(THREADS) void axpb_kernel(
+    const T *const *cols, const int64_t *lens, const double *a, const double *b,
+    int nchunks, float *const *outs) {
+  const int col = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  const T *__restrict__ x = cols[col];
+  float *__restrict__ out = outs[col];
+  const int64_t n = lens[col];
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+  const float av = (float)a[col];
+  const float bv = (float)b[col];
+  if (sizeof(T) == 4 && ((e - s) % 4 == 0) && ((s & 3) == 0)) {
+    const float4 *xv = reinterpret_cast<const float4 *>(x + s);
+    float4 *ov = reinterpret_cast<float4 *>(out + s);
+    const int64_t nv = (e - s) / 4;
+    for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
+      float4 v = xv[i];
+      v.x = (v.x - av) * bv;
+      v.y = (v.y - av) * bv;
+      v.z = (v.z - av) * bv;
+      v.w = (v.w - av) * bv;
+      ov[i] = v;
+    }
+  } else {
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS)
+      out[i] = ((float)x[i] - av) * bv;
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(THREADS) void fillnan_kernel(
+    const T *const *cols, const int64_t *lens, const double *fill, int nchunks,
+    T *const *outs) {
+  const int col = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  const T *__restrict__ x = cols[col];
+  T *__restrict__ out = outs[col];
+  const int64_t n = lens[col];
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+  const T fv = (T)fill[col];
+  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+    T v = x[i];
+    out[i] = isnan((double)v) ? fv : v;
+  }
 }
 
 // ------------------------------------------------------------------
@@ -441,16 +565,58 @@ int anovos_hll(const void *x, int64_t n, int p, int nchunks, int dtype,
 
 int anovos_row_null(const void *const *cols, int ncols, int64_t n, int dtype,
                     int32_t *out, hipStream_t stream) {
-  const int64_t rows_per_block = 32768;  // 64 KiB LDS of uint16 counters
-  int64_t nblocks = (n + rows_per_block - 1) / rows_per_block;
+  int64_t nrowchunks = (n + ROWS_PB - 1) / ROWS_PB;
+  int64_t ncolchunks = (ncols + COLS_PB - 1) / COLS_PB;
+  int64_t nblocks = nrowchunks * ncolchunks;
   if (nblocks == 0) return 0;
-  size_t lds = (size_t)rows_per_block * 2;
   if (dtype == 0)
-    hipLaunchKernelGGL(row_null_kernel<float>, dim3((uint32_t)nblocks), dim3(THREADS), lds, stream,
-                       (const float *const *)cols, ncols, n, rows_per_block, out);
+    hipLaunchKernelGGL(row_null_kernel<float>, dim3((uint32_t)nblocks), dim3(THREADS), 0, stream,
+                       (const float *const *)cols, ncols, n, out);
   else
-    hipLaunchKernelGGL(row_null_kernel<double>, dim3((uint32_t)nblocks), dim3(THREADS), lds, stream,
-                       (const double *const *)cols, ncols, n, rows_per_block, out);
+    hipLaunchKernelGGL(row_null_kernel<double>, dim3((uint32_t)nblocks), dim3(THREADS), 0, stream,
+                       (const double *const *)cols, ncols, n, out);
+  return (int)hipGetLastError();
+}
+
+int anovos_hll_multi(const void *const *cols, const int64_t *lens, int ncols,
+                     int p, int nchunks, int dtype, int32_t *regs,
+                     hipStream_t stream) {
+  size_t lds = (size_t)(1 << p) * 4;
+  dim3 grid(ncols * nchunks);
+  if (dtype == 0)
+    hipLaunchKernelGGL(hll_multi_kernel<float>, grid, dim3(THREADS), lds, stream,
+                       (const float *const *)cols, lens, p, nchunks, regs);
+  else
+    hipLaunchKernelGGL(hll_multi_kernel<double>, grid, dim3(THREADS), lds, stream,
+                       (const double *const *)cols, lens, p, nchunks, regs);
+  return (int)hipGetLastError();
+}
+
+int anovos_axpb(const void *const *cols, const int64_t *lens, int ncols,
+                const double *a, const double *b, int nchunks, int dtype,
+                float *const *outs, hipStream_t stream) {
+  dim3 grid(ncols * nchunks);
+  if (dtype == 0)
+    hipLaunchKernelGGL(axpb_kernel<float>, grid, dim3(THREADS), 0, stream,
+                       (const float *const *)cols, lens, a, b, nchunks, outs);
+  else
+    hipLaunchKernelGGL(axpb_kernel<double>, grid, dim3(THREADS), 0, stream,
+                       (const double *const *)cols, lens, a, b, nchunks, outs);
+  return (int)hipGetLastError();
+}
+
+int anovos_fillnan(const void *const *cols, const int64_t *lens, int ncols,
+                   const double *fill, int nchunks, int dtype,
+                   void *const *outs, hipStream_t stream) {
+  dim3 grid(ncols * nchunks);
+  if (dtype == 0)
+    hipLaunchKernelGGL(fillnan_kernel<float>, grid, dim3(THREADS), 0, stream,
+                       (const float *const *)cols, lens, fill, nchunks,
+                       (float *const *)outs);
+  else
+    hipLaunchKernelGGL(fillnan_kernel<double>, grid, dim3(THREADS), 0, stream,
+                       (const double *const *)cols, lens, fill, nchunks,
+                       (double *const *)outs);
   return (int)hipGetLastError();
 }
 

@@ -155,7 +155,18 @@ def approx_quantiles(
             below = float(cdf[i, b - 1]) if b > 0 else 0.0
             brackets[(i, j)] = (l + b * w, l + (b + 1) * w, target - below, float(hist[i, b]))
     for _ in range(refine):
-        brackets = _refine_pass(tensors, cols, brackets, nbins)
+        # adaptive: only brackets whose bin still holds > rel_err/2 of the
+        # rank mass need another pass (Spark guarantees 1% rank error;
+        # within-bin interpolation already bounds ours by bin_count/n)
+        need = {
+            k: v
+            for k, v in brackets.items()
+            if v[3] > max(rel_err, 1e-4) * 0.5 * max(moments[cols[k[0]]].n, 1)
+        }
+        if not need:
+            break
+        refined = _refine_pass(tensors, cols, need, nbins)
+        brackets.update(refined)
     for (i, j), (bl, bh, off, cnt) in brackets.items():
         c = cols[i]
         if cnt <= 1 or bh - bl < 1e-12 * max(1.0, abs(bl)):

@@ -15,18 +15,19 @@ from anovos_amd.core.dtypes import NULL_CODE
 
 
 def bernoulli_mask(n: int, fraction: float, seed: int, device) -> torch.Tensor:
-    g = torch.Generator(device="cpu")
+    device = torch.device(device)
+    g = torch.Generator(device=device)
     g.manual_seed(int(seed) * 1000003 + dist.rank())
-    u = torch.rand(n, generator=g)
-    return (u < fraction).to(device)
+    u = torch.rand(n, generator=g, device=device)
+    return u < fraction
 
 
 def stratified_mask(codes: torch.Tensor, fractions: Dict[int, float], seed: int) -> torch.Tensor:
     """Per-stratum Bernoulli mask over an int32 code column."""
     n = codes.shape[0]
-    g = torch.Generator(device="cpu")
+    g = torch.Generator(device=codes.device)
     g.manual_seed(int(seed) * 1000003 + dist.rank())
-    u = torch.rand(n).to(codes.device)
+    u = torch.rand(n, generator=g, device=codes.device)
     frac = torch.zeros(int(codes.max().item()) + 2 if n else 1, dtype=torch.float32, device=codes.device)
     for k, f in fractions.items():
         if 0 <= int(k) < frac.numel():

@@ -113,8 +113,10 @@ def pipeline_step(ctx, idf, source_hist, model_dir):
     drift_vals = 0.0
     import numpy as _np
 
-    for c in num_cols[:: max(1, len(num_cols) // 50)]:  # 50 representative columns
-        q_keys, q_vals = dd._bin_frequencies(binned, c + "_binned", idf.count())
+    drift_cols = num_cols[:: max(1, len(num_cols) // 50)]
+    q_freqs = dd.batched_bin_frequencies(binned, [c + "_binned" for c in drift_cols], idf.count())
+    for c in drift_cols:
+        q_keys, q_vals = q_freqs[c + "_binned"]
         pmap = source_hist.get(c, {})
         qmap = dict(zip(q_keys, q_vals))
         keys = sorted(set(pmap) | set(qmap), key=dd._key_order)
@@ -169,12 +171,10 @@ def main():
     num_cols = attributeType_segregation(idf)[0]
 
     # warmup builds the drift source snapshot (per-bin frequencies)
-    source_hist = {}
     binned = T.attribute_binning(ctx, idf, num_cols, bin_size=10, output_mode="append")
     total = idf.count()
-    for c in num_cols:
-        keys, vals = dd._bin_frequencies(binned, c + "_binned", total)
-        source_hist[c] = dict(zip(keys, vals))
+    freqs = dd.batched_bin_frequencies(binned, [c + "_binned" for c in num_cols], total)
+    source_hist = {c: dict(zip(*freqs[c + "_binned"])) for c in num_cols}
 
     for _ in range(args.warmup):
         pipeline_step(ctx, idf, source_hist, model_dir)

@@ -151,3 +151,33 @@ def test_gpu_pipeline_stats_match_cpu(ext):
             assert abs(float(pa.loc[attr, col]) - float(pb.loc[attr, col])) <= max(
                 0.01 * abs(float(pa.loc[attr, col])), 0.05
             )
+
+
+@requires_gpu
+def test_hll_multi_matches_single(ext, gpu_cols):
+    cols = [t for t in gpu_cols if t.dtype == torch.float32][:4]
+    multi = ext.hll_registers_multi(cols, 14).cpu()
+    for i, t in enumerate(cols):
+        single = ext.hll_registers(t, 14).cpu()
+        assert torch.equal(multi[i], single)
+
+
+@requires_gpu
+def test_scale_columns_vs_torch(ext, gpu_cols):
+    cols = [t for t in gpu_cols if t.dtype == torch.float32][:4]
+    a = [1.0, -2.0, 0.5, 3.0]
+    b = [2.0, 0.25, -1.0, 10.0]
+    outs = ext.scale_columns(cols, torch.tensor(a), torch.tensor(b))
+    for t, ai, bi, o in zip(cols, a, b, outs):
+        ref = (t - ai) * bi
+        assert torch.allclose(o, ref, rtol=1e-6, atol=1e-5, equal_nan=True)
+
+
+@requires_gpu
+def test_fill_nan_columns_vs_torch(ext, gpu_cols):
+    cols = [t for t in gpu_cols if t.dtype == torch.float32][:4]
+    fills = [0.0, 1.5, -3.0, 42.0]
+    outs = ext.fill_nan_columns(cols, torch.tensor(fills))
+    for t, f, o in zip(cols, fills, outs):
+        ref = torch.nan_to_num(t, nan=f)
+        assert torch.equal(o, ref)
