@@ -75,6 +75,41 @@ def index_map_from_counts(dictionary: List[str], counts: torch.Tensor, order: st
     return {k: i for i, (k, _) in enumerate(items)}
 
 
+def apply_index_maps_batch(cols: List[Column], mappings: List[Dict[str, int]]) -> List[torch.Tensor]:
+    """Batched StringIndexer apply: ONE fused LUT-gather kernel covers all
+    columns (K12). Unseen/null -> NaN float32 outputs."""
+    from anovos_amd.ops import backend
+
+    if cols and cols[0].data.is_cuda and backend.use_hip(cols[0].data):
+        ext = backend.hip_ext()
+        luts = [
+            torch.tensor(
+                [float(m.get(s, float("nan"))) for s in (c.dictionary or [])] or [float("nan")],
+                dtype=torch.float32,
+            )
+            for c, m in zip(cols, mappings)
+        ]
+        return ext.lut_apply_f32([c.data.contiguous() for c in cols], luts)
+    return [apply_index_map(c, m) for c, m in zip(cols, mappings)]
+
+
+def remap_codes_batch(cols: List[Column], luts: List[torch.Tensor]) -> List[torch.Tensor]:
+    """Batched int32 code remap (outlier_categories etc.): out = lut[code],
+    null -> -1, via one fused kernel."""
+    from anovos_amd.ops import backend
+
+    if cols and cols[0].data.is_cuda and backend.use_hip(cols[0].data):
+        ext = backend.hip_ext()
+        return ext.lut_apply_i32([c.data.contiguous() for c in cols], [l.to(torch.int32) for l in luts])
+    out = []
+    for c, lut in zip(cols, luts):
+        lut_l = torch.cat([lut.to(torch.int64), torch.tensor([NULL_CODE])]).to(c.data.device)
+        codes = c.data.to(torch.long)
+        codes = torch.where(codes == NULL_CODE, torch.full_like(codes, lut.numel()), codes)
+        out.append(lut_l[codes].to(torch.int32))
+    return out
+
+
 def apply_index_map(col: Column, mapping: Dict[str, int], unseen: int = -1) -> torch.Tensor:
     """Apply a label->index map to a categorical column; returns float32
     tensor with NaN for null/unseen (matches reference's numeric output

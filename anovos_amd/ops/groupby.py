@@ -88,8 +88,8 @@ def numeric_value_counts(idf, col: str) -> Tuple[torch.Tensor, torch.Tensor]:
 
         m = _stats.frame_moments(idf, [col])[col]
     if x.numel() and m is not None and m.min == m.min and (m.max - m.min) < 4_000_000:
-        xi = x.to(torch.int64)
-        if bool((x == xi.to(x.dtype)).all()):
+        if m.integral or (m.n_frac != m.n_frac and bool((x == torch.trunc(x)).all())):
+            xi = x.to(torch.int64)
             lo = int(m.min)
             cnts_d = torch.bincount(xi - lo, minlength=int(m.max) - lo + 1)
             nz = cnts_d.nonzero(as_tuple=True)[0]

@@ -39,6 +39,20 @@ int anovos_axpb(const void *const *cols, const int64_t *lens, int ncols,
 int anovos_fillnan(const void *const *cols, const int64_t *lens, int ncols,
                    const double *fill, int nchunks, int dtype,
                    void *const *outs, hipStream_t stream);
+int anovos_bracket_hist_grouped(const void *const *cols, const int64_t *lens,
+                                const int *bstart, int ncols, const double *lo,
+                                const double *hi, int nchunks, int dtype,
+                                uint64_t *out, hipStream_t stream);
+int anovos_bucketize_float(const void *const *cols, const int64_t *lens, int ncols,
+                           const double *cutflat, const int64_t *cutoff_off,
+                           const int *cutoff_len, int max_ncut, int nchunks, int dtype,
+                           float *const *outs, hipStream_t stream);
+int anovos_lut_apply_f32(const int32_t *const *cols, const int64_t *lens, int ncols,
+                         const float *lutflat, const int64_t *lut_off, int nchunks,
+                         float *const *outs, hipStream_t stream);
+int anovos_lut_apply_i32(const int32_t *const *cols, const int64_t *lens, int ncols,
+                         const int32_t *lutflat, const int64_t *lut_off, int nchunks,
+                         int32_t *const *outs, hipStream_t stream);
 }
 
 namespace {
@@ -82,7 +96,7 @@ int dtype_code(const torch::Tensor &t) {
 torch::Tensor column_moments(std::vector<torch::Tensor> cols) {
   TORCH_CHECK(!cols.empty(), "no columns");
   auto device = cols[0].device();
-  auto out = torch::zeros({(int64_t)cols.size(), 8},
+  auto out = torch::zeros({(int64_t)cols.size(), 9},
                           torch::TensorOptions().dtype(torch::kFloat64).device(device));
   // group by dtype, one fused launch per dtype
   for (int pass = 0; pass < 2; ++pass) {
@@ -101,9 +115,9 @@ torch::Tensor column_moments(std::vector<torch::Tensor> cols) {
     int nchunks = pick_chunks(maxn, ncols);
     auto dptr = to_device_i64(ptrs, device);
     auto dlen = to_device_i64(lens, device);
-    auto partials = torch::empty({(int64_t)ncols * nchunks, 8},
+    auto partials = torch::empty({(int64_t)ncols * nchunks, 9},
                                  torch::TensorOptions().dtype(torch::kFloat64).device(device));
-    auto sub = torch::empty({ncols, 8}, torch::TensorOptions().dtype(torch::kFloat64).device(device));
+    auto sub = torch::empty({ncols, 9}, torch::TensorOptions().dtype(torch::kFloat64).device(device));
     check_hip(anovos_moments((const void *const *)dptr.data_ptr<int64_t>(),
                              dlen.data_ptr<int64_t>(), ncols, nchunks, pass,
                              partials.data_ptr<double>(), sub.data_ptr<double>(),
@@ -376,7 +390,171 @@ std::vector<torch::Tensor> fill_nan_columns(std::vector<torch::Tensor> cols, tor
   return outs;
 }
 
+torch::Tensor bracket_histograms_grouped(std::vector<torch::Tensor> cols,
+                                         torch::Tensor bracket_col,
+                                         torch::Tensor lo, torch::Tensor hi) {
+  // brackets MUST be sorted by column index; returns [nbrackets, 512].
+  TORCH_CHECK(!cols.empty(), "no columns");
+  auto device = cols[0].device();
+  int dtype = dtype_code(cols[0]);
+  for (auto &t : cols) TORCH_CHECK(dtype_code(t) == dtype, "grouped brackets: mixed dtypes unsupported");
+  auto bc = bracket_col.to(torch::kInt64).cpu().contiguous();
+  int ncols = (int)cols.size();
+  int nb = (int)bc.numel();
+  std::vector<int> bstart(ncols + 1, 0);
+  {
+    const int64_t *b = bc.data_ptr<int64_t>();
+    for (int i = 0; i < nb; ++i) {
+      TORCH_CHECK(b[i] >= 0 && b[i] < ncols, "bad bracket col");
+      if (i) TORCH_CHECK(b[i] >= b[i - 1], "brackets must be sorted by column");
+      bstart[b[i] + 1]++;
+    }
+    for (int c = 0; c < ncols; ++c) {
+      TORCH_CHECK(bstart[c + 1] <= 16, "at most 16 brackets per column");
+      bstart[c + 1] += bstart[c];
+    }
+  }
+  std::vector<int64_t> ptrs, lens;
+  for (auto &t : cols) {
+    ptrs.push_back((int64_t)t.data_ptr());
+    lens.push_back(t.numel());
+  }
+  auto dptr = to_device_i64(ptrs, device);
+  auto dlen = to_device_i64(lens, device);
+  auto dbs = torch::from_blob(bstart.data(), {(int64_t)bstart.size()}, torch::TensorOptions().dtype(torch::kInt32)).clone().to(device);
+  auto lo_d = lo.to(torch::kFloat64).to(device);
+  auto hi_d = hi.to(torch::kFloat64).to(device);
+  int64_t maxn = *std::max_element(lens.begin(), lens.end());
+  int nchunks = pick_chunks(maxn, ncols);
+  auto out = torch::zeros({nb, 512}, torch::TensorOptions().dtype(torch::kInt64).device(device));
+  check_hip(anovos_bracket_hist_grouped((const void *const *)dptr.data_ptr<int64_t>(),
+                                        dlen.data_ptr<int64_t>(), dbs.data_ptr<int>(), ncols,
+                                        lo_d.data_ptr<double>(), hi_d.data_ptr<double>(),
+                                        nchunks, dtype, (uint64_t *)out.data_ptr<int64_t>(),
+                                        current_stream()),
+            "anovos_bracket_hist_grouped");
+  return out;
+}
+
+std::vector<torch::Tensor> bucketize_columns_float(std::vector<torch::Tensor> cols,
+                                                   std::vector<torch::Tensor> cutoffs) {
+  TORCH_CHECK(cols.size() == cutoffs.size(), "cols/cutoffs size mismatch");
+  auto device = cols[0].device();
+  std::vector<torch::Tensor> outs;
+  for (auto &t : cols)
+    outs.push_back(torch::empty(t.sizes(), torch::TensorOptions().dtype(torch::kFloat32).device(device)));
+  for (int pass = 0; pass < 2; ++pass) {
+    std::vector<int64_t> ptrs, lens, optrs, offs;
+    std::vector<double> flat;
+    std::vector<int> clens;
+    int max_ncut = 1;
+    for (size_t i = 0; i < cols.size(); ++i) {
+      if (dtype_code(cols[i]) != pass) continue;
+      ptrs.push_back((int64_t)cols[i].data_ptr());
+      lens.push_back(cols[i].numel());
+      optrs.push_back((int64_t)outs[i].data_ptr());
+      auto cc = cutoffs[i].to(torch::kFloat64).cpu().contiguous();
+      offs.push_back((int64_t)flat.size());
+      const double *cd = cc.data_ptr<double>();
+      flat.insert(flat.end(), cd, cd + cc.numel());
+      clens.push_back((int)cc.numel());
+      max_ncut = std::max(max_ncut, (int)cc.numel());
+    }
+    if (ptrs.empty()) continue;
+    int ncols = (int)ptrs.size();
+    int64_t maxn = *std::max_element(lens.begin(), lens.end());
+    int nchunks = pick_chunks(maxn, ncols);
+    auto dptr = to_device_i64(ptrs, device);
+    auto dlen = to_device_i64(lens, device);
+    auto dout = to_device_i64(optrs, device);
+    auto doff = to_device_i64(offs, device);
+    auto dflat = torch::from_blob(flat.data(), {(int64_t)std::max<size_t>(flat.size(), 1)},
+                                  torch::TensorOptions().dtype(torch::kFloat64)).clone().to(device);
+    auto dclen = torch::from_blob(clens.data(), {(int64_t)clens.size()},
+                                  torch::TensorOptions().dtype(torch::kInt32)).clone().to(device);
+    check_hip(anovos_bucketize_float((const void *const *)dptr.data_ptr<int64_t>(),
+                                     dlen.data_ptr<int64_t>(), ncols, dflat.data_ptr<double>(),
+                                     doff.data_ptr<int64_t>(), dclen.data_ptr<int>(), max_ncut,
+                                     nchunks, pass, (float *const *)dout.data_ptr<int64_t>(),
+                                     current_stream()),
+              "anovos_bucketize_float");
+  }
+  return outs;
+}
+
+std::vector<torch::Tensor> lut_apply_f32(std::vector<torch::Tensor> cols,
+                                         std::vector<torch::Tensor> luts) {
+  TORCH_CHECK(cols.size() == luts.size(), "cols/luts size mismatch");
+  auto device = cols[0].device();
+  std::vector<int64_t> ptrs, lens, optrs, offs;
+  std::vector<float> flat;
+  std::vector<torch::Tensor> outs;
+  for (size_t i = 0; i < cols.size(); ++i) {
+    TORCH_CHECK(cols[i].scalar_type() == torch::kInt32, "codes must be int32");
+    outs.push_back(torch::empty(cols[i].sizes(), torch::TensorOptions().dtype(torch::kFloat32).device(device)));
+    ptrs.push_back((int64_t)cols[i].data_ptr());
+    lens.push_back(cols[i].numel());
+    optrs.push_back((int64_t)outs[i].data_ptr());
+    auto lc = luts[i].to(torch::kFloat32).cpu().contiguous();
+    offs.push_back((int64_t)flat.size());
+    const float *ld = lc.data_ptr<float>();
+    flat.insert(flat.end(), ld, ld + lc.numel());
+  }
+  int64_t maxn = *std::max_element(lens.begin(), lens.end());
+  int nchunks = pick_chunks(maxn, (int)cols.size());
+  auto dptr = to_device_i64(ptrs, device);
+  auto dlen = to_device_i64(lens, device);
+  auto dout = to_device_i64(optrs, device);
+  auto doff = to_device_i64(offs, device);
+  auto dflat = torch::from_blob(flat.data(), {(int64_t)std::max<size_t>(flat.size(), 1)},
+                                torch::TensorOptions().dtype(torch::kFloat32)).clone().to(device);
+  check_hip(anovos_lut_apply_f32((const int32_t *const *)dptr.data_ptr<int64_t>(),
+                                 dlen.data_ptr<int64_t>(), (int)cols.size(),
+                                 dflat.data_ptr<float>(), doff.data_ptr<int64_t>(), nchunks,
+                                 (float *const *)dout.data_ptr<int64_t>(), current_stream()),
+            "anovos_lut_apply_f32");
+  return outs;
+}
+
+std::vector<torch::Tensor> lut_apply_i32(std::vector<torch::Tensor> cols,
+                                         std::vector<torch::Tensor> luts) {
+  TORCH_CHECK(cols.size() == luts.size(), "cols/luts size mismatch");
+  auto device = cols[0].device();
+  std::vector<int64_t> ptrs, lens, optrs, offs;
+  std::vector<int32_t> flat;
+  std::vector<torch::Tensor> outs;
+  for (size_t i = 0; i < cols.size(); ++i) {
+    TORCH_CHECK(cols[i].scalar_type() == torch::kInt32, "codes must be int32");
+    outs.push_back(torch::empty(cols[i].sizes(), torch::TensorOptions().dtype(torch::kInt32).device(device)));
+    ptrs.push_back((int64_t)cols[i].data_ptr());
+    lens.push_back(cols[i].numel());
+    optrs.push_back((int64_t)outs[i].data_ptr());
+    auto lc = luts[i].to(torch::kInt32).cpu().contiguous();
+    offs.push_back((int64_t)flat.size());
+    const int32_t *ld = lc.data_ptr<int32_t>();
+    flat.insert(flat.end(), ld, ld + lc.numel());
+  }
+  int64_t maxn = *std::max_element(lens.begin(), lens.end());
+  int nchunks = pick_chunks(maxn, (int)cols.size());
+  auto dptr = to_device_i64(ptrs, device);
+  auto dlen = to_device_i64(lens, device);
+  auto dout = to_device_i64(optrs, device);
+  auto doff = to_device_i64(offs, device);
+  auto dflat = torch::from_blob(flat.data(), {(int64_t)std::max<size_t>(flat.size(), 1)},
+                                torch::TensorOptions().dtype(torch::kInt32)).clone().to(device);
+  check_hip(anovos_lut_apply_i32((const int32_t *const *)dptr.data_ptr<int64_t>(),
+                                 dlen.data_ptr<int64_t>(), (int)cols.size(),
+                                 dflat.data_ptr<int32_t>(), doff.data_ptr<int64_t>(), nchunks,
+                                 (int32_t *const *)dout.data_ptr<int64_t>(), current_stream()),
+            "anovos_lut_apply_i32");
+  return outs;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("bracket_histograms_grouped", &bracket_histograms_grouped, "grouped refinement histograms (K3)");
+  m.def("bucketize_columns_float", &bucketize_columns_float, "bucketize to float bin labels (K6)");
+  m.def("lut_apply_f32", &lut_apply_f32, "fused LUT gather -> float (K12)");
+  m.def("lut_apply_i32", &lut_apply_i32, "fused LUT gather -> int32 (K12)");
   m.def("hll_registers_multi", &hll_registers_multi, "fused multi-column HLL (K4)");
   m.def("scale_columns", &scale_columns, "fused (x-a)*b scaling (K11)");
   m.def("fill_nan_columns", &fill_nan_columns, "fused NaN fill (K11)");

@@ -31,8 +31,12 @@
 // partials layout: [ncols * nchunks][8] doubles:
 //   0:n 1:s1 2:s2 3:s3 4:s4 5:min 6:max 7:zeros
 
+// 9 slots: n, s1..s4, min, max, zeros, n_frac (non-integral count — lets
+// the host pick dense-bincount exact modes for integral columns for free)
+#define NSTAT 9
+
 struct MomAcc {
-  double n, s1, s2, s3, s4, mn, mx, zn;
+  double n, s1, s2, s3, s4, mn, mx, zn, nf;
 };
 
 DEV_INLINE void mom_add(MomAcc &a, double v) {
@@ -46,6 +50,7 @@ DEV_INLINE void mom_add(MomAcc &a, double v) {
     a.mn = fmin(a.mn, v);
     a.mx = fmax(a.mx, v);
     a.zn += (v == 0.0) ? 1.0 : 0.0;
+    a.nf += (v != trunc(v)) ? 1.0 : 0.0;
   }
 }
 
@@ -78,60 +83,60 @@ __global__ __launch_bounds__(THREADS) void moments_partials_kernel(
     for (int64_t i = s + threadIdx.x; i < e; i += THREADS) mom_add(a, (double)x[i]);
   }
 
-  // block reduce through LDS (8 doubles per thread -> tree)
-  __shared__ double sm[THREADS * 8];
-  double *mine = &sm[threadIdx.x * 8];
+  // block reduce through LDS (NSTAT doubles per thread -> tree)
+  __shared__ double sm[THREADS * NSTAT];
+  double *mine = &sm[threadIdx.x * NSTAT];
   mine[0] = a.n; mine[1] = a.s1; mine[2] = a.s2; mine[3] = a.s3;
-  mine[4] = a.s4; mine[5] = a.mn; mine[6] = a.mx; mine[7] = a.zn;
+  mine[4] = a.s4; mine[5] = a.mn; mine[6] = a.mx; mine[7] = a.zn; mine[8] = a.nf;
   __syncthreads();
   for (int stride = THREADS / 2; stride > 0; stride >>= 1) {
     if (threadIdx.x < stride) {
-      double *other = &sm[(threadIdx.x + stride) * 8];
+      double *other = &sm[(threadIdx.x + stride) * NSTAT];
       mine[0] += other[0]; mine[1] += other[1]; mine[2] += other[2];
       mine[3] += other[3]; mine[4] += other[4];
       mine[5] = fmin(mine[5], other[5]);
       mine[6] = fmax(mine[6], other[6]);
-      mine[7] += other[7];
+      mine[7] += other[7]; mine[8] += other[8];
     }
     __syncthreads();
   }
   if (threadIdx.x == 0) {
-    double *out = &partials[(int64_t)blockIdx.x * 8];
-    for (int k = 0; k < 8; ++k) out[k] = sm[k];
+    double *out = &partials[(int64_t)blockIdx.x * NSTAT];
+    for (int k = 0; k < NSTAT; ++k) out[k] = sm[k];
   }
 }
 
 __global__ __launch_bounds__(THREADS) void moments_reduce_kernel(
     const double *partials, int nchunks, double *out) {
   const int col = blockIdx.x;
-  MomAcc a{0, 0, 0, 0, 0, DBL_MAX, -DBL_MAX, 0};
+  MomAcc a{0, 0, 0, 0, 0, DBL_MAX, -DBL_MAX, 0, 0};
   for (int c = threadIdx.x; c < nchunks; c += THREADS) {
-    const double *p = &partials[((int64_t)col * nchunks + c) * 8];
+    const double *p = &partials[((int64_t)col * nchunks + c) * NSTAT];
     a.n += p[0]; a.s1 += p[1]; a.s2 += p[2]; a.s3 += p[3]; a.s4 += p[4];
-    a.mn = fmin(a.mn, p[5]); a.mx = fmax(a.mx, p[6]); a.zn += p[7];
+    a.mn = fmin(a.mn, p[5]); a.mx = fmax(a.mx, p[6]); a.zn += p[7]; a.nf += p[8];
   }
-  __shared__ double sm[THREADS * 8];
-  double *mine = &sm[threadIdx.x * 8];
+  __shared__ double sm[THREADS * NSTAT];
+  double *mine = &sm[threadIdx.x * NSTAT];
   mine[0] = a.n; mine[1] = a.s1; mine[2] = a.s2; mine[3] = a.s3;
-  mine[4] = a.s4; mine[5] = a.mn; mine[6] = a.mx; mine[7] = a.zn;
+  mine[4] = a.s4; mine[5] = a.mn; mine[6] = a.mx; mine[7] = a.zn; mine[8] = a.nf;
   __syncthreads();
   for (int stride = THREADS / 2; stride > 0; stride >>= 1) {
     if (threadIdx.x < stride) {
-      double *other = &sm[(threadIdx.x + stride) * 8];
+      double *other = &sm[(threadIdx.x + stride) * NSTAT];
       mine[0] += other[0]; mine[1] += other[1]; mine[2] += other[2];
       mine[3] += other[3]; mine[4] += other[4];
       mine[5] = fmin(mine[5], other[5]);
       mine[6] = fmax(mine[6], other[6]);
-      mine[7] += other[7];
+      mine[7] += other[7]; mine[8] += other[8];
     }
     __syncthreads();
   }
   if (threadIdx.x == 0) {
-    double *o = &out[(int64_t)col * 8];
+    double *o = &out[(int64_t)col * NSTAT];
     o[0] = sm[0]; o[1] = sm[1]; o[2] = sm[2]; o[3] = sm[3]; o[4] = sm[4];
     o[5] = (sm[0] > 0) ? sm[5] : nan("");
     o[6] = (sm[0] > 0) ? sm[6] : nan("");
-    o[7] = sm[7];
+    o[7] = sm[7]; o[8] = sm[8];
   }
 }
 
@@ -214,6 +219,55 @@ __global__ __launch_bounds__(THREADS) void bracket_hist_kernel(
     if (bins[b]) atomicAdd((unsigned long long *)&g[b], (unsigned long long)bins[b]);
 }
 
+// grouped bracket histograms: one column read serves ALL of that
+// column's refinement brackets (the per-bracket kernel re-reads the
+// column once per bracket — 9x overfetch for percentile batches).
+// brackets are pre-sorted by column: col c owns [bstart[c], bstart[c+1]).
+#define RB_BINS 512
+#define RB_MAXB 16
+
+template <typename T>
+__global__ __launch_bounds__(THREADS) void bracket_hist_grouped_kernel(
+    const T *const *cols, const int64_t *lens, const int *bstart, int ncols,
+    const double *lo, const double *hi, int nchunks, uint64_t *out) {
+  __shared__ uint32_t bins[RB_MAXB * RB_BINS];
+  __shared__ double slo[RB_MAXB], sscale[RB_MAXB], shi[RB_MAXB];
+  const int col = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  const int b0 = bstart[col];
+  const int nb = bstart[col + 1] - b0;
+  if (nb == 0) return;
+  for (int i = threadIdx.x; i < nb * RB_BINS; i += THREADS) bins[i] = 0;
+  if (threadIdx.x < nb) {
+    slo[threadIdx.x] = lo[b0 + threadIdx.x];
+    shi[threadIdx.x] = hi[b0 + threadIdx.x];
+    double w = shi[threadIdx.x] - slo[threadIdx.x];
+    sscale[threadIdx.x] = (w > 0) ? (double)RB_BINS / w : 0.0;
+  }
+  __syncthreads();
+  const T *__restrict__ x = cols[col];
+  const int64_t n = lens[col];
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+    double v = (double)x[i];
+    if (isnan(v)) continue;
+    for (int b = 0; b < nb; ++b) {
+      if (v >= slo[b] && v < shi[b]) {
+        int bin = (int)((v - slo[b]) * sscale[b]);
+        bin = max(0, min(RB_BINS - 1, bin));
+        atomicAdd(&bins[b * RB_BINS + bin], 1u);
+      }
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < nb * RB_BINS; i += THREADS)
+    if (bins[i])
+      atomicAdd((unsigned long long *)&out[(int64_t)(b0 + i / RB_BINS) * RB_BINS + (i % RB_BINS)],
+                (unsigned long long)bins[i]);
+}
+
 // ------------------------------------------------------------------
 // bucketize: branchless binary search over per-column cutoffs in LDS
 // ------------------------------------------------------------------
@@ -253,6 +307,85 @@ __global__ __launch_bounds__(THREADS) void bucketize_kernel(
       len = (cuts[mid] < v) ? (len - half - 1) : half;
     }
     out[i] = lo;
+  }
+}
+
+// float-output bucketize: writes (bin index + 1) as float32 with NaN for
+// null — exactly the binned-column layout attribute_binning materializes
+// (avoids three extra elementwise passes per column on the python side).
+template <typename T>
+__global__ __launch_bounds__(THREADS) void bucketize_float_kernel(
+    const T *const *cols, const int64_t *lens,
+    const double *cutflat, const int64_t *cutoff_off, const int *cutoff_len,
+    int nchunks, float *const *outs) {
+  extern __shared__ double cuts[];
+  const int col = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  const int ncut = cutoff_len[col];
+  const double *src = &cutflat[cutoff_off[col]];
+  for (int i = threadIdx.x; i < ncut; i += THREADS) cuts[i] = src[i];
+  __syncthreads();
+  const T *__restrict__ x = cols[col];
+  float *__restrict__ out = outs[col];
+  const int64_t n = lens[col];
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+    double v = (double)x[i];
+    if (isnan(v)) {
+      out[i] = nanf("");
+      continue;
+    }
+    int lo = 0, len = ncut;
+    while (len > 0) {
+      int half = len >> 1;
+      int mid = lo + half;
+      lo = (cuts[mid] < v) ? (mid + 1) : lo;
+      len = (cuts[mid] < v) ? (len - half - 1) : half;
+    }
+    out[i] = (float)(lo + 1);
+  }
+}
+
+// ------------------------------------------------------------------
+// K12: fused dictionary LUT apply (encodings / category remaps)
+// ------------------------------------------------------------------
+// f32 variant: out = lut[code] (float), null code (-1) -> NaN.
+__global__ __launch_bounds__(THREADS) void lut_apply_f32_kernel(
+    const int32_t *const *cols, const int64_t *lens, const float *lutflat,
+    const int64_t *lut_off, int nchunks, float *const *outs) {
+  const int col = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  const int32_t *__restrict__ codes = cols[col];
+  float *__restrict__ out = outs[col];
+  const float *__restrict__ lut = &lutflat[lut_off[col]];
+  const int64_t n = lens[col];
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+    int c = codes[i];
+    out[i] = (c < 0) ? nanf("") : lut[c];
+  }
+}
+
+// i32 variant: out = lut[code] (int32 code remap), null (-1) -> -1.
+__global__ __launch_bounds__(THREADS) void lut_apply_i32_kernel(
+    const int32_t *const *cols, const int64_t *lens, const int32_t *lutflat,
+    const int64_t *lut_off, int nchunks, int32_t *const *outs) {
+  const int col = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  const int32_t *__restrict__ codes = cols[col];
+  int32_t *__restrict__ out = outs[col];
+  const int32_t *__restrict__ lut = &lutflat[lut_off[col]];
+  const int64_t n = lens[col];
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+    int c = codes[i];
+    out[i] = (c < 0) ? -1 : lut[c];
   }
 }
 
@@ -540,6 +673,52 @@ int anovos_bucketize(const void *const *cols, const int64_t *lens, int ncols,
   else
     hipLaunchKernelGGL(bucketize_kernel<double>, grid, dim3(THREADS), lds, stream,
                        (const double *const *)cols, lens, cutflat, cutoff_off, cutoff_len, nchunks, outs);
+  return (int)hipGetLastError();
+}
+
+int anovos_bracket_hist_grouped(const void *const *cols, const int64_t *lens,
+                                const int *bstart, int ncols, const double *lo,
+                                const double *hi, int nchunks, int dtype,
+                                uint64_t *out, hipStream_t stream) {
+  dim3 grid(ncols * nchunks);
+  if (dtype == 0)
+    hipLaunchKernelGGL(bracket_hist_grouped_kernel<float>, grid, dim3(THREADS), 0, stream,
+                       (const float *const *)cols, lens, bstart, ncols, lo, hi, nchunks, out);
+  else
+    hipLaunchKernelGGL(bracket_hist_grouped_kernel<double>, grid, dim3(THREADS), 0, stream,
+                       (const double *const *)cols, lens, bstart, ncols, lo, hi, nchunks, out);
+  return (int)hipGetLastError();
+}
+
+int anovos_bucketize_float(const void *const *cols, const int64_t *lens, int ncols,
+                           const double *cutflat, const int64_t *cutoff_off,
+                           const int *cutoff_len, int max_ncut, int nchunks, int dtype,
+                           float *const *outs, hipStream_t stream) {
+  dim3 grid(ncols * nchunks);
+  size_t lds = (size_t)max_ncut * 8;
+  if (lds < 8) lds = 8;
+  if (dtype == 0)
+    hipLaunchKernelGGL(bucketize_float_kernel<float>, grid, dim3(THREADS), lds, stream,
+                       (const float *const *)cols, lens, cutflat, cutoff_off, cutoff_len, nchunks, outs);
+  else
+    hipLaunchKernelGGL(bucketize_float_kernel<double>, grid, dim3(THREADS), lds, stream,
+                       (const double *const *)cols, lens, cutflat, cutoff_off, cutoff_len, nchunks, outs);
+  return (int)hipGetLastError();
+}
+
+int anovos_lut_apply_f32(const int32_t *const *cols, const int64_t *lens, int ncols,
+                         const float *lutflat, const int64_t *lut_off, int nchunks,
+                         float *const *outs, hipStream_t stream) {
+  hipLaunchKernelGGL(lut_apply_f32_kernel, dim3(ncols * nchunks), dim3(THREADS), 0, stream,
+                     cols, lens, lutflat, lut_off, nchunks, outs);
+  return (int)hipGetLastError();
+}
+
+int anovos_lut_apply_i32(const int32_t *const *cols, const int64_t *lens, int ncols,
+                         const int32_t *lutflat, const int64_t *lut_off, int nchunks,
+                         int32_t *const *outs, hipStream_t stream) {
+  hipLaunchKernelGGL(lut_apply_i32_kernel, dim3(ncols * nchunks), dim3(THREADS), 0, stream,
+                     cols, lens, lutflat, lut_off, nchunks, outs);
   return (int)hipGetLastError();
 }
 

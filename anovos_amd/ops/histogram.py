@@ -179,18 +179,20 @@ def approx_quantiles(
 
 
 def _refine_pass(tensors, cols, brackets, nbins):
-    """One narrowing pass: histogram each active bracket, all columns and
-    brackets in one fused launch (HIP) / vectorized torch loop (CPU)."""
+    """One narrowing pass: histogram each active bracket. GPU: ONE grouped
+    kernel launch reads every column once and serves all of its brackets
+    (512 sub-bins, rank resolution nbins*512 per pass). CPU: torch loop."""
     if not brackets:
         return brackets
-    keys = list(brackets.keys())
+    keys = sorted(brackets.keys())  # sorted by (col, prob) — grouped kernel needs col-major
     dev = tensors[0].device
     blo = torch.tensor([brackets[k][0] for k in keys], dtype=torch.float64)
     bhi = torch.tensor([brackets[k][1] for k in keys], dtype=torch.float64)
     colidx = torch.tensor([k[0] for k in keys], dtype=torch.int64)
     if dev.type == "cuda" and backend.use_hip(tensors[0]):
         ext = backend.hip_ext()
-        h = ext.bracket_histograms([t.contiguous() for t in tensors], colidx.to(dev), blo.to(dev), bhi.to(dev), nbins)
+        nbins = 512  # grouped kernel's fixed sub-bin count
+        h = ext.bracket_histograms_grouped([t.contiguous() for t in tensors], colidx, blo, bhi)
     else:
         h = torch.zeros(len(keys), nbins, dtype=torch.int64, device=dev)
         for kk, (i, j) in enumerate(keys):
