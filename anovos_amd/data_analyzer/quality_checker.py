@@ -131,10 +131,11 @@ def nullRows_detection(ctx, idf, list_of_cols="all", drop_cols=[], treatment=Fal
     else:
         hist = torch.bincount(key.to(torch.long), minlength=size).to(torch.float64)
     dist.all_reduce_(hist, "sum")
+    hist_l = hist.cpu().numpy().tolist()
     rows = []
     for nc in range(len(cols) + 1):
         for fl in (0, 1):
-            c = float(hist[nc * 2 + fl])
+            c = hist_l[nc * 2 + fl]
             if c > 0:
                 rows.append([nc, int(c), round(c / total, 4), fl])
     odf_print = pd.DataFrame(rows, columns=["null_cols_count", "row_count", "row_pct", "flagged"])

@@ -138,10 +138,15 @@ def mode_computation(ctx, idf, list_of_cols="all", drop_cols=[], print_impact=Fa
         warnings.warn("No Mode Computation - No discrete column(s) to analyze")
         return pd.DataFrame(columns=["attribute", "mode", "mode_rows"])
     cat_cols = [c for c in cols if idf.col(c).kind == "categorical"]
+    num_cols = [c for c in cols if idf.col(c).kind == "numerical"]
     counts = groupby_ops.cat_value_counts(idf, cat_cols) if cat_cols else {}
+    num_modes = groupby_ops.discrete_modes(idf, num_cols) if num_cols else {}
     rows = []
     for c in cols:
-        m, n = groupby_ops.mode(idf, c, counts.get(c))
+        if c in num_modes:
+            m, n = num_modes[c]
+        else:
+            m, n = groupby_ops.mode(idf, c, counts.get(c))
         rows.append([c, None if m is None else str(m), n])
     odf = pd.DataFrame(rows, columns=["attribute", "mode", "mode_rows"])
     if print_impact:

@@ -176,18 +176,18 @@ def batched_bin_frequencies(binned_idf, cols, total: int):
         tensors = [binned_idf.col(c).data for c in num_cols]
         lo = torch.full((len(num_cols),), 1.0, dtype=torch.float64)
         hi = torch.full((len(num_cols),), float(M + 1), dtype=torch.float64)
-        hist = hist_ops.global_histograms(tensors, lo, hi, M).cpu()
+        hist = hist_ops.global_histograms(tensors, lo, hi, M).cpu().numpy()
         for i, c in enumerate(num_cols):
             keys, vals = [], []
             nnull = int(total - moments[c].n)
             if nnull:
                 keys.append("-1")
                 vals.append(nnull / total)
-            h = hist[i]
-            for b in range(M):
-                if h[b] > 0:
+            h = hist[i].tolist()
+            for b, hb in enumerate(h):
+                if hb > 0:
                     keys.append(str(b + 1))
-                    vals.append(float(h[b]) / total)
+                    vals.append(hb / total)
             out[c] = (keys, vals)
     for c in cat_cols:
         out[c] = _bin_frequencies(binned_idf, c, total)
@@ -219,8 +219,8 @@ def _bin_frequencies(binned_idf, col: str, total: int):
     cnt = torch.bincount((iv + 1).clamp(min=0), minlength=mx + 2).to(torch.float64)
     dist.all_reduce_(cnt, "sum")
     keys, vals = [], []
-    for b in range(cnt.numel()):
-        if cnt[b] > 0:
+    for b, cb in enumerate(cnt.cpu().numpy().tolist()):
+        if cb > 0:
             keys.append(str(b - 1))
-            vals.append(float(cnt[b]) / total)
+            vals.append(cb / total)
     return keys, vals

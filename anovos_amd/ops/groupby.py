@@ -53,6 +53,48 @@ def cat_value_counts(idf, cols: List[str]) -> Dict[str, torch.Tensor]:
     return out
 
 
+def discrete_modes(idf, cols: List[str]) -> Dict[str, Tuple[Optional[float], int]]:
+    """Exact modes for integral numeric columns in ONE fused dense
+    histogram launch (no sorts). Columns that aren't integral or span a
+    range > 4096 fall back to per-column numeric_value_counts."""
+    from anovos_amd.ops import histogram as hist_ops
+    from anovos_amd.ops import stats as _stats
+
+    out: Dict[str, Tuple[Optional[float], int]] = {}
+    if not cols:
+        return out
+    moments = _stats.frame_moments(idf, cols)
+    dense, rest = [], []
+    for c in cols:
+        m = moments[c]
+        if m.n > 0 and m.integral and m.min == m.min and (m.max - m.min) < 4096:
+            dense.append(c)
+        else:
+            rest.append(c)
+    if dense:
+        import numpy as np
+
+        R = [int(moments[c].max - moments[c].min) + 1 for c in dense]
+        M = max(R)
+        tensors = [idf.col(c).data for c in dense]
+        lo = torch.tensor([moments[c].min for c in dense], dtype=torch.float64)
+        hi = torch.tensor([moments[c].min + r for c, r in zip(dense, R)], dtype=torch.float64)
+        hist = hist_ops.global_histograms(tensors, lo, hi, M).cpu().numpy()
+        for i, c in enumerate(dense):
+            idx = int(np.argmax(hist[i]))
+            cnt = int(hist[i][idx])
+            val = moments[c].min + int(idx * R[i] / M)
+            out[c] = (float(val), cnt) if cnt > 0 else (None, 0)
+    for c in rest:
+        vals, cnts = numeric_value_counts(idf, c)
+        if vals.numel() == 0:
+            out[c] = (None, 0)
+        else:
+            i = int(torch.argmax(cnts).item())
+            out[c] = (float(vals[i]), int(cnts[i]))
+    return out
+
+
 def mode(idf, col: str, counts: Optional[torch.Tensor] = None) -> Tuple[Optional[str], int]:
     """Global mode (value, count). Works for categorical (dictionary) and
     numeric (exact unique) columns. Ties: reference takes Spark's

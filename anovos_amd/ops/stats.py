@@ -180,8 +180,9 @@ def null_counts(idf, cols: List[str]) -> Dict[str, int]:
         for i, c in enumerate(other):
             local[i] = idf.col(c).null_mask().sum()
         dist.all_reduce_(local, "sum")
+        local_l = local.cpu().numpy().tolist()
         for i, c in enumerate(other):
-            idf.col(c).cache["nulls"] = int(local[i].item())
+            idf.col(c).cache["nulls"] = int(local_l[i])
     for c in cols:
         res[c] = idf.col(c).cache["nulls"]
     return res, total

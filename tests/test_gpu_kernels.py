@@ -38,7 +38,7 @@ def test_column_moments_vs_torch(ext, gpu_cols):
         td = t.cpu().to(torch.float64)
         valid = ~torch.isnan(td)
         x = td[valid]
-        n, s1, s2, s3, s4, mn, mx, zn = out[i].tolist()
+        n, s1, s2, s3, s4, mn, mx, zn, nf = out[i].tolist()
         assert n == int(valid.sum())
         assert abs(s1 - float(x.sum())) <= 1e-9 * max(abs(float(x.sum())), 1)
         assert abs(s2 - float((x * x).sum())) <= 1e-9 * float((x * x).sum())
