@@ -1,0 +1,105 @@
+"""Auto-detect latitude / longitude / geohash columns (reference parity:
+``anovos/data_ingest/geo_auto_detection.py`` :22-298).
+
+MI355X-native: the reference ran four separate Spark jobs *per column*
+plus per-row string UDFs. Here all numeric screening stats (max, mean,
+stddev, fractional-precision flag) come from fused tensor reductions on
+device; geohash candidacy checks run over the column *dictionary* on
+host (tiny), never over rows.
+"""
+
+from __future__ import annotations
+
+from typing import List, Tuple
+
+import torch
+
+from anovos_amd.core.frame import AnovosFrame
+from anovos_amd.data_transformer import geo_utils as gu
+
+
+def geo_to_latlong(ghs, option: int = 0):
+    """Decode geohash string(s) → lat/long (reference geo_auto_detection.py:101
+    via pygeohash; here the native bit-interleave decoder).
+    option 0 → latitude, 1 → longitude."""
+    single = isinstance(ghs, str)
+    lst = [ghs] if single else list(ghs)
+    if any(not gu.geohash_is_valid(str(g).lower()) for g in lst):
+        raise ValueError("invalid geohash")
+    ints, prec = gu.geohash_str_to_int([str(g).lower() for g in lst])
+    t = torch.from_numpy(ints)
+    lat, lon = gu.geohash_decode_int(t, prec)
+    out = lat if option == 0 else lon
+    vals = [float(v) for v in out]
+    return vals[0] if single else vals
+
+
+def latlong_to_geo(lat, long, precision: int = 9):
+    """Encode lat/long → geohash string(s) (reference geo_auto_detection.py:143)."""
+    single = not hasattr(lat, "__len__")
+    la = torch.as_tensor([lat] if single else list(lat), dtype=torch.float64)
+    lo = torch.as_tensor([long] if single else list(long), dtype=torch.float64)
+    gh = gu.geohash_encode_int(la, lo, precision=precision)
+    out = gu.geohash_int_to_str(gh, precision=precision)
+    return out[0] if single else out
+
+
+def _has_fraction(x: torch.Tensor) -> bool:
+    v = x[~torch.isnan(x)]
+    if v.numel() == 0:
+        return False
+    return bool(((v - v.trunc()).abs() > 1e-9).any())
+
+
+def ll_gh_cols(df: AnovosFrame, max_records: int = 100_000) -> Tuple[List[str], List[str], List[str]]:
+    """Detect (lat_cols, long_cols, gh_cols) with the reference's
+    heuristics (geo_auto_detection.py:177-298): name match first; else a
+    float column qualifies when it has decimal precision, |max| ≤ 90
+    (latitude) or 90 < |max| ≤ 180 (longitude), stddev ≥ 1 and
+    coefficient-of-variation < 1, and > 2 distinct values. String
+    columns of length 5-11 whose values all geohash-decode qualify as
+    geohash columns."""
+    lat_cols, long_cols, gh_cols = [], [], []
+    for name, dtype in df.dtypes:
+        c = df.col(name)
+        if c.kind == "numerical":
+            lname = name.lower()
+            if "latitude" in lname:
+                lat_cols.append(name)
+                continue
+            if "longitude" in lname:
+                long_cols.append(name)
+                continue
+            x = c.data.to(torch.float64)
+            v = x[~torch.isnan(x)]
+            if v.numel() < 3:
+                continue
+            mx = float(v.max())
+            mn_abs_max = float(v.abs().max())
+            sd = float(v.std())
+            mean = float(v.mean())
+            if not _has_fraction(x):
+                continue
+            if mn_abs_max > 180 or sd < 1 or mean == 0 or sd / abs(mean) >= 1:
+                continue
+            distinct = int(torch.unique(v).numel())
+            if distinct <= 2:
+                continue
+            if mn_abs_max <= 90:
+                lat_cols.append(name)
+            elif mn_abs_max <= 180:
+                long_cols.append(name)
+        elif c.kind == "categorical":
+            d = [s for s in (c.dictionary or []) if s][:max_records]
+            if not d:
+                continue
+            max_len = max(len(str(s)) for s in d)
+            if not (4 < max_len < 12):
+                continue
+            if len(set(d)) <= 2:
+                continue
+            if all(gu.geohash_is_valid(str(s).lower()) for s in d):
+                gh_cols.append(name)
+    if len(lat_cols) != len(long_cols):
+        lat_cols, long_cols = [], []
+    return lat_cols, long_cols, gh_cols
