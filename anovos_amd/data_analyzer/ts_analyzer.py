@@ -1,0 +1,207 @@
+"""Time-series inspection over detected timestamp columns (reference
+parity: ``anovos/data_analyzer/ts_analyzer.py`` :52-551, same on-disk
+contract: ``stats_<col>_{1,2}.csv`` and ``<ts>_<col>_<type>.csv`` under
+the output path).
+
+MI355X-native: unit extraction and lag-diff stats are int64 tensor ops
+(data_transformer/datetime.py); per-bucket aggregations run via
+scatter_reduce on the GPU; only the tiny result tables land on host.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import List, Optional
+
+import numpy as np
+import pandas as pd
+import torch
+
+from anovos_amd.core.dtypes import NULL_CODE, NULL_TS
+from anovos_amd.core.frame import AnovosFrame, Column
+from anovos_amd.data_transformer.datetime import (
+    US_PER_DAY,
+    _decompose,
+    _floor_day,
+    timeUnits_extraction,
+)
+from anovos_amd.shared.utils import attributeType_segregation, ends_with
+
+DAYPARTS = ["early_hours", "work_hours", "late_hours", "commuting_hours", "other_hours", "Missing_NA"]
+
+
+def daypart_cat(hour):
+    """Reference ts_analyzer.py:52 — scalar hour → daypart label."""
+    if hour is None:
+        return "Missing_NA"
+    if 4 <= hour < 7:
+        return "early_hours"
+    if 10 <= hour < 17:
+        return "work_hours"
+    if hour >= 23 or hour < 4:
+        return "late_hours"
+    if 7 <= hour < 10 or 17 <= hour < 20:
+        return "commuting_hours"
+    return "other_hours"
+
+
+def _daypart_codes(hh: torch.Tensor, null: torch.Tensor) -> torch.Tensor:
+    """Vectorized daypart bucketing (codes into DAYPARTS)."""
+    out = torch.full_like(hh, 4)  # other_hours
+    out = torch.where((hh >= 4) & (hh < 7), torch.zeros_like(out), out)
+    out = torch.where((hh >= 10) & (hh < 17), torch.ones_like(out), out)
+    out = torch.where((hh >= 23) | (hh < 4), torch.full_like(out, 2), out)
+    out = torch.where(((hh >= 7) & (hh < 10)) | ((hh >= 17) & (hh < 20)), torch.full_like(out, 3), out)
+    out = torch.where(null, torch.full_like(out, 5), out)
+    return out
+
+
+def ts_processed_feats(idf: AnovosFrame, col: str, id_col: str, tz: str = "local",
+                       cnt_row: Optional[int] = None, cnt_unique_id: Optional[int] = None) -> AnovosFrame:
+    """Reference ts_analyzer.py:87 — append unit columns + yyyymmdd_col,
+    daypart_cat, week_cat, dow."""
+    odf = timeUnits_extraction(idf, [col], "all", output_mode="append")
+    ts = idf.col(col).data
+    null = ts == NULL_TS
+    day = _floor_day(ts)
+    odf = odf.with_column("yyyymmdd_col", Column("yyyymmdd_col", "date",
+                                                 torch.where(null, torch.full_like(ts, NULL_TS), day * US_PER_DAY)))
+    hh = odf.col(col + "_hour").data
+    hh_null = torch.isnan(hh)
+    dp = _daypart_codes(torch.nan_to_num(hh).to(torch.int64), hh_null | null)
+    odf = odf.with_column("daypart_cat", Column("daypart_cat", "string", dp.to(torch.int32), list(DAYPARTS)))
+    dow = odf.col(col + "_dayofweek").data
+    wk = torch.where(torch.nan_to_num(dow) > 5, torch.ones_like(dow), torch.zeros_like(dow))
+    wk_codes = wk.to(torch.int32)
+    wk_codes = torch.where(null, torch.full_like(wk_codes, NULL_CODE), wk_codes)
+    odf = odf.with_column("week_cat", Column("week_cat", "string", wk_codes, ["weekday", "weekend"]))
+    odf = odf.rename({col + "_dayofweek": "dow"})
+    return odf
+
+
+def ts_eligiblity_check(ctx, idf: AnovosFrame, id_col: str, opt: int = 1, tz_offset: str = "local") -> pd.DataFrame:
+    """Reference ts_analyzer.py:160 — lag-1 date-diff stats over distinct
+    dates (opt=1) or percentile stats of id↔date pair counts (opt=2)."""
+    day = idf.col("yyyymmdd_col").data
+    valid = day != NULL_TS
+    days = torch.unique(day[valid])
+    if opt == 1:
+        if days.numel() > 1:
+            diffs = (days[1:] - days[:-1]).to(torch.float64) / US_PER_DAY
+            mean = float(diffs.mean())
+            var = float(diffs.var(unbiased=True)) if diffs.numel() > 1 else 0.0
+            sd = var ** 0.5
+            cov = sd / mean if mean else float("nan")
+        else:
+            mean = var = sd = cov = float("nan")
+        return pd.DataFrame({"mean": [mean], "variance": [var], "stdev": [sd], "coef_of_var_lag": [cov]})
+    # opt == 2: distribution of dates-per-id and ids-per-date
+    idc = idf.col(id_col)
+    id_codes = idc.data.to(torch.long) if idc.kind == "categorical" else torch.unique(idc.data, return_inverse=True)[1]
+    ok = valid & ~idc.null_mask()
+    pair = torch.stack([id_codes[ok].to(torch.float64), day[ok].to(torch.float64)], dim=1)
+    uniq_pair = torch.unique(pair, dim=0)
+    rows = []
+    for key_idx, name in ((0, "id_date_pair"), (1, "date_id_pair")):
+        keys = uniq_pair[:, key_idx]
+        _, counts = torch.unique(keys, return_counts=True)
+        c = counts.to(torch.float64)
+        qs = torch.quantile(c, torch.tensor([0.01, 0.05, 0.1, 0.25, 0.5, 0.75, 0.9, 0.95, 0.99], dtype=torch.float64))
+        rows.append([name, float(c.min()), *[float(q) for q in qs], float(c.max())])
+    return pd.DataFrame(rows, columns=["attribute", "min", "1%", "5%", "10%", "25%", "50%", "75%", "90%", "95%", "99%", "max"])
+
+
+def ts_viz_data(idf: AnovosFrame, x_col: str, y_col: str, output_mode="append", output_type="daily",
+                n_cat: int = 10, tz_offset="local") -> pd.DataFrame:
+    """Reference ts_analyzer.py:259 — per (bucket × column) aggregates:
+    counts for categorical y, min/max/mean/median for numeric y."""
+    key_map = {"daily": "yyyymmdd_col", "hourly": "daypart_cat", "weekly": "dow"}
+    k_col = key_map[output_type]
+    key = idf.col(k_col)
+    yc = idf.col(y_col)
+    if key.kind == "categorical":
+        knull = key.data == NULL_CODE
+        kcodes = key.data.to(torch.long)
+        klabels = key.dictionary
+    elif key.dtype in ("timestamp", "date"):
+        knull = key.data == NULL_TS
+        uniq, kcodes = torch.unique(key.data, return_inverse=True)
+        klabels = [str(pd.Timestamp(int(u), unit="us").date()) if int(u) != NULL_TS else None for u in uniq]
+    else:
+        knull = torch.isnan(key.data)
+        uniq, kcodes = torch.unique(torch.nan_to_num(key.data), return_inverse=True)
+        klabels = [float(u) for u in uniq]
+    G = len(klabels)
+    dev = kcodes.device
+    if yc.kind == "categorical":
+        # top-n_cat categories by count, then count per (bucket, cat)
+        ynull = yc.data == NULL_CODE
+        yv = yc.data.to(torch.long)
+        m = ~knull & ~ynull
+        cnt_y = torch.bincount(yv[m], minlength=len(yc.dictionary or []))
+        top = torch.argsort(cnt_y, descending=True)[:n_cat]
+        rows = []
+        for t in top.cpu().numpy():
+            if int(cnt_y[t]) == 0:
+                continue
+            sel = m & (yv == int(t))
+            per = torch.zeros(G, dtype=torch.float64, device=dev).scatter_reduce(
+                0, kcodes[sel], torch.ones(int(sel.sum()), dtype=torch.float64, device=dev), reduce="sum")
+            for g in range(G):
+                if float(per[g]) > 0:
+                    rows.append([yc.dictionary[int(t)], klabels[g], float(per[g])])
+        return pd.DataFrame(rows, columns=[y_col, k_col, "count"])
+    x = yc.data.to(torch.float64)
+    ynull = torch.isnan(x)
+    m = ~knull & ~ynull
+    xv, kv = x[m], kcodes[m]
+    cnt = torch.zeros(G, dtype=torch.float64, device=dev).scatter_reduce(0, kv, torch.ones_like(xv), reduce="sum")
+    s = torch.zeros(G, dtype=torch.float64, device=dev).scatter_reduce(0, kv, xv, reduce="sum")
+    mn = torch.full((G,), float("inf"), dtype=torch.float64, device=dev).scatter_reduce(0, kv, xv, reduce="amin")
+    mx = torch.full((G,), float("-inf"), dtype=torch.float64, device=dev).scatter_reduce(0, kv, xv, reduce="amax")
+    # median: sort values, stable-sort groups, middle of each segment
+    byval = torch.argsort(xv)
+    bygrp = torch.argsort(kv[byval], stable=True)
+    o = byval[bygrp]
+    gs, xs = kv[o].contiguous(), xv[o]
+    starts = torch.searchsorted(gs, torch.arange(G, device=dev))
+    mid = (starts + ((cnt.to(torch.long) - 1) // 2).clamp(min=0)).clamp(max=max(int(xs.shape[0]) - 1, 0))
+    med = xs[mid] if xs.numel() else torch.zeros(G, dtype=torch.float64)
+    pdf = pd.DataFrame(
+        {
+            k_col: klabels,
+            "min": mn.cpu().numpy(),
+            "max": mx.cpu().numpy(),
+            "mean": (s / cnt.clamp(min=1)).cpu().numpy(),
+            "median": med.cpu().numpy() if xs.numel() else np.zeros(G),
+        }
+    )
+    return pdf[cnt.cpu().numpy() > 0].reset_index(drop=True)
+
+
+def ts_analyzer(ctx, idf: AnovosFrame, id_col: str, max_days: int, output_path: str,
+                output_type: str = "daily", tz_offset: str = "local", run_type: str = "local", auth_key="NA"):
+    """Reference ts_analyzer.py:408 — driver loop: for each ts column
+    write eligibility stats (stats_<col>_{1,2}.csv) and per-attribute viz
+    aggregates (<ts>_<attr>_<type>.csv)."""
+    local_path = output_path if run_type == "local" else "report_stats"
+    os.makedirs(local_path, exist_ok=True)
+    num_cols, cat_cols, other = attributeType_segregation(idf)
+    num_cols = [x for x in num_cols if x != id_col]
+    cat_cols = [x for x in cat_cols if x != id_col]
+    ts_cols = [n for n, d in idf.dtypes if d in ("timestamp", "date")]
+    cnt_row = idf.count()
+    for i in ts_cols:
+        pdf_feats = ts_processed_feats(idf, i, id_col, tz_offset, cnt_row, None)
+        f1 = ts_eligiblity_check(ctx, pdf_feats, id_col, opt=1)
+        f1.to_csv(ends_with(local_path) + "stats_" + str(i) + "_1.csv", index=False)
+        f2 = ts_eligiblity_check(ctx, pdf_feats, id_col, opt=2)
+        f2.to_csv(ends_with(local_path) + "stats_" + str(i) + "_2.csv", index=False)
+        for cols in (num_cols, cat_cols):
+            for l in cols:
+                try:
+                    f = ts_viz_data(pdf_feats, i, l, output_type=output_type, tz_offset=tz_offset)
+                    f.to_csv(ends_with(local_path) + i + "_" + l + "_" + output_type + ".csv", index=False)
+                except Exception:
+                    continue
+    return ts_cols
