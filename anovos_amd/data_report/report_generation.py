@@ -1,0 +1,323 @@
+"""Final HTML report assembly (reference parity:
+``anovos/data_report/report_generation.py`` :3984-4413 — reads only
+*files on disk* under master_path: stats CSVs written by save_stats and
+plotly JSON chart objects written by charts_to_objects, then emits
+``ml_anovos_report.html``).
+
+The reference assembled tabs with datapane; that package is not in this
+stack, so the same tab structure (executive summary, wiki, descriptive
+statistics, quality check, attribute associations, drift & stability,
+time series, geospatial) is rendered by a small native HTML/CSS/JS
+template with plotly.js inlined — the report is a single offline file,
+as before. Chart JSONs and CSV names are consumed under the exact
+reference contract (report_generation.py:4111-4245), so artifacts
+produced by either engine interchange.
+"""
+
+from __future__ import annotations
+
+import html as _html
+import json
+import os
+from typing import Dict, List, Optional
+
+import numpy as np
+import pandas as pd
+
+import plotly.graph_objects as go
+import plotly.io as pio
+
+from anovos_amd.shared.utils import ends_with
+
+SG_tabs = [
+    "measures_of_counts", "measures_of_centralTendency", "measures_of_cardinality",
+    "measures_of_percentiles", "measures_of_dispersion", "measures_of_shape", "global_summary",
+]
+QC_tabs = [
+    "nullColumns_detection", "IDness_detection", "biasedness_detection",
+    "invalidEntries_detection", "duplicate_detection", "nullRows_detection", "outlier_detection",
+]
+AE_tabs = ["correlation_matrix", "IV_calculation", "IG_calculation", "variable_clustering"]
+drift_tab = ["drift_statistics"]
+stability_tab = ["stability_index", "stabilityIndex_metrics"]
+
+STABILITY_INTERPRETATION = pd.DataFrame(
+    [["0-1", "Very Unstable"], ["1-2", "Unstable"], ["2-3", "Marginally Stable"],
+     ["3-3.5", "Stable"], ["3.5-4", "Very Stable"]],
+    columns=["StabilityIndex", "StabilityOrder"],
+)
+
+_CSS = """
+body{font-family:'Segoe UI',Roboto,Helvetica,Arial,sans-serif;margin:0;background:#fafafa;color:#222}
+header{background:#273746;color:#fff;padding:18px 32px}
+header h1{margin:0;font-size:22px} header p{margin:4px 0 0;font-size:13px;color:#aeb6bf}
+.tabbar{display:flex;flex-wrap:wrap;background:#1f2e3d;padding:0 24px}
+.tabbar button{background:none;border:none;color:#d5dbdb;padding:12px 18px;font-size:14px;cursor:pointer;border-bottom:3px solid transparent}
+.tabbar button.active{color:#fff;border-bottom-color:#f5b041;font-weight:600}
+.tab{display:none;padding:24px 32px}.tab.active{display:block}
+h2{font-size:18px;border-bottom:2px solid #e5e8e8;padding-bottom:6px}
+h3{font-size:15px;color:#34495e;margin-top:28px}
+table.stats{border-collapse:collapse;font-size:12.5px;margin:8px 0;background:#fff}
+table.stats th{background:#273746;color:#fff;padding:6px 10px;text-align:left}
+table.stats td{border:1px solid #e5e8e8;padding:5px 10px}
+table.stats tr:nth-child(even){background:#f4f6f6}
+.grid{display:flex;flex-wrap:wrap;gap:16px}.grid>div{flex:1 1 540px;background:#fff;border:1px solid #eee;border-radius:6px;padding:6px}
+.kpi{display:inline-block;background:#fff;border:1px solid #e5e8e8;border-radius:8px;padding:14px 26px;margin:6px;text-align:center}
+.kpi .v{font-size:26px;font-weight:700;color:#273746}.kpi .l{font-size:12px;color:#7b8a8b}
+.note{font-size:12px;color:#7b8a8b}
+"""
+
+_JS = """
+function showTab(i){
+  var tabs=document.querySelectorAll('.tab');var btns=document.querySelectorAll('.tabbar button');
+  tabs.forEach(function(t,j){t.classList.toggle('active',i===j)});
+  btns.forEach(function(b,j){b.classList.toggle('active',i===j)});
+  window.dispatchEvent(new Event('resize'));
+}
+"""
+
+
+def _tbl(pdf: pd.DataFrame, max_rows: int = 200) -> str:
+    if pdf is None or len(pdf) == 0:
+        return "<p class='note'>No data.</p>"
+    p = pdf.head(max_rows).copy()
+    return p.to_html(index=False, classes="stats", border=0, float_format=lambda v: f"{v:.4f}")
+
+
+def _fig_div(fig_json_path_or_fig, div_id: str) -> str:
+    if isinstance(fig_json_path_or_fig, str):
+        with open(fig_json_path_or_fig) as f:
+            spec = json.load(f)
+    else:
+        spec = json.loads(pio.to_json(fig_json_path_or_fig))
+    data = json.dumps(spec.get("data", []))
+    layout = json.dumps(spec.get("layout", {}))
+    return (f"<div id='{div_id}' class='plt'></div>"
+            f"<script>Plotly.newPlot('{div_id}',{data},{layout},{{responsive:true}});</script>")
+
+
+def _read_csv(master_path: str, name: str) -> Optional[pd.DataFrame]:
+    p = ends_with(master_path) + name + ".csv"
+    if os.path.exists(p):
+        try:
+            return pd.read_csv(p)
+        except Exception:
+            return None
+    return None
+
+
+def _chart_files(master_path: str, prefix: str) -> List[str]:
+    return sorted(x for x in os.listdir(master_path)
+                  if x.startswith(prefix) and not x.endswith(".csv") and not x.endswith(".html"))
+
+
+def _charts_section(master_path: str, files: List[str], title: str, uid: str, limit: int = 60) -> str:
+    if not files:
+        return ""
+    out = [f"<h3>{_html.escape(title)}</h3><div class='grid'>"]
+    for i, fn in enumerate(files[:limit]):
+        try:
+            out.append("<div>" + _fig_div(os.path.join(master_path, fn), f"{uid}_{i}") + "</div>")
+        except Exception:
+            continue
+    out.append("</div>")
+    return "".join(out)
+
+
+def executive_summary_gen(master_path: str, label_col, event_label) -> str:
+    """Reference report_generation.py:524 — KPI header from global_summary."""
+    gs = _read_csv(master_path, "global_summary")
+    if gs is None:
+        return "<p class='note'>global_summary.csv not found.</p>"
+    get = lambda m: gs[gs["metric"] == m]["value"].values
+    kpis = []
+    for m, label in (("rows_count", "Rows"), ("columns_count", "Columns"),
+                     ("numcols_count", "Numerical Columns"), ("catcols_count", "Categorical Columns")):
+        v = get(m)
+        kpis.append(f"<div class='kpi'><div class='v'>{_html.escape(str(v[0])) if len(v) else '—'}</div><div class='l'>{label}</div></div>")
+    lab = f"<p>Label column: <b>{_html.escape(str(label_col))}</b>, event label: <b>{_html.escape(str(event_label))}</b></p>" if label_col else ""
+    return "".join(kpis) + lab
+
+
+def wiki_generator(master_path: str, dataDict_path=None, metricDict_path=None) -> str:
+    """Reference report_generation.py:909 — data dictionary + metric
+    definitions tab."""
+    parts = []
+    for p, title in ((dataDict_path, "Data Dictionary"), (metricDict_path, "Metric Dictionary")):
+        if p and os.path.exists(p):
+            try:
+                parts.append(f"<h3>{title}</h3>" + _tbl(pd.read_csv(p), 500))
+            except Exception:
+                pass
+    if not parts:
+        parts.append("<p class='note'>No data dictionary / metric dictionary supplied.</p>")
+    return "".join(parts)
+
+
+def descriptive_statistics(master_path: str) -> str:
+    """Reference report_generation.py:994."""
+    parts = []
+    for name in SG_tabs:
+        df = _read_csv(master_path, name)
+        if df is not None:
+            parts.append(f"<h3>{name}</h3>" + _tbl(df))
+    parts.append(_charts_section(master_path, _chart_files(master_path, "freqDist_"), "Frequency Distributions", "fd"))
+    return "".join(parts) if parts else "<p class='note'>No descriptive statistics saved.</p>"
+
+
+def quality_check(master_path: str) -> str:
+    """Reference report_generation.py:1154."""
+    parts = []
+    for name in QC_tabs:
+        df = _read_csv(master_path, name)
+        if df is not None:
+            parts.append(f"<h3>{name}</h3>" + _tbl(df))
+    parts.append(_charts_section(master_path, _chart_files(master_path, "outlier_"), "Outlier Charts", "oc"))
+    return "".join(parts) if parts else "<p class='note'>No quality-check statistics saved.</p>"
+
+
+def attribute_associations(master_path: str, label_col, event_label) -> str:
+    """Reference report_generation.py:1291 — correlation heatmap, IV/IG
+    bars, variable clustering table + event-rate charts."""
+    parts = []
+    corr = _read_csv(master_path, "correlation_matrix")
+    if corr is not None:
+        attrs = list(corr.get("attribute", corr.columns))
+        mat = corr.drop(columns=["attribute"], errors="ignore")
+        fig = go.Figure(go.Heatmap(z=mat.values, x=list(mat.columns), y=attrs, colorscale="Peach", zmin=-1, zmax=1))
+        fig.update_layout(title="Correlation Matrix", height=520)
+        parts.append("<h3>Correlation Matrix</h3>" + _fig_div(fig, "corrheat"))
+    for name, metric in (("IV_calculation", "iv"), ("IG_calculation", "ig")):
+        df = _read_csv(master_path, name)
+        if df is not None:
+            vcol = [c for c in df.columns if c != "attribute"][0]
+            fig = go.Figure(go.Bar(x=df["attribute"], y=df[vcol]))
+            fig.update_layout(title=name, height=380)
+            parts.append(f"<h3>{name}</h3>" + _fig_div(fig, f"bar_{metric}") + _tbl(df))
+    vc = _read_csv(master_path, "variable_clustering")
+    if vc is not None:
+        parts.append("<h3>Variable Clustering</h3>" + _tbl(vc))
+    parts.append(_charts_section(master_path, _chart_files(master_path, "eventDist_"), "Event-Rate Distributions", "ev"))
+    return "".join(parts) if parts else "<p class='note'>No association statistics saved.</p>"
+
+
+def data_drift_stability(master_path: str, drift_threshold_model=0.1) -> str:
+    """Reference report_generation.py:1434."""
+    parts = []
+    dd = _read_csv(master_path, "drift_statistics")
+    if dd is not None:
+        parts.append("<h3>Drift Statistics</h3>" + _tbl(dd))
+        if "flagged" in dd.columns:
+            n_drift = int(pd.to_numeric(dd["flagged"], errors="coerce").fillna(0).sum())
+            parts.insert(0, f"<div class='kpi'><div class='v'>{n_drift}</div><div class='l'>Drifted Attributes</div></div>")
+    parts.append(_charts_section(master_path, [x for x in _chart_files(master_path, "drift_") if x != "drift_statistics"], "Source vs Target Distributions", "dr"))
+    si = _read_csv(master_path, "stability_index")
+    if si is None:
+        si = _read_csv(master_path, "stabilityIndex_metrics")
+    if si is not None:
+        parts.append("<h3>Stability Index</h3>" + _tbl(si))
+        parts.append("<h3>Stability Interpretation</h3>" + _tbl(STABILITY_INTERPRETATION))
+    return "".join(parts) if parts else "<p class='note'>No drift / stability statistics saved.</p>"
+
+
+def ts_viz_generate(master_path: str) -> str:
+    """Reference report_generation.py:3091 — time-series tab from
+    ts_analyzer CSVs (stats_<col>_{1,2}.csv + <ts>_<attr>_<type>.csv)."""
+    files = [x for x in os.listdir(master_path) if x.startswith("stats_") and x.endswith(".csv")]
+    if not files and not os.path.exists(ends_with(master_path) + "ts_cols_stats.csv"):
+        return ""
+    parts = []
+    tcs = _read_csv(master_path, "ts_cols_stats")
+    if tcs is not None:
+        parts.append("<h3>Detected Timestamp Columns</h3>" + _tbl(tcs))
+    for fn in sorted(files):
+        try:
+            parts.append(f"<h3>{fn[:-4]}</h3>" + _tbl(pd.read_csv(os.path.join(master_path, fn))))
+        except Exception:
+            continue
+    viz = [x for x in os.listdir(master_path)
+           if x.endswith(("_daily.csv", "_weekly.csv", "_hourly.csv"))]
+    for fn in sorted(viz)[:40]:
+        try:
+            df = pd.read_csv(os.path.join(master_path, fn))
+            xcol = df.columns[0]
+            fig = go.Figure()
+            for c in [c for c in df.columns[1:] if pd.api.types.is_numeric_dtype(df[c])]:
+                fig.add_trace(go.Scatter(x=df[xcol].astype(str), y=df[c], mode="lines+markers", name=c))
+            fig.update_layout(title=fn[:-4], height=360)
+            parts.append(_fig_div(fig, "ts_" + fn.replace(".", "_")))
+        except Exception:
+            continue
+    return "".join(parts)
+
+
+def loc_report_gen(master_path: str) -> str:
+    """Reference report_generation.py:3902 — geospatial tab from
+    geospatial_analyzer outputs."""
+    files = os.listdir(master_path)
+    geo_csvs = [x for x in files if x.startswith(("Overall_Summary", "Top_", "cluster_output"))]
+    geo_charts = [x for x in files if x.startswith(("cluster_plot", "loc_charts"))]
+    if not geo_csvs and not geo_charts:
+        return ""
+    parts = []
+    for fn in sorted(geo_csvs)[:20]:
+        try:
+            parts.append(f"<h3>{fn[:-4]}</h3>" + _tbl(pd.read_csv(os.path.join(master_path, fn))))
+        except Exception:
+            continue
+    parts.append(_charts_section(master_path, sorted(geo_charts), "Geospatial Charts", "geo"))
+    return "".join(parts)
+
+
+def anovos_report(master_path: str, id_col="", label_col=None, corr_threshold=0.4,
+                  iv_threshold=0.02, drift_threshold_model=0.1, dataDict_path=".",
+                  metricDict_path=".", final_report_path=".", event_label=1,
+                  run_type="local", auth_key="NA", output_type=None,
+                  sg_print_impact=False, **kwargs) -> str:
+    """Reference report_generation.py:3984 — assemble ml_anovos_report.html
+    from the on-disk stats + chart objects. Returns the report path."""
+    os.makedirs(final_report_path, exist_ok=True)
+    tabs = [
+        ("Executive Summary", executive_summary_gen(master_path, label_col, event_label)),
+        ("Wiki", wiki_generator(master_path, dataDict_path if dataDict_path != "." else None,
+                                metricDict_path if metricDict_path != "." else None)),
+        ("Descriptive Statistics", descriptive_statistics(master_path)),
+        ("Quality Check", quality_check(master_path)),
+        ("Attribute Associations", attribute_associations(master_path, label_col, event_label)),
+        ("Data Drift & Data Stability", data_drift_stability(master_path, drift_threshold_model)),
+    ]
+    ts = ts_viz_generate(master_path)
+    if ts:
+        tabs.append(("Time Series", ts))
+    geo = loc_report_gen(master_path)
+    if geo:
+        tabs.append(("Geospatial", geo))
+    out = render_report(tabs, title="ML-Anovos Report")
+    path = ends_with(final_report_path) + "ml_anovos_report.html"
+    with open(path, "w") as f:
+        f.write(out)
+    return path
+
+
+def render_report(tabs, title="ML-Anovos Report") -> str:
+    """Single-file offline HTML with plotly.js inlined."""
+    from plotly.offline import get_plotlyjs
+
+    btns = "".join(
+        f"<button class='{'active' if i == 0 else ''}' onclick='showTab({i})'>{_html.escape(name)}</button>"
+        for i, (name, _) in enumerate(tabs)
+    )
+    bodies = "".join(
+        f"<div class='tab {'active' if i == 0 else ''}'><h2>{_html.escape(name)}</h2>{content}</div>"
+        for i, (name, content) in enumerate(tabs)
+    )
+    return (
+        "<!DOCTYPE html><html><head><meta charset='utf-8'>"
+        f"<title>{_html.escape(title)}</title>"
+        f"<style>{_CSS}</style>"
+        f"<script>{get_plotlyjs()}</script>"
+        f"<script>{_JS}</script></head><body>"
+        f"<header><h1>{_html.escape(title)}</h1>"
+        "<p>Generated by anovos_amd — MI355X-native feature engineering engine</p></header>"
+        f"<div class='tabbar'>{btns}</div>{bodies}</body></html>"
+    )
