@@ -64,31 +64,29 @@ def duplicate_detection(ctx, idf, list_of_cols="all", drop_cols=[], treatment=Fa
 
     h = row_hash(idf.select(cols), cols)
     idf_count = idf.count()
-    if treatment or print_impact:
-        # keep first occurrence of each hash locally; cross-rank dedup by
-        # hash ownership (hash % world owns the row)
-        uniq, first_idx = _unique_first(h)
-        if dist.is_dist():
-            owner = (uniq % dist.world_size() + dist.world_size()) % dist.world_size() == dist.rank()
-            keep_idx = first_idx[owner]
-        else:
-            keep_idx = first_idx
-        odf_tmp = idf.filter_rows(keep_idx)
-        odf_tmp_count = odf_tmp.count()
+    # keep first occurrence of each hash locally; cross-rank dedup by
+    # hash ownership (hash % world owns the row)
+    uniq, first_idx = _unique_first(h)
+    if dist.is_dist():
+        owner = (uniq % dist.world_size() + dist.world_size()) % dist.world_size() == dist.rank()
+        keep_idx = first_idx[owner]
+    else:
+        keep_idx = first_idx
+    odf_tmp = idf.filter_rows(keep_idx)
+    odf_tmp_count = odf_tmp.count()
     odf = odf_tmp if treatment else idf
+    odf_print = pd.DataFrame(
+        [
+            ["rows_count", float(idf_count)],
+            ["unique_rows_count", float(odf_tmp_count)],
+            ["duplicate_rows", float(idf_count - odf_tmp_count)],
+            ["duplicate_pct", round((idf_count - odf_tmp_count) / idf_count, 4)],
+        ],
+        columns=["metric", "value"],
+    )
     if print_impact:
-        odf_print = pd.DataFrame(
-            [
-                ["rows_count", float(idf_count)],
-                ["unique_rows_count", float(odf_tmp_count)],
-                ["duplicate_rows", float(idf_count - odf_tmp_count)],
-                ["duplicate_pct", round((idf_count - odf_tmp_count) / idf_count, 4)],
-            ],
-            columns=["metric", "value"],
-        )
         print(odf_print.to_string(index=False))
-        return odf, odf_print
-    return odf
+    return odf, odf_print
 
 
 def _unique_first(h: torch.Tensor):
@@ -280,10 +278,11 @@ def outlier_detection(
     num_cols = attributeType_segregation(idf)[0]
     treatment = _parse_bool(treatment)
     pre_existing_model = _parse_bool(pre_existing_model, "pre_existing_model")
+    _empty_stats = pd.DataFrame(columns=["attribute", "lower_outliers", "upper_outliers", "excluded_due_to_skewness"])
     if not treatment and not print_impact:
         if (not pre_existing_model and model_path == "NA") or pre_existing_model:
             warnings.warn("The original idf will be the only output. Set print_impact=True to perform detection without treatment")
-            return idf
+            return idf, _empty_stats
     if list_of_cols == "all":
         list_of_cols = num_cols
     list_of_cols = normalize_columns(idf, list_of_cols, drop_cols, restrict_to=num_cols)
@@ -401,7 +400,7 @@ def outlier_detection(
             )
             _save_model(dfm, model_path, "outlier_numcols")
             if not treatment and not print_impact:
-                return idf
+                return idf, _empty_stats
 
     odf = idf
     rows_print = []
@@ -414,10 +413,9 @@ def outlier_detection(
         if detection_side in ("upper", "both") and hi is not None:
             flag = torch.where((x > hi) & ~torch.isnan(x), torch.ones_like(flag), flag)
         flags[c] = flag
-        if print_impact:
-            lower_n = int(dist.all_reduce_scalar(int((flag == -1).sum())))
-            upper_n = int(dist.all_reduce_scalar(int((flag == 1).sum())))
-            rows_print.append([c, lower_n, upper_n, 0])
+        lower_n = int(dist.all_reduce_scalar(int((flag == -1).sum())))
+        upper_n = int(dist.all_reduce_scalar(int((flag == 1).sum())))
+        rows_print.append([c, lower_n, upper_n, 0])
         if treatment and treatment_method in ("value_replacement", "null_replacement"):
             if treatment_method == "value_replacement":
                 lo_v = float(lo) if lo is not None else float("nan")
@@ -438,11 +436,10 @@ def outlier_detection(
         odf = odf.select([c for c in column_order if c in odf.columns])
     if not treatment:
         odf = idf
+    odf_print = pd.DataFrame(rows_print + [[c, 0, 0, 1] for c in skewed_cols], columns=["attribute", "lower_outliers", "upper_outliers", "excluded_due_to_skewness"])
     if print_impact:
-        odf_print = pd.DataFrame(rows_print + [[c, 0, 0, 1] for c in skewed_cols], columns=["attribute", "lower_outliers", "upper_outliers", "excluded_due_to_skewness"])
         print(odf_print.to_string(index=False))
-        return odf, odf_print
-    return odf
+    return odf, odf_print
 
 
 def IDness_detection(ctx, idf, list_of_cols="all", drop_cols=[], treatment=False, treatment_threshold=0.8, stats_unique={}, print_impact=False):

@@ -103,7 +103,7 @@ def pipeline_step(ctx, idf, source_hist, model_dir):
     _, nullrows = qc.nullRows_detection(ctx, idf, treatment=False)  # K10 row scan
     chk += float(nullrows["row_count"].sum())
     _, biased = qc.biasedness_detection(ctx, idf, int_cols + cat_cols, treatment=False, treatment_threshold=0.9)
-    odf_out = qc.outlier_detection(
+    odf_out, _ = qc.outlier_detection(
         ctx, idf, num_cols, detection_side="both", treatment=True, treatment_method="value_replacement"
     )
     chk += float(odf_out.col(num_cols[0]).data[:8].float().nansum().item())

@@ -75,8 +75,8 @@ def test_outlier_model_roundtrip(ctx, tmp_path):
     rng = np.random.default_rng(1)
     f = AnovosFrame.from_pandas(pd.DataFrame({"x": rng.normal(0, 1, 2000)}))
     mp = str(tmp_path)
-    odf1 = qc.outlier_detection(ctx, f, ["x"], detection_side="both", model_path=mp)
-    odf2 = qc.outlier_detection(ctx, f, ["x"], detection_side="both", pre_existing_model=True, model_path=mp)
+    odf1, _ = qc.outlier_detection(ctx, f, ["x"], detection_side="both", model_path=mp, print_impact=True)
+    odf2, _ = qc.outlier_detection(ctx, f, ["x"], detection_side="both", pre_existing_model=True, model_path=mp, print_impact=True)
     assert torch.allclose(odf1.col("x").data, odf2.col("x").data, equal_nan=True)
 
 

@@ -1,0 +1,129 @@
+"""End-to-end workflow test: YAML config → full pipeline → HTML report
+(reference parity: the full-demo CI workflow running configs.yaml;
+here a compact income-like synthetic dataset on CPU)."""
+
+import os
+
+import numpy as np
+import pandas as pd
+import pytest
+import yaml
+
+from anovos_amd import workflow
+
+
+@pytest.fixture
+def income_csv(tmp_path):
+    rng = np.random.default_rng(7)
+    n = 3000
+    pdf = pd.DataFrame(
+        {
+            "ifa": [f"id_{i}" for i in range(n)],
+            "age": rng.integers(17, 90, n),
+            "workclass": rng.choice(["Private", "Self-emp", "Gov", "Other"], n),
+            "fnlwgt": rng.integers(10000, 1000000, n),
+            "education": rng.choice(["HS-grad", "Bachelors", "Masters", "Doctorate", "Some-college"], n),
+            "hours-per-week": rng.integers(1, 99, n),
+            "capital-gain": np.where(rng.random(n) < 0.9, 0, rng.integers(1, 99999, n)),
+            "income": rng.choice(["<=50K", ">50K"], n, p=[0.76, 0.24]),
+        }
+    )
+    p = tmp_path / "income.csv"
+    pdf.to_csv(p, index=False)
+    return str(p)
+
+
+def make_config(income_csv, tmp_path):
+    out = str(tmp_path / "out")
+    rep = str(tmp_path / "report")
+    return {
+        "input_dataset": {
+            "read_dataset": {"file_path": income_csv, "file_type": "csv",
+                             "file_configs": {"header": True, "inferSchema": True}},
+            "delete_column": ["fnlwgt"],
+        },
+        "stats_generator": {
+            "metric": ["global_summary", "measures_of_counts", "measures_of_centralTendency",
+                       "measures_of_cardinality", "measures_of_dispersion", "measures_of_percentiles",
+                       "measures_of_shape"],
+            "metric_args": {"list_of_cols": "all", "drop_cols": ["ifa"]},
+        },
+        "quality_checker": {
+            "duplicate_detection": {"list_of_cols": "all", "drop_cols": ["ifa"], "treatment": True},
+            "nullColumns_detection": {"list_of_cols": "all", "drop_cols": ["ifa", "income"],
+                                       "treatment": True, "treatment_method": "MMM"},
+            "biasedness_detection": {"list_of_cols": "all", "drop_cols": ["ifa", "income"], "treatment": False},
+        },
+        "association_evaluator": {
+            "IV_calculation": {"list_of_cols": "all", "drop_cols": ["ifa"],
+                               "label_col": "income", "event_label": ">50K"},
+        },
+        "transformers": {
+            "numerical_mathops": {
+                "attribute_binning": {"list_of_cols": ["age", "hours-per-week"],
+                                       "method_type": "equal_range", "bin_size": 10,
+                                       "output_mode": "append"},
+            },
+            "numerical_rescaling": {
+                "z_standardization": {"list_of_cols": ["capital-gain"], "output_mode": "append"},
+            },
+        },
+        "report_preprocessing": {
+            "master_path": rep,
+            "charts_to_objects": {"list_of_cols": "all", "drop_cols": ["ifa"],
+                                   "label_col": "income", "event_label": ">50K",
+                                   "bin_method": "equal_frequency", "bin_size": 10,
+                                   "source_path": str(tmp_path / "inter")},
+        },
+        "report_generation": {
+            "master_path": rep,
+            "final_report_path": rep,
+            "label_col": "income",
+            "event_label": ">50K",
+        },
+        "write_main": {"file_path": out, "file_type": "parquet"},
+        "write_stats": {"file_path": out, "file_type": "csv",
+                        "file_configs": {"header": True, "mode": "overwrite"}},
+    }
+
+
+def test_workflow_end_to_end(income_csv, tmp_path):
+    cfg = make_config(income_csv, tmp_path)
+    cfg_path = tmp_path / "cfg.yaml"
+    with open(cfg_path, "w") as f:
+        yaml.safe_dump(cfg, f, sort_keys=False)  # stage order = YAML order
+    df = workflow.run(str(cfg_path))
+    assert df is not None
+    assert "fnlwgt" not in df.columns
+    assert "age_binned" in df.columns
+    assert "capital-gain_scaled" in df.columns
+    rep = tmp_path / "report"
+    assert (rep / "global_summary.csv").exists()
+    assert (rep / "measures_of_counts.csv").exists()
+    assert (rep / "nullColumns_detection.csv").exists()
+    assert (rep / "IV_calculation.csv").exists()
+    assert (rep / "ml_anovos_report.html").exists()
+    html = open(rep / "ml_anovos_report.html").read()
+    assert "Attribute Associations" in html
+    # main dataset persisted
+    assert any("final_dataset" in d for d in os.listdir(tmp_path / "out"))
+
+
+def test_workflow_basic_report_short_circuit(income_csv, tmp_path):
+    cfg = {
+        "input_dataset": {
+            "read_dataset": {"file_path": income_csv, "file_type": "csv",
+                             "file_configs": {"header": True, "inferSchema": True}},
+        },
+        "anovos_basic_report": {
+            "basic_report": True,
+            "report_args": {"id_col": "ifa", "label_col": "income", "event_label": ">50K",
+                            "output_path": str(tmp_path / "basic")},
+        },
+        "stats_generator": {"metric": ["global_summary"], "metric_args": {}},
+    }
+    cfg_path = tmp_path / "cfg_basic.yaml"
+    with open(cfg_path, "w") as f:
+        yaml.safe_dump(cfg, f)
+    workflow.run(str(cfg_path))
+    assert (tmp_path / "basic" / "basic_report.html").exists()
