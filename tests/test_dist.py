@@ -1,0 +1,154 @@
+"""Multi-process distributed correctness (gloo, world_size=2 on CPU).
+
+The reference had no distributed tests (Spark semantics assumed —
+SURVEY.md §4); the rebuild must verify its RCCL partial-aggregate merge
+protocol explicitly. Each test spawns 2 processes over gloo, shards a
+frame by rank, runs the distributed op, and compares rank-0's result to
+the single-process result on the unsharded frame."""
+
+import json
+import multiprocessing as mp
+import os
+import socket
+import sys
+import tempfile
+
+import numpy as np
+import pandas as pd
+import pytest
+import torch
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def _make_pdf(n=4000, seed=0):
+    rng = np.random.default_rng(seed)
+    return pd.DataFrame(
+        {
+            "x": rng.normal(5, 2, n),
+            "y": rng.lognormal(1, 0.5, n),
+            "cat": rng.choice(["a", "b", "c", "d"], n, p=[0.4, 0.3, 0.2, 0.1]),
+            "label": rng.choice(["0", "1"], n, p=[0.7, 0.3]),
+        }
+    )
+
+
+def _worker(rank, world, port, fn_name, out_path):
+    os.environ.update(
+        {
+            "RANK": str(rank),
+            "LOCAL_RANK": str(rank),
+            "WORLD_SIZE": str(world),
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        }
+    )
+    import torch.distributed as td
+
+    from anovos_amd.core import dist
+    from anovos_amd.core.frame import AnovosFrame
+    from anovos_amd.shared.context import init_context
+
+    dist.init_from_env(timeout_s=120)
+    ctx = init_context("cpu")
+    pdf = _make_pdf()
+    half = len(pdf) // 2
+    shard = pdf.iloc[rank * half : (rank + 1) * half]
+    idf = AnovosFrame.from_pandas(shard.reset_index(drop=True), device="cpu")
+
+    result = _FNS[fn_name](ctx, idf)
+    if rank == 0:
+        with open(out_path, "w") as f:
+            json.dump(result, f)
+    td.barrier()
+    td.destroy_process_group()
+
+
+def _fn_count_moments(ctx, idf):
+    from anovos_amd.data_analyzer import stats_generator as sg
+
+    disp = sg.measures_of_dispersion(ctx, idf)
+    counts = sg.measures_of_counts(ctx, idf)
+    return {
+        "rows": idf.count(),
+        "stddev_x": float(disp[disp["attribute"] == "x"]["stddev"].iloc[0]),
+        "mean_fill": float(counts[counts["attribute"] == "x"]["fill_count"].iloc[0]),
+    }
+
+
+def _fn_quantiles_mode(ctx, idf):
+    from anovos_amd.data_analyzer import stats_generator as sg
+
+    pct = sg.measures_of_percentiles(ctx, idf)
+    ct = sg.measures_of_centralTendency(ctx, idf)
+    return {
+        "p50_x": float(pct[pct["attribute"] == "x"]["50%"].iloc[0]),
+        "mode_cat": str(ct[ct["attribute"] == "cat"]["mode"].iloc[0]),
+        "unique_cat": float(sg.uniqueCount_computation(ctx, idf)["unique_values"].iloc[2]),
+    }
+
+
+def _fn_drift(ctx, idf):
+    from anovos_amd.drift_stability import drift_detector as dd
+
+    # target = slightly shifted copy of the local shard
+    shifted = idf.copy()
+    from anovos_amd.core.frame import Column
+
+    x = shifted.col("x").data + 0.5
+    shifted = shifted.with_column("x", Column("x", "float", x))
+    # model_path must be SHARED across ranks (rank 0 writes, all read)
+    shared = os.path.join(tempfile.gettempdir(), f"anovos_dist_drift_{os.environ.get('MASTER_PORT', '0')}")
+    stats = dd.statistics(ctx, shifted, idf, list_of_cols=["x", "y"], method_type="all",
+                          use_sampling=False, model_directory=shared + "/drift")
+    row = stats[stats["attribute"] == "x"].iloc[0]
+    return {"psi_x": float(row["PSI"]), "flagged": int(pd.to_numeric(stats["flagged"]).sum())}
+
+
+_FNS = {
+    "count_moments": _fn_count_moments,
+    "quantiles_mode": _fn_quantiles_mode,
+    "drift": _fn_drift,
+}
+
+
+def _run_dist(fn_name):
+    port = _free_port()
+    out = tempfile.NamedTemporaryFile(suffix=".json", delete=False).name
+    procs = []
+    mp_ctx = mp.get_context("spawn")
+    for r in range(2):
+        p = mp_ctx.Process(target=_worker, args=(r, 2, port, fn_name, out))
+        p.start()
+        procs.append(p)
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0, f"worker failed: exit {p.exitcode}"
+    with open(out) as f:
+        return json.load(f)
+
+
+def _run_single(fn_name):
+    from anovos_amd.core.frame import AnovosFrame
+    from anovos_amd.shared.context import init_context
+
+    ctx = init_context("cpu")
+    idf = AnovosFrame.from_pandas(_make_pdf(), device="cpu")
+    return _FNS[fn_name](ctx, idf)
+
+
+@pytest.mark.parametrize("fn_name", ["count_moments", "quantiles_mode", "drift"])
+def test_dist_matches_single(fn_name):
+    dist_res = _run_dist(fn_name)
+    single = _run_single(fn_name)
+    for k, v in single.items():
+        if isinstance(v, (int, float)):
+            assert dist_res[k] == pytest.approx(v, rel=2e-2, abs=1e-6), (k, dist_res[k], v)
+        else:
+            assert dist_res[k] == v, (k, dist_res[k], v)
