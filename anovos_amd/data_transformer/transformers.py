@@ -258,6 +258,11 @@ def cat_to_num_unsupervised(
     if len(list_of_cols) == 0:
         warnings.warn("No Encoding Computation - No categorical column(s) to transform")
         return idf
+    # reference accepts 1 = label encoding, 0 = one-hot (transformers.py:506)
+    if method_type == 1:
+        method_type = "label_encoding"
+    elif method_type == 0:
+        method_type = "onehot_encoding"
     if method_type not in ("label_encoding", "onehot_encoding"):
         raise TypeError("Invalid input for method_type")
 

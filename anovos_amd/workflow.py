@@ -279,7 +279,10 @@ def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, de
                     mod = transformers if hasattr(transformers, subkey2) else transformers_advanced
                     f = getattr(mod, subkey2)
                     extra = stats_args(all_configs, subkey2)
-                    df = f(ctx, df, **value2, **extra)
+                    import inspect
+
+                    first = next(iter(inspect.signature(f).parameters))
+                    df = f(ctx, df, **value2, **extra) if first == "ctx" else f(df, **value2, **extra)
                     new = save(df, write_intermediate, "data_transformer/" + subkey2,
                                reread=bool(write_intermediate))
                     if new is not None:

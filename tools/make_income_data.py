@@ -1,0 +1,73 @@
+#!/usr/bin/env python3
+"""Generate a synthetic income-style dataset (csv + parquet) for the
+example configs under config/ (the reference shipped a real income
+dataset under data/; no datasets are fetchable in this stack)."""
+
+import argparse
+import os
+
+import numpy as np
+import pandas as pd
+
+
+def make(n: int, seed: int = 11) -> pd.DataFrame:
+    rng = np.random.default_rng(seed)
+    edu = rng.choice(["HS-grad", "Some-college", "Bachelors", "Masters", "Doctorate"], n,
+                     p=[0.35, 0.25, 0.22, 0.13, 0.05])
+    edu_num = pd.Series(edu).map({"HS-grad": 9, "Some-college": 10, "Bachelors": 13,
+                                  "Masters": 14, "Doctorate": 16}).to_numpy()
+    age = rng.integers(17, 90, n)
+    hours = np.clip(rng.normal(40, 12, n), 1, 99).round()
+    gain = np.where(rng.random(n) < 0.92, 0, rng.integers(114, 99999, n))
+    loss = np.where(rng.random(n) < 0.95, 0, rng.integers(155, 4356, n))
+    score = (age / 90 + edu_num / 16 + hours / 99 + (gain > 0) * 0.8 + rng.normal(0, 0.35, n))
+    income = np.where(score > 1.75, ">50K", "<=50K")
+    df = pd.DataFrame(
+        {
+            "ifa": [f"27520a:{i:07d}" for i in range(n)],
+            "age": age.astype(float),
+            "workclass": rng.choice(["Private", "Self-emp-not-inc", "Local-gov", "State-gov", "Federal-gov"], n,
+                                    p=[0.7, 0.1, 0.08, 0.07, 0.05]),
+            "fnlwgt": rng.integers(12285, 1484705, n),
+            "logfnl": rng.normal(11.7, 0.6, n),
+            "education": edu,
+            "education-num": edu_num.astype(float),
+            "marital-status": rng.choice(["Married-civ-spouse", "Never-married", "Divorced", "Widowed"], n,
+                                         p=[0.46, 0.33, 0.15, 0.06]),
+            "occupation": rng.choice(["Prof-specialty", "Craft-repair", "Exec-managerial", "Adm-clerical",
+                                      "Sales", "Other-service"], n),
+            "relationship": rng.choice(["Husband", "Not-in-family", "Own-child", "Unmarried", "Wife"], n,
+                                       p=[0.4, 0.26, 0.15, 0.11, 0.08]),
+            "race": rng.choice(["White", "Black", "Asian-Pac-Islander", "Other"], n, p=[0.85, 0.1, 0.03, 0.02]),
+            "sex": rng.choice(["Male", "Female"], n, p=[0.67, 0.33]),
+            "capital-gain": gain.astype(float),
+            "capital-loss": loss.astype(float),
+            "hours-per-week": hours,
+            "native-country": rng.choice(["United-States", "Mexico", "Philippines", "Germany", "India"], n,
+                                         p=[0.9, 0.04, 0.02, 0.02, 0.02]),
+            "income": income,
+        }
+    )
+    # inject some nulls
+    for c in ("workclass", "occupation", "native-country"):
+        df.loc[rng.random(n) < 0.05, c] = None
+    df.loc[rng.random(n) < 0.01, "age"] = np.nan
+    return df
+
+
+if __name__ == "__main__":
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--rows", type=int, default=32561)
+    ap.add_argument("--out", default="data/income_dataset")
+    a = ap.parse_args()
+    df = make(a.rows)
+    os.makedirs(os.path.join(a.out, "csv"), exist_ok=True)
+    os.makedirs(os.path.join(a.out, "parquet"), exist_ok=True)
+    df.to_csv(os.path.join(a.out, "csv", "part-00000.csv"), index=False)
+    df.to_parquet(os.path.join(a.out, "parquet", "part-00000.parquet"))
+    # drift source: mildly shifted snapshot
+    src = make(a.rows, seed=12)
+    src["age"] = src["age"] * 1.05
+    os.makedirs(os.path.join(a.out, "source", "csv"), exist_ok=True)
+    src.to_csv(os.path.join(a.out, "source", "csv", "part-00000.csv"), index=False)
+    print("wrote", a.out)
