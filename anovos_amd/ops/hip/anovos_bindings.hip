@@ -53,6 +53,10 @@ int anovos_lut_apply_f32(const int32_t *const *cols, const int64_t *lens, int nc
 int anovos_lut_apply_i32(const int32_t *const *cols, const int64_t *lens, int ncols,
                          const int32_t *lutflat, const int64_t *lut_off, int nchunks,
                          int32_t *const *outs, hipStream_t stream);
+int anovos_centered_gram(const void *const *cols, int64_t n, int k,
+                         const float *means, const int *pair_i,
+                         const int *pair_j, int npairs, int row_chunks,
+                         float *partials, float *gram, hipStream_t stream);
 }
 
 namespace {
@@ -550,7 +554,52 @@ std::vector<torch::Tensor> lut_apply_i32(std::vector<torch::Tensor> cols,
   return outs;
 }
 
+
+// K8: bf16 MFMA centered Gram — gram[i][j] = sum (x_i - mean_i)(x_j - mean_j)
+torch::Tensor centered_gram_bf16(std::vector<torch::Tensor> cols, torch::Tensor means) {
+  TORCH_CHECK(!cols.empty(), "no columns");
+  auto device = cols[0].device();
+  const int k = (int)cols.size();
+  std::vector<int64_t> ptrs;
+  int64_t n = cols[0].numel();
+  for (auto &t : cols) {
+    TORCH_CHECK(t.is_contiguous() && t.device() == device, "columns must be contiguous, same device");
+    TORCH_CHECK(t.scalar_type() == torch::kFloat32, "centered_gram_bf16 expects float32 columns");
+    TORCH_CHECK(t.numel() == n, "all columns must have equal length");
+    ptrs.push_back((int64_t)t.data_ptr());
+  }
+  auto means_d = means.to(device, torch::kFloat32).contiguous();
+  TORCH_CHECK(means_d.numel() == k, "means length mismatch");
+  const int kt = (k + 15) / 16;
+  std::vector<int64_t> pi64, pj64;
+  for (int i = 0; i < kt; ++i)
+    for (int j = i; j < kt; ++j) {
+      pi64.push_back(i);
+      pj64.push_back(j);
+    }
+  const int npairs = (int)pi64.size();
+  // target >= 2048 workgroups across the 8 XCDs; >= 64K rows per chunk
+  int row_chunks = (int)std::min<int64_t>(std::max<int64_t>(1, 2048 / std::max(npairs, 1)),
+                                          std::max<int64_t>(1, n >> 16));
+  row_chunks = std::max(row_chunks, 1);
+  auto opts_i32 = torch::TensorOptions().dtype(torch::kInt32).device(device);
+  auto pi = to_device_i64(pi64, device).to(torch::kInt32);
+  auto pj = to_device_i64(pj64, device).to(torch::kInt32);
+  auto dptr = to_device_i64(ptrs, device);
+  auto partials = torch::empty({(int64_t)npairs * row_chunks, 256},
+                               torch::TensorOptions().dtype(torch::kFloat32).device(device));
+  auto gram = torch::zeros({k, k}, torch::TensorOptions().dtype(torch::kFloat32).device(device));
+  check_hip(anovos_centered_gram((const void *const *)dptr.data_ptr<int64_t>(), n, k,
+                                 means_d.data_ptr<float>(), pi.data_ptr<int>(),
+                                 pj.data_ptr<int>(), npairs, row_chunks,
+                                 partials.data_ptr<float>(), gram.data_ptr<float>(),
+                                 current_stream()),
+            "anovos_centered_gram");
+  return gram;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("centered_gram_bf16", &centered_gram_bf16, "bf16 MFMA centered Gram X^T X (K8)");
   m.def("bracket_histograms_grouped", &bracket_histograms_grouped, "grouped refinement histograms (K3)");
   m.def("bucketize_columns_float", &bucketize_columns_float, "bucketize to float bin labels (K6)");
   m.def("lut_apply_f32", &lut_apply_f32, "fused LUT gather -> float (K12)");

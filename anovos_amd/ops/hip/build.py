@@ -26,6 +26,7 @@ def build(verbose: bool = True) -> str:
         sources=[
             os.path.join(HERE, "anovos_bindings.hip"),
             os.path.join(HERE, "anovos_kernels.hip"),
+            os.path.join(HERE, "corr_mfma.hip"),
         ],
         extra_cflags=["-O3"],
         extra_cuda_cflags=["-O3"],

@@ -1,0 +1,148 @@
+// K8: centered Gram matrix (correlation/covariance) on MFMA matrix cores.
+//
+// gram[i][j] = sum_r (x_i[r] - mean_i)(x_j[r] - mean_j), NaN -> 0 contribution.
+// Reference semantics: association_evaluator.py:118-123 (MLlib Correlation
+// .corr = one pass over the assembled vector column); here the pass is a
+// tall-skinny X^T X on v_mfma_f32_16x16x32_bf16 (bf16 in, fp32 accumulate).
+//
+// Shape: k columns (tens..hundreds) x n rows (millions) -> HBM-bound
+// (read n*k floats); MFMA keeps compute off the critical path.
+//
+// Tiling: columns padded to 16-col tiles; one workgroup per
+// (tile-pair (i,j<=i? no: j>=i), row-chunk). Each of the 4 waves strides
+// 32-row MFMA steps across the chunk (wave w rows r0+32w, step 128) and
+// accumulates its own f32x4 fragment; fragments are loaded STRAIGHT from
+// global (8 consecutive rows of one column = 16 B per lane, coalesced
+// across the 16 lanes of a row-group) — no LDS staging needed because
+// each element is touched once per tile-pair. Per-block partials land in
+// a [pairs*chunks][256] buffer; a second deterministic kernel reduces
+// chunks in fp64 and mirrors the symmetric half (same no-atomics
+// convention as anovos_kernels.hip).
+
+#include <hip/hip_runtime.h>
+#include <cfloat>
+#include <cstdint>
+#include <cmath>
+
+#define THREADS 256
+
+using bf16x8 = __attribute__((ext_vector_type(8))) short;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+__device__ __forceinline__ short to_bf16(float v) {
+  union {
+    float f;
+    uint32_t u;
+  } c{v};
+  // round-to-nearest-even into the upper 16 bits
+  uint32_t r = c.u + 0x7FFFu + ((c.u >> 16) & 1u);
+  return (short)(r >> 16);
+}
+
+// Load one lane's A/B fragment: 8 consecutive rows of column `col`,
+// centered, NaN->0, zero outside [0,k) x [0,limit).
+__device__ __forceinline__ bf16x8 load_frag(const float *const *cols, int k,
+                                            const float *means, int col,
+                                            int64_t row, int64_t limit) {
+  bf16x8 f;
+  if (col < k) {
+    const float *__restrict__ x = cols[col];
+    const float m = means[col];
+    if (row + 8 <= limit) {
+      const float4 *p = reinterpret_cast<const float4 *>(x + row);
+      float4 a = p[0], b = p[1];
+      float v0 = a.x - m, v1 = a.y - m, v2 = a.z - m, v3 = a.w - m;
+      float v4 = b.x - m, v5 = b.y - m, v6 = b.z - m, v7 = b.w - m;
+      f[0] = to_bf16(isnan(v0) ? 0.f : v0);
+      f[1] = to_bf16(isnan(v1) ? 0.f : v1);
+      f[2] = to_bf16(isnan(v2) ? 0.f : v2);
+      f[3] = to_bf16(isnan(v3) ? 0.f : v3);
+      f[4] = to_bf16(isnan(v4) ? 0.f : v4);
+      f[5] = to_bf16(isnan(v5) ? 0.f : v5);
+      f[6] = to_bf16(isnan(v6) ? 0.f : v6);
+      f[7] = to_bf16(isnan(v7) ? 0.f : v7);
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float v = (row + e < limit) ? x[row + e] - m : 0.f;
+        f[e] = to_bf16(isnan(v) ? 0.f : v);
+      }
+    }
+  } else {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) f[e] = 0;
+  }
+  return f;
+}
+
+__global__ __launch_bounds__(THREADS) void gram_partials_kernel(
+    const float *const *cols, int64_t n, int k, const float *means,
+    const int *pair_i, const int *pair_j, int row_chunks, float *partials) {
+  const int pair = blockIdx.x / row_chunks;
+  const int chunk = blockIdx.x % row_chunks;
+  const int i0 = pair_i[pair] * 16;
+  const int j0 = pair_j[pair] * 16;
+
+  // 32-row-aligned chunk bounds
+  const int64_t steps_total = (n + 31) / 32;
+  const int64_t steps_per = (steps_total + row_chunks - 1) / row_chunks;
+  const int64_t step_s = (int64_t)chunk * steps_per;
+  const int64_t step_e = min(steps_total, step_s + steps_per);
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int m = lane & 15;         // tile row (A) / tile col (B)
+  const int ko = (lane >> 4) * 8;  // K offset within the 32-row step
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int64_t s = step_s + wave; s < step_e; s += 4) {
+    const int64_t r = s * 32 + ko;
+    bf16x8 a = load_frag(cols, k, means, i0 + m, r, n);
+    bf16x8 b = (j0 == i0) ? a : load_frag(cols, k, means, j0 + m, r, n);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+
+  // combine the 4 waves' [16,16] fragments via LDS (C/D layout:
+  // col = lane&15, row = (lane>>4)*4 + reg)
+  __shared__ float lds[4][256];
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int row = (lane >> 4) * 4 + reg;
+    lds[wave][row * 16 + (lane & 15)] = acc[reg];
+  }
+  __syncthreads();
+  const int c = threadIdx.x;
+  partials[(int64_t)blockIdx.x * 256 + c] =
+      lds[0][c] + lds[1][c] + lds[2][c] + lds[3][c];
+}
+
+__global__ __launch_bounds__(THREADS) void gram_reduce_kernel(
+    const float *partials, int row_chunks, const int *pair_i,
+    const int *pair_j, int k, float *gram) {
+  const int pair = blockIdx.x;
+  const int c = threadIdx.x;
+  const int row = c >> 4, col = c & 15;
+  double s = 0.0;
+  for (int ch = 0; ch < row_chunks; ++ch)
+    s += (double)partials[((int64_t)pair * row_chunks + ch) * 256 + c];
+  const int gi = pair_i[pair] * 16 + row;
+  const int gj = pair_j[pair] * 16 + col;
+  if (gi < k && gj < k) {
+    gram[(int64_t)gi * k + gj] = (float)s;
+    gram[(int64_t)gj * k + gi] = (float)s;
+  }
+}
+
+extern "C" int anovos_centered_gram(const void *const *cols, int64_t n, int k,
+                                    const float *means, const int *pair_i,
+                                    const int *pair_j, int npairs,
+                                    int row_chunks, float *partials,
+                                    float *gram, hipStream_t stream) {
+  dim3 grid1((uint32_t)(npairs * row_chunks));
+  hipLaunchKernelGGL(gram_partials_kernel, grid1, dim3(THREADS), 0, stream,
+                     (const float *const *)cols, n, k, means, pair_i, pair_j,
+                     row_chunks, partials);
+  hipLaunchKernelGGL(gram_reduce_kernel, dim3((uint32_t)npairs), dim3(THREADS),
+                     0, stream, partials, row_chunks, pair_i, pair_j, k, gram);
+  return (int)hipGetLastError();
+}
