@@ -38,7 +38,8 @@ def pearson_matrix(idf, cols: List[str], moments=None, use_bf16: bool = True) ->
     tensors = [idf.col(c).data for c in cols]
     if dev.type == "cuda" and backend.use_hip(tensors[0]) and use_bf16:
         ext = backend.hip_ext()
-        gram = ext.centered_gram_bf16([t.contiguous() for t in tensors], means)  # [k,k] fp32
+        f32 = [t.contiguous() if t.dtype == torch.float32 else t.to(torch.float32).contiguous() for t in tensors]
+        gram = ext.centered_gram_bf16(f32, means)  # [k,k] fp32
     else:
         X = torch.stack([torch.nan_to_num(t.to(torch.float32), nan=float(means[i])) for i, t in enumerate(tensors)], dim=1)
         Xc = X - means.unsqueeze(0)
