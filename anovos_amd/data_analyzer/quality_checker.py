@@ -405,6 +405,45 @@ def outlier_detection(
     odf = idf
     rows_print = []
     flags: Dict[str, torch.Tensor] = {}
+
+    from anovos_amd.ops import backend as _backend
+
+    first = idf.col(list_of_cols[0]).data if list_of_cols else None
+    fused_ok = (
+        first is not None
+        and first.is_cuda
+        and _backend.use_hip(first)
+        and treatment_method in ("value_replacement", "null_replacement")
+    )
+    if fused_ok:
+        # one fused HIP launch for ALL columns: counts + clamp/null-out
+        # (K10/K11) — replaces ~10 aten launches per column
+        ext = _backend.hip_ext()
+        lo_t, hi_t = [], []
+        for (lo, hi) in params:
+            use_lo = detection_side in ("lower", "both") and lo is not None
+            use_hi = detection_side in ("upper", "both") and hi is not None
+            lo_t.append(float(lo) if use_lo else float("nan"))
+            hi_t.append(float(hi) if use_hi else float("nan"))
+        mode = 0 if not treatment else (1 if treatment_method == "value_replacement" else 2)
+        tensors = [idf.col(c).data.contiguous() for c in list_of_cols]
+        counts, outs = ext.outlier_clamp_columns(tensors, torch.tensor(lo_t), torch.tensor(hi_t), mode)
+        counts = dist.all_reduce_(counts.to(torch.float64), "sum").cpu().to(torch.int64)
+        for i, c in enumerate(list_of_cols):
+            rows_print.append([c, int(counts[i, 0]), int(counts[i, 1]), 0])
+            if mode:
+                odf = odf.with_column(c + "_outliered", Column(c + "_outliered", idf.col(c).dtype, outs[i]))
+                if output_mode == "replace":
+                    odf = odf.drop([c]).rename({c + "_outliered": c})
+        if treatment and output_mode == "replace":
+            odf = odf.select([c for c in column_order if c in odf.columns])
+        if not treatment:
+            odf = idf
+        odf_print = pd.DataFrame(rows_print + [[c, 0, 0, 1] for c in skewed_cols], columns=["attribute", "lower_outliers", "upper_outliers", "excluded_due_to_skewness"])
+        if print_impact:
+            print(odf_print.to_string(index=False))
+        return odf, odf_print
+
     for c, (lo, hi) in zip(list_of_cols, params):
         x = idf.col(c).data
         flag = torch.zeros_like(x, dtype=torch.int8)

@@ -53,6 +53,14 @@ int anovos_lut_apply_f32(const int32_t *const *cols, const int64_t *lens, int nc
 int anovos_lut_apply_i32(const int32_t *const *cols, const int64_t *lens, int ncols,
                          const int32_t *lutflat, const int64_t *lut_off, int nchunks,
                          int32_t *const *outs, hipStream_t stream);
+int anovos_code_counts_multi(const int32_t *const *cols, const int64_t *lens,
+                             const int64_t *offs, const int *sizes, int ncols,
+                             int max_slots, int nchunks, uint64_t *out,
+                             hipStream_t stream);
+int anovos_outlier_clamp(const void *const *cols, const int64_t *lens,
+                         int ncols, const double *lo, const double *hi,
+                         int nchunks, int mode, int dtype, void *const *outs,
+                         uint64_t *counts, hipStream_t stream);
 int anovos_centered_gram(const void *const *cols, int64_t n, int k,
                          const float *means, const int *pair_i,
                          const int *pair_j, int npairs, int row_chunks,
@@ -598,7 +606,102 @@ torch::Tensor centered_gram_bf16(std::vector<torch::Tensor> cols, torch::Tensor 
   return gram;
 }
 
+
+// K5 fused: multi-column code counts (+ null slot per column).
+// Returns flat int64 tensor of length sum(sizes_i + 1).
+torch::Tensor code_counts_multi(std::vector<torch::Tensor> cols, std::vector<int64_t> sizes) {
+  TORCH_CHECK(!cols.empty(), "no columns");
+  TORCH_CHECK(cols.size() == sizes.size(), "sizes mismatch");
+  auto device = cols[0].device();
+  std::vector<int64_t> ptrs, lens, offs;
+  std::vector<int64_t> sz64;
+  int64_t off = 0, maxn = 0;
+  int max_slots = 0;
+  for (size_t i = 0; i < cols.size(); ++i) {
+    auto &t = cols[i];
+    TORCH_CHECK(t.is_contiguous() && t.scalar_type() == torch::kInt32, "int32 code columns required");
+    ptrs.push_back((int64_t)t.data_ptr());
+    lens.push_back(t.numel());
+    maxn = std::max(maxn, t.numel());
+    offs.push_back(off);
+    off += sizes[i] + 1;
+    sz64.push_back(sizes[i]);
+    max_slots = std::max(max_slots, (int)sizes[i] + 1);
+  }
+  int ncols = (int)cols.size();
+  int nchunks = pick_chunks(maxn, ncols);
+  auto out = torch::zeros({off}, torch::TensorOptions().dtype(torch::kInt64).device(device));
+  auto dptr = to_device_i64(ptrs, device);
+  auto dlen = to_device_i64(lens, device);
+  auto doff = to_device_i64(offs, device);
+  auto dsz = to_device_i64(sz64, device).to(torch::kInt32);
+  check_hip(anovos_code_counts_multi((const int32_t *const *)dptr.data_ptr<int64_t>(),
+                                     dlen.data_ptr<int64_t>(), doff.data_ptr<int64_t>(),
+                                     dsz.data_ptr<int>(), ncols, max_slots, nchunks,
+                                     (uint64_t *)out.data_ptr<int64_t>(), current_stream()),
+            "anovos_code_counts_multi");
+  return out;
+}
+
+// K10/K11 fused: outlier counts + clamp/null treatment in one launch.
+// mode: 0 count only, 1 clamp to bounds, 2 null-out. Returns
+// (counts [k,2] int64, outs list — empty when mode==0).
+std::tuple<torch::Tensor, std::vector<torch::Tensor>> outlier_clamp_columns(
+    std::vector<torch::Tensor> cols, torch::Tensor lo, torch::Tensor hi, int64_t mode) {
+  TORCH_CHECK(!cols.empty(), "no columns");
+  auto device = cols[0].device();
+  auto lo_d = lo.to(torch::kFloat64).to(device).contiguous();
+  auto hi_d = hi.to(torch::kFloat64).to(device).contiguous();
+  auto counts = torch::zeros({(int64_t)cols.size(), 2},
+                             torch::TensorOptions().dtype(torch::kInt64).device(device));
+  std::vector<torch::Tensor> outs_all(cols.size());
+  for (int pass = 0; pass < 2; ++pass) {
+    std::vector<int64_t> ptrs, lens, optrs, idx;
+    std::vector<torch::Tensor> outs;
+    for (size_t i = 0; i < cols.size(); ++i) {
+      if (dtype_code(cols[i]) != pass) continue;
+      TORCH_CHECK(cols[i].is_contiguous(), "columns must be contiguous");
+      ptrs.push_back((int64_t)cols[i].data_ptr());
+      lens.push_back(cols[i].numel());
+      idx.push_back((int64_t)i);
+      if (mode) {
+        auto o = torch::empty_like(cols[i]);
+        optrs.push_back((int64_t)o.data_ptr());
+        outs.push_back(o);
+      }
+    }
+    if (ptrs.empty()) continue;
+    int ncols = (int)ptrs.size();
+    int64_t maxn = *std::max_element(lens.begin(), lens.end());
+    int nchunks = pick_chunks(maxn, ncols);
+    auto dptr = to_device_i64(ptrs, device);
+    auto dlen = to_device_i64(lens, device);
+    auto didx = to_device_i64(idx, device);
+    auto lo_sel = lo_d.index_select(0, didx);
+    auto hi_sel = hi_d.index_select(0, didx);
+    auto sub = torch::zeros({ncols, 2}, torch::TensorOptions().dtype(torch::kInt64).device(device));
+    torch::Tensor dout;
+    if (mode) dout = to_device_i64(optrs, device);
+    check_hip(anovos_outlier_clamp((const void *const *)dptr.data_ptr<int64_t>(),
+                                   dlen.data_ptr<int64_t>(), ncols,
+                                   lo_sel.data_ptr<double>(), hi_sel.data_ptr<double>(),
+                                   nchunks, (int)mode, pass,
+                                   mode ? (void *const *)dout.data_ptr<int64_t>() : nullptr,
+                                   (uint64_t *)sub.data_ptr<int64_t>(), current_stream()),
+              "anovos_outlier_clamp");
+    counts.index_copy_(0, didx, sub);
+    for (size_t j = 0; j < idx.size(); ++j)
+      if (mode) outs_all[idx[j]] = outs[j];
+  }
+  std::vector<torch::Tensor> outs_ret;
+  if (mode)
+    for (auto &o : outs_all) outs_ret.push_back(o);
+  return {counts, outs_ret};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("code_counts_multi", &code_counts_multi, "fused multi-column code counts + null slot (K5)");
+  m.def("outlier_clamp_columns", &outlier_clamp_columns, "fused outlier count/clamp (K10/K11)");
   m.def("centered_gram_bf16", &centered_gram_bf16, "bf16 MFMA centered Gram X^T X (K8)");
   m.def("bracket_histograms_grouped", &bracket_histograms_grouped, "grouped refinement histograms (K3)");
   m.def("bucketize_columns_float", &bucketize_columns_float, "bucketize to float bin labels (K6)");

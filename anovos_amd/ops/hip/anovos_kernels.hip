@@ -800,3 +800,134 @@ int anovos_fillnan(const void *const *cols, const int64_t *lens, int ncols,
 }
 
 }  // extern "C"
+
+// ------------------------------------------------------------------
+// K5 (fused): multi-column dictionary code counts + null counts.
+// One launch for ALL categorical columns: grid = ncols x nchunks;
+// per-column counts land at out[off[col] .. off[col]+size[col]] and the
+// null count (code < 0) at out[off[col]+size[col]]. LDS-staged when the
+// dictionary fits (includes the null slot).
+// ------------------------------------------------------------------
+__global__ __launch_bounds__(THREADS) void code_counts_multi_kernel(
+    const int32_t *const *cols, const int64_t *lens, const int64_t *offs,
+    const int *sizes, int nchunks, uint64_t *out) {
+  const int col = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  const int32_t *__restrict__ codes = cols[col];
+  const int64_t n = lens[col];
+  const int size = sizes[col];
+  const int slots = size + 1;  // + null slot
+  uint64_t *base = out + offs[col];
+
+  extern __shared__ uint32_t cnt[];
+  const bool use_lds = (slots <= 16384);
+  if (use_lds) {
+    for (int i = threadIdx.x; i < slots; i += THREADS) cnt[i] = 0;
+    __syncthreads();
+  }
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+  if (use_lds) {
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+      int c = codes[i];
+      atomicAdd(&cnt[(c >= 0 && c < size) ? c : size], 1u);
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < slots; i += THREADS)
+      if (cnt[i]) atomicAdd((unsigned long long *)&base[i], (unsigned long long)cnt[i]);
+  } else {
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+      int c = codes[i];
+      atomicAdd((unsigned long long *)&base[(c >= 0 && c < size) ? c : size], 1ull);
+    }
+  }
+}
+
+// ------------------------------------------------------------------
+// K10/K11 (fused): outlier flag counts + clamp/null treatment, all
+// columns in one launch. lo/hi are per-column bounds (NaN = unbounded
+// side). mode 0 = count only; 1 = clamp to bound; 2 = replace with NaN.
+// counts[col*2+0] = lower outliers, [col*2+1] = upper. Counts use integer
+// atomics (order-independent => deterministic).
+// ------------------------------------------------------------------
+template <typename T>
+__global__ __launch_bounds__(THREADS) void outlier_clamp_kernel(
+    const T *const *cols, const int64_t *lens, const double *lo,
+    const double *hi, int nchunks, int mode, T *const *outs,
+    uint64_t *counts) {
+  const int col = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  const T *__restrict__ x = cols[col];
+  T *__restrict__ y = outs ? outs[col] : nullptr;
+  const int64_t n = lens[col];
+  const double l = lo[col], h = hi[col];
+  const bool has_l = !isnan(l), has_h = !isnan(h);
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+
+  uint32_t nl = 0, nh = 0;
+  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+    double v = (double)x[i];
+    if (!isnan(v)) {
+      if (has_l && v < l) {
+        ++nl;
+        if (mode == 1) v = l;
+        else if (mode == 2) v = (double)NAN;
+      } else if (has_h && v > h) {
+        ++nh;
+        if (mode == 1) v = h;
+        else if (mode == 2) v = (double)NAN;
+      }
+    }
+    if (mode && y) y[i] = (T)v;
+  }
+  __shared__ uint32_t red[2][THREADS];
+  red[0][threadIdx.x] = nl;
+  red[1][threadIdx.x] = nh;
+  __syncthreads();
+  for (int st = THREADS / 2; st > 0; st >>= 1) {
+    if (threadIdx.x < st) {
+      red[0][threadIdx.x] += red[0][threadIdx.x + st];
+      red[1][threadIdx.x] += red[1][threadIdx.x + st];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    if (red[0][0]) atomicAdd((unsigned long long *)&counts[col * 2 + 0], (unsigned long long)red[0][0]);
+    if (red[1][0]) atomicAdd((unsigned long long *)&counts[col * 2 + 1], (unsigned long long)red[1][0]);
+  }
+}
+
+extern "C" {
+
+int anovos_code_counts_multi(const int32_t *const *cols, const int64_t *lens,
+                             const int64_t *offs, const int *sizes, int ncols,
+                             int max_slots, int nchunks, uint64_t *out,
+                             hipStream_t stream) {
+  size_t lds = (max_slots <= 16384) ? (size_t)max_slots * 4 : 0;
+  hipLaunchKernelGGL(code_counts_multi_kernel, dim3(ncols * nchunks),
+                     dim3(THREADS), lds, stream, cols, lens, offs, sizes,
+                     nchunks, out);
+  return (int)hipGetLastError();
+}
+
+int anovos_outlier_clamp(const void *const *cols, const int64_t *lens,
+                         int ncols, const double *lo, const double *hi,
+                         int nchunks, int mode, int dtype, void *const *outs,
+                         uint64_t *counts, hipStream_t stream) {
+  if (dtype == 0)
+    hipLaunchKernelGGL(outlier_clamp_kernel<float>, dim3(ncols * nchunks),
+                       dim3(THREADS), 0, stream, (const float *const *)cols,
+                       lens, lo, hi, nchunks, mode, (float *const *)outs,
+                       counts);
+  else
+    hipLaunchKernelGGL(outlier_clamp_kernel<double>, dim3(ncols * nchunks),
+                       dim3(THREADS), 0, stream, (const double *const *)cols,
+                       lens, lo, hi, nchunks, mode, (double *const *)outs,
+                       counts);
+  return (int)hipGetLastError();
+}
+
+}  // extern "C"

@@ -225,3 +225,52 @@ def test_pearson_matrix_gpu_path(ext):
     X = np.stack([cols[c].data.cpu().numpy() for c in "abc"])
     ref = np.corrcoef(X)
     assert np.allclose(corr, ref, atol=5e-3), np.abs(corr - ref).max()
+
+
+@requires_gpu
+def test_code_counts_multi_vs_bincount(ext):
+    g = torch.Generator(device="cpu").manual_seed(9)
+    sizes = [5, 40, 1]
+    cols = []
+    for s in sizes:
+        c = torch.randint(0, s, (100_000,), generator=g, dtype=torch.int32)
+        c[torch.rand(100_000, generator=g) < 0.05] = -1  # nulls
+        cols.append(c.cuda())
+    flat = ext.code_counts_multi(cols, sizes)
+    off = 0
+    for c, s in zip(cols, sizes):
+        cnt = flat[off : off + s].cpu()
+        nulls = int(flat[off + s])
+        ref = torch.bincount(c[c >= 0].long().cpu(), minlength=s)
+        assert torch.equal(cnt, ref)
+        assert nulls == int((c < 0).sum())
+        off += s + 1
+
+
+@requires_gpu
+def test_outlier_clamp_columns_vs_torch(ext):
+    g = torch.Generator(device="cpu").manual_seed(10)
+    n = 100_000
+    cols = [torch.randn(n, generator=g).cuda() * 10 for _ in range(4)]
+    cols[1][::100] = float("nan")
+    lo = torch.tensor([-5.0, float("nan"), -3.0, -1.0])
+    hi = torch.tensor([5.0, 8.0, float("nan"), 1.0])
+    counts, outs = ext.outlier_clamp_columns([c.contiguous() for c in cols], lo, hi, 1)
+    for i, c in enumerate(cols):
+        l, h = float(lo[i]), float(hi[i])
+        m = ~torch.isnan(c)
+        exp_lo = int(((c < l) & m).sum()) if l == l else 0
+        exp_hi = int(((c > h) & m).sum()) if h == h else 0
+        assert int(counts[i, 0]) == exp_lo, i
+        assert int(counts[i, 1]) == exp_hi, i
+        ref = c.clone()
+        if l == l:
+            ref = torch.where((ref < l) & m, torch.full_like(ref, l), ref)
+        if h == h:
+            ref = torch.where((ref > h) & m, torch.full_like(ref, h), ref)
+        assert torch.equal(outs[i], ref), i
+    # mode 2: null replacement
+    counts2, outs2 = ext.outlier_clamp_columns([cols[0].contiguous()], lo[:1], hi[:1], 2)
+    x = cols[0]
+    nulled = torch.isnan(outs2[0]).sum()
+    assert int(nulled) == int(((x < -5) | (x > 5)).sum())
