@@ -433,10 +433,10 @@ def outlier_detection(
             rows_print.append([c, int(counts[i, 0]), int(counts[i, 1]), 0])
             if mode:
                 odf = odf.with_column(c + "_outliered", Column(c + "_outliered", idf.col(c).dtype, outs[i]))
-                if output_mode == "replace":
-                    odf = odf.drop([c]).rename({c + "_outliered": c})
-        if treatment and output_mode == "replace":
-            odf = odf.select([c for c in column_order if c in odf.columns])
+        if mode:
+            from anovos_amd.data_transformer.transformers import _finish_output
+
+            odf = _finish_output(idf, odf, list_of_cols, "_outliered", output_mode)
         if not treatment:
             odf = idf
         odf_print = pd.DataFrame(rows_print + [[c, 0, 0, 1] for c in skewed_cols], columns=["attribute", "lower_outliers", "upper_outliers", "excluded_due_to_skewness"])

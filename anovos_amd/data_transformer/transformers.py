@@ -63,15 +63,23 @@ def _load_model(model_path: str, name: str, fmt: str = "parquet") -> pd.DataFram
 
 
 def _finish_output(idf: AnovosFrame, odf: AnovosFrame, list_of_cols, postfix: str, output_mode: str) -> AnovosFrame:
-    """replace: new '<col><postfix>' columns take the original names."""
-    if output_mode == "replace":
-        for c in list_of_cols:
-            if (c + postfix) in odf.columns:
-                tmp = odf.drop([c]).rename({c + postfix: c})
-                # keep original column order
-                order = [x if x != c + postfix else c for x in odf.columns if x != c]
-                odf = tmp.select([x for x in order if x in tmp.columns])
-    return odf
+    """replace: new '<col><postfix>' columns take the original names.
+    Single pass over the column dict (a per-column drop/rename/select
+    loop is quadratic in frame width — 60 ms at 150x350 columns)."""
+    if output_mode != "replace" or not postfix:
+        return odf
+    replaced = {c for c in list_of_cols if (c + postfix) in odf.columns}
+    out = {}
+    for name in odf.columns:
+        if name in replaced:
+            newc = odf.col(name + postfix).clone()
+            newc.name = name
+            out[name] = newc
+        elif name.endswith(postfix) and name[: -len(postfix)] in replaced:
+            continue  # consumed above
+        else:
+            out[name] = odf.col(name)
+    return AnovosFrame(out, odf.device)
 
 
 # ---------------- binning ----------------
@@ -613,9 +621,20 @@ def imputation_MMM(
                 pre = dict(zip(mdf["attribute"], mdf["mode"]))
             else:
                 pre = {}
+            # batch: one fused code-count launch for all categorical mode
+            # columns and one dense-histogram launch for numeric ones —
+            # the per-column mode() loop cost a host sync per column
+            need = [c for c in mode_cols if c not in pre or pre[c] is None]
+            cat_need = [c for c in need if c in cat_all]
+            num_need = [c for c in need if c not in cat_all]
+            if cat_need:
+                groupby_ops.cat_value_counts(idf, cat_need)  # warms per-col cache
+            num_modes = groupby_ops.discrete_modes(idf, num_need) if num_need else {}
             for c in mode_cols:
                 if c in pre and pre[c] is not None:
                     mv = pre[c]
+                elif c in num_modes:
+                    mv = num_modes[c][0]
                 else:
                     mv, _cnt = groupby_ops.mode(idf, c)
                 if c in cat_all:
