@@ -41,8 +41,9 @@ int anovos_fillnan(const void *const *cols, const int64_t *lens, int ncols,
                    void *const *outs, hipStream_t stream);
 int anovos_bracket_hist_grouped(const void *const *cols, const int64_t *lens,
                                 const int *bstart, int ncols, const double *lo,
-                                const double *hi, int nchunks, int dtype,
-                                uint64_t *out, hipStream_t stream);
+                                const double *hi, const double *p1lo,
+                                const double *p1scale, int p1bins, int nchunks,
+                                int dtype, uint64_t *out, hipStream_t stream);
 int anovos_bucketize_float(const void *const *cols, const int64_t *lens, int ncols,
                            const double *cutflat, const int64_t *cutoff_off,
                            const int *cutoff_len, int max_ncut, int nchunks, int dtype,
@@ -404,7 +405,9 @@ std::vector<torch::Tensor> fill_nan_columns(std::vector<torch::Tensor> cols, tor
 
 torch::Tensor bracket_histograms_grouped(std::vector<torch::Tensor> cols,
                                          torch::Tensor bracket_col,
-                                         torch::Tensor lo, torch::Tensor hi) {
+                                         torch::Tensor lo, torch::Tensor hi,
+                                         torch::Tensor p1lo, torch::Tensor p1scale,
+                                         int64_t p1bins) {
   // brackets MUST be sorted by column index; returns [nbrackets, 512].
   TORCH_CHECK(!cols.empty(), "no columns");
   auto device = cols[0].device();
@@ -436,13 +439,17 @@ torch::Tensor bracket_histograms_grouped(std::vector<torch::Tensor> cols,
   auto dbs = torch::from_blob(bstart.data(), {(int64_t)bstart.size()}, torch::TensorOptions().dtype(torch::kInt32)).clone().to(device);
   auto lo_d = lo.to(torch::kFloat64).to(device);
   auto hi_d = hi.to(torch::kFloat64).to(device);
+  auto p1lo_d = p1lo.to(torch::kFloat64).to(device).contiguous();
+  auto p1scale_d = p1scale.to(torch::kFloat64).to(device).contiguous();
   int64_t maxn = *std::max_element(lens.begin(), lens.end());
   int nchunks = pick_chunks(maxn, ncols);
   auto out = torch::zeros({nb, 512}, torch::TensorOptions().dtype(torch::kInt64).device(device));
   check_hip(anovos_bracket_hist_grouped((const void *const *)dptr.data_ptr<int64_t>(),
                                         dlen.data_ptr<int64_t>(), dbs.data_ptr<int>(), ncols,
                                         lo_d.data_ptr<double>(), hi_d.data_ptr<double>(),
-                                        nchunks, dtype, (uint64_t *)out.data_ptr<int64_t>(),
+                                        p1lo_d.data_ptr<double>(), p1scale_d.data_ptr<double>(),
+                                        (int)p1bins, nchunks, dtype,
+                                        (uint64_t *)out.data_ptr<int64_t>(),
                                         current_stream()),
             "anovos_bracket_hist_grouped");
   return out;

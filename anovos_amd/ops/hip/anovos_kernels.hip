@@ -163,13 +163,26 @@ __global__ __launch_bounds__(THREADS) void hist_kernel(
   const double h = hi[col];
   if (h > l) {
     const double scale = (double)nbins / (h - l);
-    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
-      double v = (double)x[i];
+    auto body = [&](double v) {
       if (!isnan(v)) {
         int b = (int)((v - l) * scale);
         b = max(0, min(nbins - 1, b));
         atomicAdd(&bins[b], 1u);
       }
+    };
+    if (sizeof(T) == 4) {
+      const int64_t nv = (e - s) / 4;
+      const float4 *xv = reinterpret_cast<const float4 *>(x + s);
+      for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
+        float4 v = xv[i];
+        body((double)v.x);
+        body((double)v.y);
+        body((double)v.z);
+        body((double)v.w);
+      }
+      for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) body((double)x[i]);
+    } else {
+      for (int64_t i = s + threadIdx.x; i < e; i += THREADS) body((double)x[i]);
     }
   } else {
     for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
@@ -220,29 +233,43 @@ __global__ __launch_bounds__(THREADS) void bracket_hist_kernel(
 }
 
 // grouped bracket histograms: one column read serves ALL of that
-// column's refinement brackets (the per-bracket kernel re-reads the
-// column once per bracket — 9x overfetch for percentile batches).
-// brackets are pre-sorted by column: col c owns [bstart[c], bstart[c+1]).
+// column's refinement brackets. Brackets live on the pass-1 bin grid, so
+// an LDS LUT (pass-1 bin -> local bracket id) resolves the bracket in
+// O(1) per element instead of looping the bracket list; loads are
+// float4-vectorized. brackets are pre-sorted by column: col c owns
+// [bstart[c], bstart[c+1]).
 #define RB_BINS 512
 #define RB_MAXB 16
 
 template <typename T>
 __global__ __launch_bounds__(THREADS) void bracket_hist_grouped_kernel(
     const T *const *cols, const int64_t *lens, const int *bstart, int ncols,
-    const double *lo, const double *hi, int nchunks, uint64_t *out) {
+    const double *lo, const double *hi, const double *p1lo,
+    const double *p1scale, int p1bins, int nchunks, uint64_t *out) {
   __shared__ uint32_t bins[RB_MAXB * RB_BINS];
+  extern __shared__ int16_t lut[];  // [p1bins] pass-1 bin -> bracket id
   __shared__ double slo[RB_MAXB], sscale[RB_MAXB], shi[RB_MAXB];
   const int col = blockIdx.x / nchunks;
   const int chunk = blockIdx.x % nchunks;
   const int b0 = bstart[col];
   const int nb = bstart[col + 1] - b0;
   if (nb == 0) return;
+  const double l1 = p1lo[col];
+  const double sc1 = p1scale[col];
   for (int i = threadIdx.x; i < nb * RB_BINS; i += THREADS) bins[i] = 0;
+  for (int i = threadIdx.x; i < p1bins; i += THREADS) lut[i] = -1;
+  __syncthreads();
   if (threadIdx.x < nb) {
-    slo[threadIdx.x] = lo[b0 + threadIdx.x];
-    shi[threadIdx.x] = hi[b0 + threadIdx.x];
-    double w = shi[threadIdx.x] - slo[threadIdx.x];
+    const double bl = lo[b0 + threadIdx.x];
+    const double bh = hi[b0 + threadIdx.x];
+    slo[threadIdx.x] = bl;
+    shi[threadIdx.x] = bh;
+    double w = bh - bl;
     sscale[threadIdx.x] = (w > 0) ? (double)RB_BINS / w : 0.0;
+    // floor with epsilon: grid-aligned bl maps exactly; a sub-bin
+    // bracket (refine >= 2) maps to the bin that CONTAINS it
+    int bin1 = (int)((bl - l1) * sc1 + 1e-6);
+    if (bin1 >= 0 && bin1 < p1bins) lut[bin1] = (int16_t)threadIdx.x;
   }
   __syncthreads();
   const T *__restrict__ x = cols[col];
@@ -250,16 +277,31 @@ __global__ __launch_bounds__(THREADS) void bracket_hist_grouped_kernel(
   const int64_t per = (n + nchunks - 1) / nchunks;
   const int64_t s = (int64_t)chunk * per;
   const int64_t e = min(n, s + per);
-  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
-    double v = (double)x[i];
-    if (isnan(v)) continue;
-    for (int b = 0; b < nb; ++b) {
-      if (v >= slo[b] && v < shi[b]) {
-        int bin = (int)((v - slo[b]) * sscale[b]);
-        bin = max(0, min(RB_BINS - 1, bin));
-        atomicAdd(&bins[b * RB_BINS + bin], 1u);
-      }
+
+  auto body = [&](double v) {
+    if (isnan(v)) return;
+    int b1 = (int)((v - l1) * sc1);
+    b1 = max(0, min(p1bins - 1, b1));
+    int br = lut[b1];
+    if (br >= 0 && v >= slo[br] && v < shi[br]) {
+      int bin = (int)((v - slo[br]) * sscale[br]);
+      bin = max(0, min(RB_BINS - 1, bin));
+      atomicAdd(&bins[br * RB_BINS + bin], 1u);
     }
+  };
+  if (sizeof(T) == 4) {
+    const int64_t nv = (e - s) / 4;
+    const float4 *xv = reinterpret_cast<const float4 *>(x + s);
+    for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
+      float4 v = xv[i];
+      body((double)v.x);
+      body((double)v.y);
+      body((double)v.z);
+      body((double)v.w);
+    }
+    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) body((double)x[i]);
+  } else {
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) body((double)x[i]);
   }
   __syncthreads();
   for (int i = threadIdx.x; i < nb * RB_BINS; i += THREADS)
@@ -432,6 +474,30 @@ DEV_INLINE uint64_t splitmix64(uint64_t x) {
   return x;
 }
 
+// hash one element: float32 hashes its raw bit pattern (no double
+// round-trip — distinctness only needs a per-column-consistent
+// injection); float64 hashes its 8 bytes.
+template <typename T>
+DEV_INLINE bool hll_hash(T v, int p, int &idx, int &rho) {
+  if (isnan((double)v)) return false;
+  uint64_t bits;
+  if (sizeof(T) == 4) {
+    uint32_t b32;
+    float f = (float)v;
+    memcpy(&b32, &f, 4);
+    bits = (uint64_t)b32;
+  } else {
+    double d = (double)v;
+    memcpy(&bits, &d, 8);
+  }
+  uint64_t h = splitmix64(bits);
+  idx = (int)(h >> (64 - p));
+  uint64_t rem = h << p;
+  rho = (rem == 0) ? (64 - p + 1) : (__clzll((long long)rem) + 1);
+  if (rho > 64 - p + 1) rho = 64 - p + 1;
+  return true;
+}
+
 template <typename T>
 __global__ __launch_bounds__(THREADS) void hll_kernel(
     const T *x, int64_t n, int p, int nchunks, int32_t *regs /*[1<<p]*/) {
@@ -444,16 +510,8 @@ __global__ __launch_bounds__(THREADS) void hll_kernel(
   const int64_t s = (int64_t)chunk * per;
   const int64_t e = min(n, s + per);
   for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
-    double v = (double)x[i];
-    if (isnan(v)) continue;
-    uint64_t bits;
-    memcpy(&bits, &v, 8);
-    uint64_t h = splitmix64(bits);
-    int idx = (int)(h >> (64 - p));
-    uint64_t rem = h << p;
-    int rho = (rem == 0) ? (64 - p + 1) : (__clzll((long long)rem) + 1);
-    if (rho > 64 - p + 1) rho = 64 - p + 1;
-    atomicMax(&sreg[idx], rho);
+    int idx, rho;
+    if (hll_hash(x[i], p, idx, rho)) atomicMax(&sreg[idx], rho);
   }
   __syncthreads();
   for (int i = threadIdx.x; i < m; i += THREADS)
@@ -535,16 +593,8 @@ __global__ __launch_bounds__(THREADS) void hll_multi_kernel(
   const int64_t s = (int64_t)chunk * per;
   const int64_t e = min(n, s + per);
   for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
-    double v = (double)x[i];
-    if (isnan(v)) continue;
-    uint64_t bits;
-    memcpy(&bits, &v, 8);
-    uint64_t h = splitmix64(bits);
-    int idx = (int)(h >> (64 - p));
-    uint64_t rem = h << p;
-    int rho = (rem == 0) ? (64 - p + 1) : (__clzll((long long)rem) + 1);
-    if (rho > 64 - p + 1) rho = 64 - p + 1;
-    atomicMax(&sreg[idx], rho);
+    int idx, rho;
+    if (hll_hash(x[i], p, idx, rho)) atomicMax(&sreg[idx], rho);
   }
   __syncthreads();
   int32_t *g = &regs[(int64_t)col * m];
@@ -678,15 +728,19 @@ int anovos_bucketize(const void *const *cols, const int64_t *lens, int ncols,
 
 int anovos_bracket_hist_grouped(const void *const *cols, const int64_t *lens,
                                 const int *bstart, int ncols, const double *lo,
-                                const double *hi, int nchunks, int dtype,
-                                uint64_t *out, hipStream_t stream) {
+                                const double *hi, const double *p1lo,
+                                const double *p1scale, int p1bins, int nchunks,
+                                int dtype, uint64_t *out, hipStream_t stream) {
   dim3 grid(ncols * nchunks);
+  size_t lds = (size_t)p1bins * sizeof(int16_t);  // bin1 -> bracket LUT
   if (dtype == 0)
-    hipLaunchKernelGGL(bracket_hist_grouped_kernel<float>, grid, dim3(THREADS), 0, stream,
-                       (const float *const *)cols, lens, bstart, ncols, lo, hi, nchunks, out);
+    hipLaunchKernelGGL(bracket_hist_grouped_kernel<float>, grid, dim3(THREADS), lds, stream,
+                       (const float *const *)cols, lens, bstart, ncols, lo, hi,
+                       p1lo, p1scale, p1bins, nchunks, out);
   else
-    hipLaunchKernelGGL(bracket_hist_grouped_kernel<double>, grid, dim3(THREADS), 0, stream,
-                       (const double *const *)cols, lens, bstart, ncols, lo, hi, nchunks, out);
+    hipLaunchKernelGGL(bracket_hist_grouped_kernel<double>, grid, dim3(THREADS), lds, stream,
+                       (const double *const *)cols, lens, bstart, ncols, lo, hi,
+                       p1lo, p1scale, p1bins, nchunks, out);
   return (int)hipGetLastError();
 }
 

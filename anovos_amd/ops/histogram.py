@@ -170,7 +170,7 @@ def approx_quantiles(
         }
         if not need:
             break
-        refined = _refine_pass(tensors, cols, need, nbins)
+        refined = _refine_pass(tensors, cols, need, nbins, lo, hi)
         brackets.update(refined)
     for (i, j), (bl, bh, off, cnt) in brackets.items():
         c = cols[i]
@@ -183,10 +183,12 @@ def approx_quantiles(
     return result
 
 
-def _refine_pass(tensors, cols, brackets, nbins):
+def _refine_pass(tensors, cols, brackets, nbins, col_lo=None, col_hi=None):
     """One narrowing pass: histogram each active bracket. GPU: ONE grouped
     kernel launch reads every column once and serves all of its brackets
-    (512 sub-bins, rank resolution nbins*512 per pass). CPU: torch loop."""
+    (512 sub-bins, rank resolution nbins*512 per pass); brackets sit on
+    the pass-1 bin grid, so the kernel resolves an element's bracket via
+    a pass-1-bin LUT in O(1). CPU: torch loop."""
     if not brackets:
         return brackets
     keys = sorted(brackets.keys())  # sorted by (col, prob) — grouped kernel needs col-major
@@ -196,8 +198,16 @@ def _refine_pass(tensors, cols, brackets, nbins):
     colidx = torch.tensor([k[0] for k in keys], dtype=torch.int64)
     if dev.type == "cuda" and backend.use_hip(tensors[0]):
         ext = backend.hip_ext()
+        p1bins = nbins
+        if col_lo is None:  # degenerate: treat each bracket as its own grid
+            col_lo = blo.clone()
+            col_hi = bhi.clone()
+        p1lo = col_lo.to(torch.float64)
+        rng = (col_hi.to(torch.float64) - p1lo).clamp(min=1e-300)
+        p1scale = float(p1bins) / rng
         nbins = 512  # grouped kernel's fixed sub-bin count
-        h = ext.bracket_histograms_grouped([t.contiguous() for t in tensors], colidx, blo, bhi)
+        h = ext.bracket_histograms_grouped([t.contiguous() for t in tensors], colidx, blo, bhi,
+                                           p1lo, p1scale, p1bins)
     else:
         h = torch.zeros(len(keys), nbins, dtype=torch.int64, device=dev)
         for kk, (i, j) in enumerate(keys):
