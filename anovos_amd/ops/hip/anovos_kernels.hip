@@ -373,20 +373,46 @@ __global__ __launch_bounds__(THREADS) void bucketize_float_kernel(
   const int64_t per = (n + nchunks - 1) / nchunks;
   const int64_t s = (int64_t)chunk * per;
   const int64_t e = min(n, s + per);
-  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
-    double v = (double)x[i];
-    if (isnan(v)) {
-      out[i] = nanf("");
-      continue;
+
+  auto place = [&](double v) -> float {
+    if (isnan(v)) return nanf("");
+    int lo;
+    if (ncut <= 32) {
+      // linear scan with a wave-uniform index: cuts[j] is an LDS
+      // broadcast (no bank conflicts), the adds are branchless — beats
+      // the divergent-index binary search for the typical 9-cut case
+      lo = 0;
+      for (int j = 0; j < ncut; ++j) lo += (cuts[j] < v) ? 1 : 0;
+    } else {
+      int len = ncut;
+      lo = 0;
+      while (len > 0) {
+        int half = len >> 1;
+        int mid = lo + half;
+        lo = (cuts[mid] < v) ? (mid + 1) : lo;
+        len = (cuts[mid] < v) ? (len - half - 1) : half;
+      }
     }
-    int lo = 0, len = ncut;
-    while (len > 0) {
-      int half = len >> 1;
-      int mid = lo + half;
-      lo = (cuts[mid] < v) ? (mid + 1) : lo;
-      len = (cuts[mid] < v) ? (len - half - 1) : half;
+    return (float)(lo + 1);
+  };
+  if (sizeof(T) == 4) {
+    const int64_t nv = (e - s) / 4;
+    const float4 *xv = reinterpret_cast<const float4 *>(x + s);
+    float4 *ov = reinterpret_cast<float4 *>(out + s);
+    for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
+      float4 v = xv[i];
+      float4 r;
+      r.x = place((double)v.x);
+      r.y = place((double)v.y);
+      r.z = place((double)v.z);
+      r.w = place((double)v.w);
+      ov[i] = r;
     }
-    out[i] = (float)(lo + 1);
+    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS)
+      out[i] = place((double)x[i]);
+  } else {
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS)
+      out[i] = place((double)x[i]);
   }
 }
 
