@@ -84,9 +84,21 @@ def approx_distinct(idf, cols: List[str]) -> Dict[str, int]:
     else:
         flat = torch.stack([hll_registers(idf.col(c).data) for c in cols]).to(torch.int32)
     dist.all_reduce_(flat, "max")
-    flat = flat.cpu()
+    regs = flat.cpu().numpy()
+    # vectorized estimate across all columns at once
+    import numpy as np
+
+    m = float(HLL_M)
+    z = np.power(2.0, -regs.astype(np.float64)).sum(axis=1)
+    alpha = 0.7213 / (1 + 1.079 / m)
+    e = alpha * m * m / z
+    zeros = (regs == 0).sum(axis=1)
+    small = (e <= 2.5 * m) & (zeros > 0)
+    with np.errstate(divide="ignore"):
+        lin = m * np.log(np.where(zeros > 0, m / np.maximum(zeros, 1), 1.0))
+    e = np.where(small, lin, e)
     for i, c in enumerate(cols):
-        out[c] = int(round(hll_estimate(flat[i])))
+        out[c] = int(round(float(e[i])))
     return out
 
 

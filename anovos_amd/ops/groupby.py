@@ -41,8 +41,8 @@ def cat_value_counts(idf, cols: List[str]) -> Dict[str, torch.Tensor]:
         gpu_cnt = {}
         off = 0
         for c, size in zip(gpu_cols, sizes):
-            gpu_cnt[c] = flat_multi[off : off + size]
-            idf.col(c).cache.setdefault("nulls_local", int(flat_multi[off + size]))
+            gpu_cnt[c] = flat_multi[off : off + size]  # stays on device
+            gpu_cnt[c + "\0null"] = flat_multi[off + size : off + size + 1]
             off += size + 1
     for c in cols:
         col = idf.col(c)
@@ -57,11 +57,16 @@ def cat_value_counts(idf, cols: List[str]) -> Dict[str, torch.Tensor]:
     if bufs:
         flat = torch.cat([b for b in bufs]) if len(bufs) > 1 else bufs[0]
         dist.all_reduce_(flat, "sum")
+        flat_host = flat.cpu()  # ONE sync for all columns
         off = 0
         for c, b in zip(cols, bufs):
-            out[c] = flat[off : off + b.numel()].cpu()
+            out[c] = flat_host[off : off + b.numel()]
             idf.col(c).cache["cat_counts"] = out[c]
             off += b.numel()
+    if gpu_cols:
+        null_host = torch.cat([gpu_cnt[c + "\0null"] for c in gpu_cols]).cpu()
+        for i, c in enumerate(gpu_cols):
+            idf.col(c).cache.setdefault("nulls_local", int(null_host[i]))
     return out
 
 

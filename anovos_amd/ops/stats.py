@@ -184,13 +184,22 @@ def null_counts(idf, cols: List[str]) -> Dict[str, int]:
             from anovos_amd.ops import groupby as groupby_ops
 
             groupby_ops.cat_value_counts(idf, cat_gpu)
-        local = torch.empty(len(other), dtype=torch.float64, device=dev)
+        # assemble host-side (a per-element GPU tensor write is a tiny
+        # kernel + sync each); only uncached non-categorical columns need
+        # a device reduction
+        vals = [None] * len(other)
+        pending = []
         for i, c in enumerate(other):
             cached_local = idf.col(c).cache.get("nulls_local")
             if cached_local is not None:
-                local[i] = float(cached_local)
+                vals[i] = float(cached_local)
             else:
-                local[i] = idf.col(c).null_mask().sum()
+                pending.append(i)
+        if pending:
+            sums = torch.stack([idf.col(other[i]).null_mask().sum() for i in pending]).cpu()
+            for k, i in enumerate(pending):
+                vals[i] = float(sums[k])
+        local = torch.tensor(vals, dtype=torch.float64, device=dev if dist.is_dist() else "cpu")
         dist.all_reduce_(local, "sum")
         local_l = local.cpu().numpy().tolist()
         for i, c in enumerate(other):
