@@ -103,3 +103,34 @@ def ll_gh_cols(df: AnovosFrame, max_records: int = 100_000) -> Tuple[List[str], 
     if len(lat_cols) != len(long_cols):
         lat_cols, long_cols = [], []
     return lat_cols, long_cols, gh_cols
+
+
+# ---------------------------------------------------------- parity helpers
+# (reference geo_auto_detection.py:22-98)
+
+def reg_lat_lon(option: str) -> str:
+    """Regex for plausible latitude/longitude strings."""
+    if option == "latitude":
+        return r"^(\+|-|)?(?:90(?:(?:\.0{1,10})?)|(?:[0-9]|[1-8][0-9])(?:(?:\.[0-9]{1,})?))$"
+    if option == "longitude":
+        return r"^(\+|-)?(?:180(?:(?:\.0{1,10})?)|(?:[0-9]|[1-9][0-9]|1[0-7][0-9])(?:(?:\.[0-9]{1,10})?))$"
+    raise ValueError("option must be latitude or longitude")
+
+
+def conv_str_plus(v):
+    """Prefix '+' onto non-negative values (regex normalization)."""
+    if v is None:
+        return None
+    if v < 0:
+        return v
+    return "+" + str(v)
+
+
+def precision_lev(v) -> int:
+    """Number of significant decimal places (0 for integral/None)."""
+    if v is None:
+        return 0
+    frac = format(float(v), ".8f").split(".")[1]
+    if float(frac) > 0:
+        return len(frac)
+    return 0

@@ -321,3 +321,51 @@ def render_report(tabs, title="ML-Anovos Report") -> str:
         "<p>Generated by anovos_amd — MI355X-native feature engineering engine</p></header>"
         f"<div class='tabbar'>{btns}</div>{bodies}</body></html>"
     )
+
+
+def plotSeasonalDecompose(series, period: int = 7, title: str = "Seasonal Decomposition"):
+    """Reference report_generation.py:1942 — trend/seasonal/residual
+    decomposition chart. statsmodels is not in this stack; the classical
+    moving-average decomposition is computed natively (centered MA trend,
+    period-mean seasonal, additive residual)."""
+    import plotly.subplots as sp
+
+    y = np.asarray(pd.to_numeric(pd.Series(series), errors="coerce").ffill().bfill(), dtype=float)
+    n = len(y)
+    if n < 2 * period:
+        period = max(2, n // 2)
+    # centered moving average trend
+    k = period
+    kernel = np.ones(k) / k
+    trend = np.convolve(y, kernel, mode="same")
+    detrended = y - trend
+    seasonal = np.array([np.nanmean(detrended[i::period]) for i in range(period)])
+    seasonal = seasonal - seasonal.mean()
+    seas_full = np.tile(seasonal, n // period + 1)[:n]
+    resid = y - trend - seas_full
+    fig = sp.make_subplots(rows=4, cols=1, shared_xaxes=True,
+                           subplot_titles=["Observed", "Trend", "Seasonal", "Residual"])
+    x = list(range(n))
+    for i, comp in enumerate([y, trend, seas_full, resid]):
+        fig.add_trace(go.Scatter(x=x, y=comp, mode="lines"), row=i + 1, col=1)
+    fig.update_layout(height=700, showlegend=False, title_text=title)
+    return fig
+
+
+def stationarity_check(series, max_lag: int = 1) -> dict:
+    """Rolling-stats stationarity heuristic standing in for the
+    reference's ADF/KPSS imports (report_generation.py:55; statsmodels
+    absent here): compares mean/variance of the two halves and the lag-1
+    autocorrelation of the differenced series."""
+    y = np.asarray(pd.to_numeric(pd.Series(series), errors="coerce").dropna(), dtype=float)
+    n = len(y)
+    if n < 10:
+        return {"stationary": None, "reason": "too few points"}
+    h1, h2 = y[: n // 2], y[n // 2 :]
+    mean_shift = abs(h1.mean() - h2.mean()) / (y.std() + 1e-12)
+    var_ratio = (h1.var() + 1e-12) / (h2.var() + 1e-12)
+    d = np.diff(y)
+    ac1 = float(np.corrcoef(d[:-1], d[1:])[0, 1]) if len(d) > 2 else 0.0
+    stationary = bool(mean_shift < 0.5 and 0.25 < var_ratio < 4.0)
+    return {"stationary": stationary, "mean_shift": float(mean_shift),
+            "variance_ratio": float(var_ratio), "diff_lag1_autocorr": ac1}

@@ -765,3 +765,22 @@ def lagged_ts(idf, list_of_cols, lag, output_type="ts", tsdiff_unit="days", part
             diff = torch.where(null, torch.full_like(diff, float("nan")), diff)
             odf = _emit(odf, i, name, Column(name, "double", diff), output_mode)
     return odf
+
+
+def argument_checker(func_name: str, args: dict):
+    """Reference datetime.py:39 — validate the common argument idiom and
+    return the normalized list_of_cols."""
+    list_of_cols = args.get("list_of_cols")
+    if isinstance(list_of_cols, str):
+        list_of_cols = [x.strip() for x in list_of_cols.split("|")]
+    if not list_of_cols or any(x == "" for x in list_of_cols):
+        raise TypeError(f"Invalid input for column(s) in {func_name}")
+    if "output_mode" in args and args["output_mode"] not in ("replace", "append"):
+        raise TypeError(f"Invalid input for output_mode in {func_name}")
+    if "units" in args:
+        bad = [u for u in args["units"] if u not in args.get("all_units", [])]
+        if bad:
+            raise TypeError(f"Invalid input for units in {func_name}: {bad}")
+    if "unit" in args and "all_units" in args and args["unit"] not in args["all_units"]:
+        raise TypeError(f"Invalid input for unit in {func_name}")
+    return list(dict.fromkeys(list_of_cols))

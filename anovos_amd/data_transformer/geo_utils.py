@@ -285,3 +285,58 @@ def point_in_country_approx(lat: torch.Tensor, lon: torch.Tensor, country: str) 
         raise ValueError(f"No approx bounding box for country '{country}'")
     s, w, n, e = COUNTRY_BBOXES[key]
     return (lat >= s) & (lat <= n) & (lon >= w) & (lon <= e)
+
+
+# ---------------------------------------------------------- parity aliases
+# (reference geo_utils.py names — same math as the tensor helpers above)
+
+def decimal_degrees_to_degrees_minutes_seconds(dd):
+    """Reference geo_utils.py:117."""
+    d, m, s = dd_to_dms(torch.as_tensor(dd, dtype=torch.float64))
+    if d.dim() == 0:
+        return float(d), float(m), float(s)
+    return d, m, s
+
+
+def to_latlon_decimal_degrees(loc, input_format, radius=EARTH_RADIUS):
+    """Reference geo_utils.py:51 — (lat, lon[, z]) in input_format → dd."""
+    if input_format == "dd":
+        return _as_tensor(loc[0]), _as_tensor(loc[1])
+    if input_format == "radian":
+        return radian_to_dd(loc[0], loc[1])
+    if input_format == "dms":
+        lat = dms_to_dd(*[_as_tensor(v) for v in loc[0]]) if isinstance(loc[0], (tuple, list)) else _as_tensor(loc[0])
+        lon = dms_to_dd(*[_as_tensor(v) for v in loc[1]]) if isinstance(loc[1], (tuple, list)) else _as_tensor(loc[1])
+        return lat, lon
+    if input_format == "cartesian":
+        return cartesian_to_dd(loc[0], loc[1], loc[2], radius=radius)
+    if input_format == "geohash":
+        ints, prec = geohash_str_to_int([loc] if isinstance(loc, str) else list(loc))
+        return geohash_decode_int(torch.from_numpy(ints), prec)
+    raise TypeError(f"Invalid input_format {input_format}")
+
+
+def from_latlon_decimal_degrees(loc, output_format, radius=EARTH_RADIUS, geohash_precision=8):
+    """Reference geo_utils.py:161 — dd (lat, lon) → output_format."""
+    lat, lon = _as_tensor(loc[0]), _as_tensor(loc[1])
+    if output_format == "dd":
+        return lat, lon
+    if output_format == "radian":
+        return dd_to_radian(lat, lon)
+    if output_format == "dms":
+        return dd_to_dms(lat), dd_to_dms(lon)
+    if output_format == "cartesian":
+        return dd_to_cartesian(lat, lon, radius=radius)
+    if output_format == "geohash":
+        gh = geohash_encode_int(lat.reshape(-1), lon.reshape(-1), precision=geohash_precision)
+        out = geohash_int_to_str(gh, precision=geohash_precision)
+        return out[0] if lat.dim() == 0 else out
+    raise TypeError(f"Invalid output_format {output_format}")
+
+
+def f_point_in_polygons(polygon_list, south_west_loc=(), north_east_loc=()):
+    """Reference geo_utils.py:530 — curried membership test."""
+    def f(lat, lon):
+        return point_in_polygons(_as_tensor(lat), _as_tensor(lon), polygon_list, south_west_loc, north_east_loc)
+
+    return f
