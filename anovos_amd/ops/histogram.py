@@ -140,25 +140,35 @@ def approx_quantiles(
     tensors = [idf.col(c).data for c in cols]
 
     result = {c: [float("nan")] * len(probs) for c in cols}
-    # bracket per (col, prob): bin containing target rank (vectorized
-    # numpy bookkeeping — per-element tensor indexing costs ~50ms/1000)
+    # bracket per (col, prob): bin containing target rank — fully
+    # vectorized over the (col x prob) grid (the per-pair python loop
+    # cost ~5 ms/step at 150x9)
     brackets = {}
     probs_np = np.asarray(probs)
-    for i, c in enumerate(cols):
-        if moments[c].n == 0 or moments[c].min != moments[c].min:
-            continue
-        l, h = float(lo[i]), float(hi[i])
-        if h <= l:
-            result[c] = [l] * len(probs)
-            continue
-        w = (h - l) / nbins
-        targets = probs_np * (moments[c].n - 1)
-        bs = np.minimum(np.searchsorted(cdf[i], targets + 0.5), nbins - 1)
-        belows = np.where(bs > 0, cdf[i][np.maximum(bs - 1, 0)], 0.0)
-        cnts = hist_np[i][bs]
-        for j in range(len(probs)):
-            b = int(bs[j])
-            brackets[(i, j)] = (l + b * w, l + (b + 1) * w, float(targets[j] - belows[j]), float(cnts[j]))
+    K, P = len(cols), len(probs)
+    ns = np.array([moments[c].n for c in cols], dtype=np.float64)
+    lows = lo.numpy().astype(np.float64)
+    highs = hi.numpy().astype(np.float64)
+    valid = (ns > 0) & (lows == lows)
+    degenerate = valid & (highs <= lows)
+    for i in np.nonzero(degenerate)[0]:
+        result[cols[i]] = [float(lows[i])] * P
+    active = np.nonzero(valid & ~degenerate)[0]
+    if active.size:
+        w = (highs[active] - lows[active]) / nbins  # [A]
+        targets = probs_np[None, :] * (ns[active, None] - 1)  # [A,P]
+        # vectorized per-row searchsorted: count of cdf entries < target
+        cdf_a = cdf[active]  # [A,nbins]
+        bs = np.minimum((cdf_a[:, :, None] < (targets + 0.5)[:, None, :]).sum(axis=1), nbins - 1)  # [A,P]
+        rows = np.arange(active.size)[:, None]
+        belows = np.where(bs > 0, cdf_a[rows, np.maximum(bs - 1, 0)], 0.0)
+        cnts = hist_np[active][rows, bs]
+        for a, i in enumerate(active):
+            l = lows[i]
+            wa = w[a]
+            for j in range(P):
+                b = int(bs[a, j])
+                brackets[(int(i), j)] = (l + b * wa, l + (b + 1) * wa, float(targets[a, j] - belows[a, j]), float(cnts[a, j]))
     for _ in range(refine):
         # adaptive: only brackets whose bin still holds > rel_err/2 of the
         # rank mass need another pass (Spark guarantees 1% rank error;
@@ -227,14 +237,20 @@ def _refine_pass(tensors, cols, brackets, nbins, col_lo=None, col_hi=None):
 
     h_np = h.cpu().numpy().astype(np.float64)
     cdf = np.cumsum(h_np, axis=1)
+    bls = np.array([brackets[k][0] for k in keys])
+    bhs = np.array([brackets[k][1] for k in keys])
+    offs = np.array([brackets[k][2] for k in keys])
+    # vectorized per-row rank search
+    bs = np.minimum((cdf < (offs + 0.5)[:, None]).sum(axis=1), nbins - 1)
+    rows = np.arange(len(keys))
+    belows = np.where(bs > 0, cdf[rows, np.maximum(bs - 1, 0)], 0.0)
+    ws = (bhs - bls) / nbins
     out = {}
     for kk, k in enumerate(keys):
-        bl, bh, off, _ = brackets[k]
-        if bh <= bl:
+        if bhs[kk] <= bls[kk]:
             out[k] = brackets[k]
             continue
-        w = (bh - bl) / nbins
-        b = min(int(np.searchsorted(cdf[kk], off + 0.5)), nbins - 1)
-        below = float(cdf[kk, b - 1]) if b > 0 else 0.0
-        out[k] = (bl + b * w, bl + (b + 1) * w, off - below, float(h_np[kk, b]))
+        b = int(bs[kk])
+        out[k] = (bls[kk] + b * ws[kk], bls[kk] + (b + 1) * ws[kk],
+                  float(offs[kk] - belows[kk]), float(h_np[kk, b]))
     return out
