@@ -38,3 +38,39 @@ def test_income_workflow_on_gpu(tmp_path, monkeypatch):
     assert out.device.type == "cuda"
     assert os.path.exists("report_stats/ml_anovos_report.html")
     assert os.path.getsize("report_stats/ml_anovos_report.html") > 1_000_000
+
+
+@requires_gpu
+def test_datetime_ops_on_gpu():
+    """Datetime civil-decompose arithmetic stays on-device."""
+    import pandas as pd
+
+    from anovos_amd.core.frame import AnovosFrame
+    from anovos_amd.data_transformer import datetime as adt
+
+    pdf = pd.DataFrame({"ts": pd.to_datetime(["2020-01-01", "2020-02-29 13:45:10", "2021-12-31"])})
+    idf = AnovosFrame.from_pandas(pdf, device="cuda:0")
+    odf = adt.timeUnits_extraction(idf, ["ts"], "all")
+    y = odf.col("ts_year")
+    assert y.data.is_cuda
+    assert y.data.cpu().tolist() == [2020.0, 2020.0, 2021.0]
+    b = adt.end_of_month(idf, ["ts"])
+    assert b.col("ts_monthEnd").data.is_cuda
+
+
+@requires_gpu
+def test_geospatial_ops_on_gpu():
+    import pandas as pd
+    import torch
+
+    from anovos_amd.core.frame import AnovosFrame
+    from anovos_amd.data_transformer import geospatial as geo
+
+    pdf = pd.DataFrame({"lat": [40.7128, 51.5074], "lon": [-74.0060, -0.1278]})
+    idf = AnovosFrame.from_pandas(pdf, device="cuda:0")
+    odf = geo.geo_format_latlon(idf, ["lat"], ["lon"], "dd", "geohash", result_prefix=["g"])
+    gh = odf.col("g_geohash")
+    assert gh.data.is_cuda
+    assert gh.dictionary[int(gh.data[0])].startswith("dr5reg")
+    d = geo.location_distance(idf, ["lat", "lon"], ["lat", "lon"], result_prefix="self")
+    assert float(d.col("self_distance").data.abs().max()) < 1e-6
