@@ -122,3 +122,20 @@ def path_ak8s_modify(path: str, auth_key: str = "NA") -> str:
     """wasbs:// path shim parity (reference shared/utils.py:157-179) — in
     this engine cloud URIs are not reachable; return as-is for local use."""
     return str(path)
+
+
+def cloud_sync(local_path: str, master_path: str, run_type: str = "local", auth_key: str = "NA", recursive: bool = False):
+    """Reference parity for the emr/ak8s side-channel copies
+    (report_preprocessing.py:97-119, ts_analyzer.py:452-460): shells
+    ``aws s3 cp`` / ``azcopy`` to push locally-written stats to cloud
+    storage. No-op for local/databricks (databricks paths are direct
+    via output_to_local)."""
+    import subprocess
+
+    if run_type == "emr":
+        cmd = ["aws", "s3", "cp"] + (["--recursive"] if recursive else []) + [ends_with(local_path) if recursive else local_path, ends_with(master_path)]
+        subprocess.check_output(cmd)
+    elif run_type == "ak8s":
+        target = ends_with(path_ak8s_modify(master_path)) + (str(auth_key) if auth_key != "NA" else "")
+        cmd = ["azcopy", "cp", local_path, target] + (["--recursive=true"] if recursive else [])
+        subprocess.check_output(cmd)

@@ -129,8 +129,11 @@ def stats_args(all_configs: Dict, func: str) -> Dict:
 
 def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, device=None):
     """Reference workflow.py:148-870 — the per-key dispatch loop."""
+    from anovos_amd.shared import mlflow_utils
+
     ctx = init_context(device)
     start_main = time.time()
+    mlflow_config = mlflow_utils.setup_mlflow(all_configs.get("mlflow"))
     write_main = all_configs.get("write_main", None)
     write_intermediate = all_configs.get("write_intermediate", None)
     write_stats = all_configs.get("write_stats", None)
@@ -314,6 +317,11 @@ def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, de
 
     if write_main:
         save(df, write_main, "final_dataset")
+        if mlflow_config is not None and mlflow_config.get("track_output"):
+            mlflow_utils.log_artifacts(write_main["file_path"], mlflow_config, "final_dataset")
+    if mlflow_config is not None and report_input_path and mlflow_config.get("track_reports"):
+        mlflow_utils.log_artifacts(report_input_path, mlflow_config, "report_stats")
+    mlflow_utils.end_run(mlflow_config)
     _log(f"workflow total: {time.time() - start_main:.3f}s")
     return df
 
