@@ -169,23 +169,67 @@ def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, de
                 df = new
 
         elif key == "geospatial_controller":
-            ga = args.get("geospatial_analyzer", args)
-            if ga:
-                from anovos_amd.data_analyzer.geospatial_analyzer import geospatial_autodetection
+            # reference workflow.py:272-421: auto-detection/analyzer plus
+            # optional geo_transformations applied to the main frame
+            from anovos_amd.data_analyzer.geospatial_analyzer import geospatial_autodetection
+            from anovos_amd.data_ingest.geo_auto_detection import ll_gh_cols
+            from anovos_amd.data_transformer.geospatial import (
+                centroid as geo_centroid,
+                geo_format_geohash,
+                geo_format_latlon,
+                location_in_country,
+                rog_calculation,
+            )
 
-                geospatial_autodetection(
+            ga = args.get("geospatial_analyzer", {}) or {}
+            gt = args.get("geo_transformations", False)
+            lat_cols, long_cols, gh_cols = [], [], []
+            if ga.get("auto_detection_analyzer", True):
+                def _grid(v, default):
+                    if isinstance(v, str):
+                        return [float(x) for x in v.split(",")]
+                    return v if v is not None else default
+
+                lat_cols, long_cols, gh_cols = geospatial_autodetection(
                     df,
                     ga.get("id_col", ""),
                     report_input_path or ".",
-                    ga.get("max_records", 100000),
+                    ga.get("max_analysis_records", ga.get("max_records", 100000)),
                     ga.get("top_geo_records", 100),
                     ga.get("max_cluster", 20),
-                    ga.get("eps", [0.2, 0.8, 0.2]),
-                    ga.get("min_samples", [25, 100, 25]),
+                    _grid(ga.get("eps"), [0.2, 0.8, 0.2]),
+                    _grid(ga.get("min_samples"), [25, 100, 25]),
                     ga.get("global_map_box_val", 0),
                     run_type,
                     auth_key_val,
                 )
+            if gt:
+                id_col = gt.get("id_col")
+                if not (lat_cols and long_cols) and not gh_cols:
+                    lat_cols = gt.get("list_of_lat", []) or []
+                    long_cols = gt.get("list_of_lon", []) or []
+                    gh_cols = gt.get("list_of_geohash", []) or []
+                if gt.get("location_in_country_detection") and lat_cols:
+                    df = location_in_country(ctx, df, lat_cols, long_cols, gt.get("country", "US"),
+                                             country_shapefile_path=gt.get("country_shapefile_path", ""),
+                                             method_type=gt.get("method_type", "approx"),
+                                             result_prefix=gt.get("result_prefix_lat_lon", []))
+                if gt.get("geo_format_conversion"):
+                    if lat_cols:
+                        df = geo_format_latlon(df, lat_cols, long_cols, gt.get("loc_input_format", "dd"),
+                                               gt.get("loc_output_format", "geohash"),
+                                               result_prefix=gt.get("result_prefix_lat_lon", []))
+                    if gh_cols:
+                        df = geo_format_geohash(df, gh_cols, "dd",
+                                                result_prefix=gt.get("result_prefix_geo", []))
+                if gt.get("centroid_calculation") and lat_cols and id_col:
+                    for idx in range(len(lat_cols)):
+                        df_ = geo_centroid(df, lat_cols[idx], long_cols[idx], id_col)
+                        df = data_ingest.join_dataset(df, df_, join_cols=id_col, join_type="inner")
+                if gt.get("rog_calculation") and lat_cols and id_col:
+                    for idx in range(len(lat_cols)):
+                        df_ = rog_calculation(df, lat_cols[idx], long_cols[idx], id_col)
+                        df = data_ingest.join_dataset(df, df_, join_cols=id_col, join_type="inner")
 
         elif key == "timeseries_analyzer":
             id_col = args.get("id_col", "")

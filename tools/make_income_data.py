@@ -55,12 +55,48 @@ def make(n: int, seed: int = 11) -> pd.DataFrame:
     return df
 
 
+def add_geo_cols(df: pd.DataFrame, seed: int = 21) -> pd.DataFrame:
+    """Append latitude/longitude (US-ish) and a geohash column."""
+    import sys, os
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from anovos_amd.data_ingest.geo_auto_detection import latlong_to_geo
+
+    rng = np.random.default_rng(seed)
+    n = len(df)
+    lat = rng.uniform(25.5, 48.5, n)
+    lon = rng.uniform(-123.0, -68.0, n)
+    df = df.copy()
+    df["latitude"] = lat
+    df["longitude"] = lon
+    df["gh7"] = latlong_to_geo(lat, lon, precision=7)
+    return df
+
+
+def add_ts_cols(df: pd.DataFrame, seed: int = 31) -> pd.DataFrame:
+    """Append transaction-date / signup-epoch timestamp candidates."""
+    rng = np.random.default_rng(seed)
+    n = len(df)
+    base = pd.Timestamp("2020-01-01")
+    df = df.copy()
+    df["txn_date"] = (base + pd.to_timedelta(rng.integers(0, 365, n), unit="D")
+                      + pd.to_timedelta(rng.integers(0, 86400, n), unit="s")).strftime("%Y-%m-%d %H:%M:%S")
+    df["signup_epoch"] = (1577836800 + rng.integers(0, 365 * 86400, n)).astype("int64")
+    return df
+
+
 if __name__ == "__main__":
     ap = argparse.ArgumentParser()
     ap.add_argument("--rows", type=int, default=32561)
     ap.add_argument("--out", default="data/income_dataset")
+    ap.add_argument("--geo", action="store_true", help="append lat/lon/geohash columns")
+    ap.add_argument("--ts", action="store_true", help="append timestamp candidate columns")
+    ap.add_argument("--snapshots", type=int, default=0, help="write N drifting snapshot datasets")
     a = ap.parse_args()
     df = make(a.rows)
+    if a.geo:
+        df = add_geo_cols(df)
+    if a.ts:
+        df = add_ts_cols(df)
     os.makedirs(os.path.join(a.out, "csv"), exist_ok=True)
     os.makedirs(os.path.join(a.out, "parquet"), exist_ok=True)
     df.to_csv(os.path.join(a.out, "csv", "part-00000.csv"), index=False)
@@ -70,4 +106,12 @@ if __name__ == "__main__":
     src["age"] = src["age"] * 1.05
     os.makedirs(os.path.join(a.out, "source", "csv"), exist_ok=True)
     src.to_csv(os.path.join(a.out, "source", "csv", "part-00000.csv"), index=False)
+    for k in range(a.snapshots):
+        snap = make(max(a.rows // 4, 1000), seed=100 + k)
+        # gradual drift across snapshots
+        snap["age"] = snap["age"] * (1 + 0.01 * k)
+        snap["hours-per-week"] = snap["hours-per-week"] + 0.3 * k
+        d = os.path.join(a.out, f"snapshot{k + 1:02d}", "csv")
+        os.makedirs(d, exist_ok=True)
+        snap.to_csv(os.path.join(d, "part-00000.csv"), index=False)
     print("wrote", a.out)
