@@ -119,6 +119,11 @@ def approx_quantiles(
     # so below this size we sort-and-index instead of sketching.
     if all(moments[c].n <= EXACT_N_THRESHOLD for c in cols):
         return _exact_quantiles(idf, cols, probs, moments, rel_err=rel_err)
+    # value cache keyed per (col, prob): sections share quantile work
+    # (percentiles computes 1..99%; outlier/imputation reuse 5/50/95%)
+    probs_l = list(probs)
+    if all(all(("q", p) in idf.col(c).cache for p in probs_l) for c in cols):
+        return {c: [idf.col(c).cache[("q", p)] for p in probs_l] for c in cols}
     dev = idf.device
     lo = torch.tensor([moments[c].min for c in cols], dtype=torch.float64)
     hi = torch.tensor([moments[c].max for c in cols], dtype=torch.float64)
@@ -190,6 +195,9 @@ def approx_quantiles(
             # interpolate inside the (now tiny) bracket by rank fraction
             frac = min(max(off / max(cnt - 1, 1e-9), 0.0), 1.0) if cnt > 1 else 0.0
             result[c][j] = bl + frac * (bh - bl)
+    for c in cols:
+        for j, p in enumerate(probs_l):
+            idf.col(c).cache[("q", p)] = result[c][j]
     return result
 
 
