@@ -48,7 +48,7 @@ def test_datetime_ops_on_gpu():
     from anovos_amd.core.frame import AnovosFrame
     from anovos_amd.data_transformer import datetime as adt
 
-    pdf = pd.DataFrame({"ts": pd.to_datetime(["2020-01-01", "2020-02-29 13:45:10", "2021-12-31"])})
+    pdf = pd.DataFrame({"ts": pd.to_datetime(["2020-01-01 00:00:00", "2020-02-29 13:45:10", "2021-12-31 23:59:59"])})
     idf = AnovosFrame.from_pandas(pdf, device="cuda:0")
     odf = adt.timeUnits_extraction(idf, ["ts"], "all")
     y = odf.col("ts_year")
