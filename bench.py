@@ -37,35 +37,50 @@ CAT_CARD = 40
 
 def make_synthetic_frame(rows: int, device, seed: int):
     """200-column synthetic tabular frame (numeric + categorical), nulls
-    included, generated directly on-device."""
+    included, generated directly on-device (a 125M-row/GPU shard — the
+    1Bx200-at-8-GPU config — is 100 GB and generates in seconds in HBM;
+    staging through host RAM would take minutes)."""
     from anovos_amd.core.frame import AnovosFrame, Column
 
-    g = torch.Generator(device="cpu")
+    dev = torch.device(device)
+    on_gpu = dev.type == "cuda"
+    g = torch.Generator(device=dev if on_gpu else "cpu")
     g.manual_seed(seed)
+
+    def rand(*a, **k):
+        return torch.rand(*a, generator=g, device=dev if on_gpu else "cpu", **k)
+
+    def randn(*a, **k):
+        return torch.randn(*a, generator=g, device=dev if on_gpu else "cpu", **k)
+
+    def randint(lo, hi, shape, **k):
+        return torch.randint(lo, hi, shape, generator=g, device=dev if on_gpu else "cpu", **k)
+
     cols = {}
     # continuous columns: varied scale/shape, ~1% nulls
-    base = torch.randn(rows, generator=g)
+    base = randn(rows)
     for i in range(N_NUM_CONT):
-        x = (base * (1 + 0.1 * i)).clone() if i % 7 == 0 else torch.randn(rows, generator=g)
+        x = (base * (1 + 0.1 * i)).clone() if i % 7 == 0 else randn(rows)
         x = x * (1.0 + i % 5) + (i % 11)
-        nulls = torch.rand(rows, generator=g) < 0.01
+        nulls = rand(rows) < 0.01
         x[nulls] = float("nan")
-        cols[f"num_{i}"] = Column(f"num_{i}", "float", x.to(device))
+        cols[f"num_{i}"] = Column(f"num_{i}", "float", x.to(dev))
+    del base
     for i in range(N_NUM_INT):
-        x = torch.randint(0, 80 + i, (rows,), generator=g).to(torch.float32)
-        nulls = torch.rand(rows, generator=g) < 0.01
+        x = randint(0, 80 + i, (rows,)).to(torch.float32)
+        nulls = rand(rows) < 0.01
         x[nulls] = float("nan")
-        cols[f"int_{i}"] = Column(f"int_{i}", "int", x.to(device))
+        cols[f"int_{i}"] = Column(f"int_{i}", "int", x.to(dev))
     for i in range(N_CAT):
-        codes = torch.randint(0, CAT_CARD, (rows,), generator=g).to(torch.int32)
-        nulls = torch.rand(rows, generator=g) < 0.01
+        codes = randint(0, CAT_CARD, (rows,)).to(torch.int32)
+        nulls = rand(rows) < 0.01
         codes[nulls] = -1
         dictionary = [f"cat{i}_v{j}" for j in range(CAT_CARD)]
-        cols[f"cat_{i}"] = Column(f"cat_{i}", "string", codes.to(device), dictionary)
+        cols[f"cat_{i}"] = Column(f"cat_{i}", "string", codes.to(dev), dictionary)
     # label column for supervised paths
-    lab = (torch.rand(rows, generator=g) < 0.25).to(torch.int32)
-    cols["label"] = Column("label", "string", lab.to(device), ["no", "yes"])
-    return AnovosFrame(cols, device)
+    lab = (rand(rows) < 0.25).to(torch.int32)
+    cols["label"] = Column("label", "string", lab.to(dev), ["no", "yes"])
+    return AnovosFrame(cols, dev)
 
 
 def pipeline_step(ctx, idf, source_hist, model_dir):
@@ -107,6 +122,7 @@ def pipeline_step(ctx, idf, source_hist, model_dir):
         ctx, idf, num_cols, detection_side="both", treatment=True, treatment_method="value_replacement"
     )
     chk += float(odf_out.col(num_cols[0]).data[:8].float().nansum().item())
+    del odf_out  # free treated copies promptly (125M-row shards: ~75 GB each)
 
     # ---- drift (PSI/JSD/HD/KS vs warmup snapshot) ----
     binned = T.attribute_binning(ctx, idf, num_cols, bin_size=10, output_mode="append")  # K6
@@ -130,14 +146,19 @@ def pipeline_step(ctx, idf, source_hist, model_dir):
     chk += drift_vals
 
     # ---- transform ----
+    del binned
     t1 = T.z_standardization(ctx, idf, num_cols[:N_NUM_CONT])  # K11
     chk += float(t1.col(num_cols[0]).data[:8].float().nansum().item())
+    del t1
     t2 = T.imputation_MMM(ctx, idf, method_type="median")
     chk += float(t2.col(num_cols[1]).data[:8].float().nansum().item())
+    del t2
     t3 = T.cat_to_num_unsupervised(ctx, idf, cat_cols[:25], method_type="label_encoding")  # K12
     chk += float(t3.col(cat_cols[0] + "_index").data[:8].float().nansum().item()) if (cat_cols[0] + "_index") in t3.columns else float(t3.col(cat_cols[0]).data[:8].float().sum().item())
+    del t3
     t4 = T.outlier_categories(ctx, idf, cat_cols[25:], max_category=20)
     chk += float(t4.col(cat_cols[25]).data[:8].float().sum().item())
+    del t4
     return chk
 
 
