@@ -167,7 +167,8 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=5)
     ap.add_argument("--warmup", type=int, default=2)
-    ap.add_argument("--rows", type=int, default=0, help="rows per GPU (default 10M on GPU, 200k on CPU)")
+    ap.add_argument("--rows", type=int, default=0,
+                    help="rows per GPU (default 125M on GPU = the 1Bx200-at-8-GPU shard, 200k on CPU)")
     args = ap.parse_args()
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
@@ -178,7 +179,10 @@ def main():
     rank = ctx.rank
     world = max(ctx.world_size, 1)
     on_gpu = ctx.device.type == "cuda"
-    rows = args.rows or (10_000_000 if on_gpu else 200_000)
+    # GPU default: 125M rows/GPU x 201 cols — at --gpus 8 this IS the
+    # BASELINE.json 1Bx200 config (100 GB resident/GPU; bigger shards
+    # also amortize host orchestration, see profiles/)
+    rows = args.rows or (125_000_000 if on_gpu else 200_000)
 
     idf = make_synthetic_frame(rows, ctx.device, seed=1234 + rank)
 
