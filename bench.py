@@ -200,6 +200,7 @@ def main():
     total = idf.count()
     freqs = dd.batched_bin_frequencies(binned, [c + "_binned" for c in num_cols], total)
     source_hist = {c: dict(zip(*freqs[c + "_binned"])) for c in num_cols}
+    del binned  # 75 GB at the 125M-row shard — must not stay resident
 
     for _ in range(args.warmup):
         pipeline_step(ctx, idf, source_hist, model_dir)
