@@ -535,9 +535,26 @@ __global__ __launch_bounds__(THREADS) void hll_kernel(
   const int64_t per = (n + nchunks - 1) / nchunks;
   const int64_t s = (int64_t)chunk * per;
   const int64_t e = min(n, s + per);
-  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
-    int idx, rho;
-    if (hll_hash(x[i], p, idx, rho)) atomicMax(&sreg[idx], rho);
+  if (sizeof(T) == 4) {
+    const float4 *xv = reinterpret_cast<const float4 *>(x + s);
+    const int64_t nv = (e - s) / 4;
+    for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
+      float4 v = xv[i];
+      int idx, rho;
+      if (hll_hash((T)v.x, p, idx, rho)) atomicMax(&sreg[idx], rho);
+      if (hll_hash((T)v.y, p, idx, rho)) atomicMax(&sreg[idx], rho);
+      if (hll_hash((T)v.z, p, idx, rho)) atomicMax(&sreg[idx], rho);
+      if (hll_hash((T)v.w, p, idx, rho)) atomicMax(&sreg[idx], rho);
+    }
+    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) {
+      int idx, rho;
+      if (hll_hash(x[i], p, idx, rho)) atomicMax(&sreg[idx], rho);
+    }
+  } else {
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+      int idx, rho;
+      if (hll_hash(x[i], p, idx, rho)) atomicMax(&sreg[idx], rho);
+    }
   }
   __syncthreads();
   for (int i = threadIdx.x; i < m; i += THREADS)
@@ -618,9 +635,26 @@ __global__ __launch_bounds__(THREADS) void hll_multi_kernel(
   const int64_t per = (n + nchunks - 1) / nchunks;
   const int64_t s = (int64_t)chunk * per;
   const int64_t e = min(n, s + per);
-  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
-    int idx, rho;
-    if (hll_hash(x[i], p, idx, rho)) atomicMax(&sreg[idx], rho);
+  if (sizeof(T) == 4) {
+    const float4 *xv = reinterpret_cast<const float4 *>(x + s);
+    const int64_t nv = (e - s) / 4;
+    for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
+      float4 v = xv[i];
+      int idx, rho;
+      if (hll_hash((T)v.x, p, idx, rho)) atomicMax(&sreg[idx], rho);
+      if (hll_hash((T)v.y, p, idx, rho)) atomicMax(&sreg[idx], rho);
+      if (hll_hash((T)v.z, p, idx, rho)) atomicMax(&sreg[idx], rho);
+      if (hll_hash((T)v.w, p, idx, rho)) atomicMax(&sreg[idx], rho);
+    }
+    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) {
+      int idx, rho;
+      if (hll_hash(x[i], p, idx, rho)) atomicMax(&sreg[idx], rho);
+    }
+  } else {
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+      int idx, rho;
+      if (hll_hash(x[i], p, idx, rho)) atomicMax(&sreg[idx], rho);
+    }
   }
   __syncthreads();
   int32_t *g = &regs[(int64_t)col * m];
@@ -647,7 +681,8 @@ __global__ __launch_bounds__(THREADS) void axpb_kernel(
   const int64_t e = min(n, s + per);
   const float av = (float)a[col];
   const float bv = (float)b[col];
-  if (sizeof(T) == 4 && ((e - s) % 4 == 0) && ((s & 3) == 0)) {
+  if (sizeof(T) == 4) {
+    // dwordx4 loads need only 4-byte alignment on CDNA — no head guard
     const float4 *xv = reinterpret_cast<const float4 *>(x + s);
     float4 *ov = reinterpret_cast<float4 *>(out + s);
     const int64_t nv = (e - s) / 4;
@@ -659,6 +694,8 @@ __global__ __launch_bounds__(THREADS) void axpb_kernel(
       v.w = (v.w - av) * bv;
       ov[i] = v;
     }
+    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS)
+      out[i] = ((float)x[i] - av) * bv;
   } else {
     for (int64_t i = s + threadIdx.x; i < e; i += THREADS)
       out[i] = ((float)x[i] - av) * bv;
@@ -678,9 +715,28 @@ __global__ __launch_bounds__(THREADS) void fillnan_kernel(
   const int64_t s = (int64_t)chunk * per;
   const int64_t e = min(n, s + per);
   const T fv = (T)fill[col];
-  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
-    T v = x[i];
-    out[i] = isnan((double)v) ? fv : v;
+  if (sizeof(T) == 4) {
+    const float4 *xv = reinterpret_cast<const float4 *>(x + s);
+    float4 *ov = reinterpret_cast<float4 *>(out + s);
+    const float ff = (float)fv;
+    const int64_t nv = (e - s) / 4;
+    for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
+      float4 v = xv[i];
+      v.x = isnan(v.x) ? ff : v.x;
+      v.y = isnan(v.y) ? ff : v.y;
+      v.z = isnan(v.z) ? ff : v.z;
+      v.w = isnan(v.w) ? ff : v.w;
+      ov[i] = v;
+    }
+    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) {
+      T v = x[i];
+      out[i] = isnan((double)v) ? fv : v;
+    }
+  } else {
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+      T v = x[i];
+      out[i] = isnan((double)v) ? fv : v;
+    }
   }
 }
 
