@@ -127,3 +127,54 @@ def test_workflow_basic_report_short_circuit(income_csv, tmp_path):
         yaml.safe_dump(cfg, f)
     workflow.run(str(cfg_path))
     assert (tmp_path / "basic" / "basic_report.html").exists()
+
+
+def test_workflow_time_series_config(tmp_path, monkeypatch):
+    """configs_time_series.yaml shape: ts auto-detect + stability over
+    snapshots (compact synthetic variant)."""
+    import subprocess
+    import sys
+
+    monkeypatch.chdir(tmp_path)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    sys.path.insert(0, os.path.join(repo, "tools"))
+    import make_income_data as mid
+
+    df = mid.add_ts_cols(mid.make(3000))
+    os.makedirs("data/income_dataset/csv", exist_ok=True)
+    df.to_csv("data/income_dataset/csv/part-00000.csv", index=False)
+    for k in range(12):
+        snap = mid.make(800, seed=100 + k)
+        snap["age"] = snap["age"] * (1 + 0.01 * k)
+        d = f"data/income_dataset/snapshot{k + 1:02d}/csv"
+        os.makedirs(d, exist_ok=True)
+        snap.to_csv(os.path.join(d, "part-00000.csv"), index=False)
+    from anovos_amd import workflow
+
+    workflow.run(os.path.join(repo, "config", "configs_time_series.yaml"))
+    assert os.path.exists("report_stats/stability_index.csv")
+    assert os.path.exists("report_stats/ts_cols_stats.csv")
+    si = __import__("pandas").read_csv("report_stats/stability_index.csv")
+    assert "stability_index" in si.columns and len(si) >= 3
+    assert os.path.exists("report_stats/ml_anovos_report.html")
+
+
+def test_workflow_geospatial_config(tmp_path, monkeypatch):
+    """configs_geospatial.yaml shape: autodetect + geo transformations."""
+    import sys
+
+    monkeypatch.chdir(tmp_path)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    sys.path.insert(0, os.path.join(repo, "tools"))
+    import make_income_data as mid
+
+    df = mid.add_geo_cols(mid.make(3000))
+    os.makedirs("data/income_dataset/csv", exist_ok=True)
+    df.to_csv("data/income_dataset/csv/part-00000.csv", index=False)
+    from anovos_amd import workflow
+
+    out = workflow.run(os.path.join(repo, "config", "configs_geospatial.yaml"))
+    assert "lat_long_in_usa" in out.columns
+    assert "lat_long_geohash" in out.columns
+    assert "radius_of_gyration" in out.columns
+    assert os.path.exists("report_stats/Overall_Summary_1_latitude_longitude.csv")
