@@ -338,7 +338,9 @@ def weighted_centroid(idf, id_col, lat_col, long_col):
     ok = gu.in_range(lat, lon) & ~torch.isnan(lat) & ~torch.isnan(lon)
     inv, valid, decode = _id_codes(idf, id_col)
     m = ok & valid
-    # weight = multiplicity of the exact (id, lat, lon) triple
+    # the reference weights distinct (lat, lon) pairs by their row
+    # multiplicity — algebraically identical to the plain per-row mean
+    # computed here (each row contributes once either way)
     x, y, z = gu.dd_to_cartesian(lat, lon, radius=1.0)
     G = int(inv.max().item()) + 1 if inv.numel() else 0
     cnt = torch.zeros(G, dtype=torch.float64, device=lat.device).scatter_reduce(0, inv[m], torch.ones_like(x[m]), reduce="sum")

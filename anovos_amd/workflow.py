@@ -19,6 +19,8 @@ import os
 import time
 from typing import Dict, Optional
 
+import logging
+
 import yaml
 
 from anovos_amd.data_ingest import data_ingest
@@ -35,8 +37,16 @@ from anovos_amd.drift_stability import stability as dstability
 from anovos_amd.shared.context import init_context
 
 
+logger = logging.getLogger("anovos_amd.workflow")
+if not logger.handlers:
+    _h = logging.StreamHandler()
+    _h.setFormatter(logging.Formatter("[%(name)s] %(message)s"))
+    logger.addHandler(_h)
+    logger.setLevel(logging.INFO)
+
+
 def _log(msg: str):
-    print(f"[anovos_amd.workflow] {msg}", flush=True)
+    logger.info(msg)
 
 
 def ETL(ctx, args: Dict):
