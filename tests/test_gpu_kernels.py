@@ -274,3 +274,37 @@ def test_outlier_clamp_columns_vs_torch(ext):
     x = cols[0]
     nulled = torch.isnan(outs2[0]).sum()
     assert int(nulled) == int(((x < -5) | (x > 5)).sum())
+
+
+@requires_gpu
+def test_approx_quantiles_sketch_path_vs_torch(ext):
+    """Large-n path: pass-1 histogram + LUT-grouped refinement must land
+    within Spark's rel-err 0.01 rank tolerance of torch.quantile."""
+    from anovos_amd.core.frame import AnovosFrame, Column
+    from anovos_amd.ops import histogram as hist_ops
+
+    g = torch.Generator(device="cpu").manual_seed(23)
+    n = 2_000_000  # >> EXACT_N_THRESHOLD -> sketch + refinement path
+    cols = {}
+    specs = {
+        "normal": torch.randn(n, generator=g) * 7 + 3,
+        "lognormal": torch.randn(n, generator=g).exp() * 10,
+        "uniform": torch.rand(n, generator=g) * 1000 - 500,
+        "heavy_dup": torch.randint(0, 20, (n,), generator=g).to(torch.float32),
+    }
+    for name, x in specs.items():
+        x[torch.rand(n, generator=g) < 0.01] = float("nan")
+        cols[name] = Column(name, "float", x.cuda())
+    idf = AnovosFrame(cols, device="cuda")
+    probs = [0.01, 0.05, 0.25, 0.5, 0.75, 0.95, 0.99]
+    got = hist_ops.approx_quantiles(idf, list(specs), probs)
+    for name in specs:
+        x = cols[name].data
+        v = x[~torch.isnan(x)]
+        vs, _ = torch.sort(v)
+        nn = vs.numel()
+        for p, q in zip(probs, got[name]):
+            # rank-tolerance check (Spark approxQuantile contract):
+            # the returned value's rank must be within rel_err of p*n
+            rank = int(torch.searchsorted(vs, torch.tensor(q, device=vs.device)))
+            assert abs(rank - p * nn) <= max(0.01 * nn, 1000), (name, p, q, rank / nn)
