@@ -209,7 +209,22 @@ def _refine_pass(tensors, cols, brackets, nbins, col_lo=None, col_hi=None):
     a pass-1-bin LUT in O(1). CPU: torch loop."""
     if not brackets:
         return brackets
-    keys = sorted(brackets.keys())  # sorted by (col, prob) — grouped kernel needs col-major
+    all_keys = sorted(brackets.keys())  # sorted by (col, prob) — grouped kernel needs col-major
+    # dedupe identical (col, range) brackets: several probs often land in
+    # the SAME pass-1 bin (heavy-duplicate data) and the kernel's
+    # bin→bracket LUT holds one entry per bin — the shared histogram
+    # serves every prob that maps to it
+    uniq = {}
+    key_to_uniq = {}
+    for k in all_keys:
+        u = (k[0], brackets[k][0], brackets[k][1])
+        if u not in uniq:
+            uniq[u] = len(uniq)
+        key_to_uniq[k] = uniq[u]
+    keys = [None] * len(uniq)
+    for k in all_keys:
+        if keys[key_to_uniq[k]] is None:
+            keys[key_to_uniq[k]] = k
     dev = tensors[0].device
     blo = torch.tensor([brackets[k][0] for k in keys], dtype=torch.float64)
     bhi = torch.tensor([brackets[k][1] for k in keys], dtype=torch.float64)
@@ -245,20 +260,23 @@ def _refine_pass(tensors, cols, brackets, nbins, col_lo=None, col_hi=None):
 
     h_np = h.cpu().numpy().astype(np.float64)
     cdf = np.cumsum(h_np, axis=1)
-    bls = np.array([brackets[k][0] for k in keys])
-    bhs = np.array([brackets[k][1] for k in keys])
-    offs = np.array([brackets[k][2] for k in keys])
-    # vectorized per-row rank search
-    bs = np.minimum((cdf < (offs + 0.5)[:, None]).sum(axis=1), nbins - 1)
-    rows = np.arange(len(keys))
-    belows = np.where(bs > 0, cdf[rows, np.maximum(bs - 1, 0)], 0.0)
+    # resolve EVERY original (col, prob) key against its (possibly
+    # shared) bracket histogram — off differs per prob even when the
+    # range is shared
+    urow = np.array([key_to_uniq[k] for k in all_keys])
+    bls = np.array([brackets[k][0] for k in all_keys])
+    bhs = np.array([brackets[k][1] for k in all_keys])
+    offs = np.array([brackets[k][2] for k in all_keys])
+    bs = np.minimum((cdf[urow] < (offs + 0.5)[:, None]).sum(axis=1), nbins - 1)
+    rows = np.arange(len(all_keys))
+    belows = np.where(bs > 0, cdf[urow, np.maximum(bs - 1, 0)], 0.0)
     ws = (bhs - bls) / nbins
     out = {}
-    for kk, k in enumerate(keys):
+    for kk, k in enumerate(all_keys):
         if bhs[kk] <= bls[kk]:
             out[k] = brackets[k]
             continue
         b = int(bs[kk])
         out[k] = (bls[kk] + b * ws[kk], bls[kk] + (b + 1) * ws[kk],
-                  float(offs[kk] - belows[kk]), float(h_np[kk, b]))
+                  float(offs[kk] - belows[kk]), float(h_np[urow[kk], b]))
     return out
