@@ -154,7 +154,7 @@ def frame_moments(idf, cols: List[str]) -> Dict[str, MomentStats]:
     if todo:
         tensors = [idf.col(c).data for c in todo]
         local = column_moments_local(tensors)
-        glob = merge_moments_global(local)
+        glob = merge_moments_global(local).numpy().tolist()  # one conversion, not 9/col
         for i, c in enumerate(todo):
             m = MomentStats(glob[i])
             idf.col(c).cache["moments"] = m
