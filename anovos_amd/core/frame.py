@@ -16,7 +16,6 @@ Columns (core/dtypes.py):
 
 from __future__ import annotations
 
-import copy as _copy
 from typing import Dict, Iterable, List, Optional, Sequence, Union
 
 import numpy as np

@@ -18,7 +18,6 @@ import json
 import os
 from typing import Dict, List, Optional
 
-import numpy as np
 import pandas as pd
 import pyarrow as pa
 import pyarrow.csv as pacsv

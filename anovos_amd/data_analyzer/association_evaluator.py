@@ -12,7 +12,6 @@ from __future__ import annotations
 
 import math
 import warnings
-from typing import Dict, List
 
 import numpy as np
 import pandas as pd

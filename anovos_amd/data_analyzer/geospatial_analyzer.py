@@ -13,7 +13,6 @@ always driver-side pandas, geospatial_analyzer.py:463-470).
 from __future__ import annotations
 
 import os
-from typing import List, Optional
 
 import numpy as np
 import pandas as pd

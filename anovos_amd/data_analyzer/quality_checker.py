@@ -19,7 +19,7 @@ from __future__ import annotations
 
 import re
 import warnings
-from typing import Dict, List, Tuple
+from typing import Dict, List
 
 import numpy as np
 import pandas as pd

@@ -13,7 +13,6 @@ all tiny driver-side tables).
 from __future__ import annotations
 
 import warnings
-from typing import Optional
 
 import numpy as np
 import pandas as pd

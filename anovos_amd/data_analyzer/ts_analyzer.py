@@ -11,7 +11,7 @@ scatter_reduce on the GPU; only the tiny result tables land on host.
 from __future__ import annotations
 
 import os
-from typing import List, Optional
+from typing import Optional
 
 import numpy as np
 import pandas as pd

@@ -15,7 +15,6 @@ import os
 import re
 from typing import List, Tuple
 
-import numpy as np
 import pandas as pd
 import torch
 

@@ -10,9 +10,7 @@ in this stack); all statistics come from the engine's fused GPU kernels.
 from __future__ import annotations
 
 import os
-from typing import Optional
 
-import pandas as pd
 
 from anovos_amd.data_analyzer import association_evaluator as ae
 from anovos_amd.data_analyzer import quality_checker as qc

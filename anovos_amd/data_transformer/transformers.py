@@ -22,7 +22,7 @@ from __future__ import annotations
 import math
 import os
 import warnings
-from typing import Dict, List, Optional, Sequence
+from typing import Dict, Optional
 
 import numpy as np
 import pandas as pd

@@ -17,7 +17,6 @@ sites match the reference module layout.
 
 from __future__ import annotations
 
-import math
 import os
 import pickle
 import warnings

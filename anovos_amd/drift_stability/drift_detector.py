@@ -12,7 +12,6 @@ arithmetic runs on the tiny (bin_size+1)-vectors.
 from __future__ import annotations
 
 import os
-from typing import List
 
 import numpy as np
 import pandas as pd

@@ -12,7 +12,6 @@ feature_stability_estimation: first/second-order Taylor propagation of
 from __future__ import annotations
 
 import os
-from typing import List
 
 import numpy as np
 import pandas as pd

@@ -9,7 +9,7 @@ binary search per value. Torch path: torch.bucketize per column.
 
 from __future__ import annotations
 
-from typing import Dict, List, Sequence
+from typing import List, Sequence
 
 import torch
 

@@ -18,7 +18,7 @@ as the column mean (contributes zero to covariance).
 
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List
 
 import numpy as np
 import torch

@@ -17,7 +17,7 @@ from __future__ import annotations
 import copy
 import os
 import time
-from typing import Dict, Optional
+from typing import Dict
 
 import logging
 
