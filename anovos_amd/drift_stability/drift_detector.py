@@ -55,6 +55,14 @@ def statistics(
     """Returns [attribute, <metrics...>, flagged] — reference
     drift_detector.py:18-371."""
     drop_cols = drop_cols or []
+    # drift requires the column in BOTH frames; columns only present in
+    # the target (e.g. joined-in side columns) are skipped with a warning
+    absent = [c for c in list_of_cols if c not in idf_source.columns]
+    if absent:
+        import warnings
+
+        warnings.warn("Columns not present in the source dataset are excluded from drift: " + ", ".join(absent))
+        list_of_cols = [c for c in list_of_cols if c not in absent]
     num_cols = attributeType_segregation(idf_target.select(list_of_cols))[0]
 
     count_target = idf_target.count()

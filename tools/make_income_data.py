@@ -106,6 +106,15 @@ if __name__ == "__main__":
     src["age"] = src["age"] * 1.05
     os.makedirs(os.path.join(a.out, "source", "csv"), exist_ok=True)
     src.to_csv(os.path.join(a.out, "source", "csv", "part-00000.csv"), index=False)
+    # join dataset (avro): ifa + duplicated demo columns (reference
+    # config joins an avro side table, configs.yaml:51-63)
+    import sys as _sys
+    _sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from anovos_amd.core.avro_codec import write_avro
+
+    join_df = df[["ifa", "age", "workclass"]].rename(columns={"age": "dupl_age", "workclass": "dupl_workclass"})
+    os.makedirs(os.path.join(a.out, "join"), exist_ok=True)
+    write_avro(join_df, os.path.join(a.out, "join", "part-00000.avro"))
     for k in range(a.snapshots):
         snap = make(max(a.rows // 4, 1000), seed=100 + k)
         # gradual drift across snapshots
