@@ -80,7 +80,18 @@ def approx_distinct(idf, cols: List[str]) -> Dict[str, int]:
     first = idf.col(cols[0]).data
     if first.is_cuda and backend.use_hip(first):
         ext = backend.hip_ext()
-        flat = ext.hll_registers_multi([idf.col(c).data.contiguous() for c in cols], HLL_P)
+        cold = [c for c in cols if "moments" not in idf.col(c).cache]
+        if len(cold) >= max(2, len(cols) // 2):
+            # fused K1/K2+K4: the same read fills the moment cache every
+            # later analyzer section will hit (ops/hip moments_hll)
+            from anovos_amd.ops import stats as stats_ops
+
+            mom_local, flat = ext.moments_hll([idf.col(c).data.contiguous() for c in cols], HLL_P)
+            glob = stats_ops.merge_moments_global(mom_local).numpy().tolist()
+            for i, c in enumerate(cols):
+                idf.col(c).cache.setdefault("moments", stats_ops.MomentStats(glob[i]))
+        else:
+            flat = ext.hll_registers_multi([idf.col(c).data.contiguous() for c in cols], HLL_P)
     else:
         flat = torch.stack([hll_registers(idf.col(c).data) for c in cols]).to(torch.int32)
     dist.all_reduce_(flat, "max")

@@ -62,6 +62,10 @@ int anovos_outlier_clamp(const void *const *cols, const int64_t *lens,
                          int ncols, const double *lo, const double *hi,
                          int nchunks, int mode, int dtype, void *const *outs,
                          uint64_t *counts, hipStream_t stream);
+int anovos_moments_hll(const void *const *cols, const int64_t *lens,
+                       int ncols, int p, int nchunks, int dtype,
+                       double *partials, double *mom_out, int32_t *regs,
+                       hipStream_t stream);
 int anovos_centered_gram(const void *const *cols, int64_t n, int k,
                          const float *means, const int *pair_i,
                          const int *pair_j, int npairs, int row_chunks,
@@ -706,9 +710,53 @@ std::tuple<torch::Tensor, std::vector<torch::Tensor>> outlier_clamp_columns(
   return {counts, outs_ret};
 }
 
+
+// K1/K2+K4 fused: moments + HLL registers in one read.
+// Returns (moments [ncols,9] f64, regs [ncols, 1<<p] i32).
+std::tuple<torch::Tensor, torch::Tensor> moments_hll(std::vector<torch::Tensor> cols, int64_t p) {
+  TORCH_CHECK(!cols.empty(), "no columns");
+  auto device = cols[0].device();
+  int64_t m = 1LL << p;
+  auto mom = torch::zeros({(int64_t)cols.size(), 9},
+                          torch::TensorOptions().dtype(torch::kFloat64).device(device));
+  auto regs = torch::zeros({(int64_t)cols.size(), m},
+                           torch::TensorOptions().dtype(torch::kInt32).device(device));
+  for (int pass = 0; pass < 2; ++pass) {
+    std::vector<int64_t> ptrs, lens, idx;
+    for (size_t i = 0; i < cols.size(); ++i) {
+      auto &t = cols[i];
+      TORCH_CHECK(t.is_contiguous() && t.device() == device, "columns must be contiguous, same device");
+      if (dtype_code(t) != pass) continue;
+      ptrs.push_back((int64_t)t.data_ptr());
+      lens.push_back(t.numel());
+      idx.push_back((int64_t)i);
+    }
+    if (ptrs.empty()) continue;
+    int ncols = (int)ptrs.size();
+    int64_t maxn = *std::max_element(lens.begin(), lens.end());
+    int nchunks = pick_chunks(maxn, ncols);
+    auto dptr = to_device_i64(ptrs, device);
+    auto dlen = to_device_i64(lens, device);
+    auto partials = torch::empty({(int64_t)ncols * nchunks, 9},
+                                 torch::TensorOptions().dtype(torch::kFloat64).device(device));
+    auto mom_sub = torch::empty({ncols, 9}, torch::TensorOptions().dtype(torch::kFloat64).device(device));
+    auto reg_sub = torch::zeros({ncols, m}, torch::TensorOptions().dtype(torch::kInt32).device(device));
+    check_hip(anovos_moments_hll((const void *const *)dptr.data_ptr<int64_t>(),
+                                 dlen.data_ptr<int64_t>(), ncols, (int)p, nchunks, pass,
+                                 partials.data_ptr<double>(), mom_sub.data_ptr<double>(),
+                                 reg_sub.data_ptr<int>(), current_stream()),
+              "anovos_moments_hll");
+    auto didx = to_device_i64(idx, device);
+    mom.index_copy_(0, didx, mom_sub);
+    regs.index_copy_(0, didx, reg_sub);
+  }
+  return {mom, regs};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("code_counts_multi", &code_counts_multi, "fused multi-column code counts + null slot (K5)");
   m.def("outlier_clamp_columns", &outlier_clamp_columns, "fused outlier count/clamp (K10/K11)");
+  m.def("moments_hll", &moments_hll, "fused moments + HLL registers (K1/K2+K4)");
   m.def("centered_gram_bf16", &centered_gram_bf16, "bf16 MFMA centered Gram X^T X (K8)");
   m.def("bracket_histograms_grouped", &bracket_histograms_grouped, "grouped refinement histograms (K3)");
   m.def("bucketize_columns_float", &bucketize_columns_float, "bucketize to float bin labels (K6)");

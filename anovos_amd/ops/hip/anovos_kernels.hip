@@ -1067,3 +1067,90 @@ int anovos_outlier_clamp(const void *const *cols, const int64_t *lens,
 }
 
 }  // extern "C"
+
+// ------------------------------------------------------------------
+// K1/K2+K4 fused: one read of each column produces BOTH the moment
+// partials and the HyperLogLog registers (moments and distinct-count
+// are the two prior-free full-frame passes of the analyzer — the
+// equal-width histogram can't join: its range comes FROM the moments).
+// ------------------------------------------------------------------
+template <typename T>
+__global__ __launch_bounds__(THREADS) void moments_hll_kernel(
+    const T *const *cols, const int64_t *lens, int p, int nchunks,
+    double *partials, int32_t *regs /*[ncols][1<<p]*/) {
+  extern __shared__ int32_t sreg[];  // 1<<p
+  const int m = 1 << p;
+  const int col = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  for (int i = threadIdx.x; i < m; i += THREADS) sreg[i] = 0;
+  __syncthreads();
+
+  const T *__restrict__ x = cols[col];
+  const int64_t n = lens[col];
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+
+  MomAcc a{0, 0, 0, 0, 0, DBL_MAX, -DBL_MAX, 0};
+  auto body = [&](T v) {
+    mom_add(a, (double)v);
+    int idx, rho;
+    if (hll_hash(v, p, idx, rho)) atomicMax(&sreg[idx], rho);
+  };
+  if (sizeof(T) == 4) {
+    const int64_t nv = (e - s) / 4;
+    const float4 *xv = reinterpret_cast<const float4 *>(x + s);
+    for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
+      float4 v = xv[i];
+      body((T)v.x);
+      body((T)v.y);
+      body((T)v.z);
+      body((T)v.w);
+    }
+    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) body(x[i]);
+  } else {
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) body(x[i]);
+  }
+
+  // moment block-reduce (same layout as moments_partials_kernel)
+  __shared__ double sm[THREADS * NSTAT];
+  double *mine = &sm[threadIdx.x * NSTAT];
+  mine[0] = a.n; mine[1] = a.s1; mine[2] = a.s2; mine[3] = a.s3;
+  mine[4] = a.s4; mine[5] = a.mn; mine[6] = a.mx; mine[7] = a.zn; mine[8] = a.nf;
+  __syncthreads();
+  for (int stride = THREADS / 2; stride > 0; stride >>= 1) {
+    if (threadIdx.x < stride) {
+      double *other = &sm[(threadIdx.x + stride) * NSTAT];
+      mine[0] += other[0]; mine[1] += other[1]; mine[2] += other[2];
+      mine[3] += other[3]; mine[4] += other[4];
+      mine[5] = fmin(mine[5], other[5]);
+      mine[6] = fmax(mine[6], other[6]);
+      mine[7] += other[7]; mine[8] += other[8];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    double *out = &partials[(int64_t)blockIdx.x * NSTAT];
+    for (int k = 0; k < NSTAT; ++k) out[k] = sm[k];
+  }
+  int32_t *g = &regs[(int64_t)col * m];
+  for (int i = threadIdx.x; i < m; i += THREADS)
+    if (sreg[i]) atomicMax(&g[i], sreg[i]);
+}
+
+extern "C" int anovos_moments_hll(const void *const *cols, const int64_t *lens,
+                                  int ncols, int p, int nchunks, int dtype,
+                                  double *partials, double *mom_out,
+                                  int32_t *regs, hipStream_t stream) {
+  dim3 grid(ncols * nchunks);
+  size_t lds = (size_t)(1 << p) * 4;
+  if (dtype == 0)
+    hipLaunchKernelGGL(moments_hll_kernel<float>, grid, dim3(THREADS), lds, stream,
+                       (const float *const *)cols, lens, p, nchunks, partials, regs);
+  else
+    hipLaunchKernelGGL(moments_hll_kernel<double>, grid, dim3(THREADS), lds, stream,
+                       (const double *const *)cols, lens, p, nchunks, partials, regs);
+  hipLaunchKernelGGL(moments_reduce_kernel, dim3(ncols), dim3(THREADS), 0, stream,
+                     partials, nchunks, mom_out);
+  return (int)hipGetLastError();
+}

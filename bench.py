@@ -103,7 +103,12 @@ def pipeline_step(ctx, idf, source_hist, model_dir):
     chk = 0.0
 
     # ---- analyzer ----
-    moments = stats_ops.frame_moments(idf, num_cols)  # K1/K2 fused pass
+    # cardinality first: its fused K1/K2+K4 kernel computes moments AND
+    # HLL registers in ONE frame read; everything below hits the moment
+    # cache (the engine's stats-reuse contract)
+    card = sg.measures_of_cardinality(ctx, idf, num_cols + cat_cols, use_approx_unique_count=True)  # K4 HLL
+    chk += float(pd.to_numeric(card["unique_values"], errors="coerce").fillna(0).sum())
+    moments = stats_ops.frame_moments(idf, num_cols)  # cached by the fused pass
     quant = hist_ops.approx_quantiles(
         idf, num_cols, [0.01, 0.05, 0.10, 0.25, 0.50, 0.75, 0.90, 0.95, 0.99], moments=moments
     )  # K3
@@ -111,8 +116,6 @@ def pipeline_step(ctx, idf, source_hist, model_dir):
     chk += float(counts["missing_count"].sum())
     ct = sg.measures_of_centralTendency(ctx, idf, int_cols + cat_cols)  # discrete modes (K5)
     chk += float(pd.to_numeric(ct["mode_rows"], errors="coerce").fillna(0).sum())
-    card = sg.measures_of_cardinality(ctx, idf, num_cols + cat_cols, use_approx_unique_count=True)  # K4 HLL
-    chk += float(pd.to_numeric(card["unique_values"], errors="coerce").fillna(0).sum())
     shape = sg.measures_of_shape(ctx, idf)
     chk += float(pd.to_numeric(shape["skewness"], errors="coerce").fillna(0).abs().sum())
     _, nullrows = qc.nullRows_detection(ctx, idf, treatment=False)  # K10 row scan
