@@ -325,3 +325,19 @@ def test_approx_quantiles_sketch_path_vs_torch(ext):
             target = p * nn
             ok = any(a - tol <= target <= b + tol for a, b in cand)
             assert ok, (name, p, q, cand, target)
+
+
+@requires_gpu
+def test_moments_hll_fused_vs_separate(ext):
+    """Fused K1/K2+K4 must match the separate moments and HLL kernels."""
+    g = torch.Generator(device="cpu").manual_seed(31)
+    cols = []
+    for i in range(5):
+        x = torch.randn(500_000, generator=g) * (i + 1)
+        x[torch.rand(500_000, generator=g) < 0.02] = float("nan")
+        cols.append(x.cuda().contiguous())
+    mom_f, regs_f = ext.moments_hll(cols, 12)
+    mom_s = ext.column_moments(cols)
+    regs_s = ext.hll_registers_multi(cols, 12)
+    assert torch.allclose(mom_f.cpu(), mom_s.cpu(), rtol=1e-12, atol=1e-9, equal_nan=True)
+    assert torch.equal(regs_f.cpu(), regs_s.cpu())
