@@ -25,8 +25,15 @@ def test_income_workflow_on_gpu(tmp_path, monkeypatch):
 
     df = make_income_data.make(20000)
     os.makedirs("data/income_dataset/csv", exist_ok=True)
+    os.makedirs("data/income_dataset/parquet", exist_ok=True)
+    os.makedirs("data/income_dataset/join", exist_ok=True)
     os.makedirs("data/income_dataset/source/csv", exist_ok=True)
     df.to_csv("data/income_dataset/csv/part-00000.csv", index=False)
+    df.to_parquet("data/income_dataset/parquet/part-00000.parquet")
+    from anovos_amd.core.avro_codec import write_avro
+
+    join_df = df[["ifa", "age", "workclass"]].rename(columns={"age": "dupl_age", "workclass": "dupl_workclass"})
+    write_avro(join_df, "data/income_dataset/join/part-00000.avro")
     src = make_income_data.make(20000, seed=12)
     src["age"] = src["age"] * 1.05
     src.to_csv("data/income_dataset/source/csv/part-00000.csv", index=False)
