@@ -39,6 +39,7 @@ from anovos_amd.ops import histogram as hist_ops
 from anovos_amd.ops import rowops
 from anovos_amd.ops import stats as stats_ops
 from anovos_amd.shared.utils import attributeType_segregation, get_dtype, normalize_columns
+from anovos_amd.shared.tracing import traced
 
 
 def _parse_bool(v, name="treatment"):
@@ -49,6 +50,7 @@ def _parse_bool(v, name="treatment"):
     raise TypeError(f"Non-Boolean input for {name}")
 
 
+@traced
 def duplicate_detection(ctx, idf, list_of_cols="all", drop_cols=[], treatment=False, print_impact=False):
     """Reference quality_checker.py:49-150. treatment=True returns the
     deduplicated frame."""
@@ -96,6 +98,7 @@ def _unique_first(h: torch.Tensor):
     return uniq, first
 
 
+@traced
 def nullRows_detection(ctx, idf, list_of_cols="all", drop_cols=[], treatment=False, treatment_threshold=0.8, print_impact=False):
     """Reference quality_checker.py:152-283. Returns (odf, odf_print) with
     odf_print schema [null_cols_count, row_count, row_pct, flagged/treated]."""
@@ -148,6 +151,7 @@ def nullRows_detection(ctx, idf, list_of_cols="all", drop_cols=[], treatment=Fal
     return odf, odf_print
 
 
+@traced
 def nullColumns_detection(
     ctx,
     idf,
@@ -247,6 +251,7 @@ def nullColumns_detection(
     return odf, odf_print.reset_index(drop=True)
 
 
+@traced
 def outlier_detection(
     ctx,
     idf,
@@ -481,6 +486,7 @@ def outlier_detection(
     return odf, odf_print
 
 
+@traced
 def IDness_detection(ctx, idf, list_of_cols="all", drop_cols=[], treatment=False, treatment_threshold=0.8, stats_unique={}, print_impact=False):
     """Reference quality_checker.py:1048-1183."""
     if list_of_cols == "all":
@@ -522,6 +528,7 @@ def IDness_detection(ctx, idf, list_of_cols="all", drop_cols=[], treatment=False
     return odf, odf_print.reset_index(drop=True)
 
 
+@traced
 def biasedness_detection(ctx, idf, list_of_cols="all", drop_cols=[], treatment=False, treatment_threshold=0.8, stats_mode={}, print_impact=False):
     """Reference quality_checker.py:1185-1340: flag columns whose mode
     covers >= threshold of non-null rows."""
@@ -603,6 +610,7 @@ def _detect_invalid_value(e, detection_type, invalid_entries, valid_entries, par
     return 0
 
 
+@traced
 def invalidEntries_detection(
     ctx,
     idf,

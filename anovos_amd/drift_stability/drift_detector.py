@@ -23,10 +23,12 @@ from anovos_amd.data_ingest.data_sampling import data_sample
 from anovos_amd.data_transformer.transformers import attribute_binning
 from anovos_amd.drift_stability.validations import check_distance_method, check_list_of_columns
 from anovos_amd.shared.utils import attributeType_segregation
+from anovos_amd.shared.tracing import traced
 
 
 @check_distance_method
 @check_list_of_columns(target_idx=1, target="idf_target")
+@traced
 def statistics(
     ctx,
     idf_target,
