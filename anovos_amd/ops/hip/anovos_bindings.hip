@@ -289,14 +289,17 @@ torch::Tensor hll_registers(torch::Tensor x, int64_t p) {
 }
 
 void row_null_counts_num(std::vector<torch::Tensor> cols, torch::Tensor out) {
+  // accepts float32/float64 (NaN null) AND int32 dictionary codes (-1 null)
   TORCH_CHECK(!cols.empty(), "no columns");
   TORCH_CHECK(out.scalar_type() == torch::kInt32, "out must be int32");
   auto device = cols[0].device();
-  for (int pass = 0; pass < 2; ++pass) {
+  for (int pass = 0; pass < 3; ++pass) {
     std::vector<int64_t> ptrs;
     for (auto &t : cols) {
-      if (dtype_code(t) != pass) continue;
+      int dc = (t.scalar_type() == torch::kInt32) ? 2 : dtype_code(t);
+      if (dc != pass) continue;
       TORCH_CHECK(t.numel() == out.numel(), "column length mismatch");
+      TORCH_CHECK(t.is_contiguous(), "columns must be contiguous");
       ptrs.push_back((int64_t)t.data_ptr());
     }
     if (ptrs.empty()) continue;

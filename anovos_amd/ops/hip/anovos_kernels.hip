@@ -572,6 +572,11 @@ __global__ __launch_bounds__(THREADS) void hll_kernel(
 #define ROWS_PB 8192
 #define COLS_PB 64
 
+// null predicate: NaN for float/double, code -1 for int32 dictionaries
+DEV_INLINE bool is_null_elem(float v) { return isnan(v); }
+DEV_INLINE bool is_null_elem(double v) { return isnan(v); }
+DEV_INLINE bool is_null_elem(int32_t v) { return v == -1; }
+
 template <typename T>
 __global__ __launch_bounds__(THREADS) void row_null_kernel(
     const T *const *cols, int ncols, int64_t n, int32_t *out) {
@@ -591,20 +596,21 @@ __global__ __launch_bounds__(THREADS) void row_null_kernel(
   for (int c = c0; c < c1; ++c) {
     const T *__restrict__ x = cols[c] + s;
     if (vec_ok) {
-      const float4 *xv = reinterpret_cast<const float4 *>(x);
+      typedef T T4 __attribute__((ext_vector_type(4)));
+      const T4 *xv = reinterpret_cast<const T4 *>(x);
       for (int i = threadIdx.x; i < nr / 4; i += THREADS) {
-        float4 v = xv[i];
+        T4 v = xv[i];
         // 4 consecutive rows per lane: one 8-B LDS read+write
         ushort4 cnt = ((ushort4 *)rc)[i];
-        cnt.x += isnan(v.x);
-        cnt.y += isnan(v.y);
-        cnt.z += isnan(v.z);
-        cnt.w += isnan(v.w);
+        cnt.x += is_null_elem((T)v.x);
+        cnt.y += is_null_elem((T)v.y);
+        cnt.z += is_null_elem((T)v.z);
+        cnt.w += is_null_elem((T)v.w);
         ((ushort4 *)rc)[i] = cnt;
       }
     } else {
       for (int i = threadIdx.x; i < nr; i += THREADS)
-        if (isnan((double)x[i])) rc[i] += 1;
+        if (is_null_elem(x[i])) rc[i] += 1;
     }
     __syncthreads();
   }
@@ -880,16 +886,18 @@ int anovos_hll(const void *x, int64_t n, int p, int nchunks, int dtype,
 
 int anovos_row_null(const void *const *cols, int ncols, int64_t n, int dtype,
                     int32_t *out, hipStream_t stream) {
-  int64_t nrowchunks = (n + ROWS_PB - 1) / ROWS_PB;
-  int64_t ncolchunks = (ncols + COLS_PB - 1) / COLS_PB;
-  int64_t nblocks = nrowchunks * ncolchunks;
-  if (nblocks == 0) return 0;
+  const int ncolchunks = (ncols + COLS_PB - 1) / COLS_PB;
+  const int64_t nrowchunks = (n + ROWS_PB - 1) / ROWS_PB;
+  dim3 grid((uint32_t)(nrowchunks * ncolchunks));
   if (dtype == 0)
-    hipLaunchKernelGGL(row_null_kernel<float>, dim3((uint32_t)nblocks), dim3(THREADS), 0, stream,
+    hipLaunchKernelGGL(row_null_kernel<float>, grid, dim3(THREADS), 0, stream,
                        (const float *const *)cols, ncols, n, out);
-  else
-    hipLaunchKernelGGL(row_null_kernel<double>, dim3((uint32_t)nblocks), dim3(THREADS), 0, stream,
+  else if (dtype == 1)
+    hipLaunchKernelGGL(row_null_kernel<double>, grid, dim3(THREADS), 0, stream,
                        (const double *const *)cols, ncols, n, out);
+  else
+    hipLaunchKernelGGL(row_null_kernel<int32_t>, grid, dim3(THREADS), 0, stream,
+                       (const int32_t *const *)cols, ncols, n, out);
   return (int)hipGetLastError();
 }
 

@@ -22,13 +22,17 @@ def row_null_counts(idf, cols: List[str]) -> torch.Tensor:
     first = idf.col(cols[0]).data if cols else None
     if first is not None and first.is_cuda and backend.use_hip(first):
         ext = backend.hip_ext()
-        num = [idf.col(c).data.contiguous() for c in cols if idf.col(c).kind == "numerical"]
-        cat = [idf.col(c).data.contiguous() for c in cols if idf.col(c).kind != "numerical"]
+        # one fused launch covers numeric (NaN) AND categorical (-1 code)
+        # columns; timestamps (INT64_MIN null) use the eager fallback
+        fused = [idf.col(c).data.contiguous() for c in cols
+                 if idf.col(c).kind == "numerical" or idf.col(c).data.dtype == torch.int32]
+        rest = [c for c in cols
+                if not (idf.col(c).kind == "numerical" or idf.col(c).data.dtype == torch.int32)]
         out = torch.zeros(n, dtype=torch.int32, device=dev)
-        if num:
-            ext.row_null_counts_num(num, out)
-        for t in cat:
-            out += (t == -1).to(torch.int32) if t.dtype == torch.int32 else torch.zeros_like(out)
+        if fused:
+            ext.row_null_counts_num(fused, out)
+        for c in rest:
+            out += idf.col(c).null_mask().to(torch.int32)
         return out
     out = torch.zeros(n, dtype=torch.int32, device=dev)
     for c in cols:

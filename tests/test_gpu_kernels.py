@@ -341,3 +341,20 @@ def test_moments_hll_fused_vs_separate(ext):
     regs_s = ext.hll_registers_multi(cols, 12)
     assert torch.allclose(mom_f.cpu(), mom_s.cpu(), rtol=1e-12, atol=1e-9, equal_nan=True)
     assert torch.equal(regs_f.cpu(), regs_s.cpu())
+
+
+@requires_gpu
+def test_row_null_counts_mixed_dtypes(ext):
+    """Fused row-null covers float32 NaN, float64 NaN AND int32 -1 codes."""
+    g = torch.Generator(device="cpu").manual_seed(41)
+    n = 300_000
+    f32 = torch.randn(n, generator=g)
+    f32[torch.rand(n, generator=g) < 0.1] = float("nan")
+    f64 = torch.randn(n, generator=g).double()
+    f64[torch.rand(n, generator=g) < 0.05] = float("nan")
+    codes = torch.randint(-1, 10, (n,), generator=g).to(torch.int32)
+    cols = [f32.cuda().contiguous(), f64.cuda().contiguous(), codes.cuda().contiguous()]
+    out = torch.zeros(n, dtype=torch.int32, device="cuda")
+    ext.row_null_counts_num(cols, out)
+    ref = torch.isnan(f32).to(torch.int32) + torch.isnan(f64).to(torch.int32) + (codes == -1).to(torch.int32)
+    assert torch.equal(out.cpu(), ref)
