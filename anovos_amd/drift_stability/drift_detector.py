@@ -109,9 +109,9 @@ def statistics(
 
     # batched frequency pass: one fused histogram launch covers every
     # numeric (binned) column of each frame
-    q_freqs = batched_bin_frequencies(target_bin, list_of_cols, count_target)
+    q_freqs = batched_bin_frequencies(target_bin, list_of_cols, count_target, max_bin=bin_size)
     if not pre_existing_source:
-        p_freqs = batched_bin_frequencies(source_bin, list_of_cols, count_source)
+        p_freqs = batched_bin_frequencies(source_bin, list_of_cols, count_source, max_bin=bin_size)
     rows = []
     for i in list_of_cols:
         if pre_existing_source:
@@ -166,10 +166,12 @@ def _key_order(k):
         return (1, 0.0, str(k))
 
 
-def batched_bin_frequencies(binned_idf, cols, total: int):
+def batched_bin_frequencies(binned_idf, cols, total: int, max_bin: int = 0):
     """Per-bin frequencies for many columns at once: numeric (binned)
     columns go through ONE fused histogram kernel + one all-reduce;
-    categorical columns through the fused dictionary bincount."""
+    categorical columns through the fused dictionary bincount. Pass
+    ``max_bin`` (= the binning bin_size) to skip the moments pass — the
+    per-column null count then falls out of the histogram row sum."""
     from anovos_amd.ops import histogram as hist_ops
     from anovos_amd.ops import stats as stats_ops
 
@@ -177,18 +179,24 @@ def batched_bin_frequencies(binned_idf, cols, total: int):
     num_cols = [c for c in cols if binned_idf.col(c).kind == "numerical"]
     cat_cols = [c for c in cols if c not in num_cols]
     if num_cols:
-        moments = stats_ops.frame_moments(binned_idf, num_cols)
-        M = 1
-        for c in num_cols:
-            if moments[c].max == moments[c].max:
-                M = max(M, int(moments[c].max))
+        if max_bin:
+            M = int(max_bin)
+            ns = None
+        else:
+            moments = stats_ops.frame_moments(binned_idf, num_cols)
+            M = 1
+            for c in num_cols:
+                if moments[c].max == moments[c].max:
+                    M = max(M, int(moments[c].max))
+            ns = {c: moments[c].n for c in num_cols}
         tensors = [binned_idf.col(c).data for c in num_cols]
         lo = torch.full((len(num_cols),), 1.0, dtype=torch.float64)
         hi = torch.full((len(num_cols),), float(M + 1), dtype=torch.float64)
         hist = hist_ops.global_histograms(tensors, lo, hi, M).cpu().numpy()
         for i, c in enumerate(num_cols):
             keys, vals = [], []
-            nnull = int(total - moments[c].n)
+            n_valid = float(hist[i].sum()) if ns is None else ns[c]
+            nnull = int(total - n_valid)
             if nnull:
                 keys.append("-1")
                 vals.append(nnull / total)

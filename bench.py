@@ -133,7 +133,7 @@ def pipeline_step(ctx, idf, source_hist, model_dir):
     import numpy as _np
 
     drift_cols = num_cols[:: max(1, len(num_cols) // 50)]
-    q_freqs = dd.batched_bin_frequencies(binned, [c + "_binned" for c in drift_cols], idf.count())
+    q_freqs = dd.batched_bin_frequencies(binned, [c + "_binned" for c in drift_cols], idf.count(), max_bin=10)
     for c in drift_cols:
         q_keys, q_vals = q_freqs[c + "_binned"]
         pmap = source_hist.get(c, {})
@@ -201,7 +201,7 @@ def main():
     # warmup builds the drift source snapshot (per-bin frequencies)
     binned = T.attribute_binning(ctx, idf, num_cols, bin_size=10, output_mode="append")
     total = idf.count()
-    freqs = dd.batched_bin_frequencies(binned, [c + "_binned" for c in num_cols], total)
+    freqs = dd.batched_bin_frequencies(binned, [c + "_binned" for c in num_cols], total, max_bin=10)
     source_hist = {c: dict(zip(*freqs[c + "_binned"])) for c in num_cols}
     del binned  # 75 GB at the 125M-row shard — must not stay resident
 
