@@ -102,13 +102,16 @@ def approx_quantiles(
     nbins: int = DEFAULT_BINS,
     refine: int = 1,
     moments: Optional[dict] = None,
-    rel_err: float = 1e-4,
+    rel_err: float = 1e-2,
 ) -> Dict[str, List[float]]:
     """Approximate quantiles for each column at the given probabilities.
 
-    Semantics follow Spark approxQuantile: returns a value whose rank is
-    within rel-err of prob*n; after one refinement pass the bracket is
-    ~(range/nbins^2) wide and we return its midpoint interpolation.
+    Semantics follow Spark approxQuantile (default rel-err 0.01 — the
+    reference's summary()/approxQuantile contract): returns a value whose
+    rank is within rel_err*n of prob*n. The adaptive refinement pass only
+    runs for brackets whose pass-1 bin still exceeds the rank tolerance
+    (at production row counts the 2048-bin pass-1 CDF already satisfies
+    0.01); pass a smaller rel_err for tighter interpolation.
     """
     from anovos_amd.ops import stats as stats_ops
 
@@ -181,7 +184,7 @@ def approx_quantiles(
         need = {
             k: v
             for k, v in brackets.items()
-            if v[3] > max(rel_err, 1e-4) * 0.5 * max(moments[cols[k[0]]].n, 1)
+            if v[3] > rel_err * 0.5 * max(moments[cols[k[0]]].n, 1)
         }
         if not need:
             break
