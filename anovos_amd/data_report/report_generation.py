@@ -139,18 +139,46 @@ def executive_summary_gen(master_path: str, label_col, event_label) -> str:
     return "".join(kpis) + lab
 
 
+_METRIC_DICT = [
+    ("fill_count / fill_pct", "rows with a non-null value for the attribute, and their share of all rows"),
+    ("missing_count / missing_pct", "rows with a null value, and their share of all rows"),
+    ("nonzero_count / nonzero_pct", "rows with a value different from zero (numerical attributes)"),
+    ("mean / median", "arithmetic mean and 50th percentile of the non-null values"),
+    ("mode / mode_rows / mode_pct", "most frequent value, its row count and its share of non-null rows"),
+    ("unique_values", "number of distinct non-null values (HyperLogLog approximation at scale)"),
+    ("IDness", "unique_values divided by non-null rows — 1.0 marks identifier-like attributes"),
+    ("stddev / variance / cov", "sample standard deviation, variance and coefficient of variation"),
+    ("IQR", "inter-quartile range (75th minus 25th percentile)"),
+    ("skewness / kurtosis", "third / fourth standardized moments (population, excess kurtosis)"),
+    ("PSI", "population stability index between source and target bin frequencies"),
+    ("JSD", "Jensen-Shannon divergence between source and target distributions"),
+    ("HD", "Hellinger distance between source and target distributions"),
+    ("KS", "Kolmogorov-Smirnov statistic: max CDF gap between source and target"),
+    ("stability_index", "weighted CV-derived 0-4 score of an attribute across snapshots (4 = most stable)"),
+    ("IV", "information value of an attribute against the binary label (WOE-weighted)"),
+    ("IG", "information gain: label entropy reduction from splitting on the attribute"),
+]
+
+
 def wiki_generator(master_path: str, dataDict_path=None, metricDict_path=None) -> str:
     """Reference report_generation.py:909 — data dictionary + metric
-    definitions tab."""
+    definitions tab. Falls back to the engine's built-in metric
+    definitions when no metricDict CSV is supplied."""
     parts = []
-    for p, title in ((dataDict_path, "Data Dictionary"), (metricDict_path, "Metric Dictionary")):
+    for p, title in ((dataDict_path, "Data Dictionary"),):
         if p and os.path.exists(p):
             try:
                 parts.append(f"<h3>{title}</h3>" + _tbl(pd.read_csv(p), 500))
             except Exception:
                 pass
-    if not parts:
-        parts.append("<p class='note'>No data dictionary / metric dictionary supplied.</p>")
+    if metricDict_path and os.path.exists(metricDict_path):
+        try:
+            parts.append("<h3>Metric Dictionary</h3>" + _tbl(pd.read_csv(metricDict_path), 500))
+        except Exception:
+            pass
+    else:
+        parts.append("<h3>Metric Dictionary</h3>"
+                     + _tbl(pd.DataFrame(_METRIC_DICT, columns=["metric", "definition"]), 100))
     return "".join(parts)
 
 
