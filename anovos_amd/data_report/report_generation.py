@@ -136,7 +136,26 @@ def executive_summary_gen(master_path: str, label_col, event_label) -> str:
         v = get(m)
         kpis.append(f"<div class='kpi'><div class='v'>{_html.escape(str(v[0])) if len(v) else '—'}</div><div class='l'>{label}</div></div>")
     lab = f"<p>Label column: <b>{_html.escape(str(label_col))}</b>, event label: <b>{_html.escape(str(event_label))}</b></p>" if label_col else ""
-    return "".join(kpis) + lab
+    extra = ""
+    if label_col:
+        # label distribution from the label's frequency chart object
+        p_lab = ends_with(master_path) + "freqDist_" + str(label_col)
+        if os.path.exists(p_lab):
+            try:
+                extra = "<h3>Label Distribution</h3>" + _fig_div(p_lab, "exec_label")
+            except Exception:
+                extra = ""
+    # quality flags roll-up
+    flags = []
+    for name, col, label in (("drift_statistics", "flagged", "drifted attributes"),
+                             ("stability_index", "flagged", "unstable attributes"),
+                             ("outlier_detection", "upper_outliers", "attributes with upper outliers")):
+        df = _read_csv(master_path, name)
+        if df is not None and col in df.columns:
+            v = pd.to_numeric(df[col], errors="coerce").fillna(0)
+            n = int((v > 0).sum()) if col != "flagged" else int(v.sum())
+            flags.append(f"<div class='kpi'><div class='v'>{n}</div><div class='l'>{label}</div></div>")
+    return "".join(kpis) + "".join(flags) + lab + extra
 
 
 _METRIC_DICT = [
