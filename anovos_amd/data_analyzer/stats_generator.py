@@ -174,7 +174,7 @@ def measures_of_centralTendency(ctx, idf, list_of_cols="all", drop_cols=[], prin
     medians = hist_ops.approx_quantiles(idf, num_cols, [0.5], moments=moments) if num_cols else {}
     nulls, total = stats_ops.null_counts(idf, cols)
     dfm = mode_computation(ctx, idf, cols)
-    mode_map = {r["attribute"]: (r["mode"], r["mode_rows"]) for _, r in dfm.iterrows()}
+    mode_map = dict(zip(dfm["attribute"], zip(dfm["mode"], dfm["mode_rows"])))
     rows = []
     for c in cols:
         fill = total - nulls[c]

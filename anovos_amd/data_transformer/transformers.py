@@ -118,7 +118,7 @@ def attribute_binning(
 
     if pre_existing_model:
         dfm = _load_model(model_path, "attribute_binning")
-        cut_map = {r["attribute"]: list(r["parameters"]) for _, r in dfm.iterrows()}
+        cut_map = {a: list(p) for a, p in zip(dfm["attribute"], dfm["parameters"])}
         bin_cutoffs = [cut_map[c] for c in list_of_cols]
     else:
         if method_type == "equal_frequency":
@@ -419,7 +419,7 @@ def z_standardization(
 
     if pre_existing_model:
         dfm = _load_model(model_path, "z_standardization")
-        params = {r["feature"]: (r["mean"], r["stddev"]) for _, r in dfm.iterrows()}
+        params = dict(zip(dfm["feature"], zip(dfm["mean"], dfm["stddev"])))
     else:
         moments = stats_ops.frame_moments(idf, list_of_cols)
         params = {c: (moments[c].mean, moments[c].stddev) for c in list_of_cols}
@@ -472,7 +472,7 @@ def IQR_standardization(
         return idf
     if pre_existing_model:
         dfm = _load_model(model_path, "IQR_standardization")
-        params = {r["feature"]: (r["p25"], r["p50"], r["p75"]) for _, r in dfm.iterrows()}
+        params = dict(zip(dfm["feature"], zip(dfm["p25"], dfm["p50"], dfm["p75"])))
     else:
         q = hist_ops.approx_quantiles(idf, list_of_cols, [0.25, 0.5, 0.75])
         params = {c: (q[c][0], q[c][1], q[c][2]) for c in list_of_cols}
@@ -528,7 +528,7 @@ def normalization(
         return idf
     if pre_existing_model:
         dfm = _load_model(model_path, "normalization")
-        params = {r["feature"]: (r["min"], r["max"]) for _, r in dfm.iterrows()}
+        params = dict(zip(dfm["feature"], zip(dfm["min"], dfm["max"])))
     else:
         moments = stats_ops.frame_moments(idf, list_of_cols)
         params = {c: (moments[c].min, moments[c].max) for c in list_of_cols}
