@@ -113,10 +113,14 @@ class AnovosFrame:
         return 0
 
     def count(self) -> int:
-        """Global row count across ranks (RCCL all-reduce when distributed)."""
-        from anovos_amd.core import dist
+        """Global row count across ranks (RCCL all-reduce when
+        distributed). Cached — a frame's row count is immutable (every
+        row-changing op returns a new frame)."""
+        if getattr(self, "_count_cache", None) is None:
+            from anovos_amd.core import dist
 
-        return dist.all_reduce_scalar(self.local_rows())
+            self._count_cache = dist.all_reduce_scalar(self.local_rows())
+        return self._count_cache
 
     # ---------------- construction ----------------
     @staticmethod
