@@ -246,10 +246,12 @@ def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, de
             out_path = report_input_path or "."
             df, ts_cols, num_cols, cat_cols = ts_auto_detection.ts_preprocess(
                 ctx, df, id_col, out_path, tz_offset=args.get("tz_offset", "local"), run_type=run_type)
-            if args.get("analysis_level") or args.get("auto_detection", True):
+            if args.get("inspection", True):
+                # reference key analysis_level ∈ {daily, weekly, hourly}
                 ts_analyzer_mod.ts_analyzer(
                     ctx, df, id_col, args.get("max_days", 90), out_path,
-                    output_type=args.get("output_type", "daily"), run_type=run_type)
+                    output_type=args.get("analysis_level", args.get("output_type", "daily")),
+                    run_type=run_type)
 
         elif key == "anovos_basic_report" and args.get("basic_report", False):
             anovos_basic_report(ctx, df, **(args.get("report_args", {}) or {}), run_type=run_type)
