@@ -244,8 +244,9 @@ def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, de
         elif key == "timeseries_analyzer":
             id_col = args.get("id_col", "")
             out_path = report_input_path or "."
-            df, ts_cols, num_cols, cat_cols = ts_auto_detection.ts_preprocess(
-                ctx, df, id_col, out_path, tz_offset=args.get("tz_offset", "local"), run_type=run_type)
+            if args.get("auto_detection", True):
+                df, ts_cols, num_cols, cat_cols = ts_auto_detection.ts_preprocess(
+                    ctx, df, id_col, out_path, tz_offset=args.get("tz_offset", "local"), run_type=run_type)
             if args.get("inspection", True):
                 # reference key analysis_level ∈ {daily, weekly, hourly}
                 ts_analyzer_mod.ts_analyzer(
