@@ -81,3 +81,29 @@ def test_geospatial_ops_on_gpu():
     assert gh.dictionary[int(gh.data[0])].startswith("dr5reg")
     d = geo.location_distance(idf, ["lat", "lon"], ["lat", "lon"], result_prefix="self")
     assert float(d.col("self_distance").data.abs().max()) < 1e-6
+
+
+@requires_gpu
+def test_edge_cases_on_gpu():
+    """Degenerate columns through the GPU kernel paths (constant,
+    all-null, inf, null-only categorical) — same assertions as the CPU
+    edge suite."""
+    from tests.test_edge_cases import edge_frame
+    from anovos_amd.data_analyzer import quality_checker as qc
+    from anovos_amd.data_analyzer import stats_generator as sg
+    from anovos_amd.data_transformer import transformers as T
+    from anovos_amd.shared.context import AnovosContext
+
+    ctx = AnovosContext("cuda:0")
+    idf = edge_frame("cuda:0")
+    m = sg.measures_of_counts(ctx, idf).set_index("attribute")
+    assert int(m.loc["all_null", "missing_count"]) == 5000
+    assert int(m.loc["null_cat", "missing_count"]) == 5000
+    d = sg.measures_of_dispersion(ctx, idf).set_index("attribute")
+    assert float(d.loc["constant", "stddev"]) == pytest.approx(0.0, abs=1e-9)
+    p = sg.measures_of_percentiles(ctx, idf, ["normal", "constant"]).set_index("attribute")
+    assert float(p.loc["constant", "50%"]) == pytest.approx(3.14, rel=1e-6)
+    odf = T.attribute_binning(ctx, idf, ["normal", "constant"], bin_size=5, output_mode="append")
+    assert "constant_binned" in odf.columns
+    _, nr = qc.nullRows_detection(ctx, idf, treatment=False)
+    assert int(nr["row_count"].sum()) == 5000
