@@ -284,6 +284,7 @@ def ts_viz_generate(master_path: str) -> str:
             continue
     viz = [x for x in os.listdir(master_path)
            if x.endswith(("_daily.csv", "_weekly.csv", "_hourly.csv"))]
+    decomposed = 0
     for fn in sorted(viz)[:40]:
         try:
             df = pd.read_csv(os.path.join(master_path, fn))
@@ -293,6 +294,16 @@ def ts_viz_generate(master_path: str) -> str:
                 fig.add_trace(go.Scatter(x=df[xcol].astype(str), y=df[c], mode="lines+markers", name=c))
             fig.update_layout(title=fn[:-4], height=360)
             parts.append(_fig_div(fig, "ts_" + fn.replace(".", "_")))
+            # seasonal decomposition + stationarity for the first few
+            # daily mean series (reference plotSeasonalDecompose + ADF/KPSS)
+            if fn.endswith("_daily.csv") and "mean" in df.columns and len(df) >= 14 and decomposed < 4:
+                series = pd.to_numeric(df["mean"], errors="coerce")
+                sfig = plotSeasonalDecompose(series, period=7, title="Seasonal Decomposition — " + fn[:-4])
+                parts.append(_fig_div(sfig, "tsdec_" + fn.replace(".", "_")))
+                st = stationarity_check(series)
+                parts.append("<p class='note'>Stationarity (rolling-stats heuristic): "
+                             + ", ".join(f"{k}={v}" for k, v in st.items()) + "</p>")
+                decomposed += 1
         except Exception:
             continue
     return "".join(parts)
