@@ -106,7 +106,8 @@ def ts_eligiblity_check(ctx, idf: AnovosFrame, id_col: str, opt: int = 1, tz_off
         keys = uniq_pair[:, key_idx]
         _, counts = torch.unique(keys, return_counts=True)
         c = counts.to(torch.float64)
-        qs = torch.quantile(c, torch.tensor([0.01, 0.05, 0.1, 0.25, 0.5, 0.75, 0.9, 0.95, 0.99], dtype=torch.float64))
+        qs = torch.quantile(c, torch.tensor([0.01, 0.05, 0.1, 0.25, 0.5, 0.75, 0.9, 0.95, 0.99],
+                                            dtype=torch.float64, device=c.device))
         rows.append([name, float(c.min()), *[float(q) for q in qs], float(c.max())])
     return pd.DataFrame(rows, columns=["attribute", "min", "1%", "5%", "10%", "25%", "50%", "75%", "90%", "95%", "99%", "max"])
 
