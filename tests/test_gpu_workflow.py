@@ -107,3 +107,43 @@ def test_edge_cases_on_gpu():
     assert "constant_binned" in odf.columns
     _, nr = qc.nullRows_detection(ctx, idf, treatment=False)
     assert int(nr["row_count"].sum()) == 5000
+
+
+@requires_gpu
+def test_geospatial_config_on_gpu(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    sys.path.insert(0, os.path.join(repo, "tools"))
+    import make_income_data as mid
+
+    df = mid.add_geo_cols(mid.make(15000))
+    os.makedirs("data/income_dataset/csv", exist_ok=True)
+    df.to_csv("data/income_dataset/csv/part-00000.csv", index=False)
+    from anovos_amd import workflow
+
+    out = workflow.run(os.path.join(repo, "config", "configs_geospatial.yaml"), device="cuda:0")
+    assert out.device.type == "cuda"
+    assert "radius_of_gyration" in out.columns
+
+
+@requires_gpu
+def test_time_series_config_on_gpu(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    sys.path.insert(0, os.path.join(repo, "tools"))
+    import make_income_data as mid
+
+    df = mid.add_ts_cols(mid.make(8000))
+    os.makedirs("data/income_dataset/csv", exist_ok=True)
+    df.to_csv("data/income_dataset/csv/part-00000.csv", index=False)
+    for k in range(12):
+        snap = mid.make(1500, seed=100 + k)
+        snap["age"] = snap["age"] * (1 + 0.01 * k)
+        d = f"data/income_dataset/snapshot{k + 1:02d}/csv"
+        os.makedirs(d, exist_ok=True)
+        snap.to_csv(os.path.join(d, "part-00000.csv"), index=False)
+    from anovos_amd import workflow
+
+    out = workflow.run(os.path.join(repo, "config", "configs_time_series.yaml"), device="cuda:0")
+    assert out.device.type == "cuda"
+    assert os.path.exists("report_stats/stability_index.csv")
