@@ -238,7 +238,7 @@ def main():
                     "higher_is_better": True,
                     "scaling": "weak",
                     "vs_baseline": None,
-                    "dtype": "fp32",
+                    "dtype": "fp32 data + fp64 accumulators",
                     "data": "synthetic",
                     "config": {
                         "model": "anovos-full-pipeline",
