@@ -293,8 +293,7 @@ def outlier_detection(
     list_of_cols = normalize_columns(idf, list_of_cols, drop_cols, restrict_to=num_cols)
     if not list_of_cols:
         warnings.warn("No Outlier Check - No numerical column to analyze")
-        empty = pd.DataFrame(columns=["attribute", "lower_outliers", "upper_outliers"])
-        return (idf, empty) if print_impact else idf
+        return idf, _empty_stats
     if any(x not in num_cols for x in list_of_cols):
         raise TypeError("Invalid input for Column(s)")
     if detection_side not in ("upper", "lower", "both"):
