@@ -56,14 +56,38 @@ def read_dataset(file_path: str, file_type: str, file_configs: Dict = None, devi
     file_configs = dict(file_configs or {})
     ft = file_type.lower()
     if ft == "csv":
-        return _read_csv(file_path, file_configs, device)
-    if ft == "parquet":
-        return _read_parquet(file_path, file_configs, device)
-    if ft == "json":
-        return _read_json(file_path, file_configs, device)
-    if ft == "avro":
-        return _read_avro(file_path, file_configs, device)
-    raise ValueError(f"unsupported file_type: {file_type}")
+        idf = _read_csv(file_path, file_configs, device)
+    elif ft == "parquet":
+        idf = _read_parquet(file_path, file_configs, device)
+    elif ft == "json":
+        idf = _read_json(file_path, file_configs, device)
+    elif ft == "avro":
+        idf = _read_avro(file_path, file_configs, device)
+    else:
+        raise ValueError(f"unsupported file_type: {file_type}")
+    return _unify_frame_dictionaries(idf)
+
+
+def _unify_frame_dictionaries(idf: AnovosFrame) -> AnovosFrame:
+    """Multi-rank reads shard rows by file part, so each rank's
+    first-seen category set can differ; every cross-rank count merge
+    assumes positionally identical dictionaries. One all-gather of the
+    (host, tiny) dictionaries at ingest unifies them for the frame's
+    lifetime."""
+    from anovos_amd.core import dist
+
+    if not dist.is_dist():
+        return idf
+    from anovos_amd.ops.encode import unify_dictionaries
+
+    cats = [n for n in idf.columns if idf.col(n).kind == "categorical"]
+    if not cats:
+        return idf
+    unified = unify_dictionaries([idf.col(n) for n in cats])
+    out = idf
+    for col in unified:
+        out = out.with_column(col.name, col)
+    return out
 
 
 def _concat_tables(tables: List[pa.Table]) -> pa.Table:

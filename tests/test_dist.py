@@ -170,3 +170,64 @@ def test_dist_matches_single(fn_name):
             assert dist_res[k] == pytest.approx(v, rel=2e-2, abs=1e-6), (k, dist_res[k], v)
         else:
             assert dist_res[k] == v, (k, dist_res[k], v)
+
+
+def _fn_skewed_dicts(ctx, idf):
+    """Rank-disjoint categories must still merge correctly (dictionary
+    unify at ingest)."""
+    from anovos_amd.data_analyzer import stats_generator as sg
+
+    ct = sg.measures_of_centralTendency(ctx, idf, ["skewcat"])
+    u = sg.uniqueCount_computation(ctx, idf, ["skewcat"])
+    return {"mode": str(ct["mode"].iloc[0]), "uniques": float(u["unique_values"].iloc[0])}
+
+
+_FNS["skewed_dicts"] = _fn_skewed_dicts
+
+
+def test_dist_rank_disjoint_dictionaries(tmp_path):
+    """Write a dataset whose category values cluster by file part (so
+    each rank's local dictionary differs), read it distributed, and
+    check mode/uniques match the single-process truth."""
+    import pandas as pd
+
+    d = tmp_path / "skew"
+    (d).mkdir()
+    # part 0: categories a,a,a,b ; part 1: categories c,c,b,d
+    pd.DataFrame({"skewcat": ["a", "a", "a", "b"] * 200}).to_csv(d / "part-00000.csv", index=False)
+    pd.DataFrame({"skewcat": ["c", "c", "b", "d"] * 200}).to_csv(d / "part-00001.csv", index=False)
+
+    port = _free_port()
+    out = tempfile.NamedTemporaryFile(suffix=".json", delete=False).name
+
+    mp_ctx = mp.get_context("spawn")
+    procs = [mp_ctx.Process(target=_dict_worker, args=(r, port, str(d), out)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0
+    res = json.load(open(out))
+    # truth over all 1600 rows: counts a=600, b=400, c=400, d=200 -> mode a, 4 uniques
+    assert res["mode"] == "a"
+    assert res["uniques"] == 4.0
+
+
+def _dict_worker(rank, port, data_dir, out):
+    os.environ.update({"RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": "2",
+                       "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port)})
+    import torch.distributed as td
+
+    from anovos_amd.core import dist
+    from anovos_amd.core.io import read_dataset
+    from anovos_amd.shared.context import init_context
+
+    dist.init_from_env(timeout_s=120)
+    ctx = init_context("cpu")
+    idf = read_dataset(data_dir, "csv", {"header": True})
+    res = _fn_skewed_dicts(ctx, idf)
+    if rank == 0:
+        with open(out, "w") as f:
+            json.dump(res, f)
+    td.barrier()
+    td.destroy_process_group()
