@@ -95,9 +95,27 @@ def join_dataset(*idfs: AnovosFrame, join_cols, join_type: str) -> AnovosFrame:
     return pairwise_reduce(lambda a, b: _join2(a, b, join_cols, join_type), idfs)
 
 
-def _join2(a: AnovosFrame, b: AnovosFrame, join_cols: List[str], how: str) -> AnovosFrame:
+def _replicate(b: AnovosFrame) -> AnovosFrame:
+    """Gather a (small, row-sharded) side frame onto every rank — the
+    engine's broadcast join: the left frame stays sharded, the right
+    frame becomes globally visible (Spark shuffles instead; side tables
+    in this workload are MBs)."""
+    from anovos_amd.core import dist
+
+    if not dist.is_dist():
+        return b
+    import pandas as pd
+
+    parts = dist.all_gather_object(b.to_pandas())
+    full = pd.concat(parts, ignore_index=True)
+    return AnovosFrame.from_pandas(full, device=b.device)
+
+
+def _join2(a: AnovosFrame, b: AnovosFrame, join_cols: List[str], how: str, _replicated: bool = False) -> AnovosFrame:
     from anovos_amd.ops.groupby import row_hash
 
+    if not _replicated:
+        b = _replicate(b)
     ha = row_hash(a, join_cols)
     hb = row_hash(b, join_cols).to(ha.device)
     sb, order_b = torch.sort(hb)
@@ -113,7 +131,21 @@ def _join2(a: AnovosFrame, b: AnovosFrame, join_cols: List[str], how: str) -> An
     elif how in ("left", "full"):
         keep_a = torch.arange(ha.numel(), device=ha.device)
     elif how == "right":
-        return _join2(b, a, join_cols, "left").select(a.columns + [c for c in b.columns if c not in join_cols])
+        # all b rows survive. Matched b rows pair with this rank's a
+        # shard; unmatched-vs-ALL-ranks b rows are emitted once, by
+        # their owner rank (b is replicated in dist mode).
+        inner = _join2(a, b, join_cols, "inner", _replicated=True)
+        extra = _unmatched_b_rows(a, b, ha, hb)
+        if extra is None or extra.local_rows() == 0:
+            return inner.select(a.columns + [c for c in b.columns if c not in join_cols])
+        cols = {}
+        for name in inner.columns:
+            if name in extra.columns:
+                cols[name] = extra.col(name)
+            else:
+                cols[name] = _all_null_column(inner.col(name), extra.local_rows())
+        filled = AnovosFrame(cols, inner.device).select(inner.columns)
+        return concatenate_dataset(inner, filled, method_type="name")
     else:
         raise ValueError(f"join_type {how} not supported")
 
@@ -131,14 +163,10 @@ def _join2(a: AnovosFrame, b: AnovosFrame, join_cols: List[str], how: str) -> An
             g.data = _null_where(g, ~bmatched)
         out = out.with_column(name, g)
     if how == "full":
-        # append unmatched b rows
-        bm = torch.zeros(hb.numel(), dtype=torch.bool, device=hb.device)
-        sa, _ = torch.sort(ha)
-        posb = torch.searchsorted(sa, hb).clamp(max=max(sa.numel() - 1, 0))
-        bm = (sa.numel() > 0) & (sa[posb] == hb) if sa.numel() else bm
-        extra_idx = (~bm).nonzero(as_tuple=True)[0]
-        if extra_idx.numel():
-            extra = b.filter_rows(extra_idx)
+        # append b rows unmatched against the GLOBAL a (ownership-split
+        # across ranks so replicated-b extras are emitted exactly once)
+        extra = _unmatched_b_rows(a, b, ha, hb)
+        if extra is not None and extra.local_rows():
             cols = {}
             for name in out.columns:
                 if name in extra.columns:
@@ -148,6 +176,31 @@ def _join2(a: AnovosFrame, b: AnovosFrame, join_cols: List[str], how: str) -> An
                     cols[name] = _all_null_column(src, extra.local_rows())
             out = concatenate_dataset(out, AnovosFrame(cols, out.device).select(out.columns), method_type="name")
     return out
+
+
+def _unmatched_b_rows(a: AnovosFrame, b: AnovosFrame, ha: torch.Tensor, hb: torch.Tensor):
+    """b rows whose key matches NO a row on ANY rank, split by owner
+    rank (b identical on all ranks in dist mode)."""
+    from anovos_amd.core import dist
+
+    if hb.numel() == 0:
+        return None
+    bm = torch.zeros(hb.numel(), dtype=torch.bool, device=hb.device)
+    if ha.numel():
+        sa, _ = torch.sort(ha)
+        posb = torch.searchsorted(sa, hb).clamp(max=max(sa.numel() - 1, 0))
+        bm = sa[posb] == hb
+    if dist.is_dist():
+        bm_f = bm.to(torch.float64)
+        dist.all_reduce_(bm_f, "max")
+        bm = bm_f > 0
+        owner = (torch.arange(hb.numel(), device=hb.device) % dist.world_size()) == dist.rank()
+        extra_idx = (~bm & owner).nonzero(as_tuple=True)[0]
+    else:
+        extra_idx = (~bm).nonzero(as_tuple=True)[0]
+    if extra_idx.numel() == 0:
+        return None
+    return b.filter_rows(extra_idx)
 
 
 def _null_where(col: Column, mask: torch.Tensor) -> torch.Tensor:
