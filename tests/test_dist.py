@@ -231,3 +231,59 @@ def _dict_worker(rank, port, data_dir, out):
             json.dump(res, f)
     td.barrier()
     td.destroy_process_group()
+
+
+def _join_worker(rank, port, data_dir, out):
+    os.environ.update({"RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": "2",
+                       "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port)})
+    import torch.distributed as td
+
+    from anovos_amd.core import dist
+    from anovos_amd.core.frame import AnovosFrame
+    from anovos_amd.core.io import read_dataset
+    from anovos_amd.data_ingest.data_ingest import join_dataset
+    from anovos_amd.shared.context import init_context
+
+    dist.init_from_env(timeout_s=120)
+    init_context("cpu")
+    main = read_dataset(os.path.join(data_dir, "main"), "csv", {"header": True})
+    side = read_dataset(os.path.join(data_dir, "side"), "csv", {"header": True})
+    joined = join_dataset(main, side, join_cols="k", join_type="inner")
+    local = int(joined.local_rows())
+    tot = dist.all_reduce_scalar(local)
+    vsum = dist.all_reduce_scalar(float(joined.col("v").data.to(torch.float64).sum()))
+    if rank == 0:
+        with open(out, "w") as f:
+            json.dump({"rows": tot, "vsum": vsum}, f)
+    td.barrier()
+    td.destroy_process_group()
+
+
+def test_dist_join_sharded_side_table(tmp_path):
+    """The side table's parts land on different ranks; the broadcast
+    join must still match every key (reference join semantics)."""
+    import pandas as pd
+
+    d = tmp_path / "jd"
+    (d / "main").mkdir(parents=True)
+    (d / "side").mkdir()
+    main = pd.DataFrame({"k": [f"k{i % 50}" for i in range(1000)], "x": range(1000)})
+    main.iloc[:500].to_csv(d / "main" / "part-00000.csv", index=False)
+    main.iloc[500:].to_csv(d / "main" / "part-00001.csv", index=False)
+    side = pd.DataFrame({"k": [f"k{i}" for i in range(50)], "v": [float(i) for i in range(50)]})
+    side.iloc[:25].to_csv(d / "side" / "part-00000.csv", index=False)
+    side.iloc[25:].to_csv(d / "side" / "part-00001.csv", index=False)
+
+    port = _free_port()
+    out = tempfile.NamedTemporaryFile(suffix=".json", delete=False).name
+    mp_ctx = mp.get_context("spawn")
+    procs = [mp_ctx.Process(target=_join_worker, args=(r, port, str(d), out)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0
+    res = json.load(open(out))
+    assert res["rows"] == 1000  # every main row matches
+    truth = sum(float(i % 50) for i in range(1000))
+    assert res["vsum"] == pytest.approx(truth)
