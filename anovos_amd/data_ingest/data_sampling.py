@@ -108,9 +108,10 @@ def data_sample(
         gcounts = _t.from_numpy(gc).to(key.device)
         pos = _t.searchsorted(guniq, uniq)
         counts = gcounts[pos]
+        smallest = float(gcounts.min())  # global smallest stratum, even if absent locally
     else:
         counts = local_counts
-    smallest = float(counts.min())
+        smallest = float(counts.min())
     frac_per = (fraction * smallest / counts).clamp(max=1.0)
     g = torch.Generator(device=base.device)
     g.manual_seed(int(seed_value) * 1000003 + dist.rank())
