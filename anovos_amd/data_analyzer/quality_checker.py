@@ -62,18 +62,16 @@ def duplicate_detection(ctx, idf, list_of_cols="all", drop_cols=[], treatment=Fa
         raise TypeError("Invalid input for Column(s)")
     treatment = _parse_bool(treatment)
 
+    from anovos_amd.ops import dedup as dedup_ops
     from anovos_amd.ops.groupby import row_hash
 
     h = row_hash(idf.select(cols), cols)
     idf_count = idf.count()
-    # keep first occurrence of each hash locally; cross-rank dedup by
-    # hash ownership (hash % world owns the row)
-    uniq, first_idx = _unique_first(h)
-    if dist.is_dist():
-        owner = (uniq % dist.world_size() + dist.world_size()) % dist.world_size() == dist.rank()
-        keep_idx = first_idx[owner]
-    else:
-        keep_idx = first_idx
+    # keep first local occurrence of each hash, then resolve cross-rank
+    # winners with the hash-exchange protocol (ops/dedup.py — a row
+    # present only on a non-owner rank must still survive)
+    uniq, first_idx = dedup_ops.unique_first(h)
+    keep_idx = first_idx[dedup_ops.global_keep_mask(uniq)]
     odf_tmp = idf.filter_rows(keep_idx)
     odf_tmp_count = odf_tmp.count()
     odf = odf_tmp if treatment else idf

@@ -287,3 +287,48 @@ def test_dist_join_sharded_side_table(tmp_path):
     assert res["rows"] == 1000  # every main row matches
     truth = sum(float(i % 50) for i in range(1000))
     assert res["vsum"] == pytest.approx(truth)
+
+
+def _dedup_worker(rank, port, out):
+    os.environ.update({"RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": "2",
+                       "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port)})
+    import torch.distributed as td
+
+    from anovos_amd.core import dist
+    from anovos_amd.core.frame import AnovosFrame
+    from anovos_amd.data_analyzer.quality_checker import duplicate_detection
+    from anovos_amd.shared.context import init_context
+
+    dist.init_from_env(timeout_s=120)
+    ctx = init_context("cpu")
+    # rank 0 holds values 0..99 (each twice); rank 1 holds 50..149 (each
+    # twice): cross-rank dupes for 50..99, rank-exclusive rows elsewhere
+    lo = 0 if rank == 0 else 50
+    vals = [float(v) for v in range(lo, lo + 100)] * 2
+    pdf = pd.DataFrame({"v": vals})
+    idf = AnovosFrame.from_pandas(pdf, device="cpu")
+    odf, stats = duplicate_detection(ctx, idf, treatment=True)
+    kept = sorted(float(x) for x in odf.col("v").data.tolist())
+    gathered = dist.all_gather_object(kept)
+    if rank == 0:
+        allkept = sorted([v for g in gathered for v in g])
+        with open(out, "w") as f:
+            json.dump({"all": allkept, "unique_rows": float(stats[stats["metric"] == "unique_rows_count"]["value"].iloc[0])}, f)
+    td.barrier()
+    td.destroy_process_group()
+
+
+def test_dist_exact_dedup_across_ranks(tmp_path):
+    port = _free_port()
+    out = tempfile.NamedTemporaryFile(suffix=".json", delete=False).name
+    mp_ctx = mp.get_context("spawn")
+    procs = [mp_ctx.Process(target=_dedup_worker, args=(r, port, out)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0
+    res = json.load(open(out))
+    # global distinct values: 0..149 — every one kept exactly once
+    assert res["all"] == [float(v) for v in range(150)]
+    assert res["unique_rows"] == 150.0
