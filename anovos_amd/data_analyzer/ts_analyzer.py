@@ -82,9 +82,16 @@ def ts_processed_feats(idf: AnovosFrame, col: str, id_col: str, tz: str = "local
 def ts_eligiblity_check(ctx, idf: AnovosFrame, id_col: str, opt: int = 1, tz_offset: str = "local") -> pd.DataFrame:
     """Reference ts_analyzer.py:160 — lag-1 date-diff stats over distinct
     dates (opt=1) or percentile stats of id↔date pair counts (opt=2)."""
+    from anovos_amd.core import dist as _dist
+
     day = idf.col("yyyymmdd_col").data
     valid = day != NULL_TS
     days = torch.unique(day[valid])
+    if _dist.is_dist():
+        gathered = _dist.all_gather_object(days.cpu().numpy())
+        import numpy as _np
+
+        days = torch.from_numpy(_np.unique(_np.concatenate(gathered))).to(day.device)
     if opt == 1:
         if days.numel() > 1:
             diffs = (days[1:] - days[:-1]).to(torch.float64) / US_PER_DAY
@@ -101,6 +108,11 @@ def ts_eligiblity_check(ctx, idf: AnovosFrame, id_col: str, opt: int = 1, tz_off
     ok = valid & ~idc.null_mask()
     pair = torch.stack([id_codes[ok].to(torch.float64), day[ok].to(torch.float64)], dim=1)
     uniq_pair = torch.unique(pair, dim=0)
+    if _dist.is_dist():
+        import numpy as _np
+
+        gp = _dist.all_gather_object(uniq_pair.cpu().numpy())
+        uniq_pair = torch.from_numpy(_np.unique(_np.concatenate(gp), axis=0)).to(day.device)
     rows = []
     for key_idx, name in ((0, "id_date_pair"), (1, "date_id_pair")):
         keys = uniq_pair[:, key_idx]
@@ -185,8 +197,13 @@ def ts_analyzer(ctx, idf: AnovosFrame, id_col: str, max_days: int, output_path: 
     """Reference ts_analyzer.py:408 — driver loop: for each ts column
     write eligibility stats (stats_<col>_{1,2}.csv) and per-attribute viz
     aggregates (<ts>_<attr>_<type>.csv)."""
+    from anovos_amd.core import dist as _dist
+
     local_path = output_path if run_type == "local" else "report_stats"
-    os.makedirs(local_path, exist_ok=True)
+    if _dist.rank() == 0:
+        os.makedirs(local_path, exist_ok=True)
+    _dist.barrier()
+    write = _dist.rank() == 0
     num_cols, cat_cols, other = attributeType_segregation(idf)
     num_cols = [x for x in num_cols if x != id_col]
     cat_cols = [x for x in cat_cols if x != id_col]
@@ -195,14 +212,16 @@ def ts_analyzer(ctx, idf: AnovosFrame, id_col: str, max_days: int, output_path: 
     for i in ts_cols:
         pdf_feats = ts_processed_feats(idf, i, id_col, tz_offset, cnt_row, None)
         f1 = ts_eligiblity_check(ctx, pdf_feats, id_col, opt=1)
-        f1.to_csv(ends_with(local_path) + "stats_" + str(i) + "_1.csv", index=False)
         f2 = ts_eligiblity_check(ctx, pdf_feats, id_col, opt=2)
-        f2.to_csv(ends_with(local_path) + "stats_" + str(i) + "_2.csv", index=False)
+        if write:
+            f1.to_csv(ends_with(local_path) + "stats_" + str(i) + "_1.csv", index=False)
+            f2.to_csv(ends_with(local_path) + "stats_" + str(i) + "_2.csv", index=False)
         for cols in (num_cols, cat_cols):
             for l in cols:
                 try:
                     f = ts_viz_data(pdf_feats, i, l, output_type=output_type, tz_offset=tz_offset)
-                    f.to_csv(ends_with(local_path) + i + "_" + l + "_" + output_type + ".csv", index=False)
+                    if write:
+                        f.to_csv(ends_with(local_path) + i + "_" + l + "_" + output_type + ".csv", index=False)
                 except Exception:
                     continue
     return ts_cols

@@ -173,8 +173,12 @@ def ts_preprocess(ctx, idf: AnovosFrame, id_col: str, output_path: str,
                   tz_offset: str = "local", run_type: "str" = "local", mlflow_config=None, auth_key="NA"):
     """Reference ts_auto_detection.py:622 — driver: convert candidate
     columns, write ``ts_cols_stats.csv``, return (odf, ts_cols)."""
+    from anovos_amd.core import dist as _dist
+
     local_path = output_path if run_type == "local" else "report_stats"
-    os.makedirs(local_path, exist_ok=True)
+    if _dist.rank() == 0:
+        os.makedirs(local_path, exist_ok=True)
+    _dist.barrier()
     lc1, lc2, lc3 = ts_loop_cols_pre(idf, id_col)
     ts_loop_cols = [lc1[i] for i, k in enumerate(lc2) if k in ("string", "string_c", "int_c", "bigint_c", "long_c")]
     pre_exist_ts_cols = [lc1[i] for i, k in enumerate(lc2) if k == "dt"]
@@ -195,5 +199,7 @@ def ts_preprocess(ctx, idf: AnovosFrame, id_col: str, output_path: str,
             "source": ["pre_existing" if c in pre_exist_ts_cols else "auto_detected" for c in ts_cols_post],
         }
     )
-    stats.to_csv(ends_with(local_path) + "ts_cols_stats.csv", index=False)
+    if _dist.rank() == 0:
+        stats.to_csv(ends_with(local_path) + "ts_cols_stats.csv", index=False)
+    _dist.barrier()
     return odf, ts_cols_post, num_cols, cat_cols

@@ -41,9 +41,13 @@ def save_stats(ctx, idf, master_path, function_name, reread=False, run_type="loc
     local_path = master_path if run_type == "local" else "report_stats"
     if mlflow_config is not None and mlflow_config.get("track_reports", False):
         local_path = os.path.join(local_path, str(mlflow_config.get("run_id", "run")))
-    Path(local_path).mkdir(parents=True, exist_ok=True)
-    pdf = idf.to_pandas() if isinstance(idf, AnovosFrame) else idf
-    pdf.to_csv(ends_with(local_path) + function_name + ".csv", index=False)
+    from anovos_amd.core import dist as _dist
+
+    if _dist.rank() == 0:
+        Path(local_path).mkdir(parents=True, exist_ok=True)
+        pdf = idf.to_pandas() if isinstance(idf, AnovosFrame) else idf
+        pdf.to_csv(ends_with(local_path) + function_name + ".csv", index=False)
+    _dist.barrier()
     if reread:
         return pd.read_csv(ends_with(local_path) + function_name + ".csv")
     return None
