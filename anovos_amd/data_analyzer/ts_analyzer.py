@@ -128,6 +128,8 @@ def ts_viz_data(idf: AnovosFrame, x_col: str, y_col: str, output_mode="append", 
                 n_cat: int = 10, tz_offset="local") -> pd.DataFrame:
     """Reference ts_analyzer.py:259 — per (bucket × column) aggregates:
     counts for categorical y, min/max/mean/median for numeric y."""
+    from anovos_amd.core import dist as _dist
+
     key_map = {"daily": "yyyymmdd_col", "hourly": "daypart_cat", "weekly": "dow"}
     k_col = key_map[output_type]
     key = idf.col(k_col)
@@ -144,6 +146,19 @@ def ts_viz_data(idf: AnovosFrame, x_col: str, y_col: str, output_mode="append", 
         knull = torch.isnan(key.data)
         uniq, kcodes = torch.unique(torch.nan_to_num(key.data), return_inverse=True)
         klabels = [float(u) for u in uniq]
+    if _dist.is_dist() and key.kind != "categorical":
+        # unify bucket keys across ranks (dates/dows differ per shard)
+        import numpy as _np
+
+        gathered = _dist.all_gather_object(uniq.cpu().numpy())
+        guniq = _np.unique(_np.concatenate(gathered))
+        gt = torch.from_numpy(guniq).to(key.data.device)
+        pos = torch.searchsorted(gt, uniq)
+        kcodes = pos[kcodes]
+        if key.dtype in ("timestamp", "date"):
+            klabels = [str(pd.Timestamp(int(u), unit="us").date()) if int(u) != NULL_TS else None for u in guniq]
+        else:
+            klabels = [float(u) for u in guniq]
     G = len(klabels)
     dev = kcodes.device
     if yc.kind == "categorical":
@@ -160,6 +175,7 @@ def ts_viz_data(idf: AnovosFrame, x_col: str, y_col: str, output_mode="append", 
             sel = m & (yv == int(t))
             per = torch.zeros(G, dtype=torch.float64, device=dev).scatter_reduce(
                 0, kcodes[sel], torch.ones(int(sel.sum()), dtype=torch.float64, device=dev), reduce="sum")
+            _dist.all_reduce_(per, "sum")
             for g in range(G):
                 if float(per[g]) > 0:
                     rows.append([yc.dictionary[int(t)], klabels[g], float(per[g])])
@@ -172,6 +188,10 @@ def ts_viz_data(idf: AnovosFrame, x_col: str, y_col: str, output_mode="append", 
     s = torch.zeros(G, dtype=torch.float64, device=dev).scatter_reduce(0, kv, xv, reduce="sum")
     mn = torch.full((G,), float("inf"), dtype=torch.float64, device=dev).scatter_reduce(0, kv, xv, reduce="amin")
     mx = torch.full((G,), float("-inf"), dtype=torch.float64, device=dev).scatter_reduce(0, kv, xv, reduce="amax")
+    _dist.all_reduce_(cnt, "sum")
+    _dist.all_reduce_(s, "sum")
+    _dist.all_reduce_(mn, "min")
+    _dist.all_reduce_(mx, "max")
     # median: sort values, stable-sort groups, middle of each segment
     byval = torch.argsort(xv)
     bygrp = torch.argsort(kv[byval], stable=True)
