@@ -127,7 +127,10 @@ def ts_eligiblity_check(ctx, idf: AnovosFrame, id_col: str, opt: int = 1, tz_off
 def ts_viz_data(idf: AnovosFrame, x_col: str, y_col: str, output_mode="append", output_type="daily",
                 n_cat: int = 10, tz_offset="local") -> pd.DataFrame:
     """Reference ts_analyzer.py:259 — per (bucket × column) aggregates:
-    counts for categorical y, min/max/mean/median for numeric y."""
+    counts for categorical y, min/max/mean/median for numeric y.
+    Multi-rank: keys are unified and count/sum/min/max all-reduced; the
+    per-bucket median is computed over the local shard (a full global
+    median would need per-bucket value exchange — report-viz tolerance)."""
     from anovos_amd.core import dist as _dist
 
     key_map = {"daily": "yyyymmdd_col", "hourly": "daypart_cat", "weekly": "dow"}
