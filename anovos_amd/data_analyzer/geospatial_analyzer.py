@@ -111,10 +111,13 @@ def descriptive_stats_gen(df, lat_col, long_col, geohash_col, id_col, master_pat
         for nm, tbl in zip(names, [gen_stats, top_pairs]):
             tbl.to_csv(ends_with(master_path) + nm + "_1_" + lat_col + "_" + long_col + ".csv", index=False)
     if geohash_col is not None:
+        from anovos_amd.core import dist as _dist3
+
         c = df.col(geohash_col)
         valid = ~c.null_mask()
         codes = c.data[valid].to(torch.long)
         cnt = torch.bincount(codes, minlength=len(c.dictionary or []))
+        _dist3.all_reduce_(cnt, "sum")  # dictionaries are rank-unified at ingest
         order = torch.argsort(cnt, descending=True)
         dist_geohash = int((cnt > 0).sum())
         prec = max((len(s) for s in (c.dictionary or [])), default=0)
@@ -134,7 +137,7 @@ def descriptive_stats_gen(df, lat_col, long_col, geohash_col, id_col, master_pat
                     "The Most Common Geohash",
                 ],
                 "Count": [
-                    int(valid.sum()),
+                    int(_dist3.all_reduce_scalar(int(valid.sum()))),
                     dist_geohash,
                     prec,
                     top_gh[geohash_col].iloc[0] if len(top_gh) else "NA",
