@@ -212,3 +212,50 @@ def test_dateformat_conversion(ctx):
     odf = adt.dateformat_conversion(ctx, idf, ["s"], input_format="%d/%m/%Y", output_format="%Y-%m-%d")
     vals = odf.col("s").to_numpy_objects()
     assert vals[0] == "2020-02-01" and vals[1] == "2021-06-15"
+
+
+def test_civil_decompose_fuzz_vs_pandas(ctx):
+    """Hinnant civil arithmetic vs pandas over random epochs 1700-2200."""
+    rng = np.random.default_rng(99)
+    epochs_us = rng.integers(
+        int(pd.Timestamp("1700-01-01").value // 1000),
+        int(pd.Timestamp("2200-01-01").value // 1000),
+        5000,
+    )
+    pdf = pd.DataFrame({"ts": pd.to_datetime(epochs_us, unit="us")})
+    idf = AnovosFrame.from_pandas(pdf, device="cpu")
+    odf = adt.timeUnits_extraction(idf, ["ts"], "all")
+    got = {u: odf.col(f"ts_{u}").data.numpy().astype(int) for u in
+           ["year", "month", "dayofmonth", "hour", "minute", "second", "dayofweek", "dayofyear", "weekofyear", "quarter"]}
+    s = pdf["ts"].dt
+    assert (got["year"] == s.year.to_numpy()).all()
+    assert (got["month"] == s.month.to_numpy()).all()
+    assert (got["dayofmonth"] == s.day.to_numpy()).all()
+    assert (got["hour"] == s.hour.to_numpy()).all()
+    assert (got["minute"] == s.minute.to_numpy()).all()
+    assert (got["second"] == s.second.to_numpy()).all()
+    # pandas dayofweek: 0=Monday; Spark: 1=Sunday..7=Saturday
+    spark_dow = ((s.dayofweek.to_numpy() + 1) % 7) + 1
+    assert (got["dayofweek"] == spark_dow).all()
+    assert (got["dayofyear"] == s.dayofyear.to_numpy()).all()
+    assert (got["weekofyear"] == s.isocalendar().week.to_numpy().astype(int)).all()
+    assert (got["quarter"] == s.quarter.to_numpy()).all()
+
+
+def test_month_boundaries_fuzz_vs_pandas(ctx):
+    rng = np.random.default_rng(5)
+    epochs_us = rng.integers(
+        int(pd.Timestamp("1900-01-01").value // 1000),
+        int(pd.Timestamp("2100-01-01").value // 1000),
+        3000,
+    )
+    pdf = pd.DataFrame({"ts": pd.to_datetime(epochs_us, unit="us")})
+    idf = AnovosFrame.from_pandas(pdf, device="cpu")
+    eo = adt.end_of_month(idf, ["ts"]).col("ts_monthEnd").data.numpy()
+    ref = (pdf["ts"] + pd.offsets.MonthEnd(0)).dt.normalize()
+    # rows already at month end: MonthEnd(0) keeps the date — same as ours
+    got_days = eo // (86400 * 10**6)
+    ref_days = ref.astype("int64").to_numpy() // (86400 * 10**9)
+    assert (got_days == ref_days).all()
+    lp = adt.is_leapYear(idf, ["ts"]).col("ts_isleapYear").data.numpy()
+    assert (lp == pdf["ts"].dt.is_leap_year.to_numpy().astype(float)).all()
