@@ -298,7 +298,14 @@ def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, de
                     continue
                 f = getattr(association_evaluator, subkey)
                 extra = stats_args(all_configs, subkey)
-                stats = f(ctx, df, **value, **extra)
+                if subkey == "correlation_matrix" and all_configs.get("cat_to_num_transformer"):
+                    # reference workflow.py:600-602: encode categoricals
+                    # on a frame copy before the correlation matrix
+                    df_corr = transformers.cat_to_num_transformer(
+                        ctx, df, **all_configs["cat_to_num_transformer"])
+                    stats = f(ctx, df_corr, **value, **extra)
+                else:
+                    stats = f(ctx, df, **value, **extra)
                 if report_input_path:
                     report_preprocessing.save_stats(ctx, stats, report_input_path, subkey, run_type=run_type)
                 _log(f"association_evaluator.{subkey}: {time.time() - start:.3f}s")
