@@ -8,7 +8,8 @@
 //   K3/K6  column_histograms   — equal-width histograms, LDS-staged bins.
 //          bracket_histograms  — quantile-refinement histograms.
 //          bucketize_columns   — branchless binary search over LDS cutoffs.
-//   K4     hll_registers       — HyperLogLog p=14, LDS-staged registers.
+//   K4     hll_registers       — HyperLogLog (p=12 default), LDS-staged
+//                                registers; moments_hll fuses K1/K2+K4.
 //   K5     code_counts         — dictionary-code bincount, LDS-staged.
 //   K10    row_null_counts     — fused row-wise NaN count across columns.
 //
