@@ -1,0 +1,116 @@
+"""Property-style invariant checks over randomized frames: whatever the
+data, the analyzer's outputs must satisfy these algebraic facts."""
+
+import numpy as np
+import pandas as pd
+import pytest
+import torch
+
+from anovos_amd.core.frame import AnovosFrame
+from anovos_amd.data_analyzer import quality_checker as qc
+from anovos_amd.data_analyzer import stats_generator as sg
+from anovos_amd.drift_stability import drift_detector as dd
+from anovos_amd.shared.context import init_context
+
+
+@pytest.fixture
+def ctx():
+    return init_context("cpu")
+
+
+def random_frame(seed, n=20_000):
+    rng = np.random.default_rng(seed)
+    pdf = pd.DataFrame(
+        {
+            "a": rng.normal(rng.uniform(-100, 100), rng.uniform(0.1, 50), n),
+            "b": rng.lognormal(2, 1, n),
+            "c": rng.integers(-5, 5, n).astype(float),
+            "d": rng.choice([f"v{i}" for i in range(rng.integers(2, 30))], n),
+        }
+    )
+    for col in ("a", "b", "c"):
+        pdf.loc[rng.random(n) < rng.uniform(0, 0.3), col] = np.nan
+    pdf.loc[rng.random(n) < 0.1, "d"] = None
+    return AnovosFrame.from_pandas(pdf, device="cpu")
+
+
+@pytest.mark.parametrize("seed", [1, 2, 3, 4])
+def test_count_invariants(ctx, seed):
+    idf = random_frame(seed)
+    total = idf.count()
+    m = sg.measures_of_counts(ctx, idf).set_index("attribute")
+    for attr in m.index:
+        assert int(m.loc[attr, "fill_count"]) + int(m.loc[attr, "missing_count"]) == total
+        assert 0 <= float(m.loc[attr, "missing_pct"]) <= 1
+
+
+@pytest.mark.parametrize("seed", [5, 6, 7])
+def test_percentile_monotonicity(ctx, seed):
+    idf = random_frame(seed, n=400_000)  # sketch path
+    p = sg.measures_of_percentiles(ctx, idf, ["a", "b", "c"]).set_index("attribute")
+    qs = ["1%", "5%", "10%", "25%", "50%", "75%", "90%", "95%", "99%"]
+    for attr in p.index:
+        vals = [float(p.loc[attr, q]) for q in qs]
+        vals = [v for v in vals if v == v]
+        assert vals == sorted(vals), (attr, vals)
+
+
+@pytest.mark.parametrize("seed", [8, 9])
+def test_dispersion_invariants(ctx, seed):
+    idf = random_frame(seed)
+    d = sg.measures_of_dispersion(ctx, idf).set_index("attribute")
+    for attr in d.index:
+        sd = float(d.loc[attr, "stddev"])
+        var = float(d.loc[attr, "variance"])
+        if sd == sd:
+            assert sd >= 0
+            assert var == pytest.approx(sd * sd, rel=5e-3)  # table rounds to 4 decimals
+        iqr = float(d.loc[attr, "IQR"])
+        if iqr == iqr:
+            assert iqr >= -1e-9
+
+
+@pytest.mark.parametrize("seed", [10, 11])
+def test_drift_invariants(ctx, seed):
+    idf = random_frame(seed)
+    rng = np.random.default_rng(seed + 100)
+    from anovos_amd.core.frame import Column
+
+    shifted = idf.with_column("a", Column("a", "float", idf.col("a").data + rng.uniform(0, 5)))
+    import tempfile
+
+    with tempfile.TemporaryDirectory() as tmp:
+        stats = dd.statistics(ctx, shifted, idf, list_of_cols=["a", "b"], method_type="all",
+                              use_sampling=False, source_path=tmp)
+    for _, r in stats.iterrows():
+        assert float(r["PSI"]) >= -1e-9
+        assert 0 <= float(r["HD"]) <= 1 + 1e-9
+        assert float(r["JSD"]) >= -1e-9
+        assert 0 <= float(r["KS"]) <= 1 + 1e-9
+    # identical frames drift to zero
+    with tempfile.TemporaryDirectory() as tmp:
+        zero = dd.statistics(ctx, idf, idf, list_of_cols=["a", "b"], method_type="all",
+                             use_sampling=False, source_path=tmp)
+    assert (pd.to_numeric(zero["PSI"]) == 0).all()
+    assert (zero["flagged"] == 0).all()
+
+
+@pytest.mark.parametrize("seed", [12, 13])
+def test_unique_counts_bounds(ctx, seed):
+    idf = random_frame(seed, n=400_000)
+    u = sg.uniqueCount_computation(ctx, idf, compute_approx_unique_count=True).set_index("attribute")
+    m = sg.measures_of_counts(ctx, idf).set_index("attribute")
+    for attr in u.index:
+        uv = float(u.loc[attr, "unique_values"])
+        fill = float(m.loc[attr, "fill_count"])
+        assert 0 <= uv <= fill * 1.05 + 10  # HLL rsd margin
+
+
+@pytest.mark.parametrize("seed", [14])
+def test_nullrows_consistency(ctx, seed):
+    idf = random_frame(seed)
+    total = idf.count()
+    _, nr = qc.nullRows_detection(ctx, idf, treatment=False)
+    assert int(nr["row_count"].sum()) == total
+    odf, _ = qc.duplicate_detection(ctx, idf, treatment=True)
+    assert odf.count() <= total
