@@ -35,12 +35,13 @@ def init_from_env(timeout_s: int = 600) -> None:
     ws = int(os.environ.get("WORLD_SIZE", "1"))
     if ws <= 1:
         return
-    backend = "nccl" if torch.cuda.is_available() else "gloo"
+    backend = os.environ.get("ANOVOS_AMD_DIST_BACKEND") or ("nccl" if torch.cuda.is_available() else "gloo")
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29517")
     td.init_process_group(backend=backend, timeout=datetime.timedelta(seconds=timeout_s))
     if torch.cuda.is_available():
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+        # ranks may oversubscribe one device in tests (gloo backend)
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")) % max(torch.cuda.device_count(), 1))
     _initialized = True
 
 
