@@ -22,7 +22,8 @@ class AnovosContext:
         dist.init_from_env()
         if device is None:
             if torch.cuda.is_available():
-                device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", "0")))
+                # modulo: tests may oversubscribe one device with 2 ranks
+                device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", "0")) % max(torch.cuda.device_count(), 1))
             else:
                 device = torch.device("cpu")
         self.device = torch.device(device)
