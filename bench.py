@@ -134,18 +134,24 @@ def pipeline_step(ctx, idf, source_hist, model_dir):
 
     drift_cols = num_cols[:: max(1, len(num_cols) // 50)]
     q_freqs = dd.batched_bin_frequencies(binned, [c + "_binned" for c in drift_cols], idf.count(), max_bin=10)
-    for c in drift_cols:
-        q_keys, q_vals = q_freqs[c + "_binned"]
-        pmap = source_hist.get(c, {})
-        qmap = dict(zip(q_keys, q_vals))
-        keys = sorted(set(pmap) | set(qmap), key=dd._key_order)
-        p = _np.array([max(pmap.get(k, 0.0001), 0.0001) for k in keys])
-        q = _np.array([max(qmap.get(k, 0.0001), 0.0001) for k in keys])
-        drift_vals += float(_np.sum((p - q) * _np.log(p / q)))
-        m = (p + q) / 2
-        drift_vals += float((_np.sum(p * _np.log(p / m)) + _np.sum(q * _np.log(q / m))) / 2)
-        drift_vals += float(_np.sqrt(_np.sum((_np.sqrt(p) - _np.sqrt(q)) ** 2) / 2))
-        drift_vals += float(_np.max(_np.abs(_np.cumsum(p) - _np.cumsum(q))))
+    # vectorize PSI/JSD/HD/KS over the whole (col x bin) matrix at once
+    key_union = sorted({k for c in drift_cols for k in source_hist.get(c, {})}
+                       | {k for c in drift_cols for k in q_freqs[c + "_binned"][0]}, key=dd._key_order)
+    kpos = {k: j for j, k in enumerate(key_union)}
+    K = len(key_union)
+    P = _np.full((len(drift_cols), K), 0.0001)
+    Q = _np.full((len(drift_cols), K), 0.0001)
+    for i, c in enumerate(drift_cols):
+        for k, v in source_hist.get(c, {}).items():
+            P[i, kpos[k]] = max(v, 0.0001)
+        qk, qv = q_freqs[c + "_binned"]
+        for k, v in zip(qk, qv):
+            Q[i, kpos[k]] = max(v, 0.0001)
+    M = (P + Q) / 2
+    drift_vals += float(_np.sum((P - Q) * _np.log(P / Q)))  # PSI
+    drift_vals += float((_np.sum(P * _np.log(P / M)) + _np.sum(Q * _np.log(Q / M))) / 2)  # JSD
+    drift_vals += float(_np.sum(_np.sqrt(_np.sum((_np.sqrt(P) - _np.sqrt(Q)) ** 2, axis=1) / 2)))  # HD
+    drift_vals += float(_np.sum(_np.max(_np.abs(_np.cumsum(P, axis=1) - _np.cumsum(Q, axis=1)), axis=1)))  # KS
     chk += drift_vals
 
     # ---- transform ----
