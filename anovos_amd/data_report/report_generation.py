@@ -223,7 +223,7 @@ def quality_check(master_path: str) -> str:
     return "".join(parts) if parts else "<p class='note'>No quality-check statistics saved.</p>"
 
 
-def attribute_associations(master_path: str, label_col, event_label) -> str:
+def attribute_associations(master_path: str, label_col, event_label, corr_threshold=0.4, iv_threshold=0.02) -> str:
     """Reference report_generation.py:1291 — correlation heatmap, IV/IG
     bars, variable clustering table + event-rate charts."""
     parts = []
@@ -239,6 +239,12 @@ def attribute_associations(master_path: str, label_col, event_label) -> str:
         if df is not None:
             vcol = [c for c in df.columns if c != "attribute"][0]
             fig = go.Figure(go.Bar(x=df["attribute"], y=df[vcol]))
+            if metric == "iv" and iv_threshold:
+                fig.add_hline(y=iv_threshold, line_dash="dash",
+                              annotation_text=f"iv_threshold={iv_threshold}")
+                n_pred = int((pd.to_numeric(df[vcol], errors="coerce") >= iv_threshold).sum())
+                parts.append(f"<div class='kpi'><div class='v'>{n_pred}</div>"
+                             f"<div class='l'>attributes with IV ≥ {iv_threshold}</div></div>")
             fig.update_layout(title=name, height=380)
             parts.append(f"<h3>{name}</h3>" + _fig_div(fig, f"bar_{metric}") + _tbl(df))
     vc = _read_csv(master_path, "variable_clustering")
@@ -341,7 +347,8 @@ def anovos_report(master_path: str, id_col="", label_col=None, corr_threshold=0.
                                 metricDict_path if metricDict_path != "." else None)),
         ("Descriptive Statistics", descriptive_statistics(master_path)),
         ("Quality Check", quality_check(master_path)),
-        ("Attribute Associations", attribute_associations(master_path, label_col, event_label)),
+        ("Attribute Associations", attribute_associations(master_path, label_col, event_label,
+                                                           corr_threshold, iv_threshold)),
         ("Data Drift & Data Stability", data_drift_stability(master_path, drift_threshold_model)),
     ]
     ts = ts_viz_generate(master_path)
