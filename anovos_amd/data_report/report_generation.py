@@ -233,6 +233,14 @@ def attribute_associations(master_path: str, label_col, event_label, corr_thresh
         mat = corr.drop(columns=["attribute"], errors="ignore")
         fig = go.Figure(go.Heatmap(z=mat.values, x=list(mat.columns), y=attrs, colorscale="Peach", zmin=-1, zmax=1))
         fig.update_layout(title="Correlation Matrix", height=520)
+        if corr_threshold:
+            import numpy as _np
+
+            v = _np.abs(mat.to_numpy(dtype=float))
+            _np.fill_diagonal(v, 0.0)
+            n_pairs = int((v >= corr_threshold).sum() // 2)
+            parts.append(f"<div class='kpi'><div class='v'>{n_pairs}</div>"
+                         f"<div class='l'>attribute pairs with |corr| ≥ {corr_threshold}</div></div>")
         parts.append("<h3>Correlation Matrix</h3>" + _fig_div(fig, "corrheat"))
     for name, metric in (("IV_calculation", "iv"), ("IG_calculation", "ig")):
         df = _read_csv(master_path, name)
