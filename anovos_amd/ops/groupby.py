@@ -103,6 +103,23 @@ def discrete_modes(idf, cols: List[str]) -> Dict[str, Tuple[Optional[float], int
             val = moments[c].min + int(idx * R[i] / M)
             out[c] = (float(val), cnt) if cnt > 0 else (None, 0)
     for c in rest:
+        m = moments[c]
+        if m.n > 20_000_000 and not m.integral:
+            # continuous column at production scale: exact mode means a
+            # 64-bit sort + cross-rank value-count merge of ~n uniques —
+            # GBs of traffic for a statistic that is almost surely a
+            # 1-count tie. Approximate with the mode BIN of a fine
+            # histogram (midpoint), like Spark users bin first.
+            import warnings
+
+            warnings.warn(f"mode of continuous column '{c}' approximated by 4096-bin histogram at n={int(m.n)}")
+            lo1 = torch.tensor([m.min], dtype=torch.float64)
+            hi1 = torch.tensor([m.max], dtype=torch.float64)
+            h = hist_ops.global_histograms([idf.col(c).data], lo1, hi1, 4096).cpu().numpy()[0]
+            i = int(np.argmax(h))
+            w = (m.max - m.min) / 4096 if m.max > m.min else 0.0
+            out[c] = (float(m.min + (i + 0.5) * w), int(h[i]))
+            continue
         vals, cnts = numeric_value_counts(idf, c)
         if vals.numel() == 0:
             out[c] = (None, 0)
