@@ -88,10 +88,12 @@ def approx_distinct(idf, cols: List[str]) -> Dict[str, int]:
             # later analyzer section will hit (ops/hip moments_hll)
             from anovos_amd.ops import stats as stats_ops
 
-            mom_local, flat = ext.moments_hll([idf.col(c).data.contiguous() for c in cols], HLL_P)
+            tensors = [idf.col(c).data.contiguous() for c in cols]
+            shifts = stats_ops.compute_column_shifts(tensors)
+            mom_local, flat = ext.moments_hll(tensors, HLL_P, shifts)
             glob = stats_ops.merge_moments_global(mom_local).numpy().tolist()
             for i, c in enumerate(cols):
-                idf.col(c).cache.setdefault("moments", stats_ops.MomentStats(glob[i]))
+                idf.col(c).cache.setdefault("moments", stats_ops.MomentStats(glob[i], shift=shifts[i]))
         else:
             flat = ext.hll_registers_multi([idf.col(c).data.contiguous() for c in cols], HLL_P)
     else:

@@ -10,9 +10,9 @@
 #include <vector>
 
 extern "C" {
-int anovos_moments(const void *const *cols, const int64_t *lens, int ncols,
-                   int nchunks, int dtype, double *partials, double *out,
-                   hipStream_t stream);
+int anovos_moments(const void *const *cols, const int64_t *lens,
+                   const double *shifts, int ncols, int nchunks, int dtype,
+                   double *partials, double *out, hipStream_t stream);
 int anovos_hist(const void *const *cols, const int64_t *lens, int ncols,
                 const double *lo, const double *hi, int nbins, int nchunks,
                 int dtype, uint64_t *out, hipStream_t stream);
@@ -63,9 +63,9 @@ int anovos_outlier_clamp(const void *const *cols, const int64_t *lens,
                          int nchunks, int mode, int dtype, void *const *outs,
                          uint64_t *counts, hipStream_t stream);
 int anovos_moments_hll(const void *const *cols, const int64_t *lens,
-                       int ncols, int p, int nchunks, int dtype,
-                       double *partials, double *mom_out, int32_t *regs,
-                       hipStream_t stream);
+                       const double *shifts, int ncols, int p, int nchunks,
+                       int dtype, double *partials, double *mom_out,
+                       int32_t *regs, hipStream_t stream);
 int anovos_centered_gram(const void *const *cols, int64_t n, int k,
                          const float *means, const int *pair_i,
                          const int *pair_j, int npairs, int row_chunks,
@@ -101,6 +101,13 @@ torch::Tensor to_device_i64(const std::vector<int64_t> &vals, const torch::Devic
   return cpu.to(dev, /*non_blocking=*/true);
 }
 
+torch::Tensor to_device_f64(const std::vector<double> &vals, const torch::Device &dev) {
+  auto cpu = torch::from_blob((void *)vals.data(), {(int64_t)vals.size()},
+                              torch::TensorOptions().dtype(torch::kFloat64))
+                 .clone();
+  return cpu.to(dev, /*non_blocking=*/true);
+}
+
 int dtype_code(const torch::Tensor &t) {
   if (t.scalar_type() == torch::kFloat32) return 0;
   if (t.scalar_type() == torch::kFloat64) return 1;
@@ -110,20 +117,23 @@ int dtype_code(const torch::Tensor &t) {
 
 }  // namespace
 
-torch::Tensor column_moments(std::vector<torch::Tensor> cols) {
+torch::Tensor column_moments(std::vector<torch::Tensor> cols, std::vector<double> shifts) {
   TORCH_CHECK(!cols.empty(), "no columns");
+  TORCH_CHECK(shifts.empty() || shifts.size() == cols.size(), "shifts must match cols");
   auto device = cols[0].device();
   auto out = torch::zeros({(int64_t)cols.size(), 9},
                           torch::TensorOptions().dtype(torch::kFloat64).device(device));
   // group by dtype, one fused launch per dtype
   for (int pass = 0; pass < 2; ++pass) {
     std::vector<int64_t> ptrs, lens, idx;
+    std::vector<double> sh;
     for (size_t i = 0; i < cols.size(); ++i) {
       auto &t = cols[i];
       TORCH_CHECK(t.is_contiguous() && t.device() == device, "columns must be contiguous, same device");
       if (dtype_code(t) != pass) continue;
       ptrs.push_back((int64_t)t.data_ptr());
       lens.push_back(t.numel());
+      sh.push_back(shifts.empty() ? 0.0 : shifts[i]);
       idx.push_back((int64_t)i);
     }
     if (ptrs.empty()) continue;
@@ -132,11 +142,13 @@ torch::Tensor column_moments(std::vector<torch::Tensor> cols) {
     int nchunks = pick_chunks(maxn, ncols);
     auto dptr = to_device_i64(ptrs, device);
     auto dlen = to_device_i64(lens, device);
+    auto dsh = to_device_f64(sh, device);
     auto partials = torch::empty({(int64_t)ncols * nchunks, 9},
                                  torch::TensorOptions().dtype(torch::kFloat64).device(device));
     auto sub = torch::empty({ncols, 9}, torch::TensorOptions().dtype(torch::kFloat64).device(device));
     check_hip(anovos_moments((const void *const *)dptr.data_ptr<int64_t>(),
-                             dlen.data_ptr<int64_t>(), ncols, nchunks, pass,
+                             dlen.data_ptr<int64_t>(), dsh.data_ptr<double>(),
+                             ncols, nchunks, pass,
                              partials.data_ptr<double>(), sub.data_ptr<double>(),
                              current_stream()),
               "anovos_moments");
@@ -716,8 +728,10 @@ std::tuple<torch::Tensor, std::vector<torch::Tensor>> outlier_clamp_columns(
 
 // K1/K2+K4 fused: moments + HLL registers in one read.
 // Returns (moments [ncols,9] f64, regs [ncols, 1<<p] i32).
-std::tuple<torch::Tensor, torch::Tensor> moments_hll(std::vector<torch::Tensor> cols, int64_t p) {
+std::tuple<torch::Tensor, torch::Tensor> moments_hll(std::vector<torch::Tensor> cols, int64_t p,
+                                                     std::vector<double> shifts) {
   TORCH_CHECK(!cols.empty(), "no columns");
+  TORCH_CHECK(shifts.empty() || shifts.size() == cols.size(), "shifts must match cols");
   auto device = cols[0].device();
   int64_t m = 1LL << p;
   auto mom = torch::zeros({(int64_t)cols.size(), 9},
@@ -726,12 +740,14 @@ std::tuple<torch::Tensor, torch::Tensor> moments_hll(std::vector<torch::Tensor> 
                            torch::TensorOptions().dtype(torch::kInt32).device(device));
   for (int pass = 0; pass < 2; ++pass) {
     std::vector<int64_t> ptrs, lens, idx;
+    std::vector<double> sh;
     for (size_t i = 0; i < cols.size(); ++i) {
       auto &t = cols[i];
       TORCH_CHECK(t.is_contiguous() && t.device() == device, "columns must be contiguous, same device");
       if (dtype_code(t) != pass) continue;
       ptrs.push_back((int64_t)t.data_ptr());
       lens.push_back(t.numel());
+      sh.push_back(shifts.empty() ? 0.0 : shifts[i]);
       idx.push_back((int64_t)i);
     }
     if (ptrs.empty()) continue;
@@ -740,12 +756,14 @@ std::tuple<torch::Tensor, torch::Tensor> moments_hll(std::vector<torch::Tensor> 
     int nchunks = pick_chunks(maxn, ncols);
     auto dptr = to_device_i64(ptrs, device);
     auto dlen = to_device_i64(lens, device);
+    auto dsh = to_device_f64(sh, device);
     auto partials = torch::empty({(int64_t)ncols * nchunks, 9},
                                  torch::TensorOptions().dtype(torch::kFloat64).device(device));
     auto mom_sub = torch::empty({ncols, 9}, torch::TensorOptions().dtype(torch::kFloat64).device(device));
     auto reg_sub = torch::zeros({ncols, m}, torch::TensorOptions().dtype(torch::kInt32).device(device));
     check_hip(anovos_moments_hll((const void *const *)dptr.data_ptr<int64_t>(),
-                                 dlen.data_ptr<int64_t>(), ncols, (int)p, nchunks, pass,
+                                 dlen.data_ptr<int64_t>(), dsh.data_ptr<double>(),
+                                 ncols, (int)p, nchunks, pass,
                                  partials.data_ptr<double>(), mom_sub.data_ptr<double>(),
                                  reg_sub.data_ptr<int>(), current_stream()),
               "anovos_moments_hll");
@@ -759,7 +777,8 @@ std::tuple<torch::Tensor, torch::Tensor> moments_hll(std::vector<torch::Tensor> 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("code_counts_multi", &code_counts_multi, "fused multi-column code counts + null slot (K5)");
   m.def("outlier_clamp_columns", &outlier_clamp_columns, "fused outlier count/clamp (K10/K11)");
-  m.def("moments_hll", &moments_hll, "fused moments + HLL registers (K1/K2+K4)");
+  m.def("moments_hll", &moments_hll, "fused moments + HLL registers (K1/K2+K4)",
+        py::arg("cols"), py::arg("p"), py::arg("shifts") = std::vector<double>());
   m.def("centered_gram_bf16", &centered_gram_bf16, "bf16 MFMA centered Gram X^T X (K8)");
   m.def("bracket_histograms_grouped", &bracket_histograms_grouped, "grouped refinement histograms (K3)");
   m.def("bucketize_columns_float", &bucketize_columns_float, "bucketize to float bin labels (K6)");
@@ -768,7 +787,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hll_registers_multi", &hll_registers_multi, "fused multi-column HLL (K4)");
   m.def("scale_columns", &scale_columns, "fused (x-a)*b scaling (K11)");
   m.def("fill_nan_columns", &fill_nan_columns, "fused NaN fill (K11)");
-  m.def("column_moments", &column_moments, "fused per-column moments (K1/K2)");
+  m.def("column_moments", &column_moments, "fused per-column moments (K1/K2)",
+        py::arg("cols"), py::arg("shifts") = std::vector<double>());
   m.def("column_histograms", &column_histograms, "fused per-column histograms (K3/K6)");
   m.def("bracket_histograms", &bracket_histograms, "quantile-refinement histograms (K3)");
   m.def("bucketize_columns", &bucketize_columns, "branchless bucketize (K6)");

@@ -40,14 +40,21 @@ struct MomAcc {
   double n, s1, s2, s3, s4, mn, mx, zn, nf;
 };
 
-DEV_INLINE void mom_add(MomAcc &a, double v) {
+// s1..s4 accumulate about a per-column pivot (shift): raw power sums of
+// x cancel catastrophically for skew/kurt when |mean| >> stddev (the
+// SURVEY hard-part on moment stability at 1e9 rows); with a pivot taken
+// from the data, d = x - shift stays O(spread) and fp64 keeps ~15
+// significant digits of the central moments. min/max/zeros/n_frac use
+// the raw value.
+DEV_INLINE void mom_add(MomAcc &a, double v, double shift) {
   if (!isnan(v)) {
     a.n += 1.0;
-    a.s1 += v;
-    double v2 = v * v;
-    a.s2 += v2;
-    a.s3 += v2 * v;
-    a.s4 += v2 * v2;
+    double d = v - shift;
+    a.s1 += d;
+    double d2 = d * d;
+    a.s2 += d2;
+    a.s3 += d2 * d;
+    a.s4 += d2 * d2;
     a.mn = fmin(a.mn, v);
     a.mx = fmax(a.mx, v);
     a.zn += (v == 0.0) ? 1.0 : 0.0;
@@ -57,11 +64,13 @@ DEV_INLINE void mom_add(MomAcc &a, double v) {
 
 template <typename T>
 __global__ __launch_bounds__(THREADS) void moments_partials_kernel(
-    const T *const *cols, const int64_t *lens, int nchunks, double *partials) {
+    const T *const *cols, const int64_t *lens, const double *shifts, int nchunks,
+    double *partials) {
   const int col = blockIdx.x / nchunks;
   const int chunk = blockIdx.x % nchunks;
   const T *__restrict__ x = cols[col];
   const int64_t n = lens[col];
+  const double shift = shifts ? shifts[col] : 0.0;
   const int64_t per = (n + nchunks - 1) / nchunks;
   const int64_t s = (int64_t)chunk * per;
   const int64_t e = min(n, s + per);
@@ -74,14 +83,14 @@ __global__ __launch_bounds__(THREADS) void moments_partials_kernel(
     const float4 *xv = reinterpret_cast<const float4 *>(x + s);
     for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
       float4 v = xv[i];
-      mom_add(a, (double)v.x);
-      mom_add(a, (double)v.y);
-      mom_add(a, (double)v.z);
-      mom_add(a, (double)v.w);
+      mom_add(a, (double)v.x, shift);
+      mom_add(a, (double)v.y, shift);
+      mom_add(a, (double)v.z, shift);
+      mom_add(a, (double)v.w, shift);
     }
-    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) mom_add(a, (double)x[i]);
+    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) mom_add(a, (double)x[i], shift);
   } else {
-    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) mom_add(a, (double)x[i]);
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) mom_add(a, (double)x[i], shift);
   }
 
   // block reduce through LDS (NSTAT doubles per thread -> tree)
@@ -755,16 +764,17 @@ __global__ __launch_bounds__(THREADS) void fillnan_kernel(
 
 extern "C" {
 
-int anovos_moments(const void *const *cols, const int64_t *lens, int ncols,
-                   int nchunks, int dtype /*0=f32 1=f64*/, double *partials,
-                   double *out, hipStream_t stream) {
+int anovos_moments(const void *const *cols, const int64_t *lens,
+                   const double *shifts, int ncols, int nchunks,
+                   int dtype /*0=f32 1=f64*/, double *partials, double *out,
+                   hipStream_t stream) {
   dim3 grid(ncols * nchunks);
   if (dtype == 0)
     hipLaunchKernelGGL(moments_partials_kernel<float>, grid, dim3(THREADS), 0, stream,
-                       (const float *const *)cols, lens, nchunks, partials);
+                       (const float *const *)cols, lens, shifts, nchunks, partials);
   else
     hipLaunchKernelGGL(moments_partials_kernel<double>, grid, dim3(THREADS), 0, stream,
-                       (const double *const *)cols, lens, nchunks, partials);
+                       (const double *const *)cols, lens, shifts, nchunks, partials);
   hipLaunchKernelGGL(moments_reduce_kernel, dim3(ncols), dim3(THREADS), 0, stream,
                      partials, nchunks, out);
   return (int)hipGetLastError();
@@ -1085,8 +1095,8 @@ int anovos_outlier_clamp(const void *const *cols, const int64_t *lens,
 // ------------------------------------------------------------------
 template <typename T>
 __global__ __launch_bounds__(THREADS) void moments_hll_kernel(
-    const T *const *cols, const int64_t *lens, int p, int nchunks,
-    double *partials, int32_t *regs /*[ncols][1<<p]*/) {
+    const T *const *cols, const int64_t *lens, const double *shifts, int p,
+    int nchunks, double *partials, int32_t *regs /*[ncols][1<<p]*/) {
   extern __shared__ int32_t sreg[];  // 1<<p
   const int m = 1 << p;
   const int col = blockIdx.x / nchunks;
@@ -1096,13 +1106,14 @@ __global__ __launch_bounds__(THREADS) void moments_hll_kernel(
 
   const T *__restrict__ x = cols[col];
   const int64_t n = lens[col];
+  const double shift = shifts ? shifts[col] : 0.0;
   const int64_t per = (n + nchunks - 1) / nchunks;
   const int64_t s = (int64_t)chunk * per;
   const int64_t e = min(n, s + per);
 
   MomAcc a{0, 0, 0, 0, 0, DBL_MAX, -DBL_MAX, 0};
   auto body = [&](T v) {
-    mom_add(a, (double)v);
+    mom_add(a, (double)v, shift);
     int idx, rho;
     if (hll_hash(v, p, idx, rho)) atomicMax(&sreg[idx], rho);
   };
@@ -1148,17 +1159,18 @@ __global__ __launch_bounds__(THREADS) void moments_hll_kernel(
 }
 
 extern "C" int anovos_moments_hll(const void *const *cols, const int64_t *lens,
-                                  int ncols, int p, int nchunks, int dtype,
-                                  double *partials, double *mom_out,
-                                  int32_t *regs, hipStream_t stream) {
+                                  const double *shifts, int ncols, int p,
+                                  int nchunks, int dtype, double *partials,
+                                  double *mom_out, int32_t *regs,
+                                  hipStream_t stream) {
   dim3 grid(ncols * nchunks);
   size_t lds = (size_t)(1 << p) * 4;
   if (dtype == 0)
     hipLaunchKernelGGL(moments_hll_kernel<float>, grid, dim3(THREADS), lds, stream,
-                       (const float *const *)cols, lens, p, nchunks, partials, regs);
+                       (const float *const *)cols, lens, shifts, p, nchunks, partials, regs);
   else
     hipLaunchKernelGGL(moments_hll_kernel<double>, grid, dim3(THREADS), lds, stream,
-                       (const double *const *)cols, lens, p, nchunks, partials, regs);
+                       (const double *const *)cols, lens, shifts, p, nchunks, partials, regs);
   hipLaunchKernelGGL(moments_reduce_kernel, dim3(ncols), dim3(THREADS), 0, stream,
                      partials, nchunks, mom_out);
   return (int)hipGetLastError();

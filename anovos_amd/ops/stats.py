@@ -25,26 +25,62 @@ from anovos_amd.ops import backend
 NSTAT = 9  # n, s1, s2, s3, s4, min, max, zeros, n_frac (non-integral count)
 
 
-def column_moments_local(tensors: Sequence[torch.Tensor]) -> torch.Tensor:
+def compute_column_shifts(tensors: Sequence[torch.Tensor]) -> List[float]:
+    """Per-column accumulation pivots for numerically stable moments.
+
+    Raw power sums Σx..Σx⁴ cancel catastrophically when |mean| >> stddev
+    (skew/kurt lose ALL digits at mean/sd ≳ 1e5 in fp64 — the SURVEY §6
+    hard-part; Spark avoids it with incremental central moments). We
+    instead accumulate about a pivot sampled from the data: the median of
+    the first 64 non-null values (robust to a leading outlier). Every
+    rank must use the SAME pivot so the s1..s4 all-reduce(sum) stays
+    valid — an all-reduce(min) over per-rank candidates picks one
+    deterministically. All-null columns shift by 0."""
+    if len(tensors) == 0:
+        return []
+    import numpy as np
+
+    heads = torch.cat([t[:64].to(torch.float64) for t in tensors]).cpu().numpy()
+    shifts, off = [], 0
+    for t in tensors:
+        k = min(64, t.numel())
+        h = heads[off : off + k]
+        off += k
+        h = h[~np.isnan(h)]
+        shifts.append(float(np.median(h)) if h.size else float("inf"))
+    if dist.is_dist():
+        dev = tensors[0].device
+        tt = torch.tensor(shifts, dtype=torch.float64,
+                          device=dev if dev.type == "cuda" else "cpu")
+        dist.all_reduce_(tt, "min")
+        shifts = tt.cpu().tolist()
+    return [0.0 if s == float("inf") else s for s in shifts]
+
+
+def column_moments_local(tensors: Sequence[torch.Tensor], shifts: Sequence[float] = None) -> torch.Tensor:
     """Local partial moment vectors for a list of numeric columns
-    (NaN = null). Returns fp64 tensor [ncols, 8] on the columns' device."""
+    (NaN = null). Returns fp64 tensor [ncols, NSTAT] on the columns'
+    device; s1..s4 are power sums of (x - shift) per column."""
     if len(tensors) == 0:
         return torch.empty(0, NSTAT, dtype=torch.float64)
+    if shifts is None:
+        shifts = [0.0] * len(tensors)
     dev = tensors[0].device
     if dev.type == "cuda" and backend.use_hip(tensors[0]):
         ext = backend.hip_ext()
-        return ext.column_moments([t.contiguous() for t in tensors])
+        return ext.column_moments([t.contiguous() for t in tensors], list(shifts))
     out = torch.empty(len(tensors), NSTAT, dtype=torch.float64, device=dev)
     for i, t in enumerate(tensors):
         td = t.to(torch.float64)
         valid = ~torch.isnan(td)
         x = torch.where(valid, td, torch.zeros_like(td))
+        d = torch.where(valid, td - shifts[i], torch.zeros_like(td))
         n = valid.sum()
-        s1 = x.sum()
-        x2 = x * x
-        s2 = x2.sum()
-        s3 = (x2 * x).sum()
-        s4 = (x2 * x2).sum()
+        s1 = d.sum()
+        d2 = d * d
+        s2 = d2.sum()
+        s3 = (d2 * d).sum()
+        s4 = (d2 * d2).sum()
         if int(n) > 0:
             mn = td[valid].min()
             mx = td[valid].max()
@@ -76,16 +112,22 @@ def merge_moments_global(local: torch.Tensor) -> torch.Tensor:
 
 
 class MomentStats:
-    """Derived statistics for one column from its global moment vector."""
+    """Derived statistics for one column from its global moment vector.
+    s1..s4 are power sums about `shift` (the accumulation pivot); the
+    central-moment conversion below is exact algebra and stays accurate
+    because |mean - shift| is O(data spread), not O(|mean|)."""
 
-    __slots__ = ("n", "s1", "s2", "s3", "s4", "min", "max", "zeros", "n_frac")
+    __slots__ = ("n", "s1", "s2", "s3", "s4", "min", "max", "zeros", "n_frac", "shift")
 
-    def __init__(self, vec):
+    def __init__(self, vec, shift: float = 0.0):
         vals = [float(v) for v in vec]
         if len(vals) == 8:  # legacy 8-slot vector
             vals.append(float("nan"))
+        if len(vals) == 10:  # shift appended host-side
+            shift = vals.pop()
         (self.n, self.s1, self.s2, self.s3, self.s4, self.min, self.max,
          self.zeros, self.n_frac) = vals
+        self.shift = shift
 
     @property
     def integral(self):
@@ -95,13 +137,15 @@ class MomentStats:
 
     @property
     def mean(self):
-        return self.s1 / self.n if self.n > 0 else float("nan")
+        return self.shift + self.s1 / self.n if self.n > 0 else float("nan")
 
     def _central(self):
-        n, m = self.n, self.mean
-        M2 = self.s2 - n * m * m
-        M3 = self.s3 - 3 * m * self.s2 + 2 * n * m**3
-        M4 = self.s4 - 4 * m * self.s3 + 6 * m * m * self.s2 - 3 * n * m**4
+        # δ = mean - shift; central moments from shifted power sums
+        n = self.n
+        d = self.s1 / n
+        M2 = self.s2 - n * d * d
+        M3 = self.s3 - 3 * d * self.s2 + 2 * n * d**3
+        M4 = self.s4 - 4 * d * self.s3 + 6 * d * d * self.s2 - 3 * n * d**4
         return max(M2, 0.0), M3, M4
 
     @property
@@ -153,10 +197,11 @@ def frame_moments(idf, cols: List[str]) -> Dict[str, MomentStats]:
             todo.append(c)
     if todo:
         tensors = [idf.col(c).data for c in todo]
-        local = column_moments_local(tensors)
+        shifts = compute_column_shifts(tensors)
+        local = column_moments_local(tensors, shifts)
         glob = merge_moments_global(local).numpy().tolist()  # one conversion, not 9/col
         for i, c in enumerate(todo):
-            m = MomentStats(glob[i])
+            m = MomentStats(glob[i], shift=shifts[i])
             idf.col(c).cache["moments"] = m
             out[c] = m
     return out

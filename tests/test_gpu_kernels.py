@@ -402,3 +402,25 @@ def test_captured_transform_graph_replay(ext):
     torch.cuda.synchronize()
     eager_t = time.perf_counter() - t0
     assert graph_t < eager_t * 1.5  # copies included; replay must not regress
+
+
+@requires_gpu
+def test_column_moments_shifted_stability(ext):
+    """Shifted accumulation keeps skew/kurt correct at |mean| >> stddev
+    (raw power sums lose every digit at offset 1e8 — SURVEY §6 hard part)."""
+    from anovos_amd.ops.stats import MomentStats, compute_column_shifts
+
+    g = torch.Generator(device="cpu").manual_seed(41)
+    for offset in (1e6, 1e8, 1e10):
+        x = (torch.randn(1_000_000, generator=g, dtype=torch.float64) + offset).cuda()
+        shifts = compute_column_shifts([x])
+        assert abs(shifts[0] - offset) < 100.0  # pivot lands inside the data
+        vec = ext.column_moments([x], shifts).cpu()[0].tolist()
+        m = MomentStats(vec, shift=shifts[0])
+        assert m.mean == pytest.approx(offset, rel=1e-9)
+        assert m.stddev == pytest.approx(1.0, rel=1e-2)
+        assert abs(m.skewness) < 0.05
+        assert abs(m.kurtosis) < 0.1
+        # fused moments+HLL kernel honors the same pivot
+        mom_f, _ = ext.moments_hll([x], 12, shifts)
+        assert torch.allclose(mom_f.cpu()[0], torch.tensor(vec, dtype=torch.float64), rtol=1e-12, atol=1e-9)
