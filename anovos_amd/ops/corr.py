@@ -33,17 +33,34 @@ def pearson_matrix(idf, cols: List[str], moments=None, use_bf16: bool = True) ->
     if moments is None:
         moments = stats_ops.frame_moments(idf, cols)
     dev = idf.device
-    means = torch.tensor([moments[c].mean for c in cols], dtype=torch.float32, device=dev)
     n = idf.count()
     tensors = [idf.col(c).data for c in cols]
+    col_means = [m if m == m else 0.0 for m in (moments[c].mean for c in cols)]
     if dev.type == "cuda" and backend.use_hip(tensors[0]) and use_bf16:
         ext = backend.hip_ext()
-        f32 = [t.contiguous() if t.dtype == torch.float32 else t.to(torch.float32).contiguous() for t in tensors]
+        # f32 columns: the kernel centers in-register (subtraction of
+        # nearby f32 values is exact, error = storage quantization).
+        # f64 columns MUST center in f64 BEFORE the downcast: casting
+        # raw values loses the whole signal when |mean| >> spread
+        # (f32 spacing at 1e8 is 8.0).
+        f32, kmeans = [], []
+        for t, m in zip(tensors, col_means):
+            if t.dtype == torch.float32:
+                f32.append(t.contiguous())
+                kmeans.append(m)
+            else:
+                f32.append((t - m).to(torch.float32).contiguous())
+                kmeans.append(0.0)
+        means = torch.tensor(kmeans, dtype=torch.float32, device=dev)
         gram = ext.centered_gram_bf16(f32, means)  # [k,k] fp32
     else:
-        X = torch.stack([torch.nan_to_num(t.to(torch.float32), nan=float(means[i])) for i, t in enumerate(tensors)], dim=1)
-        Xc = X - means.unsqueeze(0)
-        gram = Xc.T @ Xc
+        # center in f64, then downcast; NaN contributes zero (== mean)
+        X = torch.stack(
+            [torch.nan_to_num((t.to(torch.float64) - m).to(torch.float32), nan=0.0)
+             for t, m in zip(tensors, col_means)],
+            dim=1,
+        )
+        gram = X.T @ X
     gram = gram.to(torch.float64)
     if dist.is_dist():
         dist.all_reduce_(gram, "sum")
@@ -59,11 +76,13 @@ def covariance_matrix(idf, cols: List[str], moments=None) -> np.ndarray:
     computeCovariance semantics — reference association_eval_varclus.py:71-84)."""
     if moments is None:
         moments = stats_ops.frame_moments(idf, cols)
-    dev = idf.device
-    means = torch.tensor([moments[c].mean for c in cols], dtype=torch.float32, device=dev)
     tensors = [idf.col(c).data for c in cols]
-    X = torch.stack([torch.nan_to_num(t.to(torch.float32), nan=float(means[i])) for i, t in enumerate(tensors)], dim=1)
-    Xc = (X - means.unsqueeze(0)).to(torch.float64)
+    col_means = [m if m == m else 0.0 for m in (moments[c].mean for c in cols)]
+    # center in f64 before any precision loss; NaN contributes zero
+    Xc = torch.stack(
+        [torch.nan_to_num(t.to(torch.float64) - m, nan=0.0) for t, m in zip(tensors, col_means)],
+        dim=1,
+    )
     gram = Xc.T @ Xc
     if dist.is_dist():
         dist.all_reduce_(gram, "sum")

@@ -424,3 +424,20 @@ def test_column_moments_shifted_stability(ext):
         # fused moments+HLL kernel honors the same pivot
         mom_f, _ = ext.moments_hll([x], 12, shifts)
         assert torch.allclose(mom_f.cpu()[0], torch.tensor(vec, dtype=torch.float64), rtol=1e-12, atol=1e-9)
+
+
+@requires_gpu
+def test_pearson_mfma_large_offset():
+    """f64 columns with |mean| >> spread keep their correlation through
+    the bf16 MFMA path (centered in f64 before the downcast)."""
+    from anovos_amd.core.frame import AnovosFrame, Column
+    from anovos_amd.ops import corr
+
+    g = torch.Generator(device="cpu").manual_seed(43)
+    n = 1_000_000
+    z = torch.randn(n, generator=g, dtype=torch.float64)
+    a = (z + 1e9).cuda()
+    b = (0.5 * z + 0.866 * torch.randn(n, generator=g, dtype=torch.float64) + 1e9).cuda()
+    idf = AnovosFrame({"a": Column("a", "double", a), "b": Column("b", "double", b)}, device="cuda")
+    m = corr.pearson_matrix(idf, ["a", "b"])
+    assert abs(m[0, 1] - 0.5) < 0.02

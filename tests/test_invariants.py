@@ -129,3 +129,19 @@ def test_moment_stability_large_offset(ctx, offset):
     assert float(d.loc["x", "stddev"]) == pytest.approx(1.0, rel=2e-2)
     assert abs(float(s.loc["x", "skewness"])) < 0.1
     assert abs(float(s.loc["x", "kurtosis"])) < 0.2
+
+
+def test_correlation_large_offset(ctx):
+    """Pearson/covariance survive |mean| >> spread (f64 centering before
+    any f32 downcast)."""
+    from anovos_amd.core.frame import Column
+    from anovos_amd.ops import corr
+
+    g = torch.Generator().manual_seed(3)
+    n = 200_000
+    z = torch.randn(n, generator=g, dtype=torch.float64)
+    a = z + 1e9
+    b = 0.5 * z + 0.866 * torch.randn(n, generator=g, dtype=torch.float64) + 1e9
+    idf = AnovosFrame({"a": Column("a", "double", a), "b": Column("b", "double", b)}, device="cpu")
+    assert corr.pearson_matrix(idf, ["a", "b"])[0, 1] == pytest.approx(0.5, abs=0.02)
+    assert corr.covariance_matrix(idf, ["a", "b"])[0, 1] == pytest.approx(0.5, abs=0.02)
