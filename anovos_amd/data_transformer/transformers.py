@@ -688,21 +688,33 @@ def imputation_MMM(
             continue
         col = idf.col(c)
         odf = odf.with_column(c + "_imputed", Column(c + "_imputed", col.dtype, col.data.clone()))
+    # categorical mode fill: one fused launch for every column (K11)
+    cat_fill_cols, cat_fill_codes, cat_dicts = [], [], {}
     for c in cat_cols:
         v = fill_cat.get(c)
         col = idf.col(c)
         if v is None:
-            data = col.data.clone()
-            d = list(col.dictionary or [])
+            odf = odf.with_column(
+                c + "_imputed", Column(c + "_imputed", "string", col.data.clone(), list(col.dictionary or []))
+            )
+            continue
+        d = list(col.dictionary or [])
+        if v in d:
+            code = d.index(v)
         else:
-            d = list(col.dictionary or [])
-            if v in d:
-                code = d.index(v)
-            else:
-                d.append(v)
-                code = len(d) - 1
-            data = torch.where(col.data == NULL_CODE, torch.full_like(col.data, code), col.data)
-        odf = odf.with_column(c + "_imputed", Column(c + "_imputed", "string", data, d))
+            d.append(v)
+            code = len(d) - 1
+        cat_fill_cols.append(c)
+        cat_fill_codes.append(code)
+        cat_dicts[c] = d
+    if cat_fill_cols:
+        from anovos_amd.ops import elementwise
+
+        filled_codes = elementwise.fill_code_columns(
+            [idf.col(c).data for c in cat_fill_cols], cat_fill_codes
+        )
+        for c, data in zip(cat_fill_cols, filled_codes):
+            odf = odf.with_column(c + "_imputed", Column(c + "_imputed", "string", data, cat_dicts[c]))
     odf = _finish_output(idf, odf, num_cols + cat_cols, "_imputed", output_mode)
     if print_impact:
         from anovos_amd.data_analyzer.stats_generator import missingCount_computation

@@ -11,6 +11,18 @@ import torch
 from anovos_amd.ops import backend
 
 
+def fill_code_columns(tensors: Sequence[torch.Tensor], fills: Sequence[int]) -> List[torch.Tensor]:
+    """out_i = (codes_i == NULL_CODE) ? fill_i : codes_i — categorical
+    mode imputation, one fused launch for all columns (the torch
+    fallback's eq/full_like/where chain costs ~5x the traffic)."""
+    from anovos_amd.core.dtypes import NULL_CODE
+
+    if tensors and tensors[0].is_cuda and backend.use_hip(tensors[0]):
+        ext = backend.hip_ext()
+        return ext.fill_code_columns([t.contiguous() for t in tensors], [int(f) for f in fills])
+    return [torch.where(t == NULL_CODE, torch.full_like(t, int(f)), t) for t, f in zip(tensors, fills)]
+
+
 def scale_columns(tensors: Sequence[torch.Tensor], a: Sequence[float], b: Sequence[float]) -> List[torch.Tensor]:
     """out_i = (x_i - a_i) * b_i as float32; NaN propagates."""
     if tensors and tensors[0].is_cuda and backend.use_hip(tensors[0]):

@@ -54,6 +54,9 @@ int anovos_lut_apply_f32(const int32_t *const *cols, const int64_t *lens, int nc
 int anovos_lut_apply_i32(const int32_t *const *cols, const int64_t *lens, int ncols,
                          const int32_t *lutflat, const int64_t *lut_off, int nchunks,
                          int32_t *const *outs, hipStream_t stream);
+int anovos_fill_code(const int32_t *const *cols, const int64_t *lens, int ncols,
+                     const int32_t *fill, int nchunks, int32_t *const *outs,
+                     hipStream_t stream);
 int anovos_code_counts_multi(const int32_t *const *cols, const int64_t *lens,
                              const int64_t *offs, const int *sizes, int ncols,
                              int max_slots, int nchunks, uint64_t *out,
@@ -422,6 +425,39 @@ std::vector<torch::Tensor> fill_nan_columns(std::vector<torch::Tensor> cols, tor
   return outs;
 }
 
+// Fused categorical null-fill: out_i = (code == -1) ? fill_i : code.
+std::vector<torch::Tensor> fill_code_columns(std::vector<torch::Tensor> cols,
+                                             std::vector<int64_t> fills) {
+  TORCH_CHECK(!cols.empty(), "no columns");
+  TORCH_CHECK(fills.size() == cols.size(), "fills must match cols");
+  auto device = cols[0].device();
+  std::vector<torch::Tensor> outs;
+  std::vector<int64_t> ptrs, lens, optrs;
+  std::vector<int64_t> fv;
+  for (size_t i = 0; i < cols.size(); ++i) {
+    auto &t = cols[i];
+    TORCH_CHECK(t.scalar_type() == torch::kInt32 && t.is_contiguous() && t.device() == device,
+                "expected contiguous int32 code columns on one device");
+    outs.push_back(torch::empty_like(t));
+    ptrs.push_back((int64_t)t.data_ptr());
+    lens.push_back(t.numel());
+    optrs.push_back((int64_t)outs.back().data_ptr());
+    fv.push_back(fills[i]);
+  }
+  int64_t maxn = *std::max_element(lens.begin(), lens.end());
+  int nchunks = pick_chunks(maxn, (int)ptrs.size());
+  auto dptr = to_device_i64(ptrs, device);
+  auto dlen = to_device_i64(lens, device);
+  auto dout = to_device_i64(optrs, device);
+  auto df64 = to_device_i64(fv, device).to(torch::kInt32);
+  check_hip(anovos_fill_code((const int32_t *const *)dptr.data_ptr<int64_t>(),
+                             dlen.data_ptr<int64_t>(), (int)ptrs.size(),
+                             df64.data_ptr<int32_t>(), nchunks,
+                             (int32_t *const *)dout.data_ptr<int64_t>(), current_stream()),
+            "anovos_fill_code");
+  return outs;
+}
+
 torch::Tensor bracket_histograms_grouped(std::vector<torch::Tensor> cols,
                                          torch::Tensor bracket_col,
                                          torch::Tensor lo, torch::Tensor hi,
@@ -787,6 +823,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hll_registers_multi", &hll_registers_multi, "fused multi-column HLL (K4)");
   m.def("scale_columns", &scale_columns, "fused (x-a)*b scaling (K11)");
   m.def("fill_nan_columns", &fill_nan_columns, "fused NaN fill (K11)");
+  m.def("fill_code_columns", &fill_code_columns, "fused categorical null fill (K11)");
   m.def("column_moments", &column_moments, "fused per-column moments (K1/K2)",
         py::arg("cols"), py::arg("shifts") = std::vector<double>());
   m.def("column_histograms", &column_histograms, "fused per-column histograms (K3/K6)");

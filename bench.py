@@ -83,7 +83,24 @@ def make_synthetic_frame(rows: int, device, seed: int):
     return AnovosFrame(cols, dev)
 
 
-def pipeline_step(ctx, idf, source_hist, model_dir):
+import contextlib
+
+
+@contextlib.contextmanager
+def _timed(name, sections, ctx):
+    """Per-section wall timer for --sections runs (sync-bracketed so the
+    numbers are attributable; adds sync points, so only for profiling)."""
+    if sections is None:
+        yield
+        return
+    ctx.synchronize()
+    t0 = time.perf_counter()
+    yield
+    ctx.synchronize()
+    sections[name] = sections.get(name, 0.0) + (time.perf_counter() - t0) * 1000
+
+
+def pipeline_step(ctx, idf, source_hist, model_dir, sections=None):
     """One full analyzer + drift + transform pass. Returns a checksum to
     defeat dead-code elimination."""
     import pandas as pd
@@ -106,68 +123,82 @@ def pipeline_step(ctx, idf, source_hist, model_dir):
     # cardinality first: its fused K1/K2+K4 kernel computes moments AND
     # HLL registers in ONE frame read; everything below hits the moment
     # cache (the engine's stats-reuse contract)
-    card = sg.measures_of_cardinality(ctx, idf, num_cols + cat_cols, use_approx_unique_count=True)  # K4 HLL
-    chk += float(pd.to_numeric(card["unique_values"], errors="coerce").fillna(0).sum())
-    moments = stats_ops.frame_moments(idf, num_cols)  # cached by the fused pass
-    quant = hist_ops.approx_quantiles(
-        idf, num_cols, [0.01, 0.05, 0.10, 0.25, 0.50, 0.75, 0.90, 0.95, 0.99], moments=moments
-    )  # K3
-    counts = sg.measures_of_counts(ctx, idf)
-    chk += float(counts["missing_count"].sum())
-    ct = sg.measures_of_centralTendency(ctx, idf, int_cols + cat_cols)  # discrete modes (K5)
-    chk += float(pd.to_numeric(ct["mode_rows"], errors="coerce").fillna(0).sum())
-    shape = sg.measures_of_shape(ctx, idf)
-    chk += float(pd.to_numeric(shape["skewness"], errors="coerce").fillna(0).abs().sum())
-    _, nullrows = qc.nullRows_detection(ctx, idf, treatment=False)  # K10 row scan
-    chk += float(nullrows["row_count"].sum())
-    _, biased = qc.biasedness_detection(ctx, idf, int_cols + cat_cols, treatment=False, treatment_threshold=0.9)
-    odf_out, _ = qc.outlier_detection(
-        ctx, idf, num_cols, detection_side="both", treatment=True, treatment_method="value_replacement"
-    )
-    chk += float(odf_out.col(num_cols[0]).data[:8].float().nansum().item())
-    del odf_out  # free treated copies promptly (125M-row shards: ~75 GB each)
+    with _timed("cardinality_moments", sections, ctx):
+        card = sg.measures_of_cardinality(ctx, idf, num_cols + cat_cols, use_approx_unique_count=True)  # K4 HLL
+        chk += float(pd.to_numeric(card["unique_values"], errors="coerce").fillna(0).sum())
+        moments = stats_ops.frame_moments(idf, num_cols)  # cached by the fused pass
+    with _timed("quantiles", sections, ctx):
+        quant = hist_ops.approx_quantiles(
+            idf, num_cols, [0.01, 0.05, 0.10, 0.25, 0.50, 0.75, 0.90, 0.95, 0.99], moments=moments
+        )  # K3
+    with _timed("counts", sections, ctx):
+        counts = sg.measures_of_counts(ctx, idf)
+        chk += float(counts["missing_count"].sum())
+    with _timed("central_tendency", sections, ctx):
+        ct = sg.measures_of_centralTendency(ctx, idf, int_cols + cat_cols)  # discrete modes (K5)
+        chk += float(pd.to_numeric(ct["mode_rows"], errors="coerce").fillna(0).sum())
+    with _timed("shape", sections, ctx):
+        shape = sg.measures_of_shape(ctx, idf)
+        chk += float(pd.to_numeric(shape["skewness"], errors="coerce").fillna(0).abs().sum())
+    with _timed("null_rows", sections, ctx):
+        _, nullrows = qc.nullRows_detection(ctx, idf, treatment=False)  # K10 row scan
+        chk += float(nullrows["row_count"].sum())
+    with _timed("biasedness", sections, ctx):
+        _, biased = qc.biasedness_detection(ctx, idf, int_cols + cat_cols, treatment=False, treatment_threshold=0.9)
+    with _timed("outlier_detection", sections, ctx):
+        odf_out, _ = qc.outlier_detection(
+            ctx, idf, num_cols, detection_side="both", treatment=True, treatment_method="value_replacement"
+        )
+        chk += float(odf_out.col(num_cols[0]).data[:8].float().nansum().item())
+        del odf_out  # free treated copies promptly (125M-row shards: ~75 GB each)
 
     # ---- drift (PSI/JSD/HD/KS vs warmup snapshot) ----
-    binned = T.attribute_binning(ctx, idf, num_cols, bin_size=10, output_mode="append")  # K6
-    drift_vals = 0.0
-    import numpy as _np
+    with _timed("binning", sections, ctx):
+        binned = T.attribute_binning(ctx, idf, num_cols, bin_size=10, output_mode="append")  # K6
+    with _timed("drift", sections, ctx):
+        drift_vals = 0.0
+        import numpy as _np
 
-    drift_cols = num_cols[:: max(1, len(num_cols) // 50)]
-    q_freqs = dd.batched_bin_frequencies(binned, [c + "_binned" for c in drift_cols], idf.count(), max_bin=10)
-    # vectorize PSI/JSD/HD/KS over the whole (col x bin) matrix at once
-    key_union = sorted({k for c in drift_cols for k in source_hist.get(c, {})}
-                       | {k for c in drift_cols for k in q_freqs[c + "_binned"][0]}, key=dd._key_order)
-    kpos = {k: j for j, k in enumerate(key_union)}
-    K = len(key_union)
-    P = _np.full((len(drift_cols), K), 0.0001)
-    Q = _np.full((len(drift_cols), K), 0.0001)
-    for i, c in enumerate(drift_cols):
-        for k, v in source_hist.get(c, {}).items():
-            P[i, kpos[k]] = max(v, 0.0001)
-        qk, qv = q_freqs[c + "_binned"]
-        for k, v in zip(qk, qv):
-            Q[i, kpos[k]] = max(v, 0.0001)
-    M = (P + Q) / 2
-    drift_vals += float(_np.sum((P - Q) * _np.log(P / Q)))  # PSI
-    drift_vals += float((_np.sum(P * _np.log(P / M)) + _np.sum(Q * _np.log(Q / M))) / 2)  # JSD
-    drift_vals += float(_np.sum(_np.sqrt(_np.sum((_np.sqrt(P) - _np.sqrt(Q)) ** 2, axis=1) / 2)))  # HD
-    drift_vals += float(_np.sum(_np.max(_np.abs(_np.cumsum(P, axis=1) - _np.cumsum(Q, axis=1)), axis=1)))  # KS
-    chk += drift_vals
+        drift_cols = num_cols[:: max(1, len(num_cols) // 50)]
+        q_freqs = dd.batched_bin_frequencies(binned, [c + "_binned" for c in drift_cols], idf.count(), max_bin=10)
+        # vectorize PSI/JSD/HD/KS over the whole (col x bin) matrix at once
+        key_union = sorted({k for c in drift_cols for k in source_hist.get(c, {})}
+                           | {k for c in drift_cols for k in q_freqs[c + "_binned"][0]}, key=dd._key_order)
+        kpos = {k: j for j, k in enumerate(key_union)}
+        K = len(key_union)
+        P = _np.full((len(drift_cols), K), 0.0001)
+        Q = _np.full((len(drift_cols), K), 0.0001)
+        for i, c in enumerate(drift_cols):
+            for k, v in source_hist.get(c, {}).items():
+                P[i, kpos[k]] = max(v, 0.0001)
+            qk, qv = q_freqs[c + "_binned"]
+            for k, v in zip(qk, qv):
+                Q[i, kpos[k]] = max(v, 0.0001)
+        M = (P + Q) / 2
+        drift_vals += float(_np.sum((P - Q) * _np.log(P / Q)))  # PSI
+        drift_vals += float((_np.sum(P * _np.log(P / M)) + _np.sum(Q * _np.log(Q / M))) / 2)  # JSD
+        drift_vals += float(_np.sum(_np.sqrt(_np.sum((_np.sqrt(P) - _np.sqrt(Q)) ** 2, axis=1) / 2)))  # HD
+        drift_vals += float(_np.sum(_np.max(_np.abs(_np.cumsum(P, axis=1) - _np.cumsum(Q, axis=1)), axis=1)))  # KS
+        chk += drift_vals
 
     # ---- transform ----
     del binned
-    t1 = T.z_standardization(ctx, idf, num_cols[:N_NUM_CONT])  # K11
-    chk += float(t1.col(num_cols[0]).data[:8].float().nansum().item())
-    del t1
-    t2 = T.imputation_MMM(ctx, idf, method_type="median")
-    chk += float(t2.col(num_cols[1]).data[:8].float().nansum().item())
-    del t2
-    t3 = T.cat_to_num_unsupervised(ctx, idf, cat_cols[:25], method_type="label_encoding")  # K12
-    chk += float(t3.col(cat_cols[0] + "_index").data[:8].float().nansum().item()) if (cat_cols[0] + "_index") in t3.columns else float(t3.col(cat_cols[0]).data[:8].float().sum().item())
-    del t3
-    t4 = T.outlier_categories(ctx, idf, cat_cols[25:], max_category=20)
-    chk += float(t4.col(cat_cols[25]).data[:8].float().sum().item())
-    del t4
+    with _timed("z_standardization", sections, ctx):
+        t1 = T.z_standardization(ctx, idf, num_cols[:N_NUM_CONT])  # K11
+        chk += float(t1.col(num_cols[0]).data[:8].float().nansum().item())
+        del t1
+    with _timed("imputation_median", sections, ctx):
+        t2 = T.imputation_MMM(ctx, idf, method_type="median")
+        chk += float(t2.col(num_cols[1]).data[:8].float().nansum().item())
+        del t2
+    with _timed("cat_label_encoding", sections, ctx):
+        t3 = T.cat_to_num_unsupervised(ctx, idf, cat_cols[:25], method_type="label_encoding")  # K12
+        chk += float(t3.col(cat_cols[0] + "_index").data[:8].float().nansum().item()) if (cat_cols[0] + "_index") in t3.columns else float(t3.col(cat_cols[0]).data[:8].float().sum().item())
+        del t3
+    with _timed("outlier_categories", sections, ctx):
+        t4 = T.outlier_categories(ctx, idf, cat_cols[25:], max_category=20)
+        chk += float(t4.col(cat_cols[25]).data[:8].float().sum().item())
+        del t4
     return chk
 
 
@@ -176,6 +207,8 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=5)
     ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--sections", action="store_true",
+                    help="print per-section ms (adds sync points; profiling only)")
     ap.add_argument("--rows", type=int, default=0,
                     help="rows per GPU (default 125M on GPU = the 1Bx200-at-8-GPU shard, 200k on CPU)")
     args = ap.parse_args()
@@ -217,8 +250,9 @@ def main():
     dist.barrier()
     ctx.synchronize()
     t0 = time.perf_counter()
+    sections = {} if args.sections else None
     for _ in range(args.steps):
-        pipeline_step(ctx, idf, source_hist, model_dir)
+        pipeline_step(ctx, idf, source_hist, model_dir, sections=sections)
     dist.barrier()
     ctx.synchronize()
     t1 = time.perf_counter()
@@ -230,6 +264,10 @@ def main():
     total_rows = rows * world
     value = total_rows / (elapsed / args.steps)
 
+    if rank == 0 and sections is not None:
+        per = {k: round(v / args.steps, 2) for k, v in sections.items()}
+        per["SECTION_SUM"] = round(sum(per.values()), 2)
+        print(json.dumps({"sections_ms_per_step": per}))
     if rank == 0:
         print(
             json.dumps(

@@ -441,3 +441,21 @@ def test_pearson_mfma_large_offset():
     idf = AnovosFrame({"a": Column("a", "double", a), "b": Column("b", "double", b)}, device="cuda")
     m = corr.pearson_matrix(idf, ["a", "b"])
     assert abs(m[0, 1] - 0.5) < 0.02
+
+
+@requires_gpu
+def test_fill_code_columns(ext):
+    """Fused categorical null-fill matches the where() reference."""
+    g = torch.Generator(device="cpu").manual_seed(47)
+    cols, fills = [], []
+    for i in range(4):
+        n = 1_000_000 + i * 3 + 1  # odd tails exercise the scalar loop
+        c = torch.randint(0, 30, (n,), generator=g, dtype=torch.int32)
+        c[torch.rand(n, generator=g) < 0.05] = -1
+        cols.append(c.cuda().contiguous())
+        fills.append(i + 1)
+    outs = ext.fill_code_columns(cols, fills)
+    for c, f, o in zip(cols, fills, outs):
+        ref = torch.where(c == -1, torch.full_like(c, f), c)
+        assert torch.equal(o, ref)
+        assert int((o == -1).sum()) == 0

@@ -1,0 +1,109 @@
+"""Sub-phase attribution for the three host-heavy bench sections at the
+production shard (outlier_detection, imputation_MMM, cardinality) — run
+via gpurun. Mirrors the bench: analyzer sections run first so the stats
+cache is warm exactly as in pipeline_step."""
+
+import json
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, ".")
+
+from bench import make_synthetic_frame
+from anovos_amd.data_analyzer import quality_checker as qc
+from anovos_amd.data_analyzer import stats_generator as sg
+from anovos_amd.data_transformer import transformers as T
+from anovos_amd.ops import histogram as hist_ops
+from anovos_amd.ops import stats as stats_ops
+from anovos_amd.ops import groupby as groupby_ops
+from anovos_amd.ops import elementwise
+from anovos_amd.shared.context import init_context
+from anovos_amd.shared.utils import attributeType_segregation
+
+times = {}
+
+
+def sync():
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+
+
+def t(name, fn, reps=2):
+    fn()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(reps):
+        fn()
+    sync()
+    times[name] = round((time.perf_counter() - t0) / reps * 1000, 2)
+
+
+def main(rows=125_000_000):
+    ctx = init_context()
+    idf = make_synthetic_frame(rows, ctx.device, seed=99)
+    num_cols, cat_cols, _ = attributeType_segregation(idf)
+
+    # warm the cache the way the bench does (cardinality + quantiles)
+    sg.measures_of_cardinality(ctx, idf, num_cols + cat_cols, use_approx_unique_count=True)
+    moments = stats_ops.frame_moments(idf, num_cols)
+    hist_ops.approx_quantiles(idf, num_cols, [0.01, 0.05, 0.10, 0.25, 0.50, 0.75, 0.90, 0.95, 0.99], moments=moments)
+
+    # ---- outlier_detection sub-phases ----
+    from anovos_amd.data_ingest.data_sampling import data_sample
+
+    t("out.sample", lambda: data_sample(idf.select(num_cols), fraction=min(1e6 / rows, 0.5), method_type="random", seed_value=11))
+    smp = data_sample(idf.select(num_cols), fraction=min(1e6 / rows, 0.5), method_type="random", seed_value=11)
+
+    def sample_quantiles_cold():
+        for c in num_cols:
+            smp.col(c).cache.clear()
+        hist_ops.approx_quantiles(smp, num_cols, [0.05, 0.95], rel_err=0.01)
+
+    def sample_moments_cold():
+        for c in num_cols:
+            smp.col(c).cache.pop("moments", None)
+        stats_ops.frame_moments(smp, num_cols)
+
+    t("out.sample_quantiles_cold", sample_quantiles_cold)
+    t("out.sample_moments", sample_moments_cold)
+    t("out.full_quantiles_cached", lambda: hist_ops.approx_quantiles(idf, num_cols, [0.05, 0.95], rel_err=0.01))
+    t("out.whole_section", lambda: qc.outlier_detection(
+        ctx, idf, num_cols, detection_side="both", treatment=True, treatment_method="value_replacement")[1].shape)
+
+    # ---- imputation sub-phases ----
+    candidates = num_cols + cat_cols
+    t("imp.null_counts", lambda: stats_ops.null_counts(idf, candidates))
+    t("imp.median_cached", lambda: hist_ops.approx_quantiles(idf, num_cols, [0.5]))
+    t("imp.cat_counts_cached", lambda: groupby_ops.cat_value_counts(idf, cat_cols))
+    t("imp.fillnan_150", lambda: elementwise.fill_nan_columns([idf.col(c).data for c in num_cols], [0.0] * len(num_cols)))
+    def cat_where():
+        outs = []
+        for c in cat_cols:
+            d = idf.col(c).data
+            outs.append(torch.where(d == -1, torch.full_like(d, 3), d))
+        return outs
+    t("imp.cat_where_50", cat_where)
+    t("imp.whole_section", lambda: T.imputation_MMM(ctx, idf, method_type="median").count())
+
+    # ---- cardinality sub-phases ----
+    def clear_card():
+        for c in num_cols + cat_cols:
+            for k in ("moments", "cat_counts", "nulls_local", "nulls"):
+                idf.col(c).cache.pop(k, None)
+    clear_card()
+    t("card.shifts", lambda: stats_ops.compute_column_shifts([idf.col(c).data for c in num_cols]))
+    t("card.whole_cold", lambda: (clear_card(), sg.measures_of_cardinality(ctx, idf, num_cols + cat_cols, use_approx_unique_count=True))[1].shape, reps=1)
+    t("card.whole_warm", lambda: sg.measures_of_cardinality(ctx, idf, num_cols + cat_cols, use_approx_unique_count=True).shape)
+
+    # ---- binning sub-phases ----
+    t("bin.whole", lambda: T.attribute_binning(ctx, idf, num_cols, bin_size=10, output_mode="append").count())
+
+    print(json.dumps(times, indent=1))
+    with open("gpurun_out/hotsections.json", "w") as f:
+        json.dump(times, f, indent=1)
+
+
+if __name__ == "__main__":
+    main(int(sys.argv[1]) if len(sys.argv) > 1 else 125_000_000)
