@@ -26,6 +26,12 @@
 #define THREADS 256
 #define DEV_INLINE __device__ __forceinline__
 
+// native ext vectors for the streaming loops: the nontemporal builtins
+// reject HIP_vector_type; 2x unroll + nt load/store measured +5-8% over
+// plain float4 on gfx950 (tools/bwbench.hip)
+typedef int nat_i4 __attribute__((ext_vector_type(4)));
+typedef float nat_f4 __attribute__((ext_vector_type(4)));
+
 // ------------------------------------------------------------------
 // K1/K2: fused column moments
 // ------------------------------------------------------------------
@@ -407,19 +413,35 @@ __global__ __launch_bounds__(THREADS) void bucketize_float_kernel(
   };
   if (sizeof(T) == 4) {
     const int64_t nv = (e - s) / 4;
-    const float4 *xv = reinterpret_cast<const float4 *>(x + s);
-    float4 *ov = reinterpret_cast<float4 *>(out + s);
-    for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
-      float4 v = xv[i];
-      float4 r;
+    const nat_f4 *__restrict__ xv = reinterpret_cast<const nat_f4 *>(x + s);
+    nat_f4 *__restrict__ ov = reinterpret_cast<nat_f4 *>(out + s);
+    int64_t i = threadIdx.x;
+    for (; i + THREADS < nv; i += 2 * THREADS) {
+      nat_f4 v = __builtin_nontemporal_load(&xv[i]);
+      nat_f4 w = __builtin_nontemporal_load(&xv[i + THREADS]);
+      nat_f4 r, q;
+      r.x = place((double)v.x);
+      r.y = place((double)v.y);
+      r.z = place((double)v.z);
+      r.w = place((double)v.w);
+      q.x = place((double)w.x);
+      q.y = place((double)w.y);
+      q.z = place((double)w.z);
+      q.w = place((double)w.w);
+      __builtin_nontemporal_store(r, &ov[i]);
+      __builtin_nontemporal_store(q, &ov[i + THREADS]);
+    }
+    for (; i < nv; i += THREADS) {
+      nat_f4 v = xv[i];
+      nat_f4 r;
       r.x = place((double)v.x);
       r.y = place((double)v.y);
       r.z = place((double)v.z);
       r.w = place((double)v.w);
       ov[i] = r;
     }
-    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS)
-      out[i] = place((double)x[i]);
+    for (int64_t j = s + nv * 4 + threadIdx.x; j < e; j += THREADS)
+      out[j] = place((double)x[j]);
   } else {
     for (int64_t i = s + threadIdx.x; i < e; i += THREADS)
       out[i] = place((double)x[i]);
@@ -469,11 +491,7 @@ __global__ __launch_bounds__(THREADS) void lut_apply_i32_kernel(
 
 // Fused categorical null-fill (mode imputation): out = (code == -1) ?
 // fill[col] : code, one launch for every column. int4 vectorized with
-// non-temporal access (streamed once, measured +5-8% over plain float4
-// in tools/bwbench.hip).
-typedef int nat_i4 __attribute__((ext_vector_type(4)));
-typedef float nat_f4 __attribute__((ext_vector_type(4)));
-
+// non-temporal access (streamed once).
 __global__ __launch_bounds__(THREADS) void fill_code_kernel(
     const int32_t *const *cols, const int64_t *lens, const int32_t *fill,
     int nchunks, int32_t *const *outs) {
@@ -1104,8 +1122,7 @@ __global__ __launch_bounds__(THREADS) void outlier_clamp_kernel(
   const int64_t e = min(n, s + per);
 
   uint32_t nl = 0, nh = 0;
-  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
-    double v = (double)x[i];
+  auto clamp1 = [&](double v) -> double {
     if (!isnan(v)) {
       if (has_l && v < l) {
         ++nl;
@@ -1117,7 +1134,48 @@ __global__ __launch_bounds__(THREADS) void outlier_clamp_kernel(
         else if (mode == 2) v = (double)NAN;
       }
     }
-    if (mode && y) y[i] = (T)v;
+    return v;
+  };
+  if (sizeof(T) == 4) {
+    // float4 + nt streaming (the scalar loop ran at ~2.3 TB/s — half
+    // the float4 rate; this kernel touches 100 GB per bench step)
+    const nat_f4 *__restrict__ xv = reinterpret_cast<const nat_f4 *>(x + s);
+    nat_f4 *__restrict__ ov = y ? reinterpret_cast<nat_f4 *>((float *)y + s) : nullptr;
+    const int64_t nv = (e - s) / 4;
+    int64_t i = threadIdx.x;
+    for (; i + THREADS < nv; i += 2 * THREADS) {
+      nat_f4 v = __builtin_nontemporal_load(&xv[i]);
+      nat_f4 w = __builtin_nontemporal_load(&xv[i + THREADS]);
+      v.x = (float)clamp1((double)v.x);
+      v.y = (float)clamp1((double)v.y);
+      v.z = (float)clamp1((double)v.z);
+      v.w = (float)clamp1((double)v.w);
+      w.x = (float)clamp1((double)w.x);
+      w.y = (float)clamp1((double)w.y);
+      w.z = (float)clamp1((double)w.z);
+      w.w = (float)clamp1((double)w.w);
+      if (mode && y) {
+        __builtin_nontemporal_store(v, &ov[i]);
+        __builtin_nontemporal_store(w, &ov[i + THREADS]);
+      }
+    }
+    for (; i < nv; i += THREADS) {
+      nat_f4 v = xv[i];
+      v.x = (float)clamp1((double)v.x);
+      v.y = (float)clamp1((double)v.y);
+      v.z = (float)clamp1((double)v.z);
+      v.w = (float)clamp1((double)v.w);
+      if (mode && y) ov[i] = v;
+    }
+    for (int64_t j = s + nv * 4 + threadIdx.x; j < e; j += THREADS) {
+      double v = clamp1((double)x[j]);
+      if (mode && y) y[j] = (T)v;
+    }
+  } else {
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+      double v = clamp1((double)x[i]);
+      if (mode && y) y[i] = (T)v;
+    }
   }
   __shared__ uint32_t red[2][THREADS];
   red[0][threadIdx.x] = nl;

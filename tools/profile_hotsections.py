@@ -97,8 +97,54 @@ def main(rows=125_000_000):
     t("card.whole_cold", lambda: (clear_card(), sg.measures_of_cardinality(ctx, idf, num_cols + cat_cols, use_approx_unique_count=True))[1].shape, reps=1)
     t("card.whole_warm", lambda: sg.measures_of_cardinality(ctx, idf, num_cols + cat_cols, use_approx_unique_count=True).shape)
 
+    # ---- cardinality internals ----
+    from anovos_amd.ops import backend as _backend
+    from anovos_amd.ops import distinct as distinct_ops
+
+    if torch.cuda.is_available() and _backend.use_hip(idf.col(num_cols[0]).data):
+        ext = _backend.hip_ext()
+        tensors = [idf.col(c).data.contiguous() for c in num_cols]
+        shifts = stats_ops.compute_column_shifts(tensors)
+        t("card.moments_hll_kernel", lambda: ext.moments_hll(tensors, 12, shifts))
+        t("card.hll_only_kernel", lambda: ext.hll_registers_multi(tensors, 12))
+
+    def cat_exact_cold():
+        for c in cat_cols:
+            idf.col(c).cache.pop("cat_counts", None)
+        distinct_ops.exact_distinct(idf, cat_cols)
+
+    t("card.cat_exact_cold", cat_exact_cold)
+
+    def approx_cold():
+        for c in num_cols:
+            idf.col(c).cache.pop("moments", None)
+        distinct_ops.approx_distinct(idf, num_cols)
+
+    t("card.approx_distinct_cold", approx_cold)
+
+    # ---- quantiles internals ----
+    moments2 = stats_ops.frame_moments(idf, num_cols)
+    lo = torch.tensor([moments2[c].min for c in num_cols], dtype=torch.float64)
+    hi = torch.tensor([moments2[c].max for c in num_cols], dtype=torch.float64)
+    t("quant.pass1_hist", lambda: hist_ops.global_histograms([idf.col(c).data for c in num_cols], lo, hi, 2048))
+
+    def quant_cold():
+        for c in num_cols:
+            for k in list(idf.col(c).cache):
+                if isinstance(k, tuple) and k and k[0] in ("q", "hist"):
+                    idf.col(c).cache.pop(k)
+            idf.col(c).cache.pop("p1", None)
+        hist_ops.approx_quantiles(idf, num_cols, [0.01, 0.05, 0.10, 0.25, 0.50, 0.75, 0.90, 0.95, 0.99], moments=moments2)
+
+    t("quant.whole_cold", quant_cold)
+
     # ---- binning sub-phases ----
     t("bin.whole", lambda: T.attribute_binning(ctx, idf, num_cols, bin_size=10, output_mode="append").count())
+    t("out.clamp_kernel_only", lambda: _backend.hip_ext().outlier_clamp_columns(
+        [idf.col(c).data for c in num_cols],
+        torch.full((len(num_cols),), -3.0, dtype=torch.float64),
+        torch.full((len(num_cols),), 3.0, dtype=torch.float64), 1)[0].sum()
+      if torch.cuda.is_available() else None)
 
     print(json.dumps(times, indent=1))
     with open("gpurun_out/hotsections.json", "w") as f:
