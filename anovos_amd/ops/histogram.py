@@ -63,6 +63,33 @@ def _exact_quantiles(idf, cols, probs, moments, rel_err=1e-4):
     return out
 
 
+def _exact_integral_quantiles(idf, cols, probs, moments):
+    """Exact quantiles for integer-valued columns via one dense
+    integer-aligned histogram (bin width exactly 1.0): rank = ceil(p*n),
+    value = first integer whose CDF reaches the rank — the same rank
+    convention as the exact sort path (Spark GK returns data elements)."""
+    import math as _math
+
+    import numpy as np
+
+    R = [int(moments[c].max - moments[c].min) + 1 for c in cols]
+    M = max(R)
+    lo = torch.tensor([moments[c].min - 0.5 for c in cols], dtype=torch.float64)
+    hi = torch.tensor([moments[c].min - 0.5 + M for c in cols], dtype=torch.float64)
+    h = global_histograms([idf.col(c).data for c in cols], lo, hi, M).cpu().numpy()
+    out = {}
+    for i, c in enumerate(cols):
+        n = int(moments[c].n)
+        cdf = np.cumsum(h[i])
+        vals = []
+        for p in probs:
+            r = min(max(int(_math.ceil(p * n)), 1), n)
+            b = int(np.searchsorted(cdf, r, side="left"))
+            vals.append(float(moments[c].min + min(b, R[i] - 1)))
+        out[c] = vals
+    return out
+
+
 def column_histograms(
     tensors: Sequence[torch.Tensor],
     lo: torch.Tensor,
@@ -132,6 +159,30 @@ def approx_quantiles(
     if all(all(("q", p) in idf.col(c).cache for p in probs_l) for c in cols):
         return {c: [idf.col(c).cache[("q", p)] for p in probs_l] for c in cols}
     dev = idf.device
+    # integral columns with bounded range: EXACT quantiles from one
+    # dense integer-aligned histogram (the generic 2048-bin pass-1 puts
+    # each integer in a spike bin that always exceeds the rank tolerance
+    # and forces a refinement read; this path is one read, exact, and
+    # returns actual data elements like Spark's GK sketch)
+    int_cols = [
+        c for c in cols
+        if moments[c].n > 0 and moments[c].integral
+        and moments[c].min == moments[c].min
+        and 0 < (moments[c].max - moments[c].min) < 4096
+        and any(("q", p) not in idf.col(c).cache for p in probs_l)
+    ]
+    if int_cols:
+        dense = _exact_integral_quantiles(idf, int_cols, probs_l, moments)
+        for c in int_cols:
+            for j, p in enumerate(probs_l):
+                idf.col(c).cache[("q", p)] = dense[c][j]
+        rest = [c for c in cols if c not in set(int_cols)]
+        if not rest:
+            return {c: [idf.col(c).cache[("q", p)] for p in probs_l] for c in cols}
+        out = approx_quantiles(idf, rest, probs_l, nbins=nbins, refine=refine,
+                               moments=moments, rel_err=rel_err)
+        out.update(dense)
+        return {c: out[c] for c in cols}
     lo = torch.tensor([moments[c].min for c in cols], dtype=torch.float64)
     hi = torch.tensor([moments[c].max for c in cols], dtype=torch.float64)
     probs = list(probs)
