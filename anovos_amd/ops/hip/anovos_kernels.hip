@@ -485,9 +485,21 @@ __global__ __launch_bounds__(THREADS) void lut_apply_f32_kernel(
   const int64_t per = (n + nchunks - 1) / nchunks;
   const int64_t s = (int64_t)chunk * per;
   const int64_t e = min(n, s + per);
-  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
-    int c = codes[i];
-    out[i] = (c < 0) ? nanf("") : lut[c];
+  const nat_i4 *__restrict__ cv = reinterpret_cast<const nat_i4 *>(codes + s);
+  nat_f4 *__restrict__ ov = reinterpret_cast<nat_f4 *>(out + s);
+  const int64_t nv = (e - s) / 4;
+  for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
+    nat_i4 c = __builtin_nontemporal_load(&cv[i]);
+    nat_f4 r;
+    r.x = (c.x < 0) ? nanf("") : lut[c.x];
+    r.y = (c.y < 0) ? nanf("") : lut[c.y];
+    r.z = (c.z < 0) ? nanf("") : lut[c.z];
+    r.w = (c.w < 0) ? nanf("") : lut[c.w];
+    __builtin_nontemporal_store(r, &ov[i]);
+  }
+  for (int64_t j = s + nv * 4 + threadIdx.x; j < e; j += THREADS) {
+    int c = codes[j];
+    out[j] = (c < 0) ? nanf("") : lut[c];
   }
 }
 
@@ -504,9 +516,21 @@ __global__ __launch_bounds__(THREADS) void lut_apply_i32_kernel(
   const int64_t per = (n + nchunks - 1) / nchunks;
   const int64_t s = (int64_t)chunk * per;
   const int64_t e = min(n, s + per);
-  for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
-    int c = codes[i];
-    out[i] = (c < 0) ? -1 : lut[c];
+  const nat_i4 *__restrict__ cv = reinterpret_cast<const nat_i4 *>(codes + s);
+  nat_i4 *__restrict__ ov = reinterpret_cast<nat_i4 *>(out + s);
+  const int64_t nv = (e - s) / 4;
+  for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
+    nat_i4 c = __builtin_nontemporal_load(&cv[i]);
+    nat_i4 r;
+    r.x = (c.x < 0) ? -1 : lut[c.x];
+    r.y = (c.y < 0) ? -1 : lut[c.y];
+    r.z = (c.z < 0) ? -1 : lut[c.z];
+    r.w = (c.w < 0) ? -1 : lut[c.w];
+    __builtin_nontemporal_store(r, &ov[i]);
+  }
+  for (int64_t j = s + nv * 4 + threadIdx.x; j < e; j += THREADS) {
+    int c = codes[j];
+    out[j] = (c < 0) ? -1 : lut[c];
   }
 }
 
@@ -712,8 +736,11 @@ __global__ __launch_bounds__(THREADS) void row_null_kernel(
       for (int i = threadIdx.x; i < nr; i += THREADS)
         if (is_null_elem(x[i])) rc[i] += 1;
     }
-    __syncthreads();
+    // no per-column barrier: each thread touches the same rc[] slots for
+    // every column (the i -> thread mapping is column-invariant), so
+    // there is no cross-thread hazard until the final flush
   }
+  __syncthreads();
   if (ncolchunks == 1) {
     for (int i = threadIdx.x; i < nr; i += THREADS) out[s + i] += (int32_t)rc[i];
   } else {
@@ -1104,7 +1131,18 @@ __global__ __launch_bounds__(THREADS) void code_counts_multi_kernel(
   const int64_t s = (int64_t)chunk * per;
   const int64_t e = min(n, s + per);
   if (use_lds) {
-    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+    // int4 vectorized reads (the scalar loop measured ~3.3 TB/s; dword
+    // loads are 1/4 the per-instruction bytes of dwordx4)
+    const int64_t nv = (e - s) / 4;
+    const nat_i4 *__restrict__ cv = reinterpret_cast<const nat_i4 *>(codes + s);
+    for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
+      nat_i4 c = __builtin_nontemporal_load(&cv[i]);
+      atomicAdd(&cnt[(c.x >= 0 && c.x < size) ? c.x : size], 1u);
+      atomicAdd(&cnt[(c.y >= 0 && c.y < size) ? c.y : size], 1u);
+      atomicAdd(&cnt[(c.z >= 0 && c.z < size) ? c.z : size], 1u);
+      atomicAdd(&cnt[(c.w >= 0 && c.w < size) ? c.w : size], 1u);
+    }
+    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) {
       int c = codes[i];
       atomicAdd(&cnt[(c >= 0 && c < size) ? c : size], 1u);
     }
