@@ -182,23 +182,33 @@ def pipeline_step(ctx, idf, source_hist, model_dir, sections=None):
         chk += drift_vals
 
     # ---- transform ----
+    # The transforms need NO stats from the device (fills/cutoffs/LUTs
+    # all come from the cached analyzer results), so their checksums
+    # accumulate ON-DEVICE and ONE sync closes the step: the host builds
+    # frame N+1's launch while the GPU still streams frame N's kernels.
+    # `del` stays prompt — freeing is stream-ordered, no sync needed.
     del binned
+    chk_dev = torch.zeros((), dtype=torch.float32, device=idf.device)
     with _timed("z_standardization", sections, ctx):
         t1 = T.z_standardization(ctx, idf, num_cols[:N_NUM_CONT])  # K11
-        chk += float(t1.col(num_cols[0]).data[:8].float().nansum().item())
+        chk_dev += t1.col(num_cols[0]).data[:8].float().nansum()
         del t1
     with _timed("imputation_median", sections, ctx):
         t2 = T.imputation_MMM(ctx, idf, method_type="median")
-        chk += float(t2.col(num_cols[1]).data[:8].float().nansum().item())
+        chk_dev += t2.col(num_cols[1]).data[:8].float().nansum()
         del t2
     with _timed("cat_label_encoding", sections, ctx):
         t3 = T.cat_to_num_unsupervised(ctx, idf, cat_cols[:25], method_type="label_encoding")  # K12
-        chk += float(t3.col(cat_cols[0] + "_index").data[:8].float().nansum().item()) if (cat_cols[0] + "_index") in t3.columns else float(t3.col(cat_cols[0]).data[:8].float().sum().item())
+        if (cat_cols[0] + "_index") in t3.columns:
+            chk_dev += t3.col(cat_cols[0] + "_index").data[:8].float().nansum()
+        else:
+            chk_dev += t3.col(cat_cols[0]).data[:8].float().sum()
         del t3
     with _timed("outlier_categories", sections, ctx):
         t4 = T.outlier_categories(ctx, idf, cat_cols[25:], max_category=20)
-        chk += float(t4.col(cat_cols[25]).data[:8].float().sum().item())
+        chk_dev += t4.col(cat_cols[25]).data[:8].float().sum()
         del t4
+    chk += float(chk_dev.item())  # the step's ONE transform-phase sync
     return chk
 
 
