@@ -27,26 +27,55 @@ HLL_P = 12
 HLL_M = 1 << HLL_P
 
 
-def hll_registers(t: torch.Tensor) -> torch.Tensor:
-    """Local HLL registers (uint8 as int16 tensor [HLL_M]) for one numeric
-    column; NaN skipped. float32 hashes its raw bit pattern (matches the
-    HIP kernel); float64 hashes its 8 bytes."""
+def hll_registers(t: torch.Tensor, p: int = None) -> torch.Tensor:
+    """Local HLL registers (uint8 as int16 tensor [2^p]) for one numeric
+    column; NaN skipped. float32 hashes its raw 32 bits through the
+    murmur3 finalizer (bit-identical to the HIP kernel's int32 fast
+    path); float64 hashes its 8 bytes via splitmix64."""
+    p = HLL_P if p is None else p
+    m = 1 << p
     if t.is_cuda and backend.use_hip(t):
         ext = backend.hip_ext()
-        return ext.hll_registers(t.contiguous(), HLL_P)
+        return ext.hll_registers(t.contiguous(), p)
     v = t[~torch.isnan(t)]
     if v.dtype == torch.float32:
         x = v.view(torch.int32).to(torch.int64) & 0xFFFFFFFF
+        h = _fmix32(x)
+        idx = h >> (32 - p)
+        rem = (h << p) & 0xFFFFFFFF
+        rho = _clz32(rem) + 1
+        rho = torch.clamp(rho, max=32 - p + 1)
     else:
         x = v.to(torch.float64).view(torch.int64)
-    h = _mix64(x)
-    idx = (h >> (64 - HLL_P)) & (HLL_M - 1)
-    rem = h << HLL_P  # wrapping
-    rho = _clz64(rem) + 1
-    rho = torch.clamp(rho, max=64 - HLL_P + 1)
-    regs = torch.zeros(HLL_M, dtype=torch.int64, device=t.device)
+        h = _mix64(x)
+        idx = (h >> (64 - p)) & (m - 1)
+        rem = h << p  # wrapping
+        rho = _clz64(rem) + 1
+        rho = torch.clamp(rho, max=64 - p + 1)
+    regs = torch.zeros(m, dtype=torch.int64, device=t.device)
     regs.scatter_reduce_(0, idx, rho, reduce="amax")
     return regs.to(torch.int16)
+
+
+_M32 = 0xFFFFFFFF
+
+
+def _fmix32(x: torch.Tensor) -> torch.Tensor:
+    """murmur3 32-bit finalizer on int64 tensors holding uint32 values
+    (int64 mul wraps; masking keeps the low 32 bits, which is exactly
+    the uint32 product)."""
+    x = (x ^ (x >> 16)) & _M32
+    x = (x * 0x85EBCA6B) & _M32
+    x = (x ^ (x >> 13)) & _M32
+    x = (x * 0xC2B2AE35) & _M32
+    x = (x ^ (x >> 16)) & _M32
+    return x
+
+
+def _clz32(x: torch.Tensor) -> torch.Tensor:
+    """Count leading zeros of the low 32 bits (values in [0, 2^32))."""
+    clz = 31 - torch.floor(torch.log2(x.to(torch.float64) + 0.5)).to(torch.int64)
+    return torch.where(x == 0, torch.full_like(clz, 32), clz.clamp(0, 31))
 
 
 def _clz64(x: torch.Tensor) -> torch.Tensor:
