@@ -332,3 +332,70 @@ def test_dist_exact_dedup_across_ranks(tmp_path):
     # global distinct values: 0..149 — every one kept exactly once
     assert res["all"] == [float(v) for v in range(150)]
     assert res["unique_rows"] == 150.0
+
+
+def _stable_worker(rank, world, port, out_path):
+    os.environ.update(
+        {"RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": str(world),
+         "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port)}
+    )
+    import torch.distributed as td
+
+    from anovos_amd.core import dist
+    from anovos_amd.core.frame import AnovosFrame, Column
+    from anovos_amd.ops import histogram as hist_ops
+    from anovos_amd.ops import stats as stats_ops
+    from anovos_amd.shared.context import init_context
+
+    dist.init_from_env(timeout_s=120)
+    init_context("cpu")
+    rng = np.random.default_rng(123)
+    n = 300_000  # above the exact-sort threshold -> sketch / dense-int paths
+    off_full = rng.normal(0, 1, n) + 1e9
+    k_full = rng.integers(0, 43, n).astype(np.float32)
+    half = n // 2
+    sl = slice(rank * half, (rank + 1) * half)
+    idf = AnovosFrame(
+        {"off": Column("off", "double", torch.tensor(off_full[sl])),
+         "k": Column("k", "float", torch.tensor(k_full[sl]))},
+        device="cpu",
+    )
+    m = stats_ops.frame_moments(idf, ["off", "k"])["off"]
+    q = hist_ops.approx_quantiles(idf, ["k", "off"], [0.25, 0.5, 0.9])
+    if rank == 0:
+        with open(out_path, "w") as f:
+            json.dump({"stddev": m.stddev, "skew": m.skewness, "kurt": m.kurtosis,
+                       "mean": m.mean, "k_q": q["k"], "off_q": q["off"]}, f)
+    td.barrier()
+    td.destroy_process_group()
+
+
+def test_dist_stable_moments_and_integral_quantiles():
+    """2-rank gloo: the shared accumulation pivot keeps skew/kurt correct
+    at |mean| >> stddev, and integral-column quantiles stay EXACT across
+    the cross-rank dense-histogram merge."""
+    port = _free_port()
+    out = tempfile.NamedTemporaryFile(suffix=".json", delete=False).name
+    mp_ctx = mp.get_context("spawn")
+    procs = [mp_ctx.Process(target=_stable_worker, args=(r, 2, port, out)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0, f"worker failed: exit {p.exitcode}"
+    with open(out) as f:
+        res = json.load(f)
+    rng = np.random.default_rng(123)
+    n = 300_000
+    off_full = rng.normal(0, 1, n) + 1e9
+    k_full = rng.integers(0, 43, n).astype(np.float32)
+    assert res["mean"] == pytest.approx(float(off_full.mean()), rel=1e-12)
+    assert res["stddev"] == pytest.approx(float(off_full.std(ddof=1)), rel=1e-3)
+    assert abs(res["skew"]) < 0.05 and abs(res["kurt"]) < 0.1
+    import math
+
+    ks = np.sort(k_full)
+    exact_k = [float(ks[min(max(math.ceil(p * n), 1), n) - 1]) for p in (0.25, 0.5, 0.9)]
+    assert res["k_q"] == exact_k  # dense integral path is exact
+    for got, p in zip(res["off_q"], (0.25, 0.5, 0.9)):
+        assert got == pytest.approx(float(np.quantile(off_full, p)), abs=0.05)
