@@ -1,7 +1,9 @@
 """Fused multi-column statistics (kernel K1/K2 of SURVEY.md §2.10).
 
 One pass over all numeric columns produces, per column, the fp64 partial
-vector [n_valid, sum, sum2, sum3, sum4, min, max, zero_count]. The
+vector [n_valid, s1, s2, s3, s4, min, max, zero_count, n_frac] where
+s1..s4 are power sums about a per-column pivot sampled from the data
+(numerically stable skew/kurt at |mean| >> stddev). The
 reference computes these with one Spark job per column per statistic
 (stats_generator.py:485-494 — the dominant anti-pattern); here every
 column is covered by a single kernel launch and ONE batched RCCL
@@ -94,7 +96,7 @@ def column_moments_local(tensors: Sequence[torch.Tensor], shifts: Sequence[float
 
 
 def merge_moments_global(local: torch.Tensor) -> torch.Tensor:
-    """RCCL merge of [ncols, 8] partials: sum for n/s1..s4/zeros, min/max
+    """RCCL merge of [ncols, NSTAT] partials: sum for n/s1..s4/zeros/n_frac, min/max
     for the extrema — batched into two fused all-reduces."""
     if not dist.is_dist():
         return local.cpu()
