@@ -19,8 +19,40 @@ from anovos_amd.data_report.report_generation import _tbl, render_report
 from anovos_amd.shared.utils import ends_with
 
 
-def _remove_u_score(s: str) -> str:
-    return " ".join(w.title() for w in s.split("_"))
+def remove_u_score(col: str) -> str:
+    """`_`-separated name -> title-cased words (reference
+    basic_report_generation.py:236)."""
+    return " ".join(w.title() for w in str(col).split("_"))
+
+
+_remove_u_score = remove_u_score
+
+
+def stats_args(path, func):
+    """kwargs that wire pre-saved analyzer CSVs into a quality-checker
+    call (reference basic_report_generation.py:55-92): stats_unique /
+    stats_mode / stats_missing map to the saved cardinality /
+    centralTendency / counts CSVs under `path` so detectors reuse the
+    already-computed statistics instead of re-scanning the frame."""
+    mainfunc_to_args = {
+        "biasedness_detection": ["stats_mode"],
+        "IDness_detection": ["stats_unique"],
+        "nullColumns_detection": ["stats_unique", "stats_mode", "stats_missing"],
+        "variable_clustering": ["stats_mode"],
+    }
+    args_to_statsfunc = {
+        "stats_unique": "measures_of_cardinality",
+        "stats_mode": "measures_of_centralTendency",
+        "stats_missing": "measures_of_counts",
+    }
+    out = {}
+    for arg in mainfunc_to_args.get(func, []):
+        out[arg] = {
+            "file_path": ends_with(path) + args_to_statsfunc[arg] + ".csv",
+            "file_type": "csv",
+            "file_configs": {"header": True, "inferSchema": True},
+        }
+    return out
 
 
 def anovos_basic_report(ctx, idf, id_col="", label_col="", event_label="",

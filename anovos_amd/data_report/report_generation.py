@@ -29,6 +29,13 @@ import plotly.io as pio
 
 from anovos_amd.shared.utils import ends_with
 
+
+def remove_u_score(col: str) -> str:
+    """`_`-separated name -> title-cased words for display labels
+    (reference report_generation.py:78)."""
+    return " ".join(w.title() for w in str(col).split("_"))
+
+
 SG_tabs = [
     "measures_of_counts", "measures_of_centralTendency", "measures_of_cardinality",
     "measures_of_percentiles", "measures_of_dispersion", "measures_of_shape", "global_summary",
