@@ -66,6 +66,40 @@ _SEED_CORPUS = [
 ]
 
 
+def detect_model_path() -> str:
+    """Local sentence-transformers model path, if one was pre-downloaded
+    (reference featrec_init.py:11-33). The engine itself embeds with the
+    built-in TF-IDF embedder and never requires this model; the path is
+    honored when a user drops the reference's `all-mpnet-base-v2` cache
+    in the conventional location."""
+    transformers_path = os.getenv("SENTENCE_TRANSFORMERS_HOME")
+    if transformers_path is None:
+        torch_home = os.path.expanduser(
+            os.getenv("TORCH_HOME", os.path.join(os.getenv("XDG_CACHE_HOME", "~/.cache"), "torch"))
+        )
+        transformers_path = os.path.join(torch_home, "sentence_transformers")
+    return os.path.join(transformers_path, "sentence-transformers_all-mpnet-base-v2")
+
+
+def model_download():
+    """Reference featrec_init.py:36 downloads `all-mpnet-base-v2`. This
+    deployment has no package for it and typically no egress; the
+    recommender runs on the built-in TF-IDF embedder instead, so the
+    download is optional. Raises with that explanation when the
+    sentence-transformers package is unavailable."""
+    try:
+        from sentence_transformers import SentenceTransformer  # noqa: F401
+    except ImportError as e:
+        raise RuntimeError(
+            "sentence-transformers is not installed in this environment; "
+            "the feature recommender uses its built-in TF-IDF embedder "
+            "(TfidfEmbedder/semantic_search) and does not need the download."
+        ) from e
+    print("Starting the Semantic Model download")
+    SentenceTransformer("all-mpnet-base-v2")
+    print("Model downloading finished")
+
+
 def set_corpus_path(path: Optional[str]):
     """Point the recommender at a custom corpus CSV; None resets to the
     built-in seed corpus."""
