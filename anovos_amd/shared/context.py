@@ -57,3 +57,23 @@ def init_context(device=None) -> AnovosContext:
 
 def get_context() -> AnovosContext:
     return init_context()
+
+
+def init_spark(*args, **kwargs):
+    """Migration shim for the reference's shared/spark.py:26 entry point.
+
+    This engine has no Spark: compute runs on the MI355X column store
+    (one process per GPU, torch.distributed over RCCL). Returns the
+    triple shape callers destructure — (ctx, None, None) — so
+    `spark, sc, sqlContext = init_spark(...)` keeps working, with `ctx`
+    standing in for the session everywhere the reference passed `spark`.
+    All engine APIs accept this context as their first argument."""
+    import warnings
+
+    warnings.warn(
+        "init_spark(): no Spark in anovos_amd — returning (AnovosContext, None, None); "
+        "pass the context where the reference passed `spark`.",
+        stacklevel=2,
+    )
+    ctx = init_context()
+    return ctx, None, None
