@@ -346,6 +346,10 @@ def outlier_detection(
         _cached_full = all(
             ("q", p) in idf.col(c).cache for c in list_of_cols for p in _need_q
         ) and ("stdev" not in methodologies or all("moments" in idf.col(c).cache for c in list_of_cols))
+        if dist.is_dist():
+            # the branch chooses between different collective sequences —
+            # take it only when EVERY rank has the cache (all-reduce min)
+            _cached_full = bool(dist.all_reduce_scalar(1.0 if _cached_full else 0.0, "min"))
         if _cached_full:
             idf_sample = idf
         elif idf_count > sample_size:
