@@ -91,6 +91,18 @@ def discrete_modes(idf, cols: List[str]) -> Dict[str, Tuple[Optional[float], int
     if dense:
         import numpy as np
 
+        # cached dense per-integer counts (filled by the exact integral
+        # quantile path — same single frame read serves both)
+        cached = [c for c in dense if ("inthist",) in idf.col(c).cache]
+        for c in cached:
+            h, mn = idf.col(c).cache[("inthist",)]
+            idx = int(np.argmax(h))
+            cnt = int(h[idx])
+            out[c] = (float(mn + idx), cnt) if cnt > 0 else (None, 0)
+        dense = [c for c in dense if c not in set(cached)]
+    if dense:
+        import numpy as np
+
         R = [int(moments[c].max - moments[c].min) + 1 for c in dense]
         M = max(R)
         tensors = [idf.col(c).data for c in dense]

@@ -80,6 +80,9 @@ def _exact_integral_quantiles(idf, cols, probs, moments):
     out = {}
     for i, c in enumerate(cols):
         n = int(moments[c].n)
+        # the dense per-integer counts serve discrete_modes too — one
+        # frame read covers exact quantiles AND exact modes
+        idf.col(c).cache[("inthist",)] = (h[i][: R[i]].copy(), float(moments[c].min))
         cdf = np.cumsum(h[i])
         vals = []
         for p in probs:
