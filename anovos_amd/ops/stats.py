@@ -9,7 +9,7 @@ reference computes these with one Spark job per column per statistic
 column is covered by a single kernel launch and ONE batched RCCL
 all-reduce merges the 8-double-per-column partials across ranks.
 
-HIP path: ops/hip/stats_kernels.hip (partials per (col, chunk) block,
+HIP path: ops/hip/anovos_kernels.hip (partials per (col, chunk) block,
 deterministic two-kernel reduce, fp64 accumulators).
 Torch path: reference implementation, also the CPU backend.
 """
