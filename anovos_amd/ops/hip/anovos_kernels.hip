@@ -708,9 +708,11 @@ __global__ __launch_bounds__(THREADS) void hll_kernel(
 // K10: fused row-wise NaN count over numeric columns
 // ------------------------------------------------------------------
 // Grid: (row_chunk, col_chunk). Each block owns ROWS_PB rows x <=COLS_PB
-// columns; per-row counters live in LDS as uint16, loads are float4, and
+// columns; in the vectorized path each thread owns fixed row-quads and
+// counts in VGPRs (no LDS traffic in the loop), loads are float4 nt, and
 // the block adds its partial counts to global with one atomic per row
-// only when col chunking splits a row.
+// only when col chunking splits a row. LDS uint16 counters remain for
+// the unaligned-tail fallback.
 
 #define ROWS_PB 8192
 #define COLS_PB 64
@@ -732,32 +734,60 @@ __global__ __launch_bounds__(THREADS) void row_null_kernel(
   const int nr = (int)(e - s);
   const int c0 = colchunk * COLS_PB;
   const int c1 = min(ncols, c0 + COLS_PB);
+  const bool vec_ok = (sizeof(T) == 4) && (nr % 4 == 0) && ((s & 3) == 0);
+  if (vec_ok) {
+    // register accumulation: each thread owns QPT fixed row-quads and
+    // keeps their counters in VGPRs — the LDS read-modify-write per
+    // float4 bounded the old loop at ~3.9 TB/s
+    constexpr int QPT = ROWS_PB / 4 / THREADS;  // quads per thread
+    ushort4 acc[QPT];
+#pragma unroll
+    for (int q = 0; q < QPT; ++q) acc[q] = ushort4{0, 0, 0, 0};
+    const int nq = nr / 4;
+    typedef T T4 __attribute__((ext_vector_type(4)));
+    for (int c = c0; c < c1; ++c) {
+      const T4 *__restrict__ xv = reinterpret_cast<const T4 *>(cols[c] + s);
+#pragma unroll
+      for (int q = 0; q < QPT; ++q) {
+        const int i = threadIdx.x + q * THREADS;
+        if (i < nq) {
+          T4 v = __builtin_nontemporal_load(&xv[i]);
+          acc[q].x += is_null_elem((T)v.x);
+          acc[q].y += is_null_elem((T)v.y);
+          acc[q].z += is_null_elem((T)v.z);
+          acc[q].w += is_null_elem((T)v.w);
+        }
+      }
+    }
+#pragma unroll
+    for (int q = 0; q < QPT; ++q) {
+      const int i = threadIdx.x + q * THREADS;
+      if (i < nq) {
+        const int64_t r = s + 4LL * i;
+        if (ncolchunks == 1) {
+          // out accumulates across per-dtype launches: read-modify-write
+          out[r] += (int32_t)acc[q].x;
+          out[r + 1] += (int32_t)acc[q].y;
+          out[r + 2] += (int32_t)acc[q].z;
+          out[r + 3] += (int32_t)acc[q].w;
+        } else {
+          if (acc[q].x) atomicAdd(&out[r], (int32_t)acc[q].x);
+          if (acc[q].y) atomicAdd(&out[r + 1], (int32_t)acc[q].y);
+          if (acc[q].z) atomicAdd(&out[r + 2], (int32_t)acc[q].z);
+          if (acc[q].w) atomicAdd(&out[r + 3], (int32_t)acc[q].w);
+        }
+      }
+    }
+    return;
+  }
   for (int i = threadIdx.x; i < ROWS_PB / 4; i += THREADS)
     ((uint64_t *)rc)[i] = 0;  // 4 counters per store
   __syncthreads();
-  const bool vec_ok = (sizeof(T) == 4) && (nr % 4 == 0) && ((s & 3) == 0);
   for (int c = c0; c < c1; ++c) {
     const T *__restrict__ x = cols[c] + s;
-    if (vec_ok) {
-      typedef T T4 __attribute__((ext_vector_type(4)));
-      const T4 *xv = reinterpret_cast<const T4 *>(x);
-      for (int i = threadIdx.x; i < nr / 4; i += THREADS) {
-        T4 v = xv[i];
-        // 4 consecutive rows per lane: one 8-B LDS read+write
-        ushort4 cnt = ((ushort4 *)rc)[i];
-        cnt.x += is_null_elem((T)v.x);
-        cnt.y += is_null_elem((T)v.y);
-        cnt.z += is_null_elem((T)v.z);
-        cnt.w += is_null_elem((T)v.w);
-        ((ushort4 *)rc)[i] = cnt;
-      }
-    } else {
-      for (int i = threadIdx.x; i < nr; i += THREADS)
-        if (is_null_elem(x[i])) rc[i] += 1;
-    }
-    // no per-column barrier: each thread touches the same rc[] slots for
-    // every column (the i -> thread mapping is column-invariant), so
-    // there is no cross-thread hazard until the final flush
+    for (int i = threadIdx.x; i < nr; i += THREADS)
+      if (is_null_elem(x[i])) rc[i] += 1;
+    // no per-column barrier: the i -> thread mapping is column-invariant
   }
   __syncthreads();
   if (ncolchunks == 1) {
