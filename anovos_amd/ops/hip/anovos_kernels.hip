@@ -684,19 +684,19 @@ __global__ __launch_bounds__(THREADS) void hll_kernel(
     for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
       float4 v = xv[i];
       int idx, rho;
-      if (hll_hash((T)v.x, p, idx, rho)) atomicMax(&sreg[idx], rho);
-      if (hll_hash((T)v.y, p, idx, rho)) atomicMax(&sreg[idx], rho);
-      if (hll_hash((T)v.z, p, idx, rho)) atomicMax(&sreg[idx], rho);
-      if (hll_hash((T)v.w, p, idx, rho)) atomicMax(&sreg[idx], rho);
+      if (hll_hash((T)v.x, p, idx, rho)) { if (sreg[idx] < rho) atomicMax(&sreg[idx], rho); }
+      if (hll_hash((T)v.y, p, idx, rho)) { if (sreg[idx] < rho) atomicMax(&sreg[idx], rho); }
+      if (hll_hash((T)v.z, p, idx, rho)) { if (sreg[idx] < rho) atomicMax(&sreg[idx], rho); }
+      if (hll_hash((T)v.w, p, idx, rho)) { if (sreg[idx] < rho) atomicMax(&sreg[idx], rho); }
     }
     for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) {
       int idx, rho;
-      if (hll_hash(x[i], p, idx, rho)) atomicMax(&sreg[idx], rho);
+      if (hll_hash(x[i], p, idx, rho)) { if (sreg[idx] < rho) atomicMax(&sreg[idx], rho); }
     }
   } else {
     for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
       int idx, rho;
-      if (hll_hash(x[i], p, idx, rho)) atomicMax(&sreg[idx], rho);
+      if (hll_hash(x[i], p, idx, rho)) { if (sreg[idx] < rho) atomicMax(&sreg[idx], rho); }
     }
   }
   __syncthreads();
@@ -823,19 +823,19 @@ __global__ __launch_bounds__(THREADS) void hll_multi_kernel(
     for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
       float4 v = xv[i];
       int idx, rho;
-      if (hll_hash((T)v.x, p, idx, rho)) atomicMax(&sreg[idx], rho);
-      if (hll_hash((T)v.y, p, idx, rho)) atomicMax(&sreg[idx], rho);
-      if (hll_hash((T)v.z, p, idx, rho)) atomicMax(&sreg[idx], rho);
-      if (hll_hash((T)v.w, p, idx, rho)) atomicMax(&sreg[idx], rho);
+      if (hll_hash((T)v.x, p, idx, rho)) { if (sreg[idx] < rho) atomicMax(&sreg[idx], rho); }
+      if (hll_hash((T)v.y, p, idx, rho)) { if (sreg[idx] < rho) atomicMax(&sreg[idx], rho); }
+      if (hll_hash((T)v.z, p, idx, rho)) { if (sreg[idx] < rho) atomicMax(&sreg[idx], rho); }
+      if (hll_hash((T)v.w, p, idx, rho)) { if (sreg[idx] < rho) atomicMax(&sreg[idx], rho); }
     }
     for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) {
       int idx, rho;
-      if (hll_hash(x[i], p, idx, rho)) atomicMax(&sreg[idx], rho);
+      if (hll_hash(x[i], p, idx, rho)) { if (sreg[idx] < rho) atomicMax(&sreg[idx], rho); }
     }
   } else {
     for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
       int idx, rho;
-      if (hll_hash(x[i], p, idx, rho)) atomicMax(&sreg[idx], rho);
+      if (hll_hash(x[i], p, idx, rho)) { if (sreg[idx] < rho) atomicMax(&sreg[idx], rho); }
     }
   }
   __syncthreads();
@@ -1361,7 +1361,7 @@ __global__ __launch_bounds__(THREADS) void moments_hll_kernel(
   auto body = [&](T v) {
     mom_add(a, (double)v, shift);
     int idx, rho;
-    if (hll_hash(v, p, idx, rho)) atomicMax(&sreg[idx], rho);
+    if (hll_hash(v, p, idx, rho)) { if (sreg[idx] < rho) atomicMax(&sreg[idx], rho); }
   };
   if (sizeof(T) == 4) {
     const int64_t nv = (e - s) / 4;
