@@ -402,7 +402,22 @@ __global__ __launch_bounds__(THREADS) void bucketize_float_kernel(
   const int chunk = blockIdx.x % nchunks;
   const int ncut = cutoff_len[col];
   const double *src = &cutflat[cutoff_off[col]];
-  for (int i = threadIdx.x; i < ncut; i += THREADS) cuts[i] = src[i];
+  // f32 columns scan f32 cuts with f64-EXACT placement: for an f32
+  // value v, (cut < (double)v) <=> (cutf < v) where cutf is (float)cut
+  // adjusted DOWN to the largest f32 whose double is <= ... i.e. if
+  // (double)(float)cut > cut, step one ulp down. Halves the LDS reads
+  // and removes the per-element f64 convert/compare chain.
+  float *cutsf = reinterpret_cast<float *>(cuts);
+  if (sizeof(T) == 4) {
+    for (int i = threadIdx.x; i < ncut; i += THREADS) {
+      double c = src[i];
+      float cf = (float)c;
+      if ((double)cf > c) cf = nextafterf(cf, -(float)INFINITY);
+      cutsf[i] = cf;
+    }
+  } else {
+    for (int i = threadIdx.x; i < ncut; i += THREADS) cuts[i] = src[i];
+  }
   __syncthreads();
   const T *__restrict__ x = cols[col];
   float *__restrict__ out = outs[col];
@@ -432,6 +447,24 @@ __global__ __launch_bounds__(THREADS) void bucketize_float_kernel(
     }
     return (float)(lo + 1);
   };
+  auto placef = [&](float v) -> float {
+    if (isnan(v)) return nanf("");
+    int lo;
+    if (ncut <= 32) {
+      lo = 0;
+      for (int j = 0; j < ncut; ++j) lo += (cutsf[j] < v) ? 1 : 0;
+    } else {
+      int len = ncut;
+      lo = 0;
+      while (len > 0) {
+        int half = len >> 1;
+        int mid = lo + half;
+        lo = (cutsf[mid] < v) ? (mid + 1) : lo;
+        len = (cutsf[mid] < v) ? (len - half - 1) : half;
+      }
+    }
+    return (float)(lo + 1);
+  };
   if (sizeof(T) == 4) {
     const int64_t nv = (e - s) / 4;
     const nat_f4 *__restrict__ xv = reinterpret_cast<const nat_f4 *>(x + s);
@@ -441,28 +474,28 @@ __global__ __launch_bounds__(THREADS) void bucketize_float_kernel(
       nat_f4 v = __builtin_nontemporal_load(&xv[i]);
       nat_f4 w = __builtin_nontemporal_load(&xv[i + THREADS]);
       nat_f4 r, q;
-      r.x = place((double)v.x);
-      r.y = place((double)v.y);
-      r.z = place((double)v.z);
-      r.w = place((double)v.w);
-      q.x = place((double)w.x);
-      q.y = place((double)w.y);
-      q.z = place((double)w.z);
-      q.w = place((double)w.w);
+      r.x = placef(v.x);
+      r.y = placef(v.y);
+      r.z = placef(v.z);
+      r.w = placef(v.w);
+      q.x = placef(w.x);
+      q.y = placef(w.y);
+      q.z = placef(w.z);
+      q.w = placef(w.w);
       __builtin_nontemporal_store(r, &ov[i]);
       __builtin_nontemporal_store(q, &ov[i + THREADS]);
     }
     for (; i < nv; i += THREADS) {
       nat_f4 v = xv[i];
       nat_f4 r;
-      r.x = place((double)v.x);
-      r.y = place((double)v.y);
-      r.z = place((double)v.z);
-      r.w = place((double)v.w);
+      r.x = placef(v.x);
+      r.y = placef(v.y);
+      r.z = placef(v.z);
+      r.w = placef(v.w);
       ov[i] = r;
     }
     for (int64_t j = s + nv * 4 + threadIdx.x; j < e; j += THREADS)
-      out[j] = place((double)x[j]);
+      out[j] = placef((float)x[j]);
   } else {
     for (int64_t i = s + threadIdx.x; i < e; i += THREADS)
       out[i] = place((double)x[i]);
@@ -1244,8 +1277,25 @@ __global__ __launch_bounds__(THREADS) void outlier_clamp_kernel(
     return v;
   };
   if (sizeof(T) == 4) {
-    // float4 + nt streaming (the scalar loop ran at ~2.3 TB/s — half
-    // the float4 rate; this kernel touches 100 GB per bench step)
+    // f32 fast path with f64-EXACT semantics: for an f32 value v,
+    // (double)v < l  <=>  v < lf  where lf is the smallest f32 > l (or
+    // l itself when representable); likewise v > h <=> v > hf with hf
+    // the largest-f32-below adjustment. This drops the per-element
+    // f64 convert/compare chain that ALU-bound the kernel at ~3 TB/s.
+    float lf = (float)l;
+    if ((double)lf < l) lf = nextafterf(lf, (float)INFINITY);
+    float hf = (float)h;
+    if ((double)hf > h) hf = nextafterf(hf, -(float)INFINITY);
+    const float lrepf = (mode == 2) ? nanf("") : (float)l;
+    const float hrepf = (mode == 2) ? nanf("") : (float)h;
+    auto clamp1f = [&](float v) -> float {
+      const bool low = has_l && (v < lf);
+      const bool high = has_h && (v > hf);
+      nl += low ? 1u : 0u;
+      nh += high ? 1u : 0u;
+      if (mode) v = low ? lrepf : (high ? hrepf : v);
+      return v;
+    };
     const nat_f4 *__restrict__ xv = reinterpret_cast<const nat_f4 *>(x + s);
     nat_f4 *__restrict__ ov = y ? reinterpret_cast<nat_f4 *>((float *)y + s) : nullptr;
     const int64_t nv = (e - s) / 4;
@@ -1253,14 +1303,14 @@ __global__ __launch_bounds__(THREADS) void outlier_clamp_kernel(
     for (; i + THREADS < nv; i += 2 * THREADS) {
       nat_f4 v = __builtin_nontemporal_load(&xv[i]);
       nat_f4 w = __builtin_nontemporal_load(&xv[i + THREADS]);
-      v.x = (float)clamp1((double)v.x);
-      v.y = (float)clamp1((double)v.y);
-      v.z = (float)clamp1((double)v.z);
-      v.w = (float)clamp1((double)v.w);
-      w.x = (float)clamp1((double)w.x);
-      w.y = (float)clamp1((double)w.y);
-      w.z = (float)clamp1((double)w.z);
-      w.w = (float)clamp1((double)w.w);
+      v.x = clamp1f(v.x);
+      v.y = clamp1f(v.y);
+      v.z = clamp1f(v.z);
+      v.w = clamp1f(v.w);
+      w.x = clamp1f(w.x);
+      w.y = clamp1f(w.y);
+      w.z = clamp1f(w.z);
+      w.w = clamp1f(w.w);
       if (mode && y) {
         __builtin_nontemporal_store(v, &ov[i]);
         __builtin_nontemporal_store(w, &ov[i + THREADS]);
@@ -1268,14 +1318,14 @@ __global__ __launch_bounds__(THREADS) void outlier_clamp_kernel(
     }
     for (; i < nv; i += THREADS) {
       nat_f4 v = xv[i];
-      v.x = (float)clamp1((double)v.x);
-      v.y = (float)clamp1((double)v.y);
-      v.z = (float)clamp1((double)v.z);
-      v.w = (float)clamp1((double)v.w);
+      v.x = clamp1f(v.x);
+      v.y = clamp1f(v.y);
+      v.z = clamp1f(v.z);
+      v.w = clamp1f(v.w);
       if (mode && y) ov[i] = v;
     }
     for (int64_t j = s + nv * 4 + threadIdx.x; j < e; j += THREADS) {
-      double v = clamp1((double)x[j]);
+      float v = clamp1f((float)x[j]);
       if (mode && y) y[j] = (T)v;
     }
   } else {
