@@ -39,6 +39,23 @@ _RE_EPOCH10 = re.compile(r"^\d{10}$")
 _RE_EPOCH13 = re.compile(r"^\d{13}$")
 _RE_NONDATE = re.compile(r"^[a-zA-Z]+$|@|://")  # words, emails, urls never parse
 
+# label -> regex map for candidate screening, same contract as the
+# reference's REGEX_PATTERNS_PARSERS table (ts_auto_detection.py:151):
+# a column qualifies as a timestamp candidate when its sampled values
+# match one of these shapes before the strptime/dateutil parse attempts.
+REGEX_PATTERNS_PARSERS = {
+    "epoch_sec": r"^\d{10}$",
+    "epoch_ms": r"^\d{13}$",
+    "yyyy": r"^\d{4}$",
+    "yyyymm": r"^\d{6}$",
+    "yyyymmdd": r"^\d{8}$",
+    "iso_date": r"^\d{4}-\d{2}-\d{2}$",
+    "iso_datetime": r"^\d{4}-\d{2}-\d{2}[T ]\d{2}:\d{2}(:\d{2})?$",
+    "slash_date": r"^\d{4}/\d{2}/\d{2}$|^\d{2}/\d{2}/\d{4}$",
+    "dash_dmy": r"^\d{2}-\d{2}-\d{4}$",
+    "month_name": r"^\d{1,2}[- ][A-Za-z]{3}[- ]\d{4}$|^[A-Za-z]{3} \d{1,2}, \d{4}$",
+}
+
 
 def _parse_one(s: str):
     """Parse a single string to epoch-us, or None."""
