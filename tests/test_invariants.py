@@ -145,3 +145,61 @@ def test_correlation_large_offset(ctx):
     idf = AnovosFrame({"a": Column("a", "double", a), "b": Column("b", "double", b)}, device="cpu")
     assert corr.pearson_matrix(idf, ["a", "b"])[0, 1] == pytest.approx(0.5, abs=0.02)
     assert corr.covariance_matrix(idf, ["a", "b"])[0, 1] == pytest.approx(0.5, abs=0.02)
+
+
+@pytest.mark.parametrize("seed", [21, 22])
+def test_binning_invariants(ctx, seed):
+    """Equal-range bin labels live in 1..bin_size (NaN for null); equal-
+    frequency bins hold roughly n/bin_size rows each at the sketch's
+    rank tolerance."""
+    from anovos_amd.data_transformer import transformers as T
+
+    idf = random_frame(seed, n=300_000)
+    for method in ("equal_range", "equal_frequency"):
+        odf = T.attribute_binning(ctx, idf, ["a", "b"], method_type=method,
+                                  bin_size=10, output_mode="append")
+        for c in ("a_binned", "b_binned"):
+            v = odf.col(c).data
+            vv = v[~torch.isnan(v)]
+            assert float(vv.min()) >= 1 and float(vv.max()) <= 10
+            if method == "equal_frequency":
+                counts = torch.bincount(vv.long(), minlength=11)[1:]
+                n = int(vv.numel())
+                # each bucket within ~3x rank tolerance of n/10
+                assert int(counts.max()) < n / 10 + 3 * 0.01 * n + 10
+
+
+@pytest.mark.parametrize("seed", [23])
+def test_z_standardization_roundtrip(ctx, seed):
+    """Standardized columns have mean~0/sd~1 and invert back to the
+    original values through the saved mean/sd."""
+    from anovos_amd.data_analyzer import stats_generator as sg
+    from anovos_amd.data_transformer import transformers as T
+
+    idf = random_frame(seed)
+    d0 = sg.measures_of_dispersion(ctx, idf, ["a"]).set_index("attribute")
+    m0 = sg.measures_of_centralTendency(ctx, idf, ["a"]).set_index("attribute")
+    odf = T.z_standardization(ctx, idf, ["a"], output_mode="append")
+    z = odf.col("a_scaled").data
+    zz = z[~torch.isnan(z)].to(torch.float64)
+    assert abs(float(zz.mean())) < 1e-3
+    assert float(zz.std(unbiased=True)) == pytest.approx(1.0, rel=1e-2)
+    back = zz * float(d0.loc["a", "stddev"]) + float(m0.loc["a", "mean"])
+    orig = idf.col("a").data
+    oo = orig[~torch.isnan(orig)].to(torch.float64)
+    assert float((back - oo).abs().max()) < 1e-3 * max(1.0, float(oo.abs().max()))
+
+
+def test_duplicate_detection_idempotent(ctx):
+    """Deduplicating twice equals deduplicating once."""
+    import pandas as pd
+    from anovos_amd.data_analyzer import quality_checker as qc
+
+    rng = np.random.default_rng(31)
+    pdf = pd.DataFrame({"a": rng.integers(0, 50, 5000).astype(float),
+                        "b": rng.choice(["x", "y"], 5000)})
+    idf = AnovosFrame.from_pandas(pdf, device="cpu")
+    once, _ = qc.duplicate_detection(ctx, idf, treatment=True)
+    twice, stats2 = qc.duplicate_detection(ctx, once, treatment=True)
+    assert once.count() == twice.count()
+    assert once.count() == len(pdf.drop_duplicates())
