@@ -449,3 +449,172 @@ def stationarity_check(series, max_lag: int = 1) -> dict:
     stationary = bool(mean_shift < 0.5 and 0.25 < var_ratio < 4.0)
     return {"stationary": stationary, "mean_shift": float(mean_shift),
             "variance_ratio": float(var_ratio), "diff_lag1_autocorr": ac1}
+
+
+# ---------------------------------------------------------------------------
+# Reference-named section builders (report_generation.py:99-3812). The
+# reference returns datapane objects; here each returns the native html
+# fragment (or a DataFrame where the reference tabulates) built from the
+# same master_path file contract, so custom-report composers migrating
+# from the reference keep their entry points.
+# ---------------------------------------------------------------------------
+
+def list_ts_remove_append(l, opt):
+    """Remove (opt==1) or append (else) the `_ts` suffix on each name
+    (reference report_generation.py:2308)."""
+    if opt == 1:
+        return [x[:-3] if str(x).endswith("_ts") else x for x in l]
+    return [x if str(x).endswith("_ts") else str(x) + "_ts" for x in l]
+
+
+def drift_stability_ind(missing_recs_drift, drift_tab, missing_recs_stability, stability_tab):
+    """(drift_ind, stability_ind) flags from which stat files are missing
+    (reference report_generation.py:440): 0 = tab absent, 1 = full,
+    stability 0.5 = index present but per-attribute metrics missing."""
+    drift_ind = 0 if len(missing_recs_drift) == len(drift_tab) else 1
+    if len(missing_recs_stability) == len(stability_tab):
+        stability_ind = 0
+    elif ("stabilityIndex_metrics" in missing_recs_stability
+          and "stability_index" not in missing_recs_stability):
+        stability_ind = 0.5
+    else:
+        stability_ind = 1
+    return drift_ind, stability_ind
+
+
+def data_analyzer_output(master_path, avl_recs_tab=None, tab_name="stats_generator"):
+    """html section for one analyzer tab from its saved CSVs (reference
+    report_generation.py:233). `avl_recs_tab` limits to specific file
+    stems; default renders everything the tab saved."""
+    tab_map = {
+        "stats_generator": SG_tabs,
+        "quality_checker": QC_tabs,
+        "association_evaluator": AE_tabs,
+    }
+    names = tab_map.get(tab_name)
+    if names is None:
+        raise ValueError(f"tab_name must be one of {sorted(tab_map)}")
+    if avl_recs_tab:
+        names = [n for n in names if n in set(avl_recs_tab)]
+    parts = []
+    for n in names:
+        df = _read_csv(master_path, n)
+        if df is not None:
+            parts.append(f"<h3>{remove_u_score(n)}</h3>" + _tbl(df))
+    return "".join(parts)
+
+
+def chart_gen_list(master_path, chart_type, type_col=None):
+    """List of html chart divs for every saved `<chart_type>_<col>` plotly
+    JSON (reference report_generation.py:475); `type_col` filters to the
+    given column names."""
+    files = _chart_files(master_path, chart_type)
+    if type_col:
+        allow = {str(c) for c in type_col}
+        files = [f for f in files
+                 if f[len(chart_type) + 1:].rsplit(".", 1)[0] in allow]
+    out = []
+    for i, fn in enumerate(files):
+        try:
+            out.append(_fig_div(os.path.join(master_path, fn), f"{chart_type}_{i}"))
+        except Exception:
+            continue
+    return out
+
+
+def line_chart_gen_stability(df1, df2, col):
+    """html line chart of a column's per-snapshot stability metrics with
+    its summarized CV/SI annotation (reference report_generation.py:99).
+    df1 = summarized stability metrics (attribute, stability_index, ...),
+    df2 = per-snapshot metrics (attribute, idx, mean, stddev, kurtosis)."""
+    d = df2[df2["attribute"] == col] if "attribute" in df2.columns else df2
+    fig = go.Figure()
+    xcol = "idx" if "idx" in d.columns else d.columns[0]
+    for metric in ("mean", "stddev", "kurtosis"):
+        if metric in d.columns:
+            fig.add_trace(go.Scatter(x=d[xcol], y=pd.to_numeric(d[metric], errors="coerce"),
+                                     mode="lines+markers", name=metric))
+    title = f"Stability — {col}"
+    if df1 is not None and "attribute" in getattr(df1, "columns", []):
+        row = df1[df1["attribute"] == col]
+        if len(row) and "stability_index" in row.columns:
+            title += f" (SI={float(row['stability_index'].iloc[0]):.2f})"
+    fig.update_layout(title=title, height=360)
+    return _fig_div(fig, "stab_" + str(col))
+
+
+def ts_landscape(base_path, ts_cols=None, id_col=None):
+    """Detected-timestamp landscape table as html (reference
+    report_generation.py:2636): the ts_cols_stats.csv summary, filtered
+    to `ts_cols` when given."""
+    tcs = _read_csv(base_path, "ts_cols_stats")
+    if tcs is None:
+        return ""
+    if ts_cols:
+        namecol = tcs.columns[0]
+        tcs = tcs[tcs[namecol].isin(set(ts_cols) | {id_col})]
+    return "<h3>Time-Series Landscape</h3>" + _tbl(tcs)
+
+
+def ts_stats(base_path):
+    """All saved time-series stat tables as one html block (reference
+    report_generation.py:3051) — same file contract as ts_viz_generate."""
+    return ts_viz_generate(base_path)
+
+
+def overall_stats_gen(lat_col_list, long_col_list, geohash_col_list):
+    """Summary DataFrame of the detected geospatial fields (reference
+    report_generation.py:3210)."""
+    return pd.DataFrame(
+        {
+            "Field Category": ["Latitude", "Longitude", "Geohash"],
+            "Count": [len(lat_col_list or []), len(long_col_list or []), len(geohash_col_list or [])],
+            "Columns": [", ".join(lat_col_list or []), ", ".join(long_col_list or []),
+                        ", ".join(geohash_col_list or [])],
+        }
+    )
+
+
+def loc_field_stats(lat_col_list, long_col_list, geohash_col_list, max_records=100):
+    """Per-field listing of the detected location columns (reference
+    report_generation.py:3250)."""
+    rows = [("latitude", c) for c in (lat_col_list or [])]
+    rows += [("longitude", c) for c in (long_col_list or [])]
+    rows += [("geohash", c) for c in (geohash_col_list or [])]
+    return pd.DataFrame(rows[:max_records], columns=["field_type", "column"])
+
+
+def read_stats_ll_geo(lat_col, long_col, geohash_col, master_path, top_geo_records=100):
+    """html of the saved lat/long + geohash descriptive stats CSVs
+    (reference report_generation.py:3298)."""
+    parts = []
+    for fn in sorted(os.listdir(master_path)):
+        if fn.endswith(".csv") and fn.startswith(("Overall_Summary", "Top_")):
+            try:
+                df = pd.read_csv(os.path.join(master_path, fn)).head(top_geo_records)
+                parts.append(f"<h3>{fn[:-4]}</h3>" + _tbl(df))
+            except Exception:
+                continue
+    return "".join(parts)
+
+
+def read_cluster_stats_ll_geo(lat_col, long_col, geohash_col, master_path):
+    """html of the saved cluster outputs + cluster plots (reference
+    report_generation.py:3535)."""
+    parts = []
+    for fn in sorted(os.listdir(master_path)):
+        if fn.startswith("cluster_output") and fn.endswith(".csv"):
+            try:
+                parts.append(f"<h3>{fn[:-4]}</h3>" + _tbl(pd.read_csv(os.path.join(master_path, fn))))
+            except Exception:
+                continue
+    plots = [x for x in sorted(os.listdir(master_path)) if x.startswith("cluster_plot")]
+    parts.append(_charts_section(master_path, plots, "Cluster Plots", "clus"))
+    return "".join(parts)
+
+
+def read_loc_charts(master_path):
+    """html of the saved location charts (reference
+    report_generation.py:3812)."""
+    plots = [x for x in sorted(os.listdir(master_path)) if x.startswith("loc_charts")]
+    return _charts_section(master_path, plots, "Location Charts", "locc")
