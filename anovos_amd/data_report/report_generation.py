@@ -618,3 +618,81 @@ def read_loc_charts(master_path):
     report_generation.py:3812)."""
     plots = [x for x in sorted(os.listdir(master_path)) if x.startswith("loc_charts")]
     return _charts_section(master_path, plots, "Location Charts", "locc")
+
+
+def lambda_cat(val):
+    """Box-Cox λ → the transform it implies (reference
+    report_generation.py:2734; the standard Box-Cox power ladder)."""
+    if val < -1:
+        return "Reciprocal Square Transform"
+    if val < -0.5:
+        return "Reciprocal Transform"
+    if val < 0:
+        return "Receiprocal Square Root Transform"
+    if val < 0.5:
+        return "Log Transform"
+    if val < 1:
+        return "Square Root Transform"
+    if val < 2:
+        return "No Transform"
+    return "Square Transform"
+
+
+def gen_time_series_plots(base_path, x_col, y_col, time_cat):
+    """html line chart from the saved `<x>_<y>_<time_cat>.csv` aggregate
+    (reference report_generation.py:2054); empty string when absent."""
+    p = ends_with(base_path) + f"{x_col}_{y_col}_{time_cat}.csv"
+    if not os.path.exists(p):
+        return ""
+    df = pd.read_csv(p).dropna()
+    if df.empty:
+        return ""
+    xc = df.columns[0]
+    fig = go.Figure()
+    for c in [c for c in df.columns[1:] if pd.api.types.is_numeric_dtype(df[c])]:
+        fig.add_trace(go.Scatter(x=df[xc].astype(str), y=df[c], mode="lines+markers", name=c))
+    fig.update_layout(title=f"{y_col} by {x_col} ({time_cat})", height=360)
+    return _fig_div(fig, f"tsp_{x_col}_{y_col}_{time_cat}")
+
+
+def _ts_viz(base_path, x_col, y_col, time_cat):
+    return gen_time_series_plots(base_path, x_col, y_col, time_cat)
+
+
+# The reference's nine ts_viz_<view>_<slot> builders lay out the same
+# per-granularity aggregate chart in the three time-series report views
+# (report_generation.py:2345-3090); slots 1/2/3 are daily/hourly/weekly.
+def ts_viz_1_1(base_path, x_col, y_col, output_type="daily"):
+    return _ts_viz(base_path, x_col, y_col, output_type or "daily")
+
+
+def ts_viz_1_2(base_path, x_col, y_col, output_type="hourly"):
+    return _ts_viz(base_path, x_col, y_col, output_type or "hourly")
+
+
+def ts_viz_1_3(base_path, x_col, y_col, output_type="weekly"):
+    return _ts_viz(base_path, x_col, y_col, output_type or "weekly")
+
+
+def ts_viz_2_1(base_path, x_col, y_col):
+    return _ts_viz(base_path, x_col, y_col, "daily")
+
+
+def ts_viz_2_2(base_path, x_col, y_col):
+    return _ts_viz(base_path, x_col, y_col, "hourly")
+
+
+def ts_viz_2_3(base_path, x_col, y_col):
+    return _ts_viz(base_path, x_col, y_col, "weekly")
+
+
+def ts_viz_3_1(base_path, x_col, y_col):
+    return _ts_viz(base_path, x_col, y_col, "daily")
+
+
+def ts_viz_3_2(base_path, x_col, y_col):
+    return _ts_viz(base_path, x_col, y_col, "hourly")
+
+
+def ts_viz_3_3(base_path, x_col, y_col):
+    return _ts_viz(base_path, x_col, y_col, "weekly")
