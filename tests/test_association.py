@@ -62,3 +62,25 @@ def test_quartimax_rotation():
     # rotation preserves column space / frobenius norm
     assert abs(np.linalg.norm(R) - np.linalg.norm(L)) < 1e-8
     assert np.sum(R**4) >= np.sum(L**4) - 1e-9  # criterion not decreased
+
+
+def test_correlation_matrix_algebraic_invariants(ctx, assoc_frame):
+    """Symmetric, unit diagonal, entries in [-1,1], positive
+    semi-definite up to numerical tolerance."""
+    out = ae.correlation_matrix(ctx, assoc_frame, list_of_cols=["x", "y", "z", "w"])
+    m = out.set_index("attribute").loc[["x", "y", "z", "w"], ["x", "y", "z", "w"]].to_numpy(dtype=float)
+    assert np.allclose(m, m.T, atol=1e-9)
+    assert np.allclose(np.diag(m), 1.0)
+    assert (np.abs(m) <= 1 + 1e-9).all()
+    assert np.linalg.eigvalsh(m).min() > -1e-6
+
+
+def test_iv_separability_ordering(ctx, assoc_frame):
+    """The label-driving attribute must dominate IV and IG; an
+    independent attribute must land near zero (IV < 0.1 = weak per the
+    reference's own interpretation bands)."""
+    iv = ae.IV_calculation(ctx, assoc_frame, list_of_cols=["x", "z"], label_col="label", event_label=">50K").set_index("attribute")
+    ig = ae.IG_calculation(ctx, assoc_frame, list_of_cols=["x", "z"], label_col="label", event_label=">50K").set_index("attribute")
+    assert float(iv.loc["x", "iv"]) > 10 * max(float(iv.loc["z", "iv"]), 1e-6)
+    assert float(iv.loc["z", "iv"]) < 0.1
+    assert float(ig.loc["x", "ig"]) > float(ig.loc["z", "ig"])
