@@ -64,20 +64,13 @@ def descriptive_stats_gen(df, lat_col, long_col, geohash_col, id_col, master_pat
         pair = torch.stack([la, lo], dim=1)
         uniq, counts = torch.unique(pair, dim=0, return_counts=True)
         if _dist.is_dist():
-            import numpy as _np
-
             k = int(max_val) * 4
             local_order = torch.argsort(counts, descending=True)[:k]
-            cand = uniq[local_order].cpu().numpy()
-            ccnt = counts[local_order].cpu().numpy()
-            gathered = _dist.all_gather_object((cand, ccnt))
-            av = _np.concatenate([g[0] for g in gathered])
-            ac = _np.concatenate([g[1] for g in gathered])
-            gu, ginv = _np.unique(av, axis=0, return_inverse=True)
-            gc = _np.zeros(len(gu), dtype=_np.int64)
-            _np.add.at(gc, ginv, ac)
-            uniq = torch.from_numpy(gu).to(la.device)
-            counts = torch.from_numpy(gc).to(la.device)
+            av = torch.cat(_dist.all_gather_tensor(uniq[local_order]))
+            ac = torch.cat(_dist.all_gather_tensor(counts[local_order]))
+            uniq, ginv = torch.unique(av, dim=0, return_inverse=True)
+            counts = torch.zeros(uniq.shape[0], dtype=torch.int64, device=av.device)
+            counts.index_add_(0, ginv, ac)
         order = torch.argsort(counts, descending=True)
         topn = order[: int(max_val)]
         top_pairs = pd.DataFrame(

@@ -464,6 +464,7 @@ def outlier_detection(
             print(odf_print.to_string(index=False))
         return odf, odf_print
 
+    local_lo_hi = []  # batched cross-rank count merge: ONE collective below
     for c, (lo, hi) in zip(list_of_cols, params):
         x = idf.col(c).data
         flag = torch.zeros_like(x, dtype=torch.int8)
@@ -472,9 +473,7 @@ def outlier_detection(
         if detection_side in ("upper", "both") and hi is not None:
             flag = torch.where((x > hi) & ~torch.isnan(x), torch.ones_like(flag), flag)
         flags[c] = flag
-        lower_n = int(dist.all_reduce_scalar(int((flag == -1).sum())))
-        upper_n = int(dist.all_reduce_scalar(int((flag == 1).sum())))
-        rows_print.append([c, lower_n, upper_n, 0])
+        local_lo_hi.extend([int((flag == -1).sum()), int((flag == 1).sum())])
         if treatment and treatment_method in ("value_replacement", "null_replacement"):
             if treatment_method == "value_replacement":
                 lo_v = float(lo) if lo is not None else float("nan")
@@ -486,6 +485,9 @@ def outlier_detection(
             odf = odf.with_column(c + "_outliered", Column(c + "_outliered", idf.col(c).dtype, y))
             if output_mode == "replace":
                 odf = odf.drop([c]).rename({c + "_outliered": c})
+    merged = dist.all_reduce_scalars(local_lo_hi) if local_lo_hi else []
+    for i, c in enumerate(list_of_cols):
+        rows_print.append([c, int(merged[2 * i]), int(merged[2 * i + 1]), 0])
     if treatment and treatment_method == "row_removal":
         keep = torch.ones(idf.local_rows(), dtype=torch.bool, device=idf.device)
         for c in list_of_cols:

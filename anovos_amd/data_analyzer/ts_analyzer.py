@@ -88,10 +88,7 @@ def ts_eligiblity_check(ctx, idf: AnovosFrame, id_col: str, opt: int = 1, tz_off
     valid = day != NULL_TS
     days = torch.unique(day[valid])
     if _dist.is_dist():
-        gathered = _dist.all_gather_object(days.cpu().numpy())
-        import numpy as _np
-
-        days = torch.from_numpy(_np.unique(_np.concatenate(gathered))).to(day.device)
+        days = torch.unique(torch.cat(_dist.all_gather_tensor(days)))
     if opt == 1:
         if days.numel() > 1:
             diffs = (days[1:] - days[:-1]).to(torch.float64) / US_PER_DAY
@@ -109,10 +106,7 @@ def ts_eligiblity_check(ctx, idf: AnovosFrame, id_col: str, opt: int = 1, tz_off
     pair = torch.stack([id_codes[ok].to(torch.float64), day[ok].to(torch.float64)], dim=1)
     uniq_pair = torch.unique(pair, dim=0)
     if _dist.is_dist():
-        import numpy as _np
-
-        gp = _dist.all_gather_object(uniq_pair.cpu().numpy())
-        uniq_pair = torch.from_numpy(_np.unique(_np.concatenate(gp), axis=0)).to(day.device)
+        uniq_pair = torch.unique(torch.cat(_dist.all_gather_tensor(uniq_pair)), dim=0)
     rows = []
     for key_idx, name in ((0, "id_date_pair"), (1, "date_id_pair")):
         keys = uniq_pair[:, key_idx]
@@ -151,11 +145,8 @@ def ts_viz_data(idf: AnovosFrame, x_col: str, y_col: str, output_mode="append", 
         klabels = [float(u) for u in uniq]
     if _dist.is_dist() and key.kind != "categorical":
         # unify bucket keys across ranks (dates/dows differ per shard)
-        import numpy as _np
-
-        gathered = _dist.all_gather_object(uniq.cpu().numpy())
-        guniq = _np.unique(_np.concatenate(gathered))
-        gt = torch.from_numpy(guniq).to(key.data.device)
+        gt = torch.unique(torch.cat(_dist.all_gather_tensor(uniq)))
+        guniq = gt.cpu().numpy()
         pos = torch.searchsorted(gt, uniq)
         kcodes = pos[kcodes]
         if key.dtype in ("timestamp", "date"):

@@ -94,19 +94,12 @@ def data_sample(
     uniq, inv = torch.unique(key, return_inverse=True)
     local_counts = torch.bincount(inv, minlength=uniq.numel()).to(torch.float64)
     if dist.is_dist():
-        gathered = dist.all_gather_object((uniq.cpu().numpy(), local_counts.cpu().numpy()))
-        import numpy as np
-
-        av = np.concatenate([g[0] for g in gathered])
-        ac = np.concatenate([g[1] for g in gathered])
-        gu, ginv = np.unique(av, return_inverse=True)
-        gc = np.zeros(len(gu), dtype=np.float64)
-        np.add.at(gc, ginv, ac)
-        import torch as _t
-
-        guniq = _t.from_numpy(gu).to(key.device)
-        gcounts = _t.from_numpy(gc).to(key.device)
-        pos = _t.searchsorted(guniq, uniq)
+        av = torch.cat(dist.all_gather_tensor(uniq))
+        ac = torch.cat(dist.all_gather_tensor(local_counts))
+        guniq, ginv = torch.unique(av, return_inverse=True)
+        gcounts = torch.zeros(guniq.numel(), dtype=torch.float64, device=av.device)
+        gcounts.index_add_(0, ginv, ac)
+        pos = torch.searchsorted(guniq, uniq)
         counts = gcounts[pos]
         smallest = float(gcounts.min())  # global smallest stratum, even if absent locally
     else:

@@ -193,15 +193,14 @@ def numeric_value_counts(idf, col: str) -> Tuple[torch.Tensor, torch.Tensor]:
 
 
 def _merge_value_counts(vals: torch.Tensor, cnts: torch.Tensor):
-    gathered = dist.all_gather_object((vals.cpu().numpy(), cnts.cpu().numpy()))
-    import numpy as np
-
-    av = np.concatenate([g[0] for g in gathered])
-    ac = np.concatenate([g[1] for g in gathered])
-    uv, inv = np.unique(av, return_inverse=True)
-    uc = np.zeros(len(uv), dtype=np.int64)
-    np.add.at(uc, inv, ac)
-    return torch.from_numpy(uv), torch.from_numpy(uc)
+    # tensorized cross-rank merge (device-resident under RCCL): gather
+    # both varlen vectors, then unique+index_add on device
+    av = torch.cat(dist.all_gather_tensor(vals))
+    ac = torch.cat(dist.all_gather_tensor(cnts.to(torch.int64)))
+    uv, inv = torch.unique(av, return_inverse=True)
+    uc = torch.zeros(uv.numel(), dtype=torch.int64, device=av.device)
+    uc.index_add_(0, inv, ac)
+    return uv.cpu(), uc.cpu()
 
 
 def duplicate_row_count(idf, cols: Optional[List[str]] = None) -> int:
@@ -212,10 +211,7 @@ def duplicate_row_count(idf, cols: Optional[List[str]] = None) -> int:
     h = row_hash(idf, cols)
     uniq = torch.unique(h)
     if dist.is_dist():
-        gathered = dist.all_gather_object(uniq.cpu().numpy())
-        import numpy as np
-
-        nuniq = len(np.unique(np.concatenate(gathered)))
+        nuniq = torch.unique(torch.cat(dist.all_gather_tensor(uniq))).numel()
     else:
         nuniq = uniq.numel()
     total = idf.count()

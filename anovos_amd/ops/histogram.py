@@ -39,11 +39,9 @@ def _exact_quantiles(idf, cols, probs, moments, rel_err=1e-4):
         t = idf.col(c).data
         x = t[~torch.isnan(t)]
         if _dist.is_dist():
-            gathered = _dist.all_gather_object(x.cpu().numpy())
-            import numpy as _np
-
-            allv = _np.concatenate(gathered)
-            xs = torch.from_numpy(_np.sort(allv))
+            # tensorized gather (small columns only — EXACT_N_THRESHOLD
+            # gates this path); sort on device after the merge
+            xs, _ = torch.sort(torch.cat(_dist.all_gather_tensor(x)))
         else:
             xs, _ = torch.sort(x)
         n = xs.numel()

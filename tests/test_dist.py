@@ -334,6 +334,65 @@ def test_dist_exact_dedup_across_ranks(tmp_path):
     assert res["unique_rows"] == 150.0
 
 
+def _wrappers_worker(rank, world, port, out_path):
+    """Exercise every dist wrapper with rank-asymmetric payloads: varlen
+    all_gather_tensor (1-D + 2-D + empty-on-one-rank), batched scalar
+    reduce, broadcast_."""
+    os.environ.update(
+        {"RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": str(world),
+         "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port)}
+    )
+    import torch.distributed as td
+
+    from anovos_amd.core import dist
+
+    dist.init_from_env(timeout_s=120)
+    # 1-D varlen: rank 0 sends 5 values, rank 1 sends 3
+    t = torch.arange(5 - 2 * rank, dtype=torch.float64) + 10 * rank
+    parts = dist.all_gather_tensor(t)
+    gathered_1d = [p.tolist() for p in parts]
+    # empty on rank 1
+    e = torch.arange(4) if rank == 0 else torch.empty(0, dtype=torch.int64)
+    eparts = [p.tolist() for p in dist.all_gather_tensor(e)]
+    # 2-D pairs, different row counts
+    m = torch.full((2 + rank, 2), float(rank))
+    mparts = [p.shape for p in dist.all_gather_tensor(m)]
+    # batched scalars
+    sc = dist.all_reduce_scalars([float(rank), 1.0, float(rank) * 2], "sum")
+    mx = dist.all_reduce_scalars([float(rank)], "max")
+    # broadcast tensor
+    b = torch.tensor([3.25, -1.5]) if rank == 0 else torch.zeros(2)
+    dist.broadcast_(b)
+    if rank == 0:
+        json.dump(
+            {"g1": gathered_1d, "ge": eparts, "gm": [list(s) for s in mparts],
+             "sc": sc, "mx": mx, "b": b.tolist(), "backend": dist.backend()},
+            open(out_path, "w"),
+        )
+    td.barrier()
+    td.destroy_process_group()
+
+
+def test_dist_wrapper_primitives_2rank():
+    port = _free_port()
+    out = tempfile.NamedTemporaryFile(suffix=".json", delete=False).name
+    mp_ctx = mp.get_context("spawn")
+    procs = [mp_ctx.Process(target=_wrappers_worker, args=(r, 2, port, out)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0
+    res = json.load(open(out))
+    assert res["backend"] == "gloo"
+    assert res["g1"] == [[0.0, 1.0, 2.0, 3.0, 4.0], [10.0, 11.0, 12.0]]
+    assert res["ge"] == [[0, 1, 2, 3], []]
+    assert res["gm"] == [[2, 2], [3, 2]]
+    assert res["sc"] == [1.0, 2.0, 2.0]
+    assert res["mx"] == [1.0]
+    assert res["b"] == [3.25, -1.5]
+
+
 def _stable_worker(rank, world, port, out_path):
     os.environ.update(
         {"RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": str(world),
