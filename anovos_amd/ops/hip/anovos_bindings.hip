@@ -73,6 +73,11 @@ int anovos_centered_gram(const void *const *cols, int64_t n, int k,
                          const float *means, const int *pair_i,
                          const int *pair_j, int npairs, int row_chunks,
                          float *partials, float *gram, hipStream_t stream);
+int anovos_label_counts_multi(const void *const *cols, const uint8_t *label,
+                              const int64_t *lens, const int64_t *offs,
+                              const int *sizes, const int *dtypes, int ncols,
+                              int max_slots, int nchunks, uint64_t *out,
+                              hipStream_t stream);
 }
 
 namespace {
@@ -705,6 +710,56 @@ torch::Tensor code_counts_multi(std::vector<torch::Tensor> cols, std::vector<int
   return out;
 }
 
+// K9 fused: label-conditioned histograms, ALL columns in one launch.
+// Per column at out[off]: [slots] totals then [slots] event counts.
+// Column dtype drives the slot mapping (f32 binned vs int32 codes).
+torch::Tensor label_counts_multi(std::vector<torch::Tensor> cols,
+                                 torch::Tensor label,
+                                 std::vector<int64_t> sizes) {
+  TORCH_CHECK(!cols.empty(), "no columns");
+  TORCH_CHECK(cols.size() == sizes.size(), "sizes mismatch");
+  TORCH_CHECK(label.is_contiguous() && label.scalar_type() == torch::kUInt8,
+              "uint8 label required");
+  auto device = cols[0].device();
+  std::vector<int64_t> ptrs, lens, offs, dt, sz;
+  int64_t off = 0, maxn = 0;
+  int max_slots = 0;
+  for (size_t i = 0; i < cols.size(); ++i) {
+    auto &t = cols[i];
+    TORCH_CHECK(t.is_contiguous(), "contiguous columns required");
+    int dtype;
+    if (t.scalar_type() == torch::kFloat) dtype = 0;
+    else if (t.scalar_type() == torch::kInt) dtype = 1;
+    else TORCH_CHECK(false, "f32 or int32 columns required");
+    TORCH_CHECK(t.numel() == label.numel(), "column/label length mismatch");
+    TORCH_CHECK(sizes[i] >= 1 && sizes[i] <= 8192, "slot count out of LDS range");
+    ptrs.push_back((int64_t)t.data_ptr());
+    lens.push_back(t.numel());
+    offs.push_back(off);
+    off += 2 * sizes[i];
+    dt.push_back(dtype);
+    sz.push_back(sizes[i]);
+    maxn = std::max(maxn, t.numel());
+    max_slots = std::max(max_slots, (int)sizes[i]);
+  }
+  int ncols = (int)cols.size();
+  int nchunks = pick_chunks(maxn, ncols);
+  auto out = torch::zeros({off}, torch::TensorOptions().dtype(torch::kInt64).device(device));
+  auto dptr = to_device_i64(ptrs, device);
+  auto dlen = to_device_i64(lens, device);
+  auto doff = to_device_i64(offs, device);
+  auto dsz = to_device_i64(sz, device).to(torch::kInt32);
+  auto ddt = to_device_i64(dt, device).to(torch::kInt32);
+  check_hip(anovos_label_counts_multi(
+                (const void *const *)dptr.data_ptr<int64_t>(),
+                label.data_ptr<uint8_t>(), dlen.data_ptr<int64_t>(),
+                doff.data_ptr<int64_t>(), dsz.data_ptr<int>(),
+                ddt.data_ptr<int>(), ncols, max_slots, nchunks,
+                (uint64_t *)out.data_ptr<int64_t>(), current_stream()),
+            "anovos_label_counts_multi");
+  return out;
+}
+
 // K10/K11 fused: outlier counts + clamp/null treatment in one launch.
 // mode: 0 count only, 1 clamp to bounds, 2 null-out. Returns
 // (counts [k,2] int64, outs list — empty when mode==0).
@@ -813,6 +868,7 @@ std::tuple<torch::Tensor, torch::Tensor> moments_hll(std::vector<torch::Tensor> 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("code_counts_multi", &code_counts_multi, "fused multi-column code counts + null slot (K5)");
   m.def("outlier_clamp_columns", &outlier_clamp_columns, "fused outlier count/clamp (K10/K11)");
+  m.def("label_counts_multi", &label_counts_multi, "fused multi-column label-conditioned histograms (K9)");
   m.def("moments_hll", &moments_hll, "fused moments + HLL registers (K1/K2+K4)",
         py::arg("cols"), py::arg("p"), py::arg("shifts") = std::vector<double>());
   m.def("centered_gram_bf16", &centered_gram_bf16, "bf16 MFMA centered Gram X^T X (K8)");

@@ -1453,3 +1453,92 @@ extern "C" int anovos_moments_hll(const void *const *cols, const int64_t *lens,
                      partials, nchunks, mom_out);
   return (int)hipGetLastError();
 }
+
+// ------------------------------------------------------------------
+// K9 (fused): label-conditioned histograms for ALL columns in one
+// launch (IV/IG per-bin event counts — reference
+// association_evaluator.py:368-409, :533-572). Layout per column at
+// out[off[col]]: [slots] total counts then [slots] event counts
+// (n0 = total - event). dtypes: 0 = f32 binned values (NaN -> slot 0,
+// v -> trunc(v)+1, clamped — matches the host torch fallback exactly);
+// 1 = int32 dictionary codes (null/invalid -> last slot).
+// ------------------------------------------------------------------
+__global__ __launch_bounds__(THREADS) void label_counts_multi_kernel(
+    const void *const *cols, const uint8_t *__restrict__ label,
+    const int64_t *lens, const int64_t *offs, const int *sizes,
+    const int *dtypes, int nchunks, uint64_t *out) {
+  const int col = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  const int slots = sizes[col];
+  uint64_t *base = out + offs[col];
+  extern __shared__ uint32_t lds2[];  // [2*slots]: totals then events
+  uint32_t *tot = lds2;
+  uint32_t *evt = lds2 + slots;
+  for (int i = threadIdx.x; i < 2 * slots; i += THREADS) lds2[i] = 0;
+  __syncthreads();
+  const int64_t n = lens[col];
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+  if (dtypes[col] == 0) {
+    const float *__restrict__ x = (const float *)cols[col];
+    const int64_t nv = (e - s) / 4;
+    const nat_f4 *__restrict__ xv = reinterpret_cast<const nat_f4 *>(x + s);
+    for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
+      nat_f4 v = __builtin_nontemporal_load(&xv[i]);
+      const int64_t r = s + i * 4;
+      float vv[4] = {v.x, v.y, v.z, v.w};
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        int slot = (vv[k] != vv[k]) ? 0 : (int)vv[k] + 1;
+        slot = (slot < 0) ? 0 : ((slot >= slots) ? slots - 1 : slot);
+        atomicAdd(&tot[slot], 1u);
+        if (label[r + k]) atomicAdd(&evt[slot], 1u);
+      }
+    }
+    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) {
+      float v = x[i];
+      int slot = (v != v) ? 0 : (int)v + 1;
+      slot = (slot < 0) ? 0 : ((slot >= slots) ? slots - 1 : slot);
+      atomicAdd(&tot[slot], 1u);
+      if (label[i]) atomicAdd(&evt[slot], 1u);
+    }
+  } else {
+    const int32_t *__restrict__ codes = (const int32_t *)cols[col];
+    const int64_t nv = (e - s) / 4;
+    const nat_i4 *__restrict__ cv = reinterpret_cast<const nat_i4 *>(codes + s);
+    for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
+      nat_i4 c = __builtin_nontemporal_load(&cv[i]);
+      const int64_t r = s + i * 4;
+      int cc[4] = {c.x, c.y, c.z, c.w};
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        int slot = (cc[k] >= 0 && cc[k] < slots - 1) ? cc[k] : slots - 1;
+        atomicAdd(&tot[slot], 1u);
+        if (label[r + k]) atomicAdd(&evt[slot], 1u);
+      }
+    }
+    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) {
+      int c = codes[i];
+      int slot = (c >= 0 && c < slots - 1) ? c : slots - 1;
+      atomicAdd(&tot[slot], 1u);
+      if (label[i]) atomicAdd(&evt[slot], 1u);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < slots; i += THREADS) {
+    if (tot[i]) atomicAdd((unsigned long long *)&base[i], (unsigned long long)tot[i]);
+    if (evt[i]) atomicAdd((unsigned long long *)&base[slots + i], (unsigned long long)evt[i]);
+  }
+}
+
+extern "C" int anovos_label_counts_multi(
+    const void *const *cols, const uint8_t *label, const int64_t *lens,
+    const int64_t *offs, const int *sizes, const int *dtypes, int ncols,
+    int max_slots, int nchunks, uint64_t *out, hipStream_t stream) {
+  size_t lds = (size_t)max_slots * 2 * sizeof(uint32_t);
+  hipLaunchKernelGGL(label_counts_multi_kernel, dim3(ncols * nchunks),
+                     dim3(THREADS), lds, stream, cols, label, lens, offs,
+                     sizes, dtypes, nchunks, out);
+  return (int)hipGetLastError();
+}

@@ -9,10 +9,18 @@ one full pipeline pass over the resident frame:
   analyzer   : measures_of_counts, centralTendency (discrete cols),
                cardinality (HLL), dispersion, percentiles, shape,
                nullRows detection, biasedness, outlier detection
+  association: correlation_matrix (bf16 MFMA Gram over ALL 150 numeric
+               cols), IV + IG over all 200 attributes (fused K9
+               label-histogram kernel; equal-frequency binning inside)
   drift      : attribute_binning (source model) + PSI/JSD/HD/KS vs the
-               warmup snapshot histograms
+               warmup snapshot histograms — ALL numeric columns
   transform  : attribute_binning, z_standardization, imputation_MMM,
-               cat_to_num label encoding, outlier_categories
+               cat_to_num label encoding (all 50 cats), outlier_categories
+               (all 50 cats)
+
+Nothing is strided or capped: every section runs over its full column
+set, and per-transform checksums are whole-column
+device reductions (not 8-element probes).
 
 Data: synthetic, random-init, generated on-device before timing (no
 network). Weak scaling: per-GPU rows fixed as N grows; value = aggregate
@@ -105,6 +113,7 @@ def pipeline_step(ctx, idf, source_hist, model_dir, sections=None):
     defeat dead-code elimination."""
     import pandas as pd
 
+    from anovos_amd.data_analyzer import association_evaluator as ae
     from anovos_amd.data_analyzer import quality_checker as qc
     from anovos_amd.data_analyzer import stats_generator as sg
     from anovos_amd.data_transformer import transformers as T
@@ -149,8 +158,18 @@ def pipeline_step(ctx, idf, source_hist, model_dir, sections=None):
         odf_out, _ = qc.outlier_detection(
             ctx, idf, num_cols, detection_side="both", treatment=True, treatment_method="value_replacement"
         )
-        chk += float(odf_out.col(num_cols[0]).data[:8].float().nansum().item())
+        chk += float(odf_out.col(num_cols[0]).data.float().nansum().item())
         del odf_out  # free treated copies promptly (125M-row shards: ~75 GB each)
+
+    # ---- association (the MFMA Gram + IV/IG are
+    # part of the "full analyzer" headline) ----
+    with _timed("correlation", sections, ctx):
+        corr = ae.correlation_matrix(ctx, idf, num_cols)  # K8 bf16 MFMA Gram
+        chk += float(np.nansum(corr[sorted(num_cols)].to_numpy()))
+    with _timed("iv_ig", sections, ctx):
+        iv = ae.IV_calculation(ctx, idf, label_col="label", event_label="yes")  # K9
+        ig = ae.IG_calculation(ctx, idf, label_col="label", event_label="yes")
+        chk += float(iv["iv"].sum()) + float(ig["ig"].sum())
 
     # ---- drift (PSI/JSD/HD/KS vs warmup snapshot) ----
     with _timed("binning", sections, ctx):
@@ -159,7 +178,7 @@ def pipeline_step(ctx, idf, source_hist, model_dir, sections=None):
         drift_vals = 0.0
         import numpy as _np
 
-        drift_cols = num_cols[:: max(1, len(num_cols) // 50)]
+        drift_cols = num_cols  # ALL numeric columns (no stride)
         q_freqs = dd.batched_bin_frequencies(binned, [c + "_binned" for c in drift_cols], idf.count(), max_bin=10)
         # vectorize PSI/JSD/HD/KS over the whole (col x bin) matrix at once
         key_union = sorted({k for c in drift_cols for k in source_hist.get(c, {})}
@@ -191,22 +210,22 @@ def pipeline_step(ctx, idf, source_hist, model_dir, sections=None):
     chk_dev = torch.zeros((), dtype=torch.float32, device=idf.device)
     with _timed("z_standardization", sections, ctx):
         t1 = T.z_standardization(ctx, idf, num_cols[:N_NUM_CONT])  # K11
-        chk_dev += t1.col(num_cols[0]).data[:8].float().nansum()
+        chk_dev += t1.col(num_cols[0]).data.float().nansum()  # whole column
         del t1
     with _timed("imputation_median", sections, ctx):
         t2 = T.imputation_MMM(ctx, idf, method_type="median")
-        chk_dev += t2.col(num_cols[1]).data[:8].float().nansum()
+        chk_dev += t2.col(num_cols[1]).data.float().nansum()
         del t2
     with _timed("cat_label_encoding", sections, ctx):
-        t3 = T.cat_to_num_unsupervised(ctx, idf, cat_cols[:25], method_type="label_encoding")  # K12
+        t3 = T.cat_to_num_unsupervised(ctx, idf, cat_cols, method_type="label_encoding")  # K12, all 50
         if (cat_cols[0] + "_index") in t3.columns:
-            chk_dev += t3.col(cat_cols[0] + "_index").data[:8].float().nansum()
+            chk_dev += t3.col(cat_cols[0] + "_index").data.float().nansum()
         else:
-            chk_dev += t3.col(cat_cols[0]).data[:8].float().sum()
+            chk_dev += t3.col(cat_cols[0]).data.float().sum()
         del t3
     with _timed("outlier_categories", sections, ctx):
-        t4 = T.outlier_categories(ctx, idf, cat_cols[25:], max_category=20)
-        chk_dev += t4.col(cat_cols[25]).data[:8].float().sum()
+        t4 = T.outlier_categories(ctx, idf, cat_cols, max_category=20)  # all 50
+        chk_dev += t4.col(cat_cols[0]).data.float().sum()
         del t4
     chk += float(chk_dev.item())  # the step's ONE transform-phase sync
     return chk
