@@ -209,11 +209,24 @@ class AnovosFrame:
         out = {n: Column(c.name, c.dtype, c.data.to(dev), c.dictionary) for n, c in self._cols.items()}
         return AnovosFrame(out, dev)
 
+    @property
+    def aux_cache(self) -> dict:
+        """Frame-instance-scoped cache for derived aggregates that span
+        multiple columns (e.g. IV/IG shared binned label counts). Not
+        inherited by derived frames (with_column/select return new
+        instances), cleared with clear_stats_cache."""
+        d = getattr(self, "_aux_cache", None)
+        if d is None:
+            d = {}
+            self._aux_cache = d
+        return d
+
     def clear_stats_cache(self) -> None:
         """Drop all cached derived statistics (forces recomputation —
         used by benchmarks to keep timed steps honest)."""
         for c in self._cols.values():
             c.cache.clear()
+        self._aux_cache = {}
 
     def persist(self) -> "AnovosFrame":  # Spark-parity no-op: tensors are already resident
         return self

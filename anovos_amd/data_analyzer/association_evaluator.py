@@ -102,6 +102,41 @@ def _binned_label_counts(idf, col: str, label: torch.Tensor):
     return out[col]
 
 
+def _encoded_label_counts(ctx, idf, cols, label_col, event_label, label, encoding_configs):
+    """Shared IV/IG front end: bin numerics per encoding_configs, then one
+    fused multi-column label-count pass. The result is memoized in the
+    frame's aux_cache so IG reuses IV's binning AND counting when called
+    with the same inputs (the reference recomputes both; the engine's
+    stats-reuse contract makes the pair one pass)."""
+    num_cols = attributeType_segregation(idf.select(cols))[0]
+    key = (
+        "ivig_counts",
+        label_col,
+        str(event_label),
+        repr(sorted(encoding_configs.items())) if encoding_configs else "",
+        tuple(cols),
+    )
+    hit = idf.aux_cache.get(key)
+    if hit is not None:
+        return hit
+    nmax = None
+    if len(num_cols) > 0 and bool(encoding_configs):
+        bin_size = encoding_configs["bin_size"]
+        bin_method = encoding_configs["bin_method"]
+        if encoding_configs.get("monotonicity_check", 0) == 1:
+            idf_encoded = monotonic_binning(ctx, idf, num_cols, [], label_col, event_label, bin_method, bin_size)
+            cap = max(int(bin_size), 20)
+        else:
+            idf_encoded = attribute_binning(ctx, idf, num_cols, [], bin_method, bin_size)
+            cap = int(bin_size)
+        nmax = {c: cap for c in num_cols}
+    else:
+        idf_encoded = idf
+    counts = _binned_label_counts_multi(idf_encoded, cols, label, numeric_max=nmax)
+    idf.aux_cache[key] = counts
+    return counts
+
+
 def _binned_label_counts_multi(idf, cols, label: torch.Tensor, numeric_max=None):
     """Per-group (n0, n1) counts for MANY columns in one fused K9 launch
     (GPU: anovos_label_counts_multi — one frame read, LDS-staged
@@ -216,21 +251,7 @@ def IV_calculation(
     label = _event_indicator(idf, label_col, event_label)
     if int(dist.all_reduce_scalar(int(label.sum()))) == 0:
         raise TypeError("Invalid input for Event Label Value")
-    num_cols = attributeType_segregation(idf.select(cols))[0]
-    if len(num_cols) > 0 and bool(encoding_configs):
-        bin_size = encoding_configs["bin_size"]
-        bin_method = encoding_configs["bin_method"]
-        if encoding_configs.get("monotonicity_check", 0) == 1:
-            idf_encoded = monotonic_binning(ctx, idf, num_cols, [], label_col, event_label, bin_method, bin_size)
-        else:
-            idf_encoded = attribute_binning(ctx, idf, num_cols, [], bin_method, bin_size)
-    else:
-        idf_encoded = idf
-    nmax = None
-    if len(num_cols) > 0 and bool(encoding_configs):
-        cap = max(int(encoding_configs["bin_size"]), 20 if encoding_configs.get("monotonicity_check", 0) == 1 else 0)
-        nmax = {c: cap for c in num_cols}
-    counts = _binned_label_counts_multi(idf_encoded, cols, label, numeric_max=nmax)
+    counts = _encoded_label_counts(ctx, idf, cols, label_col, event_label, label, encoding_configs)
     rows = []
     for col in cols:
         n0, n1 = counts[col]
@@ -281,23 +302,9 @@ def IG_calculation(
     total_events = int(dist.all_reduce_scalar(int(label.sum())))
     if total_events == 0:
         raise TypeError("Invalid input for Event Label Value")
-    num_cols = attributeType_segregation(idf.select(cols))[0]
-    if len(num_cols) > 0 and bool(encoding_configs):
-        bin_size = encoding_configs["bin_size"]
-        bin_method = encoding_configs["bin_method"]
-        if encoding_configs.get("monotonicity_check", 0) == 1:
-            idf_encoded = monotonic_binning(ctx, idf, num_cols, [], label_col, event_label, bin_method, bin_size)
-        else:
-            idf_encoded = attribute_binning(ctx, idf, num_cols, [], bin_method, bin_size)
-    else:
-        idf_encoded = idf
     total_event = total_events / total_rows
     total_entropy = -(total_event * math.log2(total_event) + (1 - total_event) * math.log2(1 - total_event))
-    nmax = None
-    if len(num_cols) > 0 and bool(encoding_configs):
-        cap = max(int(encoding_configs["bin_size"]), 20 if encoding_configs.get("monotonicity_check", 0) == 1 else 0)
-        nmax = {c: cap for c in num_cols}
-    counts = _binned_label_counts_multi(idf_encoded, cols, label, numeric_max=nmax)
+    counts = _encoded_label_counts(ctx, idf, cols, label_col, event_label, label, encoding_configs)
     rows = []
     for col in cols:
         n0, n1 = counts[col]

@@ -73,6 +73,10 @@ int anovos_centered_gram(const void *const *cols, int64_t n, int k,
                          const float *means, const int *pair_i,
                          const int *pair_j, int npairs, int row_chunks,
                          float *partials, float *gram, hipStream_t stream);
+int anovos_centered_gram_sr(const void *const *cols, int64_t n, int k,
+                            const float *means, const int *pair_i,
+                            const int *pair_j, int npairs, int row_chunks,
+                            float *partials, float *gram, hipStream_t stream);
 int anovos_label_counts_multi(const void *const *cols, const uint8_t *label,
                               const int64_t *lens, const int64_t *offs,
                               const int *sizes, const int *dtypes, int ncols,
@@ -653,17 +657,32 @@ torch::Tensor centered_gram_bf16(std::vector<torch::Tensor> cols, torch::Tensor 
       pj64.push_back(j);
     }
   const int npairs = (int)pi64.size();
-  // target >= 2048 workgroups across the 8 XCDs; >= 64K rows per chunk
-  int row_chunks = (int)std::min<int64_t>(std::max<int64_t>(1, 2048 / std::max(npairs, 1)),
-                                          std::max<int64_t>(1, n >> 16));
-  row_chunks = std::max(row_chunks, 1);
-  auto opts_i32 = torch::TensorOptions().dtype(torch::kInt32).device(device);
   auto pi = to_device_i64(pi64, device).to(torch::kInt32);
   auto pj = to_device_i64(pj64, device).to(torch::kInt32);
   auto dptr = to_device_i64(ptrs, device);
+  auto gram = torch::zeros({k, k}, torch::TensorOptions().dtype(torch::kFloat32).device(device));
+  if (kt <= 13) {
+    // single-read kernel: one block stages a 32-row slab of ALL columns
+    // through LDS; HBM traffic = n*k*4 bytes (vs ~kt x for pair-parallel)
+    const int64_t steps_total = (n + 31) / 32;
+    int row_chunks = (int)std::min<int64_t>(2048, std::max<int64_t>(1, steps_total));
+    auto partials = torch::empty({(int64_t)row_chunks * npairs, 256},
+                                 torch::TensorOptions().dtype(torch::kFloat32).device(device));
+    check_hip(anovos_centered_gram_sr((const void *const *)dptr.data_ptr<int64_t>(), n, k,
+                                      means_d.data_ptr<float>(), pi.data_ptr<int>(),
+                                      pj.data_ptr<int>(), npairs, row_chunks,
+                                      partials.data_ptr<float>(), gram.data_ptr<float>(),
+                                      current_stream()),
+              "anovos_centered_gram_sr");
+    return gram;
+  }
+  // wide matrices: pair-parallel kernel (re-reads columns, but register
+  // budget no longer admits the all-pairs accumulator set)
+  int row_chunks = (int)std::min<int64_t>(std::max<int64_t>(1, 2048 / std::max(npairs, 1)),
+                                          std::max<int64_t>(1, n >> 16));
+  row_chunks = std::max(row_chunks, 1);
   auto partials = torch::empty({(int64_t)npairs * row_chunks, 256},
                                torch::TensorOptions().dtype(torch::kFloat32).device(device));
-  auto gram = torch::zeros({k, k}, torch::TensorOptions().dtype(torch::kFloat32).device(device));
   check_hip(anovos_centered_gram((const void *const *)dptr.data_ptr<int64_t>(), n, k,
                                  means_d.data_ptr<float>(), pi.data_ptr<int>(),
                                  pj.data_ptr<int>(), npairs, row_chunks,

@@ -146,3 +146,158 @@ extern "C" int anovos_centered_gram(const void *const *cols, int64_t n, int k,
                      0, stream, partials, row_chunks, pair_i, pair_j, k, gram);
   return (int)hipGetLastError();
 }
+
+// ------------------------------------------------------------------
+// Single-read Gram variant. The pair-parallel kernel above re-reads
+// every column once per tile-pair it appears in (~kt times, 11.7x HBM
+// traffic at k=150 — measured 197 ms in the bench step). Here ONE block
+// stages a 32-row slab of ALL kt*16 columns through LDS once (centered
+// bf16), then its 4 waves compute EVERY 16x16 tile-pair from LDS:
+// HBM traffic = n*k*4 bytes exactly.
+//
+// LDS layout: [4 ko-groups][ktot cols][8 rows] bf16 — one 16-byte block
+// per (ko-group, col), so an MFMA A/B fragment (8 K-rows of one column)
+// is a single aligned ds_read_b128; consecutive columns are 16 B apart
+// (minimum-phase bank pattern for the 16 lanes of a row-group).
+//
+// Pairs are assigned to waves in CONTIGUOUS chunks of the i-major
+// upper-tri order so a wave's consecutive pairs share the A fragment
+// (compiler CSEs the identical LDS reads). Accumulators live in
+// registers via a compile-time-unrolled pair loop (MAXPW per wave:
+// 14 covers k<=160, 23 covers k<=208; larger k falls back).
+// Deterministic: fixed block/step split, fp64 block-order reduce.
+// ------------------------------------------------------------------
+
+typedef short bf16x4_t __attribute__((ext_vector_type(4)));
+
+template <int MAXPW>
+__global__ __launch_bounds__(THREADS) void gram_singleread_kernel(
+    const float *const *cols, int64_t n, int k, int ktot, const float *means,
+    const int *pair_i, const int *pair_j, int npairs, int row_chunks,
+    float *partials) {
+  extern __shared__ short slab[];  // [4][ktot][8] bf16
+  const int block = blockIdx.x;
+  const int64_t steps_total = (n + 31) / 32;
+  const int64_t steps_per = (steps_total + row_chunks - 1) / row_chunks;
+  const int64_t step_s = (int64_t)block * steps_per;
+  const int64_t step_e = min(steps_total, step_s + steps_per);
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int m = lane & 15;    // tile row (A) / tile col (B)
+  const int kog = lane >> 4;  // this lane's 8-row K group
+
+  const int ppw = (npairs + 3) / 4;  // contiguous pair chunk per wave
+  const int p0 = wave * ppw;
+  int i0s[MAXPW], j0s[MAXPW];
+#pragma unroll
+  for (int u = 0; u < MAXPW; ++u) {
+    const int p = p0 + u;
+    const bool act = (u < ppw) && (p < npairs);
+    i0s[u] = act ? pair_i[p] * 16 : -1;
+    j0s[u] = act ? pair_j[p] * 16 : 0;
+  }
+  f32x4 acc[MAXPW];
+#pragma unroll
+  for (int u = 0; u < MAXPW; ++u) acc[u] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  for (int64_t s = step_s; s < step_e; ++s) {
+    const int64_t r0 = s * 32;
+    __syncthreads();  // previous iteration's LDS reads complete
+    for (int idx = threadIdx.x; idx < ktot * 8; idx += THREADS) {
+      const int c = idx >> 3;
+      const int seg = idx & 7;  // 4-row segment of the 32-row slab
+      const int64_t r = r0 + (int64_t)seg * 4;
+      float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
+      if (c < k) {
+        const float *__restrict__ x = cols[c];
+        const float mean = means[c];
+        if (r + 4 <= n) {
+          const float4 t = *reinterpret_cast<const float4 *>(x + r);
+          v0 = t.x - mean; v1 = t.y - mean; v2 = t.z - mean; v3 = t.w - mean;
+        } else {
+          v0 = (r + 0 < n) ? x[r + 0] - mean : 0.f;
+          v1 = (r + 1 < n) ? x[r + 1] - mean : 0.f;
+          v2 = (r + 2 < n) ? x[r + 2] - mean : 0.f;
+          v3 = (r + 3 < n) ? x[r + 3] - mean : 0.f;
+        }
+      }
+      bf16x4_t pack;
+      pack.x = to_bf16(isnan(v0) ? 0.f : v0);
+      pack.y = to_bf16(isnan(v1) ? 0.f : v1);
+      pack.z = to_bf16(isnan(v2) ? 0.f : v2);
+      pack.w = to_bf16(isnan(v3) ? 0.f : v3);
+      short *dst = slab + ((size_t)(seg >> 1) * ktot + c) * 8 + (size_t)(seg & 1) * 4;
+      *reinterpret_cast<bf16x4_t *>(dst) = pack;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int u = 0; u < MAXPW; ++u) {
+      if (i0s[u] >= 0) {
+        const bf16x8 a = *reinterpret_cast<const bf16x8 *>(
+            slab + ((size_t)kog * ktot + i0s[u] + m) * 8);
+        const bf16x8 b = *reinterpret_cast<const bf16x8 *>(
+            slab + ((size_t)kog * ktot + j0s[u] + m) * 8);
+        acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[u], 0, 0, 0);
+      }
+    }
+  }
+
+  // each pair is owned by exactly one wave: write fragments directly
+  // (C layout: row = (lane>>4)*4 + reg, col = lane&15)
+  const int row = (lane >> 4) * 4;
+#pragma unroll
+  for (int u = 0; u < MAXPW; ++u) {
+    const int p = p0 + u;
+    if (u < ppw && p < npairs) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg)
+        partials[((int64_t)block * npairs + p) * 256 + (row + reg) * 16 + m] =
+            acc[u][reg];
+    }
+  }
+}
+
+__global__ __launch_bounds__(THREADS) void gram_reduce_sr_kernel(
+    const float *partials, int nblocks, int npairs, const int *pair_i,
+    const int *pair_j, int k, float *gram) {
+  const int pair = blockIdx.x;
+  const int c = threadIdx.x;
+  double s = 0.0;
+  for (int ch = 0; ch < nblocks; ++ch)
+    s += (double)partials[((int64_t)ch * npairs + pair) * 256 + c];
+  const int row = c >> 4, col = c & 15;
+  const int gi = pair_i[pair] * 16 + row;
+  const int gj = pair_j[pair] * 16 + col;
+  if (gi < k && gj < k) {
+    gram[(int64_t)gi * k + gj] = (float)s;
+    gram[(int64_t)gj * k + gi] = (float)s;
+  }
+}
+
+extern "C" int anovos_centered_gram_sr(const void *const *cols, int64_t n,
+                                       int k, const float *means,
+                                       const int *pair_i, const int *pair_j,
+                                       int npairs, int row_chunks,
+                                       float *partials, float *gram,
+                                       hipStream_t stream) {
+  const int kt = (k + 15) / 16;
+  const int ktot = kt * 16;
+  const size_t lds = (size_t)4 * ktot * 16;
+  if (kt <= 10)
+    hipLaunchKernelGGL((gram_singleread_kernel<14>), dim3(row_chunks),
+                       dim3(THREADS), lds, stream, (const float *const *)cols,
+                       n, k, ktot, means, pair_i, pair_j, npairs, row_chunks,
+                       partials);
+  else if (kt <= 13)
+    hipLaunchKernelGGL((gram_singleread_kernel<23>), dim3(row_chunks),
+                       dim3(THREADS), lds, stream, (const float *const *)cols,
+                       n, k, ktot, means, pair_i, pair_j, npairs, row_chunks,
+                       partials);
+  else
+    return -2;  // caller dispatches the pair-parallel kernel instead
+  hipLaunchKernelGGL(gram_reduce_sr_kernel, dim3((uint32_t)npairs),
+                     dim3(THREADS), 0, stream, partials, row_chunks, npairs,
+                     pair_i, pair_j, k, gram);
+  return (int)hipGetLastError();
+}
