@@ -119,6 +119,57 @@ def _encoded_label_counts(ctx, idf, cols, label_col, event_label, label, encodin
     hit = idf.aux_cache.get(key)
     if hit is not None:
         return hit
+    from anovos_amd.ops import backend as _backend
+
+    use_fused = (
+        bool(encoding_configs)
+        and encoding_configs.get("monotonicity_check", 0) == 0
+        and len(num_cols) > 0
+        and idf.device.type == "cuda"
+        and all(idf.col(c).data.is_cuda for c in cols)
+        and _backend.use_hip(idf.col(num_cols[0]).data)
+        and all(
+            len(idf.col(c).dictionary or []) + 1 <= 8192
+            for c in cols
+            if idf.col(c).kind == "categorical"
+        )
+    )
+    if use_fused:
+        # fused K6+K9: place raw values against the cutoffs inside the
+        # counting kernel — the binned frame is never materialized
+        from anovos_amd.data_transformer.transformers import compute_bin_cutoffs
+
+        bin_size = int(encoding_configs["bin_size"])
+        bin_method = encoding_configs["bin_method"]
+        kept, cutlists = compute_bin_cutoffs(ctx, idf, num_cols, bin_method, bin_size)
+        cutmap = dict(zip(kept, cutlists))
+        numset = set(num_cols)
+        ncols_order = [c for c in cols if c in numset]
+        ccols_order = [c for c in cols if c not in numset]
+        ext = _backend.hip_ext()
+        lab_u8 = label.to(torch.uint8).contiguous()
+        nsizes = [bin_size + 2] * len(ncols_order)
+        tensors = [idf.col(c).data.contiguous() for c in ncols_order]
+        cuts = [torch.tensor(cutmap.get(c, []), dtype=torch.float64) for c in ncols_order]
+        parts = [ext.bucketize_label_counts(tensors, cuts, lab_u8, nsizes)]
+        csizes = []
+        if ccols_order:
+            ctensors = []
+            for c in ccols_order:
+                t = idf.col(c).data
+                ctensors.append(t.contiguous() if t.dtype == torch.int32 else t.to(torch.int32).contiguous())
+            csizes = [len(idf.col(c).dictionary or []) + 1 for c in ccols_order]
+            parts.append(ext.label_counts_multi(ctensors, lab_u8, csizes))
+        flat = torch.cat(parts) if len(parts) > 1 else parts[0]
+        flat = dist.all_reduce_(flat.to(torch.float64), "sum").cpu().numpy()
+        out, off = {}, 0
+        for c, s in zip(ncols_order + ccols_order, nsizes + csizes):
+            tot = flat[off : off + s]
+            n1 = flat[off + s : off + 2 * s]
+            out[c] = (tot - n1, n1)
+            off += 2 * s
+        idf.aux_cache[key] = out
+        return out
     nmax = None
     if len(num_cols) > 0 and bool(encoding_configs):
         bin_size = encoding_configs["bin_size"]

@@ -85,6 +85,34 @@ def _finish_output(idf: AnovosFrame, odf: AnovosFrame, list_of_cols, postfix: st
 
 # ---------------- binning ----------------
 @traced
+def compute_bin_cutoffs(ctx, idf, list_of_cols, method_type, bin_size):
+    """Bin cutoffs exactly as attribute_binning computes them (reference
+    transformers.py:210-232): equal_frequency = j/bin_size quantiles
+    (rel_err 0.01, same float chain as the GK rank ceil), equal_range =
+    min + j*(max-min)/bin_size with all-null columns warned and dropped.
+    Returns (kept_cols, cutoff lists). Shared with the fused
+    bucketize+label-count IV/IG path (no binned materialization)."""
+    if method_type == "equal_frequency":
+        pctile_width = 1 / bin_size
+        probs = [j * pctile_width for j in range(1, bin_size)]
+        q = hist_ops.approx_quantiles(idf, list_of_cols, probs, rel_err=0.01)
+        return list_of_cols, [q[c] for c in list_of_cols]
+    moments = stats_ops.frame_moments(idf, list_of_cols)
+    bin_cutoffs = []
+    dropped = []
+    for c in list_of_cols:
+        m = moments[c]
+        if m.max != m.max:  # all-null column
+            dropped.append(c)
+            continue
+        w = (m.max - m.min) / bin_size
+        bin_cutoffs.append([m.min + j * w for j in range(1, bin_size)])
+    if dropped:
+        warnings.warn("Columns contains too much null values. Dropping " + ", ".join(dropped))
+        list_of_cols = [c for c in list_of_cols if c not in dropped]
+    return list_of_cols, bin_cutoffs
+
+
 def attribute_binning(
     ctx,
     idf,
@@ -121,27 +149,7 @@ def attribute_binning(
         cut_map = {a: list(p) for a, p in zip(dfm["attribute"], dfm["parameters"])}
         bin_cutoffs = [cut_map[c] for c in list_of_cols]
     else:
-        if method_type == "equal_frequency":
-            # reference computes j * (1/bin_size) — keep the same float
-            # chain so the GK rank ceil matches (transformers.py:210-215)
-            pctile_width = 1 / bin_size
-            probs = [j * pctile_width for j in range(1, bin_size)]
-            q = hist_ops.approx_quantiles(idf, list_of_cols, probs, rel_err=0.01)
-            bin_cutoffs = [q[c] for c in list_of_cols]
-        else:
-            moments = stats_ops.frame_moments(idf, list_of_cols)
-            bin_cutoffs = []
-            dropped = []
-            for c in list_of_cols:
-                m = moments[c]
-                if m.max != m.max:  # all-null column
-                    dropped.append(c)
-                    continue
-                w = (m.max - m.min) / bin_size
-                bin_cutoffs.append([m.min + j * w for j in range(1, bin_size)])
-            if dropped:
-                warnings.warn("Columns contains too much null values. Dropping " + ", ".join(dropped))
-                list_of_cols = [c for c in list_of_cols if c not in dropped]
+        list_of_cols, bin_cutoffs = compute_bin_cutoffs(ctx, idf, list_of_cols, method_type, bin_size)
         if model_path != "NA":
             dfm = pd.DataFrame({"attribute": list_of_cols, "parameters": bin_cutoffs})
             _save_model(dfm, model_path, "attribute_binning")

@@ -77,6 +77,12 @@ int anovos_centered_gram_sr(const void *const *cols, int64_t n, int k,
                             const float *means, const int *pair_i,
                             const int *pair_j, int npairs, int row_chunks,
                             float *partials, float *gram, hipStream_t stream);
+int anovos_bucketize_label_counts(const void *const *cols, const uint8_t *label,
+                                  const int64_t *lens, const double *cutflat,
+                                  const int64_t *cutoff_off, const int *cutoff_len,
+                                  const int64_t *offs, const int *sizes, int ncols,
+                                  int max_ncut, int max_slots, int nchunks,
+                                  int dtype, uint64_t *out, hipStream_t stream);
 int anovos_label_counts_multi(const void *const *cols, const uint8_t *label,
                               const int64_t *lens, const int64_t *offs,
                               const int *sizes, const int *dtypes, int ncols,
@@ -779,6 +785,73 @@ torch::Tensor label_counts_multi(std::vector<torch::Tensor> cols,
   return out;
 }
 
+// K6+K9 fused: bucketize against per-column cutoffs + label-conditioned
+// counts in one read of the RAW numeric columns (no binned
+// materialization). Returns flat int64: per column [slots] totals then
+// [slots] events at offsets in the ORIGINAL column order.
+torch::Tensor bucketize_label_counts(std::vector<torch::Tensor> cols,
+                                     std::vector<torch::Tensor> cutoffs,
+                                     torch::Tensor label,
+                                     std::vector<int64_t> sizes) {
+  TORCH_CHECK(!cols.empty(), "no columns");
+  TORCH_CHECK(cols.size() == cutoffs.size() && cols.size() == sizes.size(), "size mismatch");
+  TORCH_CHECK(label.is_contiguous() && label.scalar_type() == torch::kUInt8, "uint8 label required");
+  auto device = cols[0].device();
+  std::vector<int64_t> out_offs(cols.size());
+  int64_t total = 0;
+  for (size_t i = 0; i < cols.size(); ++i) {
+    TORCH_CHECK(sizes[i] >= 1 && sizes[i] <= 8192, "slot count out of LDS range");
+    TORCH_CHECK(cols[i].numel() == label.numel(), "column/label length mismatch");
+    out_offs[i] = total;
+    total += 2 * sizes[i];
+  }
+  auto out = torch::zeros({total}, torch::TensorOptions().dtype(torch::kInt64).device(device));
+  for (int pass = 0; pass < 2; ++pass) {
+    std::vector<int64_t> ptrs, lens, offs, coffs;
+    std::vector<double> flat;
+    std::vector<int> clens;
+    std::vector<int64_t> sz;
+    int max_ncut = 1, max_slots = 1;
+    for (size_t i = 0; i < cols.size(); ++i) {
+      if (dtype_code(cols[i]) != pass) continue;
+      TORCH_CHECK(cols[i].is_contiguous(), "contiguous columns required");
+      ptrs.push_back((int64_t)cols[i].data_ptr());
+      lens.push_back(cols[i].numel());
+      offs.push_back(out_offs[i]);
+      auto cc = cutoffs[i].to(torch::kFloat64).cpu().contiguous();
+      coffs.push_back((int64_t)flat.size());
+      const double *cd = cc.data_ptr<double>();
+      flat.insert(flat.end(), cd, cd + cc.numel());
+      clens.push_back((int)cc.numel());
+      sz.push_back(sizes[i]);
+      max_ncut = std::max(max_ncut, (int)cc.numel());
+      max_slots = std::max(max_slots, (int)sizes[i]);
+    }
+    if (ptrs.empty()) continue;
+    int ncols = (int)ptrs.size();
+    int64_t maxn = *std::max_element(lens.begin(), lens.end());
+    int nchunks = pick_chunks(maxn, ncols);
+    auto dptr = to_device_i64(ptrs, device);
+    auto dlen = to_device_i64(lens, device);
+    auto doff = to_device_i64(offs, device);
+    auto dcoff = to_device_i64(coffs, device);
+    auto dflat = torch::from_blob(flat.data(), {(int64_t)std::max<size_t>(flat.size(), 1)},
+                                  torch::TensorOptions().dtype(torch::kFloat64)).clone().to(device);
+    auto dclen = torch::from_blob(clens.data(), {(int64_t)clens.size()},
+                                  torch::TensorOptions().dtype(torch::kInt32)).clone().to(device);
+    auto dsz = to_device_i64(sz, device).to(torch::kInt32);
+    check_hip(anovos_bucketize_label_counts(
+                  (const void *const *)dptr.data_ptr<int64_t>(),
+                  label.data_ptr<uint8_t>(), dlen.data_ptr<int64_t>(),
+                  dflat.data_ptr<double>(), dcoff.data_ptr<int64_t>(),
+                  dclen.data_ptr<int>(), doff.data_ptr<int64_t>(),
+                  dsz.data_ptr<int>(), ncols, max_ncut, max_slots, nchunks,
+                  pass, (uint64_t *)out.data_ptr<int64_t>(), current_stream()),
+              "anovos_bucketize_label_counts");
+  }
+  return out;
+}
+
 // K10/K11 fused: outlier counts + clamp/null treatment in one launch.
 // mode: 0 count only, 1 clamp to bounds, 2 null-out. Returns
 // (counts [k,2] int64, outs list — empty when mode==0).
@@ -888,6 +961,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("code_counts_multi", &code_counts_multi, "fused multi-column code counts + null slot (K5)");
   m.def("outlier_clamp_columns", &outlier_clamp_columns, "fused outlier count/clamp (K10/K11)");
   m.def("label_counts_multi", &label_counts_multi, "fused multi-column label-conditioned histograms (K9)");
+  m.def("bucketize_label_counts", &bucketize_label_counts, "fused bucketize + label counts, no binned materialization (K6+K9)");
   m.def("moments_hll", &moments_hll, "fused moments + HLL registers (K1/K2+K4)",
         py::arg("cols"), py::arg("p"), py::arg("shifts") = std::vector<double>());
   m.def("centered_gram_bf16", &centered_gram_bf16, "bf16 MFMA centered Gram X^T X (K8)");

@@ -1542,3 +1542,143 @@ extern "C" int anovos_label_counts_multi(
                      sizes, dtypes, nchunks, out);
   return (int)hipGetLastError();
 }
+
+// ------------------------------------------------------------------
+// K6+K9 (fused): bucketize + label-conditioned counts WITHOUT
+// materializing the binned column. IV/IG only consume per-bin label
+// counts (reference association_evaluator.py:368-409), so binning the
+// frame (75 GB write + re-read at the bench shard) is pure waste: this
+// kernel reads the RAW numeric column once, places each value against
+// the cutoffs (identical placement to bucketize_float_kernel, incl.
+// the nextafter-adjusted f32 fast path), and accumulates (total,event)
+// per slot. Slot mapping matches the python fallback over a binned
+// column: NaN -> 0, bin b=lo+1 -> slot b+1. Layout per column at
+// out[off]: [slots] totals then [slots] events.
+// ------------------------------------------------------------------
+template <typename T>
+__global__ __launch_bounds__(THREADS) void bucketize_label_counts_kernel(
+    const T *const *cols, const uint8_t *__restrict__ label,
+    const int64_t *lens, const double *cutflat, const int64_t *cutoff_off,
+    const int *cutoff_len, const int64_t *offs, const int *sizes,
+    int max_ncut, int nchunks, uint64_t *out) {
+  extern __shared__ double smem[];
+  const int col = blockIdx.x / nchunks;
+  const int chunk = blockIdx.x % nchunks;
+  const int ncut = cutoff_len[col];
+  const int slots = sizes[col];
+  double *cuts = smem;  // [max_ncut] doubles (f32 path reuses as floats)
+  float *cutsf = reinterpret_cast<float *>(cuts);
+  uint32_t *cnt = reinterpret_cast<uint32_t *>(smem + max_ncut);
+  uint32_t *tot = cnt;
+  uint32_t *evt = cnt + slots;
+  const double *src = &cutflat[cutoff_off[col]];
+  if (sizeof(T) == 4) {
+    for (int i = threadIdx.x; i < ncut; i += THREADS) {
+      double c = src[i];
+      float cf = (float)c;
+      if ((double)cf > c) cf = nextafterf(cf, -(float)INFINITY);
+      cutsf[i] = cf;
+    }
+  } else {
+    for (int i = threadIdx.x; i < ncut; i += THREADS) cuts[i] = src[i];
+  }
+  for (int i = threadIdx.x; i < 2 * slots; i += THREADS) cnt[i] = 0;
+  __syncthreads();
+
+  const T *__restrict__ x = cols[col];
+  const int64_t n = lens[col];
+  const int64_t per = (n + nchunks - 1) / nchunks;
+  const int64_t s = (int64_t)chunk * per;
+  const int64_t e = min(n, s + per);
+
+  auto slot_f32 = [&](float v) -> int {
+    if (v != v) return 0;
+    int lo = 0;
+    if (ncut <= 32) {
+      for (int j = 0; j < ncut; ++j) lo += (cutsf[j] < v) ? 1 : 0;
+    } else {
+      int len = ncut;
+      while (len > 0) {
+        int half = len >> 1;
+        int mid = lo + half;
+        lo = (cutsf[mid] < v) ? (mid + 1) : lo;
+        len = (cutsf[mid] < v) ? (len - half - 1) : half;
+      }
+    }
+    const int sl = lo + 2;  // bin (lo+1) -> slot bin+1
+    return sl < slots ? sl : slots - 1;
+  };
+  auto slot_f64 = [&](double v) -> int {
+    if (v != v) return 0;
+    int lo = 0;
+    if (ncut <= 32) {
+      for (int j = 0; j < ncut; ++j) lo += (cuts[j] < v) ? 1 : 0;
+    } else {
+      int len = ncut;
+      while (len > 0) {
+        int half = len >> 1;
+        int mid = lo + half;
+        lo = (cuts[mid] < v) ? (mid + 1) : lo;
+        len = (cuts[mid] < v) ? (len - half - 1) : half;
+      }
+    }
+    const int sl = lo + 2;
+    return sl < slots ? sl : slots - 1;
+  };
+
+  if (sizeof(T) == 4) {
+    const int64_t nv = (e - s) / 4;
+    const nat_f4 *__restrict__ xv = reinterpret_cast<const nat_f4 *>((const float *)x + s);
+    for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
+      nat_f4 v = __builtin_nontemporal_load(&xv[i]);
+      const int64_t r = s + i * 4;
+      const float vv[4] = {v.x, v.y, v.z, v.w};
+#pragma unroll
+      for (int k2 = 0; k2 < 4; ++k2) {
+        const int slot = slot_f32(vv[k2]);
+        atomicAdd(&tot[slot], 1u);
+        if (label[r + k2]) atomicAdd(&evt[slot], 1u);
+      }
+    }
+    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) {
+      const int slot = slot_f32((float)x[i]);
+      atomicAdd(&tot[slot], 1u);
+      if (label[i]) atomicAdd(&evt[slot], 1u);
+    }
+  } else {
+    for (int64_t i = s + threadIdx.x; i < e; i += THREADS) {
+      const int slot = slot_f64((double)x[i]);
+      atomicAdd(&tot[slot], 1u);
+      if (label[i]) atomicAdd(&evt[slot], 1u);
+    }
+  }
+  __syncthreads();
+  uint64_t *base = out + offs[col];
+  for (int i = threadIdx.x; i < slots; i += THREADS) {
+    if (tot[i]) atomicAdd((unsigned long long *)&base[i], (unsigned long long)tot[i]);
+    if (evt[i]) atomicAdd((unsigned long long *)&base[slots + i], (unsigned long long)evt[i]);
+  }
+}
+
+extern "C" int anovos_bucketize_label_counts(
+    const void *const *cols, const uint8_t *label, const int64_t *lens,
+    const double *cutflat, const int64_t *cutoff_off, const int *cutoff_len,
+    const int64_t *offs, const int *sizes, int ncols, int max_ncut,
+    int max_slots, int nchunks, int dtype, uint64_t *out,
+    hipStream_t stream) {
+  size_t lds = (size_t)max_ncut * sizeof(double) +
+               (size_t)max_slots * 2 * sizeof(uint32_t);
+  if (dtype == 0)
+    hipLaunchKernelGGL(bucketize_label_counts_kernel<float>,
+                       dim3(ncols * nchunks), dim3(THREADS), lds, stream,
+                       (const float *const *)cols, label, lens, cutflat,
+                       cutoff_off, cutoff_len, offs, sizes, max_ncut, nchunks,
+                       out);
+  else
+    hipLaunchKernelGGL(bucketize_label_counts_kernel<double>,
+                       dim3(ncols * nchunks), dim3(THREADS), lds, stream,
+                       (const double *const *)cols, label, lens, cutflat,
+                       cutoff_off, cutoff_len, offs, sizes, max_ncut, nchunks,
+                       out);
+  return (int)hipGetLastError();
+}
