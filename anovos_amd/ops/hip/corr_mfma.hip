@@ -169,18 +169,25 @@ extern "C" int anovos_centered_gram(const void *const *cols, int64_t n, int k,
 // ------------------------------------------------------------------
 
 typedef short bf16x4_t __attribute__((ext_vector_type(4)));
+typedef float nat_f4c __attribute__((ext_vector_type(4)));
 
+// STAGES MFMA k-steps (32 rows each) are staged per barrier round: each
+// column contributes STAGES*128 contiguous bytes per round (DRAM-page
+// friendly — with 32-row rounds every column read was a lone 128 B
+// touch on a distinct page and measured ~1.5 TB/s; 512 B touches
+// amortize row activation), and barrier count drops 4x.
 template <int MAXPW>
 __global__ __launch_bounds__(THREADS) void gram_singleread_kernel(
     const float *const *cols, int64_t n, int k, int ktot, const float *means,
     const int *pair_i, const int *pair_j, int npairs, int row_chunks,
     float *partials) {
-  extern __shared__ short slab[];  // [4][ktot][8] bf16
+  constexpr int STAGES = 4;  // 128-row macro-slab
+  extern __shared__ short slab[];  // [STAGES][4][ktot][8] bf16
   const int block = blockIdx.x;
-  const int64_t steps_total = (n + 31) / 32;
-  const int64_t steps_per = (steps_total + row_chunks - 1) / row_chunks;
-  const int64_t step_s = (int64_t)block * steps_per;
-  const int64_t step_e = min(steps_total, step_s + steps_per);
+  const int64_t macro_total = (n + STAGES * 32 - 1) / (STAGES * 32);
+  const int64_t macro_per = (macro_total + row_chunks - 1) / row_chunks;
+  const int64_t mac_s = (int64_t)block * macro_per;
+  const int64_t mac_e = min(macro_total, mac_s + macro_per);
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -201,19 +208,20 @@ __global__ __launch_bounds__(THREADS) void gram_singleread_kernel(
 #pragma unroll
   for (int u = 0; u < MAXPW; ++u) acc[u] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  for (int64_t s = step_s; s < step_e; ++s) {
-    const int64_t r0 = s * 32;
-    __syncthreads();  // previous iteration's LDS reads complete
-    for (int idx = threadIdx.x; idx < ktot * 8; idx += THREADS) {
-      const int c = idx >> 3;
-      const int seg = idx & 7;  // 4-row segment of the 32-row slab
+  for (int64_t s = mac_s; s < mac_e; ++s) {
+    const int64_t r0 = s * (STAGES * 32);
+    __syncthreads();  // previous round's LDS reads complete
+    for (int idx = threadIdx.x; idx < ktot * 8 * STAGES; idx += THREADS) {
+      const int c = idx >> (3 + 2);            // / (8*STAGES)
+      const int seg = idx & (8 * STAGES - 1);  // 4-row segment in the macro-slab
       const int64_t r = r0 + (int64_t)seg * 4;
       float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
       if (c < k) {
         const float *__restrict__ x = cols[c];
         const float mean = means[c];
         if (r + 4 <= n) {
-          const float4 t = *reinterpret_cast<const float4 *>(x + r);
+          const nat_f4c t = __builtin_nontemporal_load(
+              reinterpret_cast<const nat_f4c *>(x + r));
           v0 = t.x - mean; v1 = t.y - mean; v2 = t.z - mean; v3 = t.w - mean;
         } else {
           v0 = (r + 0 < n) ? x[r + 0] - mean : 0.f;
@@ -227,18 +235,25 @@ __global__ __launch_bounds__(THREADS) void gram_singleread_kernel(
       pack.y = to_bf16(isnan(v1) ? 0.f : v1);
       pack.z = to_bf16(isnan(v2) ? 0.f : v2);
       pack.w = to_bf16(isnan(v3) ? 0.f : v3);
-      short *dst = slab + ((size_t)(seg >> 1) * ktot + c) * 8 + (size_t)(seg & 1) * 4;
+      const int st = seg >> 3;        // stage (32-row step) 0..STAGES-1
+      const int si = seg & 7;         // 4-row segment within the stage
+      short *dst = slab + ((size_t)st * 4 * ktot + (size_t)(si >> 1) * ktot + c) * 8 +
+                   (size_t)(si & 1) * 4;
       *reinterpret_cast<bf16x4_t *>(dst) = pack;
     }
     __syncthreads();
 #pragma unroll
-    for (int u = 0; u < MAXPW; ++u) {
-      if (i0s[u] >= 0) {
-        const bf16x8 a = *reinterpret_cast<const bf16x8 *>(
-            slab + ((size_t)kog * ktot + i0s[u] + m) * 8);
-        const bf16x8 b = *reinterpret_cast<const bf16x8 *>(
-            slab + ((size_t)kog * ktot + j0s[u] + m) * 8);
-        acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[u], 0, 0, 0);
+    for (int st = 0; st < STAGES; ++st) {
+      const short *sbase = slab + (size_t)st * 4 * ktot * 8;
+#pragma unroll
+      for (int u = 0; u < MAXPW; ++u) {
+        if (i0s[u] >= 0) {
+          const bf16x8 a = *reinterpret_cast<const bf16x8 *>(
+              sbase + ((size_t)kog * ktot + i0s[u] + m) * 8);
+          const bf16x8 b = *reinterpret_cast<const bf16x8 *>(
+              sbase + ((size_t)kog * ktot + j0s[u] + m) * 8);
+          acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[u], 0, 0, 0);
+        }
       }
     }
   }
@@ -283,7 +298,7 @@ extern "C" int anovos_centered_gram_sr(const void *const *cols, int64_t n,
                                        hipStream_t stream) {
   const int kt = (k + 15) / 16;
   const int ktot = kt * 16;
-  const size_t lds = (size_t)4 * ktot * 16;
+  const size_t lds = (size_t)4 /*STAGES*/ * 4 * ktot * 16;
   if (kt <= 10)
     hipLaunchKernelGGL((gram_singleread_kernel<14>), dim3(row_chunks),
                        dim3(THREADS), lds, stream, (const float *const *)cols,
