@@ -181,7 +181,7 @@ __global__ __launch_bounds__(THREADS) void gram_singleread_kernel(
     const float *const *cols, int64_t n, int k, int ktot, const float *means,
     const int *pair_i, const int *pair_j, int npairs, int row_chunks,
     float *partials) {
-  constexpr int STAGES = 4;  // 128-row macro-slab
+  constexpr int STAGES = 8;  // 256-row macro-slab (1 KB per column per round)
   extern __shared__ short slab[];  // [STAGES][4][ktot][8] bf16
   const int block = blockIdx.x;
   const int64_t macro_total = (n + STAGES * 32 - 1) / (STAGES * 32);
@@ -212,8 +212,8 @@ __global__ __launch_bounds__(THREADS) void gram_singleread_kernel(
     const int64_t r0 = s * (STAGES * 32);
     __syncthreads();  // previous round's LDS reads complete
     for (int idx = threadIdx.x; idx < ktot * 8 * STAGES; idx += THREADS) {
-      const int c = idx >> (3 + 2);            // / (8*STAGES)
-      const int seg = idx & (8 * STAGES - 1);  // 4-row segment in the macro-slab
+      const int c = idx / (8 * STAGES);
+      const int seg = idx % (8 * STAGES);  // 4-row segment in the macro-slab
       const int64_t r = r0 + (int64_t)seg * 4;
       float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
       if (c < k) {
@@ -298,18 +298,31 @@ extern "C" int anovos_centered_gram_sr(const void *const *cols, int64_t n,
                                        hipStream_t stream) {
   const int kt = (k + 15) / 16;
   const int ktot = kt * 16;
-  const size_t lds = (size_t)4 /*STAGES*/ * 4 * ktot * 16;
-  if (kt <= 10)
+  const size_t lds = (size_t)8 /*STAGES*/ * 4 * ktot * 16;
+  // > 64 KB dynamic LDS needs the explicit opt-in (MI355X has 160 KB/CU)
+  if (kt <= 10) {
+    static int attr_done10 = 0;
+    if (!attr_done10) {
+      hipFuncSetAttribute((const void *)&gram_singleread_kernel<14>,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+      attr_done10 = 1;
+    }
     hipLaunchKernelGGL((gram_singleread_kernel<14>), dim3(row_chunks),
                        dim3(THREADS), lds, stream, (const float *const *)cols,
                        n, k, ktot, means, pair_i, pair_j, npairs, row_chunks,
                        partials);
-  else if (kt <= 13)
+  } else if (kt <= 13) {
+    static int attr_done13 = 0;
+    if (!attr_done13) {
+      hipFuncSetAttribute((const void *)&gram_singleread_kernel<23>,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+      attr_done13 = 1;
+    }
     hipLaunchKernelGGL((gram_singleread_kernel<23>), dim3(row_chunks),
                        dim3(THREADS), lds, stream, (const float *const *)cols,
                        n, k, ktot, means, pair_i, pair_j, npairs, row_chunks,
                        partials);
-  else
+  } else
     return -2;  // caller dispatches the pair-parallel kernel instead
   hipLaunchKernelGGL(gram_reduce_sr_kernel, dim3((uint32_t)npairs),
                      dim3(THREADS), 0, stream, partials, row_chunks, npairs,
