@@ -108,12 +108,21 @@ def set_corpus_path(path: Optional[str]):
     _STATE.clear()
 
 
+def _default_corpus_path() -> Optional[str]:
+    """The full 1,085-entry corpus shipped with the package (carried over
+    from the reference's data/feature_recommender/flatten_fr_db.csv —
+    see data/README.md for provenance)."""
+    p = os.path.join(os.path.dirname(os.path.abspath(__file__)), "data", "flatten_fr_db.csv")
+    return p if os.path.exists(p) else None
+
+
 def init_input_fer() -> pd.DataFrame:
     """Reference featrec_init.py — load the flattened feature corpus."""
     if "df" in _STATE:
         return _STATE["df"]
-    if _CORPUS_PATH:
-        df = pd.read_csv(_CORPUS_PATH)
+    path = _CORPUS_PATH or _default_corpus_path()
+    if path:
+        df = pd.read_csv(path)
     else:
         df = pd.DataFrame(_SEED_CORPUS, columns=["Feature Name", "Feature Description", "Industry", "Usecase"])
     df["Industry"] = df["Industry"].astype(str).str.strip().str.lower()
@@ -186,9 +195,60 @@ class TfidfEmbedder:
         return m / np.maximum(n, 1e-12)
 
 
+class TransformersEmbedder:
+    """Optional LOCAL semantic-model path: mean-pooled hidden states from
+    a transformers checkpoint on disk (e.g. a pre-downloaded
+    all-mpnet-base-v2). Used automatically when `detect_model_path()`
+    exists (or ANOVOS_FR_MODEL_PATH points at a model dir); there is no
+    egress in this stack so nothing is ever downloaded. TF-IDF remains
+    the default offline backend."""
+
+    def __init__(self, model_path: str):
+        from transformers import AutoModel, AutoTokenizer  # local files only
+
+        self._tok = AutoTokenizer.from_pretrained(model_path, local_files_only=True)
+        self._model = AutoModel.from_pretrained(model_path, local_files_only=True)
+        self._model.eval()
+
+    def fit(self, texts):
+        return self
+
+    def encode(self, texts: List[str]) -> np.ndarray:
+        import torch as _t
+
+        outs = []
+        with _t.no_grad():
+            for i in range(0, len(texts), 64):
+                batch = self._tok(texts[i : i + 64], padding=True, truncation=True,
+                                  max_length=128, return_tensors="pt")
+                h = self._model(**batch).last_hidden_state
+                mask = batch["attention_mask"].unsqueeze(-1).float()
+                emb = (h * mask).sum(1) / mask.sum(1).clamp(min=1e-9)
+                outs.append(emb.cpu().numpy())
+        m = np.concatenate(outs)
+        n = np.linalg.norm(m, axis=1, keepdims=True)
+        return m / np.maximum(n, 1e-12)
+
+
+def _local_model_path() -> Optional[str]:
+    p = os.getenv("ANOVOS_FR_MODEL_PATH") or detect_model_path()
+    return p if p and os.path.isdir(p) else None
+
+
+def get_embedder(corpus_texts: List[str], query_texts: List[str]):
+    """Semantic model when a local checkpoint exists, else TF-IDF."""
+    mp = _local_model_path()
+    if mp:
+        try:
+            return TransformersEmbedder(mp)
+        except Exception:
+            pass
+    return TfidfEmbedder().fit(list(corpus_texts) + list(query_texts))
+
+
 def semantic_search(query_texts: List[str], corpus_texts: List[str], top_k: int = 5):
     """util.semantic_search equivalent: per query, top_k (idx, score)."""
-    emb = TfidfEmbedder().fit(list(corpus_texts) + list(query_texts))
+    emb = get_embedder(corpus_texts, query_texts)
     q = emb.encode(list(query_texts))
     c = emb.encode(list(corpus_texts))
     sims = q @ c.T

@@ -37,11 +37,37 @@ def list_all_pair() -> pd.DataFrame:
     return pairs
 
 
+def _lcs_len(a: str, b: str) -> int:
+    """Longest common subsequence length (both strings are short)."""
+    dp = [0] * (len(b) + 1)
+    for ch in a:
+        prev = 0
+        for j, cb in enumerate(b, 1):
+            cur = dp[j]
+            dp[j] = prev + 1 if ch == cb else max(dp[j], dp[j - 1])
+            prev = cur
+    return dp[len(b)]
+
+
 def _best_match(value: str, options) -> Tuple[str, float]:
-    hits = semantic_search([str(value).strip().lower()], list(options), top_k=1)[0]
+    """Nearest option by TF-IDF cosine PLUS a subsequence bonus: the
+    reference's sentence model snaps abbreviations ('telco' ->
+    'telecommunication', 'bank' -> 'banking ...') that character
+    n-grams alone under-score; LCS(query, option)/len(query) recovers
+    exactly that class of match while nonsense strings stay low."""
+    value = str(value).strip().lower()
+    options = list(options)
+    hits = semantic_search([value], options, top_k=len(options))[0]
     if not hits:
         return value, 0.0
-    return list(options)[hits[0]["corpus_id"]], hits[0]["score"]
+    best, best_score = value, 0.0
+    for h in hits:
+        opt = options[h["corpus_id"]]
+        bonus = 0.3 * (_lcs_len(value, opt) / max(len(value), 1)) if len(value) >= 4 else 0.0
+        s = h["score"] + bonus
+        if s > best_score:
+            best, best_score = opt, s
+    return best, best_score
 
 
 def process_usecase(usecase: str, semantic: bool = True) -> str:
