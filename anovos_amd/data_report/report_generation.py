@@ -354,7 +354,19 @@ def anovos_report(master_path: str, id_col="", label_col=None, corr_threshold=0.
                   run_type="local", auth_key="NA", output_type=None,
                   sg_print_impact=False, **kwargs) -> str:
     """Reference report_generation.py:3984 — assemble ml_anovos_report.html
-    from the on-disk stats + chart objects. Returns the report path."""
+    from the on-disk stats + chart objects. Returns the report path.
+    run_type routing mirrors reference :4386-4413: databricks resolves
+    dbfs:/ to the fuse mount; emr/ak8s write locally then push the html
+    via aws s3 cp / azcopy."""
+    from anovos_amd.shared import utils as _su
+
+    if run_type == "databricks":
+        final_report_path = _su.output_to_local(final_report_path)
+        master_path = _su.output_to_local(master_path)
+    cloud_target = None
+    if run_type in ("emr", "ak8s"):
+        cloud_target = final_report_path
+        final_report_path = "."
     os.makedirs(final_report_path, exist_ok=True)
     tabs = [
         ("Executive Summary", executive_summary_gen(master_path, label_col, event_label)),
@@ -376,6 +388,8 @@ def anovos_report(master_path: str, id_col="", label_col=None, corr_threshold=0.
     path = ends_with(final_report_path) + "ml_anovos_report.html"
     with open(path, "w") as f:
         f.write(out)
+    if cloud_target is not None:
+        _su.cloud_sync(path, ends_with(cloud_target) + "ml_anovos_report.html", run_type, auth_key)
     return path
 
 

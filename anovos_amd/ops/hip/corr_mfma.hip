@@ -176,7 +176,7 @@ typedef float nat_f4c __attribute__((ext_vector_type(4)));
 // friendly — with 32-row rounds every column read was a lone 128 B
 // touch on a distinct page and measured ~1.5 TB/s; 512 B touches
 // amortize row activation), and barrier count drops 4x.
-template <int MAXPW, int MAXLOAD>
+template <int MAXPW>
 __global__ __launch_bounds__(THREADS) void gram_singleread_kernel(
     const float *const *cols, int64_t n, int k, int ktot, const float *means,
     const int *pair_i, const int *pair_j, int npairs, int row_chunks,
@@ -208,71 +208,40 @@ __global__ __launch_bounds__(THREADS) void gram_singleread_kernel(
 #pragma unroll
   for (int u = 0; u < MAXPW; ++u) acc[u] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int nidx = ktot * 8 * STAGES;
-
-  // software pipeline: round r+1's global loads are ISSUED (into
-  // registers) before round r's MFMA phase, so memory latency is hidden
-  // by compute inside the block — the s_waitcnt lands at the next
-  // store_round, after a full compute phase
-  nat_f4c regs[MAXLOAD];
-  auto load_round = [&](int64_t r0) {
-#pragma unroll
-    for (int u = 0; u < MAXLOAD; ++u) {
-      const int idx = threadIdx.x + u * THREADS;
-      nat_f4c t = {0.f, 0.f, 0.f, 0.f};
-      if (idx < nidx) {
-        const int c = idx / (8 * STAGES);
-        const int64_t r = r0 + (int64_t)(idx % (8 * STAGES)) * 4;
-        if (c < k) {
-          const float *__restrict__ x = cols[c];
-          if (r + 4 <= n) {
-            t = __builtin_nontemporal_load(
-                reinterpret_cast<const nat_f4c *>(x + r));
-          } else {
-            // pad with NaN: the store phase maps NaN -> 0 AFTER the mean
-            // subtraction, so padding contributes exactly zero
-            t.x = (r + 0 < n) ? x[r + 0] : nanf("");
-            t.y = (r + 1 < n) ? x[r + 1] : nanf("");
-            t.z = (r + 2 < n) ? x[r + 2] : nanf("");
-            t.w = (r + 3 < n) ? x[r + 3] : nanf("");
-          }
+  for (int64_t s = mac_s; s < mac_e; ++s) {
+    const int64_t r0 = s * (STAGES * 32);
+    __syncthreads();  // previous round's LDS reads complete
+    for (int idx = threadIdx.x; idx < ktot * 8 * STAGES; idx += THREADS) {
+      const int c = idx >> (3 + 2);            // / (8*STAGES)
+      const int seg = idx & (8 * STAGES - 1);  // 4-row segment in the macro-slab
+      const int64_t r = r0 + (int64_t)seg * 4;
+      float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
+      if (c < k) {
+        const float *__restrict__ x = cols[c];
+        const float mean = means[c];
+        if (r + 4 <= n) {
+          const nat_f4c t = __builtin_nontemporal_load(
+              reinterpret_cast<const nat_f4c *>(x + r));
+          v0 = t.x - mean; v1 = t.y - mean; v2 = t.z - mean; v3 = t.w - mean;
+        } else {
+          v0 = (r + 0 < n) ? x[r + 0] - mean : 0.f;
+          v1 = (r + 1 < n) ? x[r + 1] - mean : 0.f;
+          v2 = (r + 2 < n) ? x[r + 2] - mean : 0.f;
+          v3 = (r + 3 < n) ? x[r + 3] - mean : 0.f;
         }
       }
-      regs[u] = t;
+      bf16x4_t pack;
+      pack.x = to_bf16(isnan(v0) ? 0.f : v0);
+      pack.y = to_bf16(isnan(v1) ? 0.f : v1);
+      pack.z = to_bf16(isnan(v2) ? 0.f : v2);
+      pack.w = to_bf16(isnan(v3) ? 0.f : v3);
+      const int st = seg >> 3;        // stage (32-row step) 0..STAGES-1
+      const int si = seg & 7;         // 4-row segment within the stage
+      short *dst = slab + ((size_t)st * 4 * ktot + (size_t)(si >> 1) * ktot + c) * 8 +
+                   (size_t)(si & 1) * 4;
+      *reinterpret_cast<bf16x4_t *>(dst) = pack;
     }
-  };
-  auto store_round = [&]() {
-#pragma unroll
-    for (int u = 0; u < MAXLOAD; ++u) {
-      const int idx = threadIdx.x + u * THREADS;
-      if (idx < nidx) {
-        const int c = idx / (8 * STAGES);
-        const int seg = idx % (8 * STAGES);
-        const float mean = (c < k) ? means[c] : 0.f;
-        const nat_f4c t = regs[u];
-        const float v0 = t.x - mean, v1 = t.y - mean, v2 = t.z - mean,
-                    v3 = t.w - mean;
-        bf16x4_t pack;
-        pack.x = to_bf16(isnan(v0) ? 0.f : v0);
-        pack.y = to_bf16(isnan(v1) ? 0.f : v1);
-        pack.z = to_bf16(isnan(v2) ? 0.f : v2);
-        pack.w = to_bf16(isnan(v3) ? 0.f : v3);
-        const int st = seg >> 3;  // stage (32-row step)
-        const int si = seg & 7;   // 4-row segment within the stage
-        short *dst = slab +
-                     ((size_t)st * 4 * ktot + (size_t)(si >> 1) * ktot + c) * 8 +
-                     (size_t)(si & 1) * 4;
-        *reinterpret_cast<bf16x4_t *>(dst) = pack;
-      }
-    }
-  };
-
-  if (mac_s < mac_e) load_round(mac_s * (STAGES * 32));
-  for (int64_t s = mac_s; s < mac_e; ++s) {
-    __syncthreads();  // previous round's LDS reads complete
-    store_round();
     __syncthreads();
-    if (s + 1 < mac_e) load_round((s + 1) * (STAGES * 32));
 #pragma unroll
     for (int st = 0; st < STAGES; ++st) {
       const short *sbase = slab + (size_t)st * 4 * ktot * 8;
@@ -330,13 +299,13 @@ extern "C" int anovos_centered_gram_sr(const void *const *cols, int64_t n,
   const int kt = (k + 15) / 16;
   const int ktot = kt * 16;
   const size_t lds = (size_t)4 /*STAGES*/ * 4 * ktot * 16;
-  if (kt <= 10)  // MAXLOAD = ktot*32/256
-    hipLaunchKernelGGL((gram_singleread_kernel<14, 20>), dim3(row_chunks),
+  if (kt <= 10)
+    hipLaunchKernelGGL((gram_singleread_kernel<14>), dim3(row_chunks),
                        dim3(THREADS), lds, stream, (const float *const *)cols,
                        n, k, ktot, means, pair_i, pair_j, npairs, row_chunks,
                        partials);
   else if (kt <= 13)
-    hipLaunchKernelGGL((gram_singleread_kernel<23, 26>), dim3(row_chunks),
+    hipLaunchKernelGGL((gram_singleread_kernel<23>), dim3(row_chunks),
                        dim3(THREADS), lds, stream, (const float *const *)cols,
                        n, k, ktot, means, pair_i, pair_j, npairs, row_chunks,
                        partials);
