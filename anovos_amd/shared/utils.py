@@ -133,9 +133,11 @@ def cloud_sync(local_path: str, master_path: str, run_type: str = "local", auth_
     import subprocess
 
     if run_type == "emr":
-        cmd = ["aws", "s3", "cp"] + (["--recursive"] if recursive else []) + [ends_with(local_path) if recursive else local_path, ends_with(master_path)]
+        tgt = ends_with(master_path) if recursive else master_path
+        cmd = ["aws", "s3", "cp"] + (["--recursive"] if recursive else []) + [ends_with(local_path) if recursive else local_path, tgt]
         subprocess.check_output(cmd)
     elif run_type == "ak8s":
-        target = ends_with(path_ak8s_modify(master_path)) + (str(auth_key) if auth_key != "NA" else "")
+        base = ends_with(path_ak8s_modify(master_path)) if recursive else path_ak8s_modify(master_path)
+        target = base + (str(auth_key) if auth_key != "NA" else "")
         cmd = ["azcopy", "cp", local_path, target] + (["--recursive=true"] if recursive else [])
         subprocess.check_output(cmd)

@@ -143,6 +143,11 @@ def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, de
 
     ctx = init_context(device)
     start_main = time.time()
+    # cloud credential side-channel (reference workflow.py:153-158):
+    # the last value in auth_key_val is the auth key (e.g. azure SAS)
+    auth_key = "NA"
+    for _k, _v in (auth_key_val or {}).items():
+        auth_key = _v
     mlflow_config = mlflow_utils.setup_mlflow(all_configs.get("mlflow"))
     write_main = all_configs.get("write_main", None)
     write_intermediate = all_configs.get("write_intermediate", None)
@@ -246,16 +251,16 @@ def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, de
             out_path = report_input_path or "."
             if args.get("auto_detection", True):
                 df, ts_cols, num_cols, cat_cols = ts_auto_detection.ts_preprocess(
-                    ctx, df, id_col, out_path, tz_offset=args.get("tz_offset", "local"), run_type=run_type)
+                    ctx, df, id_col, out_path, tz_offset=args.get("tz_offset", "local"), run_type=run_type, auth_key=auth_key)
             if args.get("inspection", True):
                 # reference key analysis_level ∈ {daily, weekly, hourly}
                 ts_analyzer_mod.ts_analyzer(
                     ctx, df, id_col, args.get("max_days", 90), out_path,
                     output_type=args.get("analysis_level", args.get("output_type", "daily")),
-                    run_type=run_type)
+                    run_type=run_type, auth_key=auth_key)
 
         elif key == "anovos_basic_report" and args.get("basic_report", False):
-            anovos_basic_report(ctx, df, **(args.get("report_args", {}) or {}), run_type=run_type)
+            anovos_basic_report(ctx, df, **(args.get("report_args", {}) or {}), run_type=run_type, auth_key=auth_key)
             _log("anovos_basic_report completed — skipping remaining stages (reference workflow.py:468-486)")
             return df
 
@@ -264,7 +269,7 @@ def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, de
                 f = getattr(stats_generator, m)
                 stats = f(ctx, df, **args.get("metric_args", {}), print_impact=False)
                 if report_input_path:
-                    report_preprocessing.save_stats(ctx, stats, report_input_path, m, run_type=run_type)
+                    report_preprocessing.save_stats(ctx, stats, report_input_path, m, run_type=run_type, auth_key=auth_key)
                 if write_stats:
                     save_df = stats
                     wc = copy.deepcopy(write_stats)
@@ -288,7 +293,7 @@ def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, de
                 if new is not None:
                     df = new
                 if report_input_path:
-                    report_preprocessing.save_stats(ctx, df_stats, report_input_path, subkey, run_type=run_type)
+                    report_preprocessing.save_stats(ctx, df_stats, report_input_path, subkey, run_type=run_type, auth_key=auth_key)
                 _log(f"quality_checker.{subkey}: {time.time() - start:.3f}s")
                 start = time.time()
 
@@ -307,7 +312,7 @@ def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, de
                 else:
                     stats = f(ctx, df, **value, **extra)
                 if report_input_path:
-                    report_preprocessing.save_stats(ctx, stats, report_input_path, subkey, run_type=run_type)
+                    report_preprocessing.save_stats(ctx, stats, report_input_path, subkey, run_type=run_type, auth_key=auth_key)
                 _log(f"association_evaluator.{subkey}: {time.time() - start:.3f}s")
                 start = time.time()
 
@@ -324,7 +329,7 @@ def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, de
                         idf_source = None
                     stats = ddetector.statistics(ctx, df, idf_source, **configs)
                     if report_input_path:
-                        report_preprocessing.save_stats(ctx, stats, report_input_path, "drift_statistics", run_type=run_type)
+                        report_preprocessing.save_stats(ctx, stats, report_input_path, "drift_statistics", run_type=run_type, auth_key=auth_key)
                 elif subkey == "stability_index":
                     configs = copy.deepcopy(value.get("configs", {}))
                     idfs = []
@@ -332,7 +337,7 @@ def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, de
                         idfs.append(ETL(ctx, value[k]))
                     stats = dstability.stability_index_computation(ctx, *idfs, **configs)
                     if report_input_path:
-                        report_preprocessing.save_stats(ctx, stats, report_input_path, "stability_index", run_type=run_type)
+                        report_preprocessing.save_stats(ctx, stats, report_input_path, "stability_index", run_type=run_type, auth_key=auth_key)
                 _log(f"drift_detector.{subkey}: {time.time() - start:.3f}s")
                 start = time.time()
 
@@ -364,12 +369,12 @@ def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, de
                 if subkey == "charts_to_objects":
                     extra = stats_args(all_configs, subkey)
                     report_preprocessing.charts_to_objects(
-                        ctx, df, **value, **extra, master_path=args["master_path"], run_type=run_type)
+                        ctx, df, **value, **extra, master_path=args["master_path"], run_type=run_type, auth_key=auth_key)
                     _log(f"report_preprocessing.charts_to_objects: {time.time() - start:.3f}s")
                     start = time.time()
 
         elif key == "report_generation":
-            anovos_report(**args, run_type=run_type)
+            anovos_report(**args, run_type=run_type, auth_key=auth_key)
             _log(f"report_generation: {time.time() - start:.3f}s")
 
         elif key == "write_feast_features":
