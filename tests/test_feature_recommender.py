@@ -156,3 +156,31 @@ def test_feast_config_validation():
         feast_exporter.check_feast_configuration({}, 1)
     with pytest.raises(ValueError):
         feast_exporter.check_feast_configuration({"file_path": "x", "entity": {}, "file_source": {}, "feature_view": {}}, 2)
+
+
+def test_feature_retrieval_demo(tmp_path, monkeypatch):
+    """feature_retrieval demo (reference feature_retrieval.py:8): with
+    feast absent, the point-in-time fallback reads the exported parquet
+    source and returns one row per entity."""
+    import numpy as np
+
+    from anovos_amd.feature_store import feature_retrieval as fr
+
+    repo = tmp_path / "feast_repo"
+    repo.mkdir()
+    n = 6
+    pdf = pd.DataFrame(
+        {
+            "ifa": ["27a", "30a", "475a", "965a", "1678a", "zzz"],
+            "income": np.arange(n, dtype=float),
+            "event_time": pd.Timestamp("2020-01-01"),
+        }
+    )
+    pdf.to_parquet(repo / "income_features.parquet")
+    df, df2 = fr.retrieve_historical_feature_demo(
+        str(repo), entity_ids=["27a", "965a", "missing"], features=["income_view:income"]
+    )
+    assert list(df["ifa"]) == ["27a", "965a", "missing"]
+    assert float(df["income"].iloc[0]) == 0.0
+    assert float(df["income"].iloc[1]) == 3.0
+    assert pd.isna(df["income"].iloc[2])
