@@ -88,3 +88,87 @@ def test_autoencoder_latent_features(ctx, frame):
                                         output_mode="append")
     latents = [c for c in odf.columns if c.startswith("latent_")]
     assert len(latents) >= 1
+
+
+def test_imputation_sklearn_device_apply_matches_sklearn():
+    """The tensorized KNN / IterativeImputer apply (K15) must reproduce
+    sklearn's own .transform on the same fitted model (the round-1 host
+    loop was the reference; now it IS the device path run on CPU)."""
+    import numpy as np
+    import torch
+    from sklearn.experimental import enable_iterative_imputer  # noqa: F401
+    from sklearn.impute import IterativeImputer, KNNImputer
+
+    from anovos_amd.core.frame import AnovosFrame
+    from anovos_amd.data_transformer import transformers_advanced as TA
+    from anovos_amd.shared.context import init_context
+
+    rng = np.random.default_rng(3)
+    n = 4000
+    pdf = pd.DataFrame(
+        {
+            "a": rng.normal(10, 3, n),
+            "b": rng.normal(-5, 2, n),
+            "c": rng.normal(0, 1, n),
+            "d": rng.normal(100, 20, n),
+        }
+    )
+    pdf["b"] = pdf["a"] * 0.5 + rng.normal(0, 0.3, n)  # correlated for regression
+    for c in pdf.columns:
+        pdf.loc[rng.choice(n, 160, replace=False), c] = np.nan
+    ctx = init_context("cpu")
+    idf = AnovosFrame.from_pandas(pdf, device="cpu")
+    cols = list(pdf.columns)
+
+    for method, klass in (("KNN", KNNImputer), ("regression", IterativeImputer)):
+        out = TA.imputation_sklearn(ctx, idf, cols, method_type=method,
+                                    use_sampling=True, sample_size=1500, sample_seed=42)
+        # rebuild the same fitted model (same sample path) for the truth
+        got = np.column_stack([out.col(c).data.numpy() for c in cols])
+        X = pdf.to_numpy()
+        assert not np.isnan(got).any()
+        # fit on the same sampled data the engine used: easiest faithful
+        # check — refit on FULL data and compare against sklearn's own
+        # transform of the engine's model is internal; instead verify
+        # against sklearn transform with an identical fit sample
+        from anovos_amd.data_ingest.data_sampling import data_sample
+
+        sub = data_sample(idf.select(cols), strata_cols="all", fraction=1500 / n,
+                          method_type="random", stratified_type="population", seed_value=42)
+        local = np.column_stack([sub.col(c).data.numpy().astype("float64") for c in cols])
+        model = klass(n_neighbors=5) if method == "KNN" else klass(max_iter=10, random_state=42)
+        model.fit(local)
+        truth = model.transform(X)
+        assert np.allclose(got, truth.astype(np.float32), rtol=2e-4, atol=2e-3), (
+            method, np.nanmax(np.abs(got - truth)))
+
+
+def test_matrix_factorization_batched_solver():
+    """Row-batched ALS (no n*k*k materialization) still reconstructs a
+    low-rank matrix's missing entries."""
+    import numpy as np
+    import torch
+
+    from anovos_amd.core.frame import AnovosFrame
+    from anovos_amd.data_transformer import transformers_advanced as TA
+    from anovos_amd.shared.context import init_context
+
+    rng = np.random.default_rng(5)
+    n, m, k = 3000, 8, 3
+    U = rng.normal(0, 1, (n, k))
+    V = rng.normal(0, 1, (m, k))
+    X = U @ V.T
+    pdf = pd.DataFrame(X, columns=[f"c{j}" for j in range(m)])
+    mask = rng.random((n, m)) < 0.15
+    truth = pdf.copy()
+    pdf = pdf.mask(mask)
+    ctx = init_context("cpu")
+    idf = AnovosFrame.from_pandas(pdf, device="cpu")
+    out = TA.imputation_matrixFactorization(ctx, idf, "all", rank=5, max_iter=15)
+    err_num, err_den = 0.0, 0.0
+    for j in range(m):
+        got = out.col(f"c{j}").data.numpy()
+        sel = mask[:, j]
+        err_num += float(np.sum((got[sel] - truth.iloc[:, j].to_numpy()[sel]) ** 2))
+        err_den += float(np.sum(truth.iloc[:, j].to_numpy()[sel] ** 2))
+    assert err_num / max(err_den, 1e-9) < 0.05  # <5% relative MSE on held-out entries
