@@ -75,22 +75,5 @@ def global_keep_mask(uniq: torch.Tensor) -> torch.Tensor:
 
 
 def _exchange(send_parts, dev):
-    """all_to_all of variable-length 1-D tensors. NCCL uses the native
-    collective; gloo (CPU tests) emulates it with an object all-gather
-    (send everything, pick your column)."""
-    world = dist.world_size()
-    rank = dist.rank()
-    if td.get_backend() == "nccl":
-        counts = torch.tensor([p.numel() for p in send_parts], dtype=torch.int64, device=dev)
-        mat = [torch.zeros_like(counts) for _ in range(world)]
-        td.all_gather(mat, counts)
-        recv = [torch.empty(int(mat[r][rank]), dtype=send_parts[0].dtype, device=dev) for r in range(world)]
-        td.all_to_all(recv, [p.to(dev) for p in send_parts])
-        return recv
-    gathered = dist.all_gather_object([p.cpu().numpy() for p in send_parts])
-    out = []
-    for r in range(world):
-        import numpy as np
-
-        out.append(torch.from_numpy(np.ascontiguousarray(gathered[r][rank])).to(dev))
-    return out
+    """all_to_all of variable-length 1-D tensors (dist.all_to_all_tensor)."""
+    return [t.to(dev) for t in dist.all_to_all_tensor(list(send_parts))]

@@ -116,21 +116,15 @@ def discrete_modes(idf, cols: List[str]) -> Dict[str, Tuple[Optional[float], int
             out[c] = (float(val), cnt) if cnt > 0 else (None, 0)
     for c in rest:
         m = moments[c]
-        if m.n > 20_000_000 and not m.integral:
-            # continuous column at production scale: exact mode means a
-            # 64-bit sort + cross-rank value-count merge of ~n uniques —
-            # GBs of traffic for a statistic that is almost surely a
-            # 1-count tie. Approximate with the mode BIN of a fine
-            # histogram (midpoint), like Spark users bin first.
-            import warnings
-
-            warnings.warn(f"mode of continuous column '{c}' approximated by 4096-bin histogram at n={int(m.n)}")
-            lo1 = torch.tensor([m.min], dtype=torch.float64)
-            hi1 = torch.tensor([m.max], dtype=torch.float64)
-            h = hist_ops.global_histograms([idf.col(c).data], lo1, hi1, 4096).cpu().numpy()[0]
-            i = int(np.argmax(h))
-            w = (m.max - m.min) / 4096 if m.max > m.min else 0.0
-            out[c] = (float(m.min + (i + 0.5) * w), int(h[i]))
+        if dist.is_dist() and m.n > 1_000_000 and not m.integral:
+            # EXACT continuous mode at scale (replaces the r1 histogram
+            # approximation): local unique+count, then hash-partitioned
+            # all-to-all so each unique value crosses the fabric exactly
+            # once (not an all-gather of ~n values to every rank)
+            t = idf.col(c).data
+            x = t[~torch.isnan(t)]
+            vals, cnts = torch.unique(x, return_counts=True)
+            out[c] = _global_mode_partitioned(vals, cnts)
             continue
         vals, cnts = numeric_value_counts(idf, c)
         if vals.numel() == 0:
@@ -139,6 +133,35 @@ def discrete_modes(idf, cols: List[str]) -> Dict[str, Tuple[Optional[float], int
             i = int(torch.argmax(cnts).item())
             out[c] = (float(vals[i]), int(cnts[i]))
     return out
+
+
+def _global_mode_partitioned(vals: torch.Tensor, cnts: torch.Tensor):
+    """Exact global (mode, count) from per-rank local value counts via a
+    hash-partitioned exchange: rank q owns the values whose bit-hash maps
+    to q, merges their counts, picks its partition's top-1; the world's
+    top-1 is an all-gather of world_size tiny candidates. Deterministic
+    tie-break: highest count, then smallest value."""
+    ws = dist.world_size()
+    v64 = vals.to(torch.float64)
+    v64 = torch.where(v64 == 0, torch.zeros_like(v64), v64)  # canonicalize -0.0
+    part = _mix64(v64.view(torch.int64)).remainder(ws)
+    send_v = [v64[part == q] for q in range(ws)]
+    send_c = [cnts.to(torch.int64)[part == q] for q in range(ws)]
+    rv = torch.cat([t for t in dist.all_to_all_tensor(send_v)])
+    rc = torch.cat([t for t in dist.all_to_all_tensor(send_c)])
+    if rv.numel():
+        uv, inv = torch.unique(rv, return_inverse=True)
+        uc = torch.zeros(uv.numel(), dtype=torch.int64, device=rv.device)
+        uc.index_add_(0, inv, rc)
+        mx = int(uc.max())
+        cand = float(uv[uc == mx].min())
+        local_best = (mx, cand)
+    else:
+        local_best = (0, float("nan"))
+    best = max(dist.all_gather_object(local_best), key=lambda t: (t[0], -t[1] if t[1] == t[1] else float("-inf")))
+    if best[0] == 0:
+        return (None, 0)
+    return (best[1], best[0])
 
 
 def mode(idf, col: str, counts: Optional[torch.Tensor] = None) -> Tuple[Optional[str], int]:

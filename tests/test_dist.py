@@ -458,3 +458,49 @@ def test_dist_stable_moments_and_integral_quantiles():
     assert res["k_q"] == exact_k  # dense integral path is exact
     for got, p in zip(res["off_q"], (0.25, 0.5, 0.9)):
         assert got == pytest.approx(float(np.quantile(off_full, p)), abs=0.05)
+
+
+def _mode_worker(rank, port, out):
+    os.environ.update({"RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": "2",
+                       "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port)})
+    import torch.distributed as td
+
+    from anovos_amd.core import dist
+    from anovos_amd.core.frame import AnovosFrame, Column
+    from anovos_amd.ops import groupby
+    from anovos_amd.shared.context import init_context
+
+    dist.init_from_env(timeout_s=120)
+    init_context("cpu")
+    rng = np.random.default_rng(77)
+    n = 2_000_000  # above the partitioned-exchange threshold
+    full = rng.normal(0, 1, n)
+    # plant the true mode: one value repeated across BOTH shards
+    planted = 0.123456789
+    full[rng.choice(n, 500, replace=False)] = planted
+    half = n // 2
+    shard = full[rank * half : (rank + 1) * half]
+    idf = AnovosFrame({"x": Column("x", "double", torch.tensor(shard))}, device="cpu")
+    res = groupby.discrete_modes(idf, ["x"])["x"]
+    if rank == 0:
+        json.dump({"mode": res[0], "count": res[1]}, open(out, "w"))
+    td.barrier()
+    td.destroy_process_group()
+
+
+def test_dist_exact_continuous_mode_partitioned():
+    """High-cardinality continuous column: the exact hash-partitioned
+    mode must find the planted duplicate exactly (VERDICT r01 item 7 —
+    no histogram approximation)."""
+    port = _free_port()
+    out = tempfile.NamedTemporaryFile(suffix=".json", delete=False).name
+    mp_ctx = mp.get_context("spawn")
+    procs = [mp_ctx.Process(target=_mode_worker, args=(r, port, out)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=300)
+        assert p.exitcode == 0
+    res = json.load(open(out))
+    assert res["mode"] == pytest.approx(0.123456789, abs=1e-12)
+    assert res["count"] == 500
