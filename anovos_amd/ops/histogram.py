@@ -291,15 +291,49 @@ def _refine_pass(tensors, cols, brackets, nbins, col_lo=None, col_hi=None):
     if dev.type == "cuda" and backend.use_hip(tensors[0]):
         ext = backend.hip_ext()
         p1bins = nbins
-        if col_lo is None:  # degenerate: treat each bracket as its own grid
+        degenerate = col_lo is None
+        if degenerate:  # treat each bracket as its own grid
             col_lo = blo.clone()
             col_hi = bhi.clone()
         p1lo = col_lo.to(torch.float64)
         rng = (col_hi.to(torch.float64) - p1lo).clamp(min=1e-300)
         p1scale = float(p1bins) / rng
         nbins = 512  # grouped kernel's fixed sub-bin count
-        h = ext.bracket_histograms_grouped([t.contiguous() for t in tensors], colidx, blo, bhi,
-                                           p1lo, p1scale, p1bins)
+        # Partition brackets into launch groups honoring the kernel's two
+        # structural limits: <= 16 brackets per column per
+        # launch, and no two brackets of a column in the SAME pass-1 bin
+        # (the LDS bin->bracket LUT holds one entry per bin — a collision
+        # silently zeroes the losing bracket). Spiky distributions with
+        # many refinement brackets go through extra launches instead of
+        # crashing / silently degrading.
+        groups, seen = [], []
+        for kk, k in enumerate(keys):
+            ci = k[0]
+            if degenerate:
+                bin1 = 0
+            else:
+                bin1 = int((float(blo[kk]) - float(p1lo[ci])) * float(p1scale[ci]) + 1e-6)
+            placed = False
+            for gi, g in enumerate(groups):
+                bins_here = seen[gi].setdefault(ci, set())
+                if len(bins_here) < 16 and bin1 not in bins_here:
+                    g.append(kk)
+                    bins_here.add(bin1)
+                    placed = True
+                    break
+            if not placed:
+                groups.append([kk])
+                seen.append({ci: {bin1}})
+        tens_c = [t.contiguous() for t in tensors]
+        if len(groups) == 1:
+            h = ext.bracket_histograms_grouped(tens_c, colidx, blo, bhi, p1lo, p1scale, p1bins)
+        else:
+            h = torch.zeros(len(keys), nbins, dtype=torch.int64, device=dev)
+            for g in groups:
+                gi_t = torch.tensor(g, dtype=torch.long)
+                hg = ext.bracket_histograms_grouped(tens_c, colidx[gi_t], blo[gi_t], bhi[gi_t],
+                                                    p1lo, p1scale, p1bins)
+                h[gi_t] = hg
     else:
         h = torch.zeros(len(keys), nbins, dtype=torch.int64, device=dev)
         for kk, (i, j) in enumerate(keys):
