@@ -131,8 +131,73 @@ def _charts_section(master_path: str, files: List[str], title: str, uid: str, li
     return "".join(out)
 
 
-def executive_summary_gen(master_path: str, label_col, event_label) -> str:
-    """Reference report_generation.py:524 — KPI header from global_summary."""
+def _diagnosis_matrix(master_path: str, corr_threshold, iv_threshold) -> Optional[pd.DataFrame]:
+    """The reference's data-diagnosis matrix (report_generation.py:617-790):
+    attribute x {High Variance, Positive/Negative Skewness, High/Low
+    Kurtosis, Low Fill Rates, High Biasedness, Outliers, High
+    Correlation, Significant Attributes} with check/cross marks."""
+    import numpy as np
+
+    checks = []
+
+    def q(name, expr, metric):
+        df = _read_csv(master_path, name)
+        try:
+            vals = list(df.query(expr)["attribute"].values) if df is not None else []
+        except Exception:
+            vals = []
+        checks.append((metric, vals))
+
+    q("measures_of_dispersion", "`cov`>1", "High Variance")
+    q("measures_of_shape", "`skewness`>0", "Positive Skewness")
+    q("measures_of_shape", "`skewness`<0", "Negative Skewness")
+    q("measures_of_shape", "`kurtosis`>0", "High Kurtosis")
+    q("measures_of_shape", "`kurtosis`<0", "Low Kurtosis")
+    q("measures_of_counts", "`fill_pct`<0.7", "Low Fill Rates")
+    bd = _read_csv(master_path, "biasedness_detection")
+    try:
+        col = "treated" if (bd is not None and "treated" in bd.columns) else "flagged"
+        checks.append(("High Biasedness", list(bd.query(f"`{col}`>0")["attribute"].values) if bd is not None else []))
+    except Exception:
+        checks.append(("High Biasedness", []))
+    od = _read_csv(master_path, "outlier_detection")
+    checks.append(("Outliers", list(od["attribute"].values) if od is not None else []))
+    corr = _read_csv(master_path, "correlation_matrix")
+    hi_corr = []
+    if corr is not None and "attribute" in corr.columns:
+        try:
+            mat = corr[list(corr["attribute"].values)]
+            upper = mat.where(np.triu(np.ones(mat.shape), k=1).astype(bool))
+            hi_corr = [c for c in upper.columns if (upper[c] > corr_threshold).any()]
+        except Exception:
+            hi_corr = []
+    checks.append(("High Correlation", hi_corr))
+    iv = _read_csv(master_path, "IV_calculation")
+    try:
+        checks.append(("Significant Attributes",
+                       list(iv.query(f"`iv`>{iv_threshold}")["attribute"].values) if iv is not None else []))
+    except Exception:
+        checks.append(("Significant Attributes", []))
+
+    attrs = sorted({a for _, vals in checks for a in vals})
+    if not attrs:
+        return None
+    # every attribute seen anywhere gets a row; metric column order is
+    # the reference's (report_generation.py:773-787)
+    order = ["Outliers", "Significant Attributes", "Positive Skewness", "Negative Skewness",
+             "High Variance", "High Correlation", "High Kurtosis", "Low Kurtosis"]
+    rows = []
+    bymetric = dict(checks)
+    for a in attrs:
+        rows.append([a] + [("✔" if a in bymetric.get(mname, []) else "✘") for mname in order])
+    return pd.DataFrame(rows, columns=["Attribute"] + order)
+
+
+def executive_summary_gen(master_path: str, label_col, event_label,
+                          corr_threshold=0.4, iv_threshold=0.02, drift_threshold_model=0.1) -> str:
+    """Reference report_generation.py:524-908 — narrative summary, label
+    distribution pie, the data-diagnosis matrix, and the drift/stability
+    health BigNumbers."""
     gs = _read_csv(master_path, "global_summary")
     if gs is None:
         return "<p class='note'>global_summary.csv not found.</p>"
@@ -142,27 +207,60 @@ def executive_summary_gen(master_path: str, label_col, event_label) -> str:
                      ("numcols_count", "Numerical Columns"), ("catcols_count", "Categorical Columns")):
         v = get(m)
         kpis.append(f"<div class='kpi'><div class='v'>{_html.escape(str(v[0])) if len(v) else '—'}</div><div class='l'>{label}</div></div>")
-    lab = f"<p>Label column: <b>{_html.escape(str(label_col))}</b>, event label: <b>{_html.escape(str(event_label))}</b></p>" if label_col else ""
+    # narrative line (reference :585-595)
+    def _num(m):
+        v = get(m)
+        try:
+            return int(float(v[0]))
+        except Exception:
+            return 0
+    narrative = (f"<p><b>Key Report Highlights</b></p><p>The dataset contains <b>{_num('rows_count'):,}</b> "
+                 f"records and <b>{_num('numcols_count') + _num('catcols_count')}</b> attributes "
+                 f"(<b>{_num('numcols_count')}</b> numerical + <b>{_num('catcols_count')}</b> categorical).</p>")
+    lab = (f"<p>Target variable is <b>{_html.escape(str(label_col))}</b> "
+           f"(event label: <b>{_html.escape(str(event_label))}</b>)</p>"
+           if label_col else "<p>There is <b>no</b> target variable in the dataset</p>")
     extra = ""
     if label_col:
-        # label distribution from the label's frequency chart object
+        # label distribution pie from the label's frequency chart object
+        # (reference :556-580 rebuilds a pie from freqDist_<label>)
         p_lab = ends_with(master_path) + "freqDist_" + str(label_col)
         if os.path.exists(p_lab):
             try:
-                extra = "<h3>Label Distribution</h3>" + _fig_div(p_lab, "exec_label")
+                obj = json.load(open(p_lab))
+                tr = obj["data"][0] if isinstance(obj, dict) and "data" in obj else None
+                if tr is not None and "x" in tr and "y" in tr:
+                    pie = go.Figure(go.Pie(labels=tr["x"], values=tr["y"], textinfo="label+percent",
+                                           insidetextorientation="radial", pull=[0, 0.1]))
+                    pie.update_traces(textposition="inside", textinfo="percent+label")
+                    pie.update_layout(height=360, legend=dict(orientation="h", x=0.5, xanchor="center"))
+                    extra = "<h3>Label Distribution</h3>" + _fig_div(pie, "exec_label")
+                else:
+                    extra = "<h3>Label Distribution</h3>" + _fig_div(p_lab, "exec_label")
             except Exception:
                 extra = ""
-    # quality flags roll-up
+    diag = _diagnosis_matrix(master_path, corr_threshold, iv_threshold)
+    diag_html = ""
+    if diag is not None:
+        diag_html = "<h3>Data Diagnosis</h3>" + _tbl(diag, 500)
+    # drift / stability health BigNumbers (reference :793-860)
     flags = []
-    for name, col, label in (("drift_statistics", "flagged", "drifted attributes"),
-                             ("stability_index", "flagged", "unstable attributes"),
-                             ("outlier_detection", "upper_outliers", "attributes with upper outliers")):
-        df = _read_csv(master_path, name)
-        if df is not None and col in df.columns:
-            v = pd.to_numeric(df[col], errors="coerce").fillna(0)
-            n = int((v > 0).sum()) if col != "flagged" else int(v.sum())
-            flags.append(f"<div class='kpi'><div class='v'>{n}</div><div class='l'>{label}</div></div>")
-    return "".join(kpis) + "".join(flags) + lab + extra
+    dd = _read_csv(master_path, "drift_statistics")
+    if dd is not None and "flagged" in dd.columns:
+        v = pd.to_numeric(dd["flagged"], errors="coerce").fillna(0)
+        n, tot = int(v.sum()), int(len(v))
+        pct = round(100.0 * n / max(tot, 1), 2)
+        flags.append(f"<div class='kpi'><div class='v'>{n} / {tot}</div><div class='l'># Drifted Attributes</div></div>")
+        flags.append(f"<div class='kpi'><div class='v'>{pct}%</div><div class='l'>% Drifted Attributes</div></div>")
+    si = _read_csv(master_path, "stability_index")
+    if si is not None and "flagged" in si.columns:
+        v = pd.to_numeric(si["flagged"], errors="coerce").fillna(0)
+        n, tot = int((v > 0).sum()), int(len(v))
+        pct = round(100.0 * n / max(tot, 1), 2)
+        flags.append(f"<div class='kpi'><div class='v'>{n} / {tot}</div><div class='l'># Unstable Attributes</div></div>")
+        flags.append(f"<div class='kpi'><div class='v'>{pct}%</div><div class='l'>% Unstable Attributes</div></div>")
+    health = ("<h3>Data Health (Drift &amp; Stability)</h3>" + "".join(flags)) if flags else ""
+    return "".join(kpis) + narrative + lab + extra + diag_html + health
 
 
 _METRIC_DICT = [
@@ -190,13 +288,31 @@ def wiki_generator(master_path: str, dataDict_path=None, metricDict_path=None) -
     """Reference report_generation.py:909 — data dictionary + metric
     definitions tab. Falls back to the engine's built-in metric
     definitions when no metricDict CSV is supplied."""
-    parts = []
-    for p, title in ((dataDict_path, "Data Dictionary"),):
-        if p and os.path.exists(p):
-            try:
-                parts.append(f"<h3>{title}</h3>" + _tbl(pd.read_csv(p), 500))
-            except Exception:
-                pass
+    parts = ["<p><i>A quick reference to the attributes from the dataset (Data "
+             "Dictionary) and the metrics computed in the report (Metric "
+             "Dictionary).</i></p>"]
+    # data dictionary OUTER-merged with the observed data types
+    # (reference :931-948: dataDict ⟗ data_type.csv on attribute)
+    datatype_df = _read_csv(master_path, "data_type")
+    dd_df = None
+    if dataDict_path and os.path.exists(str(dataDict_path)):
+        try:
+            dd_df = pd.read_csv(dataDict_path)
+            dd_df.columns = [c.strip().lower() if c.strip().lower() == "attribute" else c for c in dd_df.columns]
+            if "attribute" not in dd_df.columns and len(dd_df.columns):
+                dd_df = dd_df.rename(columns={dd_df.columns[0]: "attribute"})
+        except Exception:
+            dd_df = None
+    if dd_df is not None and datatype_df is not None:
+        try:
+            merged = dd_df.merge(datatype_df, how="outer", on="attribute")
+        except Exception:
+            merged = dd_df
+        parts.append("<h3>Data Dictionary</h3>" + _tbl(merged, 500))
+    elif dd_df is not None:
+        parts.append("<h3>Data Dictionary</h3>" + _tbl(dd_df, 500))
+    elif datatype_df is not None:
+        parts.append("<h3>Data Dictionary</h3>" + _tbl(datatype_df, 500))
     if metricDict_path and os.path.exists(metricDict_path):
         try:
             parts.append("<h3>Metric Dictionary</h3>" + _tbl(pd.read_csv(metricDict_path), 500))
@@ -208,24 +324,61 @@ def wiki_generator(master_path: str, dataDict_path=None, metricDict_path=None) -
     return "".join(parts)
 
 
+def _split_charts_by_kind(master_path: str, prefix: str):
+    """Split chart objects into numerical / categorical grids using
+    data_type.csv (reference descriptive_statistics keeps separate
+    all_charts_num_ / all_charts_cat_ grids)."""
+    files = _chart_files(master_path, prefix)
+    dt = _read_csv(master_path, "data_type")
+    if dt is None:
+        return files, []
+    kind = {str(a): str(t) for a, t in zip(dt["attribute"], dt["data_type"])}
+    num, cat = [], []
+    for fn in files:
+        attr = fn[len(prefix):]
+        t = kind.get(attr, "")
+        (cat if t in ("string", "categorical") else num).append(fn)
+    return num, cat
+
+
 def descriptive_statistics(master_path: str) -> str:
-    """Reference report_generation.py:994."""
+    """Reference report_generation.py:994-1153 — global summary
+    narrative, per-metric stat tables, numerical/categorical chart
+    grids."""
     parts = []
+    gs = _read_csv(master_path, "global_summary")
+    if gs is not None:
+        # the reference renders global_summary as a bullet list
+        items = "".join(f"<li><b>{_html.escape(str(m))}</b>: {_html.escape(str(v))}</li>"
+                        for m, v in zip(gs["metric"], gs["value"]))
+        parts.append(f"<h3>Global Summary</h3><ul>{items}</ul>")
     for name in SG_tabs:
         df = _read_csv(master_path, name)
         if df is not None:
-            parts.append(f"<h3>{name}</h3>" + _tbl(df))
-    parts.append(_charts_section(master_path, _chart_files(master_path, "freqDist_"), "Frequency Distributions", "fd"))
+            parts.append(f"<h3>{remove_u_score(name)}</h3>" + _tbl(df))
+    num, cat = _split_charts_by_kind(master_path, "freqDist_")
+    parts.append(_charts_section(master_path, num, "Numerical Attribute Distributions", "fdn"))
+    parts.append(_charts_section(master_path, cat, "Categorical Attribute Frequencies", "fdc"))
     return "".join(parts) if parts else "<p class='note'>No descriptive statistics saved.</p>"
 
 
+_QC_ROW_LEVEL = ["duplicate_detection", "nullRows_detection"]
+
+
 def quality_check(master_path: str) -> str:
-    """Reference report_generation.py:1154."""
-    parts = []
+    """Reference report_generation.py:1154-1290 — row-level and
+    column-level sub-sections + outlier violin charts."""
+    row_parts, col_parts = [], []
     for name in QC_tabs:
         df = _read_csv(master_path, name)
         if df is not None:
-            parts.append(f"<h3>{name}</h3>" + _tbl(df))
+            (row_parts if name in _QC_ROW_LEVEL else col_parts).append(
+                f"<h3>{remove_u_score(name)}</h3>" + _tbl(df))
+    parts = []
+    if row_parts:
+        parts.append("<h2>Row-Level Checks</h2>" + "".join(row_parts))
+    if col_parts:
+        parts.append("<h2>Column-Level Checks</h2>" + "".join(col_parts))
     parts.append(_charts_section(master_path, _chart_files(master_path, "outlier_"), "Outlier Charts", "oc"))
     return "".join(parts) if parts else "<p class='note'>No quality-check statistics saved.</p>"
 
@@ -280,10 +433,22 @@ def data_drift_stability(master_path: str, drift_threshold_model=0.1) -> str:
             parts.insert(0, f"<div class='kpi'><div class='v'>{n_drift}</div><div class='l'>Drifted Attributes</div></div>")
     parts.append(_charts_section(master_path, [x for x in _chart_files(master_path, "drift_") if x != "drift_statistics"], "Source vs Target Distributions", "dr"))
     si = _read_csv(master_path, "stability_index")
+    si_metrics = _read_csv(master_path, "stabilityIndex_metrics")
     if si is None:
-        si = _read_csv(master_path, "stabilityIndex_metrics")
+        si = si_metrics
     if si is not None:
         parts.append("<h3>Stability Index</h3>" + _tbl(si))
+        # per-attribute metric trajectories across snapshots (reference
+        # line_chart_gen_stability grid, report_generation.py:99)
+        if si_metrics is not None and "attribute" in si_metrics.columns:
+            charts = []
+            for col in list(dict.fromkeys(si_metrics["attribute"]))[:40]:
+                try:
+                    charts.append("<div>" + line_chart_gen_stability(si, si_metrics, col) + "</div>")
+                except Exception:
+                    continue
+            if charts:
+                parts.append("<h3>Stability Trajectories</h3><div class='grid'>" + "".join(charts) + "</div>")
         parts.append("<h3>Stability Interpretation</h3>" + _tbl(STABILITY_INTERPRETATION))
     return "".join(parts) if parts else "<p class='note'>No drift / stability statistics saved.</p>"
 
@@ -369,7 +534,8 @@ def anovos_report(master_path: str, id_col="", label_col=None, corr_threshold=0.
         final_report_path = "."
     os.makedirs(final_report_path, exist_ok=True)
     tabs = [
-        ("Executive Summary", executive_summary_gen(master_path, label_col, event_label)),
+        ("Executive Summary", executive_summary_gen(master_path, label_col, event_label,
+                                                    corr_threshold, iv_threshold, drift_threshold_model)),
         ("Wiki", wiki_generator(master_path, dataDict_path if dataDict_path != "." else None,
                                 metricDict_path if metricDict_path != "." else None)),
         ("Descriptive Statistics", descriptive_statistics(master_path)),

@@ -338,6 +338,18 @@ def main(all_configs: Dict, run_type: str = "local", auth_key_val: Dict = {}, de
                     stats = dstability.stability_index_computation(ctx, *idfs, **configs)
                     if report_input_path:
                         report_preprocessing.save_stats(ctx, stats, report_input_path, "stability_index", run_type=run_type, auth_key=auth_key)
+                        # per-snapshot metric history feeds the report's
+                        # stability line charts (reference workflow.py:700-721)
+                        amp = (configs or {}).get("appended_metric_path", "")
+                        if amp:
+                            try:
+                                from anovos_amd.core.io import read_dataset as _rd
+
+                                df_metrics = _rd(amp, "csv", {"header": True}).to_pandas()
+                                report_preprocessing.save_stats(ctx, df_metrics, report_input_path,
+                                                                "stabilityIndex_metrics", run_type=run_type, auth_key=auth_key)
+                            except Exception as e:
+                                logger.warning(f"stabilityIndex_metrics not persisted: {e}")
                 _log(f"drift_detector.{subkey}: {time.time() - start:.3f}s")
                 start = time.time()
 
