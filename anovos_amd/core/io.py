@@ -174,7 +174,10 @@ def write_dataset(idf: AnovosFrame, file_path: str, file_type: str, file_configs
         os.makedirs(file_path, exist_ok=True)
     dist.barrier()
     part = os.path.join(file_path, f"part-{dist.rank():05d}")
-    pdf = idf.to_pandas()
+    # vectorized materialization: categoricals stay dictionary-encoded
+    # (parquet stores the dictionary; re-ingest is O(n)), timestamps stay
+    # datetime64 — object-array churn dominated save/reread stages before
+    pdf = idf.to_pandas_io() if ft in ("csv", "parquet", "json") else idf.to_pandas()
     if ft == "csv":
         header = str(file_configs.get("header", True)).lower() in ("true", "1")
         delim = file_configs.get("delimiter", ",")

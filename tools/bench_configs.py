@@ -53,7 +53,9 @@ def main():
     ap.add_argument("--rows", type=int, default=2_000_000)
     ap.add_argument("--workdir", default="")
     a = ap.parse_args()
-    wd = a.workdir or os.path.join("gpurun_out", "cfgbench_" + os.path.basename(a.config).replace(".yaml", ""))
+    # /tmp, NOT gpurun_out: generated datasets would blow the 64 MiB
+    # copy-back budget of the GPU harness
+    wd = a.workdir or os.path.join("/tmp", "cfgbench_" + os.path.basename(a.config).replace(".yaml", ""))
     os.makedirs(wd, exist_ok=True)
     cfg_path = os.path.join(REPO, "config", a.config) if not os.path.isabs(a.config) else a.config
     os.chdir(wd)
