@@ -68,9 +68,18 @@ def ETL(ctx, args: Dict):
 
 def save(data, write_configs, folder_name, reread=False):
     """Reference workflow.py:64-88 — write the frame under
-    <file_path>/<folder_name>, optionally re-read it (stage checkpoint)."""
+    <file_path>/<folder_name>, optionally re-read it (stage checkpoint).
+
+    ANOVOS_AMD_INMEMORY_PIPELINE=1 skips the per-stage intermediate
+    materialization and keeps the frame HBM-resident (SURVEY §7 'hard
+    parts': the save/reread barrier must be optional without changing
+    results — each stage's output is identical, only the disk round-trip
+    and its restart point are elided). write_main / write_stats are
+    unaffected."""
     if not write_configs:
         return data if reread else None
+    if reread and os.environ.get("ANOVOS_AMD_INMEMORY_PIPELINE", "") in ("1", "true"):
+        return data  # only stage checkpoints reread; final/stat writes don't
     if "file_path" not in write_configs:
         raise TypeError("file path missing for writing data")
     write = copy.deepcopy(write_configs)

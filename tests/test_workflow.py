@@ -178,3 +178,37 @@ def test_workflow_geospatial_config(tmp_path, monkeypatch):
     assert "lat_long_geohash" in out.columns
     assert "radius_of_gyration" in out.columns
     assert os.path.exists("report_stats/Overall_Summary_1_latitude_longitude.csv")
+
+
+def test_workflow_inmemory_pipeline_matches_materialized(income_csv, tmp_path, monkeypatch):
+    """ANOVOS_AMD_INMEMORY_PIPELINE=1 skips the per-stage save/reread
+    barrier; the final frame and stats must be identical to the
+    materialized run (the barrier is optional, not semantic)."""
+    import pandas as pd
+
+    cfg = make_config(income_csv, tmp_path)
+    cfg_path = tmp_path / "cfg2.yaml"
+    with open(cfg_path, "w") as f:
+        yaml.safe_dump(cfg, f, sort_keys=False)
+    df_mat = workflow.run(str(cfg_path))
+    iv_mat = pd.read_csv(tmp_path / "report" / "IV_calculation.csv")
+
+    monkeypatch.setenv("ANOVOS_AMD_INMEMORY_PIPELINE", "1")
+    # rerun into fresh dirs
+    cfg2 = make_config(income_csv, tmp_path / "mem")
+    (tmp_path / "mem").mkdir(exist_ok=True)
+    cfg2_path = tmp_path / "cfg3.yaml"
+    with open(cfg2_path, "w") as f:
+        yaml.safe_dump(cfg2, f, sort_keys=False)
+    df_mem = workflow.run(str(cfg2_path))
+    iv_mem = pd.read_csv(tmp_path / "mem" / "report" / "IV_calculation.csv")
+
+    assert sorted(df_mat.columns) == sorted(df_mem.columns)
+    m = iv_mat.merge(iv_mem, on="attribute", suffixes=("_a", "_b"))
+    assert (abs(m["iv_a"] - m["iv_b"]) < 1e-9).all()
+    for c in df_mat.columns:
+        a, b = df_mat.col(c), df_mem.col(c)
+        if a.kind == "numerical":
+            import torch as _t
+
+            assert float(_t.nan_to_num(a.data.double() - b.data.double()).abs().max()) < 1e-6, c
