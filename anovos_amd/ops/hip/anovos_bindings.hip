@@ -481,9 +481,21 @@ torch::Tensor bracket_histograms_grouped(std::vector<torch::Tensor> cols,
   // brackets MUST be sorted by column index; returns [nbrackets, 512].
   TORCH_CHECK(!cols.empty(), "no columns");
   auto device = cols[0].device();
-  int dtype = dtype_code(cols[0]);
-  for (auto &t : cols) TORCH_CHECK(dtype_code(t) == dtype, "grouped brackets: mixed dtypes unsupported");
   auto bc = bracket_col.to(torch::kInt64).cpu().contiguous();
+  // dtype uniformity only matters for REFERENCED columns (the host
+  // splits mixed-dtype bracket sets into per-dtype launches; columns
+  // without brackets in this launch are never read)
+  int dtype = -1;
+  {
+    const int64_t *b = bc.data_ptr<int64_t>();
+    for (int i = 0; i < (int)bc.numel(); ++i) {
+      TORCH_CHECK(b[i] >= 0 && b[i] < (int64_t)cols.size(), "bad bracket col");
+      int d = dtype_code(cols[b[i]]);
+      if (dtype < 0) dtype = d;
+      TORCH_CHECK(d == dtype, "grouped brackets: mixed dtypes in one launch");
+    }
+    if (dtype < 0) dtype = dtype_code(cols[0]);
+  }
   int ncols = (int)cols.size();
   int nb = (int)bc.numel();
   std::vector<int> bstart(ncols + 1, 0);

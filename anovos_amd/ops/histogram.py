@@ -325,15 +325,23 @@ def _refine_pass(tensors, cols, brackets, nbins, col_lo=None, col_hi=None):
                 groups.append([kk])
                 seen.append({ci: {bin1}})
         tens_c = [t.contiguous() for t in tensors]
-        if len(groups) == 1:
-            h = ext.bracket_histograms_grouped(tens_c, colidx, blo, bhi, p1lo, p1scale, p1bins)
-        else:
-            h = torch.zeros(len(keys), nbins, dtype=torch.int64, device=dev)
-            for g in groups:
-                gi_t = torch.tensor(g, dtype=torch.long)
-                hg = ext.bracket_histograms_grouped(tens_c, colidx[gi_t], blo[gi_t], bhi[gi_t],
-                                                    p1lo, p1scale, p1bins)
-                h[gi_t] = hg
+        # the grouped kernel launches one dtype at a time — split each
+        # launch group by column dtype (mixed f32/f64 frames arise when
+        # the pipeline runs HBM-resident without a parquet round-trip)
+        split_groups = []
+        for g in groups:
+            f32 = [kk for kk in g if tens_c[keys[kk][0]].dtype == torch.float32]
+            f64 = [kk for kk in g if tens_c[keys[kk][0]].dtype != torch.float32]
+            if f32:
+                split_groups.append(f32)
+            if f64:
+                split_groups.append(f64)
+        h = torch.zeros(len(keys), nbins, dtype=torch.int64, device=dev)
+        for g in split_groups:
+            gi_t = torch.tensor(g, dtype=torch.long)
+            hg = ext.bracket_histograms_grouped(tens_c, colidx[gi_t], blo[gi_t], bhi[gi_t],
+                                                p1lo, p1scale, p1bins)
+            h[gi_t] = hg
     else:
         h = torch.zeros(len(keys), nbins, dtype=torch.int64, device=dev)
         for kk, (i, j) in enumerate(keys):

@@ -571,12 +571,19 @@ def test_advanced_imputers_on_device(ext):
     pdf["b"] = pdf["a"] * 0.5 + rng.normal(0, 0.3, n)
     for c in pdf.columns:
         pdf.loc[rng.choice(n, n // 100, replace=False), c] = np.nan
+    import tempfile
+
     ctx = init_context("cuda")
     gpu_f = AnovosFrame.from_pandas(pdf, device="cuda")
     cpu_f = AnovosFrame.from_pandas(pdf, device="cpu")
     for method in ("KNN", "regression"):
-        g = TA.imputation_sklearn(ctx, gpu_f, "all", method_type=method, sample_size=4000)
-        c0 = TA.imputation_sklearn(ctx, cpu_f, "all", method_type=method, sample_size=4000)
+        # fit ONCE (CPU), share the pickled model: isolates the APPLY
+        # path (device sampling differs, so independent fits would
+        # legitimately produce different models)
+        mp = tempfile.mkdtemp(prefix=f"imp_{method}_")
+        TA.imputation_sklearn(ctx, cpu_f, "all", method_type=method, sample_size=4000, model_path=mp)
+        g = TA.imputation_sklearn(ctx, gpu_f, "all", method_type=method, pre_existing_model=True, model_path=mp)
+        c0 = TA.imputation_sklearn(ctx, cpu_f, "all", method_type=method, pre_existing_model=True, model_path=mp)
         for col in pdf.columns:
             gv = g.col(col).data.cpu().numpy()
             cv = c0.col(col).data.numpy()
