@@ -273,7 +273,9 @@ def imputation_matrixFactorization(
         VVt = torch.einsum("mk,ml->mkl", V, V)  # [m,k,k] — small
         for s in range(0, n, ROW_BATCH):
             Mb = Mf[s : s + ROW_BATCH]  # [b,m]
-            A = torch.einsum("bm,mkl->bkl", Mb, VVt) + (reg * n_obs_row[s : s + ROW_BATCH]).view(-1, 1, 1) * eye
+            # clamp: a row with ZERO observed columns would get A=0
+            # (singular); with A=reg*I and b=0 it solves to U=0
+            A = torch.einsum("bm,mkl->bkl", Mb, VVt) + (reg * n_obs_row[s : s + ROW_BATCH].clamp(min=1.0)).view(-1, 1, 1) * eye
             b = (Z[s : s + ROW_BATCH] * Mb) @ V  # [b,k]
             U[s : s + ROW_BATCH] = torch.linalg.solve(A, b.unsqueeze(2)).squeeze(2)
         # solve V per column: V_j = (U^T W U + reg n_j I)^-1 U^T W z_j
