@@ -1626,56 +1626,7 @@ __global__ __launch_bounds__(THREADS) void bucketize_label_counts_kernel(
     return sl < slots ? sl : slots - 1;
   };
 
-  if (sizeof(T) == 4 && slots <= 16) {
-    // wave-ballot aggregation: binned columns have <= bin_size+2 slots;
-    // per-element LDS atomics serialize on ~12 hot slots and per-thread
-    // register histograms cost ~32 VALU/element. A ballot per slot per
-    // 64-lane group costs ~16*5 wave-instructions per 64 ELEMENTS
-    // (~1.2/element), and the popcounts are wave-uniform so the
-    // accumulators live in (effectively scalar) registers; lane 0
-    // flushes once at the end.
-    uint32_t rtot[16], revt[16];
-#pragma unroll
-    for (int s2 = 0; s2 < 16; ++s2) rtot[s2] = revt[s2] = 0;
-    const int lane = threadIdx.x & 63;
-    const int64_t nv = (e - s) / 4;
-    const nat_f4 *__restrict__ xv = reinterpret_cast<const nat_f4 *>((const float *)x + s);
-    for (int64_t i = threadIdx.x; i < nv; i += THREADS) {
-      nat_f4 v = __builtin_nontemporal_load(&xv[i]);
-      const int64_t r = s + i * 4;
-      const float vv[4] = {v.x, v.y, v.z, v.w};
-#pragma unroll
-      for (int k2 = 0; k2 < 4; ++k2) {
-        const int slot = slot_f32(vv[k2]);
-        const uint64_t lab_mask = __ballot(label[r + k2] != 0);
-#pragma unroll
-        for (int s2 = 0; s2 < 16; ++s2) {
-          const uint64_t m = __ballot(slot == s2);
-          rtot[s2] += (uint32_t)__popcll(m);
-          revt[s2] += (uint32_t)__popcll(m & lab_mask);
-        }
-      }
-    }
-    for (int64_t i = s + nv * 4 + threadIdx.x; i < e; i += THREADS) {
-      const int slot = slot_f32((float)x[i]);
-      const uint64_t lab_mask = __ballot(label[i] != 0);
-#pragma unroll
-      for (int s2 = 0; s2 < 16; ++s2) {
-        const uint64_t m = __ballot(slot == s2);
-        rtot[s2] += (uint32_t)__popcll(m);
-        revt[s2] += (uint32_t)__popcll(m & lab_mask);
-      }
-    }
-    // ballot results are wave-uniform: one lane per wave flushes
-    if (lane == 0) {
-#pragma unroll
-      for (int s2 = 0; s2 < 16; ++s2) {
-        if (s2 < slots) {
-          if (rtot[s2]) atomicAdd(&tot[s2], rtot[s2]);
-          if (revt[s2]) atomicAdd(&evt[s2], revt[s2]);
-        }
-      }
-    }
+  if (false) {  // slots<=16 specializations measured slower than plain LDS atomics
   } else if (sizeof(T) == 4) {
     const int64_t nv = (e - s) / 4;
     const nat_f4 *__restrict__ xv = reinterpret_cast<const nat_f4 *>((const float *)x + s);
