@@ -182,7 +182,13 @@ __global__ __launch_bounds__(THREADS) void gram_singleread_kernel(
     const int *pair_i, const int *pair_j, int npairs, int row_chunks,
     float *partials) {
   constexpr int STAGES = 4;  // 128-row macro-slab
-  extern __shared__ short slab[];  // [STAGES][4][ktot][8] bf16
+  // kstride pads each (stage, ko-group) plane by 3 column slots: the
+  // un-padded plane stride (ktot*16 B = 640 words for kt=10) is 0 mod
+  // 32 banks, so the 16 planes' writes all landed in the same 4 banks
+  // (16-way conflict — measured 2.1 TB/s for a 75 GB pass). +3 slots
+  // shifts consecutive planes by 12 banks (worst case 2-way).
+  const int kstride = ktot + 3;
+  extern __shared__ short slab[];  // [STAGES][4][kstride][8] bf16
   const int block = blockIdx.x;
   const int64_t macro_total = (n + STAGES * 32 - 1) / (STAGES * 32);
   const int64_t macro_per = (macro_total + row_chunks - 1) / row_chunks;
@@ -244,14 +250,14 @@ __global__ __launch_bounds__(THREADS) void gram_singleread_kernel(
     __syncthreads();
 #pragma unroll
     for (int st = 0; st < STAGES; ++st) {
-      const short *sbase = slab + (size_t)st * 4 * ktot * 8;
+      const short *sbase = slab + (size_t)st * 4 * kstride * 8;
 #pragma unroll
       for (int u = 0; u < MAXPW; ++u) {
         if (i0s[u] >= 0) {
           const bf16x8 a = *reinterpret_cast<const bf16x8 *>(
-              sbase + ((size_t)kog * ktot + i0s[u] + m) * 8);
+              sbase + ((size_t)kog * kstride + i0s[u] + m) * 8);
           const bf16x8 b = *reinterpret_cast<const bf16x8 *>(
-              sbase + ((size_t)kog * ktot + j0s[u] + m) * 8);
+              sbase + ((size_t)kog * kstride + j0s[u] + m) * 8);
           acc[u] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[u], 0, 0, 0);
         }
       }
@@ -298,7 +304,7 @@ extern "C" int anovos_centered_gram_sr(const void *const *cols, int64_t n,
                                        hipStream_t stream) {
   const int kt = (k + 15) / 16;
   const int ktot = kt * 16;
-  const size_t lds = (size_t)4 /*STAGES*/ * 4 * ktot * 16;
+  const size_t lds = (size_t)4 /*STAGES*/ * 4 * (ktot + 3) * 16;
   if (kt <= 10)
     hipLaunchKernelGGL((gram_singleread_kernel<14>), dim3(row_chunks),
                        dim3(THREADS), lds, stream, (const float *const *)cols,
