@@ -243,7 +243,7 @@ __global__ __launch_bounds__(THREADS) void gram_singleread_kernel(
       pack.w = to_bf16(isnan(v3) ? 0.f : v3);
       const int st = seg >> 3;        // stage (32-row step) 0..STAGES-1
       const int si = seg & 7;         // 4-row segment within the stage
-      short *dst = slab + ((size_t)st * 4 * ktot + (size_t)(si >> 1) * ktot + c) * 8 +
+      short *dst = slab + ((size_t)(st * 4 + (si >> 1)) * kstride + c) * 8 +
                    (size_t)(si & 1) * 4;
       *reinterpret_cast<bf16x4_t *>(dst) = pack;
     }
