@@ -221,7 +221,7 @@ def pipeline_step(ctx, idf, source_hist, model_dir, sections=None):
         if (cat_cols[0] + "_index") in t3.columns:
             chk_dev += t3.col(cat_cols[0] + "_index").data.float().nansum()
         else:
-            chk_dev += t3.col(cat_cols[0]).data.float().sum()
+            chk_dev += t3.col(cat_cols[0]).data.float().nansum()  # nulls are NaN after encoding
         del t3
     with _timed("outlier_categories", sections, ctx):
         t4 = T.outlier_categories(ctx, idf, cat_cols, max_category=20)  # all 50
