@@ -110,29 +110,33 @@ def geohash_encode_int(lat: torch.Tensor, lon: torch.Tensor, precision: int = 8)
 
 
 def geohash_int_to_str(gh: torch.Tensor, precision: int = 8) -> List[str]:
-    v = gh.cpu().numpy()
-    out = []
-    for g in v:
-        chars = []
-        for i in range(precision):
-            shift = 5 * (precision - 1 - i)
-            chars.append(_GH_BASE32[(int(g) >> shift) & 31])
-        out.append("".join(chars))
-    return out
+    """Packed int64 → base-32 strings, vectorized (high-cardinality
+    geohash dictionaries reach ~n entries; the python per-char loop was
+    a hotspot in the geospatial config)."""
+    v = gh.cpu().numpy().astype(np.int64)
+    shifts = 5 * np.arange(precision - 1, -1, -1, dtype=np.int64)
+    idx = (v[:, None] >> shifts[None, :]) & 31  # [n, precision]
+    chars = np.array(list(_GH_BASE32), dtype="U1")
+    mat = np.ascontiguousarray(chars[idx])  # [n, precision] U1
+    return mat.view(f"U{precision}").ravel().tolist()
 
 
 def geohash_str_to_int(ghs: Sequence[str]) -> Tuple[np.ndarray, int]:
-    """Decode base-32 strings to packed int64 (host: runs over the
-    column dictionary only). Returns (ints, precision)."""
+    """Decode base-32 strings to packed int64, vectorized over the
+    (possibly ~n-sized) dictionary. Short strings are left-aligned and
+    zero-padded — identical to the scalar shift semantics. Returns
+    (ints, precision)."""
     prec = max((len(g) for g in ghs if g), default=0)
-    out = np.zeros(len(ghs), dtype=np.int64)
-    for j, g in enumerate(ghs):
-        acc = 0
-        for c in str(g)[:prec]:
-            acc = (acc << 5) | _GH_DECODE.get(c, 0)
-        acc <<= 5 * (prec - min(len(str(g)), prec))
-        out[j] = acc
-    return out, prec
+    if prec == 0 or not len(ghs):
+        return np.zeros(len(ghs), dtype=np.int64), prec
+    arr = np.array(["" if g is None else str(g)[:prec] for g in ghs], dtype=f"U{prec}")
+    cp = arr.view(np.uint32).reshape(len(arr), prec)  # UCS4 codepoints, 0-padded
+    lut = np.zeros(1024, dtype=np.int64)  # unknown chars decode to 0 (as before)
+    for ch, val in _GH_DECODE.items():
+        lut[ord(ch)] = val
+    vals = lut[np.clip(cp, 0, 1023)]  # [n, prec]; pad cp=0 -> 0
+    weights = (np.int64(1) << (5 * np.arange(prec - 1, -1, -1, dtype=np.int64)))
+    return (vals * weights[None, :]).sum(axis=1), prec
 
 
 def geohash_decode_int(gh: torch.Tensor, precision: int) -> Tuple[torch.Tensor, torch.Tensor]:
