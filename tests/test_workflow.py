@@ -105,6 +105,13 @@ def test_workflow_end_to_end(income_csv, tmp_path):
     assert (rep / "ml_anovos_report.html").exists()
     html = open(rep / "ml_anovos_report.html").read()
     assert "Attribute Associations" in html
+    # round-2 content-parity elements (docs/REPORT_PARITY.md)
+    assert "Key Report Highlights" in html        # exec-summary narrative
+    assert "Data Diagnosis" in html               # ✔/✘ matrix
+    assert "Label Distribution" in html           # label pie
+    assert "Metric Dictionary" in html            # wiki tab
+    assert "Row-Level Checks" in html and "Column-Level Checks" in html
+    assert "Global Summary" in html
     # main dataset persisted
     assert any("final_dataset" in d for d in os.listdir(tmp_path / "out"))
 

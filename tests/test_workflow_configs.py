@@ -59,6 +59,10 @@ def test_workflow_sales_supervised_config(in_tmp):
     assert os.path.exists("report_stats/stability_index.csv")
     si = pd.read_csv("report_stats/stability_index.csv")
     assert "stability_index" in si.columns
+    # per-snapshot metric history persisted for the trajectory charts
+    assert os.path.exists("report_stats/stabilityIndex_metrics.csv")
+    html = open("report_stats/ml_anovos_report.html").read()
+    assert "Stability Trajectories" in html
     iv = pd.read_csv("report_stats/IV_calculation.csv")
     assert len(iv) > 0
     # supervised target-rate encoding happened before the corr matrix
