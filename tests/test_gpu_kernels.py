@@ -595,3 +595,41 @@ def test_advanced_imputers_on_device(ext):
     mf = TA.imputation_matrixFactorization(ctx, gpu_f, "all", rank=3, max_iter=8)
     for col in pdf.columns:
         assert not torch.isnan(mf.col(col).data).any()
+
+
+@requires_gpu
+def test_bucketize_label_counts_edges(ext):
+    """Fused K6+K9 edge conditions: empty cutoffs, f64 input, all-null
+    column, all-events label, slots exactly 16 (private-counter bound)."""
+    n = 1_000_001
+    g = torch.Generator(device="cpu").manual_seed(71)
+    lab1 = torch.ones(n, dtype=torch.uint8).cuda()
+    x = torch.randn(n, generator=g).cuda()
+    # empty cutoffs: every non-null value lands in bin 1 -> slot 2
+    flat = ext.bucketize_label_counts([x.contiguous()], [torch.empty(0, dtype=torch.float64)], lab1, [4]).cpu()
+    assert int(flat[2]) == n and int(flat[4 + 2]) == n
+    # all-null f32 column -> slot 0
+    xn = torch.full((n,), float("nan")).cuda()
+    flat = ext.bucketize_label_counts([xn], [torch.tensor([0.0])], lab1, [4]).cpu()
+    assert int(flat[0]) == n and int(flat[4 + 0]) == n
+    # f64 column vs torch reference, 14 cutoffs -> slots 16 exactly
+    xd = (torch.randn(n, generator=g, dtype=torch.float64) * 3).cuda()
+    cuts = torch.linspace(-4, 4, 14, dtype=torch.float64)
+    lab = (torch.rand(n, generator=g) < 0.5).to(torch.uint8).cuda()
+    flat = ext.bucketize_label_counts([xd.contiguous()], [cuts], lab.contiguous(), [16]).cpu()
+    ref_bins = torch.bucketize(xd.cpu(), cuts, right=False) + 1  # 1..15
+    slots = (ref_bins + 1).clamp(max=15)
+    tot_ref = torch.bincount(slots, minlength=16)
+    evt_ref = torch.bincount(slots, weights=lab.cpu().double(), minlength=16)
+    assert torch.equal(flat[:16], tot_ref.to(torch.int64))
+    assert torch.equal(flat[16:32].double(), evt_ref)
+
+
+@requires_gpu
+def test_label_counts_multi_empty_dictionary(ext):
+    """Categorical column with an empty dictionary (all nulls)."""
+    n = 100_000
+    codes = torch.full((n,), -1, dtype=torch.int32).cuda()
+    lab = torch.zeros(n, dtype=torch.uint8).cuda()
+    flat = ext.label_counts_multi([codes], lab, [1]).cpu()
+    assert int(flat[0]) == n and int(flat[1]) == 0
