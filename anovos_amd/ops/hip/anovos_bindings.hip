@@ -266,6 +266,7 @@ std::vector<torch::Tensor> bucketize_columns(std::vector<torch::Tensor> cols,
       max_ncut = std::max(max_ncut, (int)cc.numel());
     }
     if (ptrs.empty()) continue;
+    if (flat.empty()) flat.push_back(0.0);  // from_blob on empty data() segfaults
     int ncols = (int)ptrs.size();
     int64_t maxn = *std::max_element(lens.begin(), lens.end());
     int nchunks = pick_chunks(maxn, ncols);
@@ -562,6 +563,7 @@ std::vector<torch::Tensor> bucketize_columns_float(std::vector<torch::Tensor> co
       max_ncut = std::max(max_ncut, (int)cc.numel());
     }
     if (ptrs.empty()) continue;
+    if (flat.empty()) flat.push_back(0.0);  // from_blob on empty data() segfaults
     int ncols = (int)ptrs.size();
     int64_t maxn = *std::max_element(lens.begin(), lens.end());
     int nchunks = pick_chunks(maxn, ncols);
@@ -840,6 +842,7 @@ torch::Tensor bucketize_label_counts(std::vector<torch::Tensor> cols,
       max_slots = std::max(max_slots, (int)sizes[i]);
     }
     if (ptrs.empty()) continue;
+    if (flat.empty()) flat.push_back(0.0);  // from_blob on an empty vector's data() segfaults
     int ncols = (int)ptrs.size();
     int64_t maxn = *std::max_element(lens.begin(), lens.end());
     int nchunks = pick_chunks(maxn, ncols);
