@@ -8,6 +8,7 @@
 #include <hip/hip_runtime.h>
 
 #include <vector>
+#include <cstdlib>
 
 extern "C" {
 int anovos_moments(const void *const *cols, const int64_t *lens,
@@ -685,7 +686,9 @@ torch::Tensor centered_gram_bf16(std::vector<torch::Tensor> cols, torch::Tensor 
     // single-read kernel: one block stages a 32-row slab of ALL columns
     // through LDS; HBM traffic = n*k*4 bytes (vs ~kt x for pair-parallel)
     const int64_t steps_total = (n + 31) / 32;
-    int row_chunks = (int)std::min<int64_t>(2048, std::max<int64_t>(1, steps_total));
+    int64_t target_chunks = 2048;
+    if (const char *e = getenv("ANOVOS_GRAM_CHUNKS")) target_chunks = atoll(e);
+    int row_chunks = (int)std::min<int64_t>(target_chunks, std::max<int64_t>(1, steps_total));
     auto partials = torch::empty({(int64_t)row_chunks * npairs, 256},
                                  torch::TensorOptions().dtype(torch::kFloat32).device(device));
     check_hip(anovos_centered_gram_sr((const void *const *)dptr.data_ptr<int64_t>(), n, k,
