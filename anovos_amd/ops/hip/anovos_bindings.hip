@@ -78,6 +78,7 @@ int anovos_centered_gram_sr(const void *const *cols, int64_t n, int k,
                             const float *means, const int *pair_i,
                             const int *pair_j, int npairs, int row_chunks,
                             float *partials, float *gram, hipStream_t stream);
+int anovos_gram_sr_grid(int k);
 int anovos_bucketize_label_counts(const void *const *cols, const uint8_t *label,
                                   const int64_t *lens, const double *cutflat,
                                   const int64_t *cutoff_off, const int *cutoff_len,
@@ -686,7 +687,9 @@ torch::Tensor centered_gram_bf16(std::vector<torch::Tensor> cols, torch::Tensor 
     // single-read kernel: one block stages a 32-row slab of ALL columns
     // through LDS; HBM traffic = n*k*4 bytes (vs ~kt x for pair-parallel)
     const int64_t steps_total = (n + 31) / 32;
-    int64_t target_chunks = 2048;
+    // one full wave of resident blocks (occupancy x CUs): maximizes each
+    // block's contiguous per-column streams (see anovos_gram_sr_grid)
+    int64_t target_chunks = anovos_gram_sr_grid(k);
     if (const char *e = getenv("ANOVOS_GRAM_CHUNKS")) target_chunks = atoll(e);
     int row_chunks = (int)std::min<int64_t>(target_chunks, std::max<int64_t>(1, steps_total));
     auto partials = torch::empty({(int64_t)row_chunks * npairs, 256},

@@ -322,3 +322,33 @@ extern "C" int anovos_centered_gram_sr(const void *const *cols, int64_t n,
                      pair_i, pair_j, k, gram);
   return (int)hipGetLastError();
 }
+
+extern "C" int anovos_gram_sr_grid(int k) {
+  // Preferred grid = EXACTLY the resident block capacity (one wave of
+  // blocks): each block then owns the longest possible contiguous
+  // per-column row range, which is what the 150-stream-per-block access
+  // pattern needs for DRAM locality (sweep at 20M x 150: 256 blocks
+  // 0.95 TB/s, 768 = capacity 2.20 TB/s, 2048 1.93, 16384 1.01).
+  const int kt = (k + 15) / 16;
+  const int ktot = kt * 16;
+  const size_t lds = (size_t)4 * 4 * (ktot + 3) * 16;
+  static int cached_k = -1, cached_grid = 0;
+  if (k == cached_k) return cached_grid;
+  int per_cu = 0;
+  if (kt <= 10) {
+    auto *fn = gram_singleread_kernel<14, 20>;
+    hipOccupancyMaxActiveBlocksPerMultiprocessor(&per_cu, reinterpret_cast<const void *>(fn), THREADS, lds);
+  } else {
+    auto *fn = gram_singleread_kernel<23, 26>;
+    hipOccupancyMaxActiveBlocksPerMultiprocessor(&per_cu, reinterpret_cast<const void *>(fn), THREADS, lds);
+  }
+  if (per_cu <= 0) per_cu = 3;
+  hipDeviceProp_t prop;
+  int dev = 0;
+  hipGetDevice(&dev);
+  hipGetDeviceProperties(&prop, dev);
+  int grid = per_cu * (prop.multiProcessorCount > 0 ? prop.multiProcessorCount : 256);
+  cached_k = k;
+  cached_grid = grid;
+  return grid;
+}
