@@ -335,13 +335,12 @@ extern "C" int anovos_gram_sr_grid(int k) {
   static int cached_k = -1, cached_grid = 0;
   if (k == cached_k) return cached_grid;
   int per_cu = 0;
-  if (kt <= 10) {
-    auto *fn = gram_singleread_kernel<14, 20>;
-    hipOccupancyMaxActiveBlocksPerMultiprocessor(&per_cu, reinterpret_cast<const void *>(fn), THREADS, lds);
-  } else {
-    auto *fn = gram_singleread_kernel<23, 26>;
-    hipOccupancyMaxActiveBlocksPerMultiprocessor(&per_cu, reinterpret_cast<const void *>(fn), THREADS, lds);
-  }
+  using KFn = void (*)(const float *const *, int64_t, int, int, const float *,
+                       const int *, const int *, int, int, float *);
+  KFn fn = (kt <= 10) ? (KFn)gram_singleread_kernel<14>
+                      : (KFn)gram_singleread_kernel<23>;
+  (void)hipOccupancyMaxActiveBlocksPerMultiprocessor(
+      &per_cu, reinterpret_cast<const void *>(fn), THREADS, lds);
   if (per_cu <= 0) per_cu = 3;
   hipDeviceProp_t prop;
   int dev = 0;
