@@ -124,7 +124,12 @@ def _to_frame(table: pa.Table, device) -> AnovosFrame:
             if pa.types.is_dictionary(t):
                 darr = arr
                 dictionary = [("" if v is None else str(v)) for v in darr.dictionary.to_pylist()]
-                codes = darr.indices.to_numpy(zero_copy_only=False).astype(np.int64, copy=True)
+                raw = darr.indices.to_numpy(zero_copy_only=False)
+                # null indices surface as float NaN: normalize BEFORE the
+                # int cast (NaN->int is platform-defined)
+                if raw.dtype.kind == "f":
+                    raw = np.nan_to_num(raw, nan=-1.0)
+                codes = raw.astype(np.int64, copy=True)
                 null = np.asarray(pc.is_null(darr).to_numpy(zero_copy_only=False))
                 cols[name] = _sorted_dict_column(name, codes, dictionary, null, dev)
                 continue
