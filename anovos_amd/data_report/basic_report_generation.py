@@ -76,6 +76,9 @@ def anovos_basic_report(ctx, idf, id_col="", label_col="", event_label="",
     for f in SG_funcs:
         try:
             stats = f(ctx, idf, drop_cols=drop)
+            # persist like the reference (:167-210): later checkers reuse
+            # these via the stats_args wiring instead of re-scanning
+            stats.to_csv(ends_with(local_path) + f.__name__ + ".csv", index=False)
             if print_impact:
                 print(f.__name__, "\n", stats.to_string(index=False))
             sg_parts.append(f"<h3>{_remove_u_score(f.__name__)}</h3>" + _tbl(stats))
@@ -84,7 +87,8 @@ def anovos_basic_report(ctx, idf, id_col="", label_col="", event_label="",
     qc_parts = []
     for f in QC_rows_funcs + QC_cols_funcs:
         try:
-            out = f(ctx, idf, drop_cols=drop) if f not in QC_rows_funcs else f(ctx, idf)
+            extra = stats_args(local_path, f.__name__)
+            out = f(ctx, idf, drop_cols=drop, **extra) if f not in QC_rows_funcs else f(ctx, idf)
             stats = out[1] if isinstance(out, tuple) else out
             if print_impact:
                 print(f.__name__, "\n", stats.to_string(index=False))
@@ -94,11 +98,11 @@ def anovos_basic_report(ctx, idf, id_col="", label_col="", event_label="",
     aa_parts = []
     for f in AA_funcs + AT_funcs:
         try:
-            kwargs = {}
+            kwargs = dict(stats_args(local_path, f.__name__))
             if f in AT_funcs:
                 if not label_col:
                     continue
-                kwargs = {"label_col": label_col, "event_label": event_label}
+                kwargs.update({"label_col": label_col, "event_label": event_label})
             stats = f(ctx, idf, drop_cols=drop, **kwargs)
             aa_parts.append(f"<h3>{_remove_u_score(f.__name__)}</h3>" + _tbl(stats))
         except Exception as e:
