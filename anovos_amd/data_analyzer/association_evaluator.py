@@ -102,6 +102,20 @@ def _binned_label_counts(idf, col: str, label: torch.Tensor):
     return out[col]
 
 
+
+def _cached_event_indicator(idf, label_col, event_label):
+    """(label tensor, global event count) memoized on the frame: IV and
+    IG share one label reduction + host sync per step."""
+    key = ("event_indicator", label_col, str(event_label))
+    hit = idf.aux_cache.get(key)
+    if hit is not None:
+        return hit
+    label = _event_indicator(idf, label_col, event_label)
+    total = int(dist.all_reduce_scalar(int(label.sum())))
+    idf.aux_cache[key] = (label, total)
+    return label, total
+
+
 def _encoded_label_counts(ctx, idf, cols, label_col, event_label, label, encoding_configs):
     """Shared IV/IG front end: bin numerics per encoding_configs, then one
     fused multi-column label-count pass. The result is memoized in the
@@ -299,8 +313,8 @@ def IV_calculation(
     cols = [e for e in dict.fromkeys(list_of_cols) if e not in (list(drop_cols) + [label_col])]
     if any(x not in idf.columns for x in cols) or len(cols) == 0:
         raise TypeError("Invalid input for Column(s)")
-    label = _event_indicator(idf, label_col, event_label)
-    if int(dist.all_reduce_scalar(int(label.sum()))) == 0:
+    label, _n_events = _cached_event_indicator(idf, label_col, event_label)
+    if _n_events == 0:
         raise TypeError("Invalid input for Event Label Value")
     counts = _encoded_label_counts(ctx, idf, cols, label_col, event_label, label, encoding_configs)
     rows = []
@@ -348,9 +362,8 @@ def IG_calculation(
     cols = [e for e in dict.fromkeys(list_of_cols) if e not in (list(drop_cols) + [label_col])]
     if any(x not in idf.columns for x in cols) or len(cols) == 0:
         raise TypeError("Invalid input for Column(s)")
-    label = _event_indicator(idf, label_col, event_label)
+    label, total_events = _cached_event_indicator(idf, label_col, event_label)
     total_rows = idf.count()
-    total_events = int(dist.all_reduce_scalar(int(label.sum())))
     if total_events == 0:
         raise TypeError("Invalid input for Event Label Value")
     total_event = total_events / total_rows

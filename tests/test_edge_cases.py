@@ -106,3 +106,83 @@ def test_high_cardinality_int(ctx):
     u = sg.uniqueCount_computation(ctx, idf, compute_approx_unique_count=True)
     est = float(u["unique_values"][0])
     assert abs(est - n) / n < 0.1  # nearly all distinct
+
+
+def test_association_on_edges(ctx):
+    """IV/IG with an all-null column, correlation with a constant column,
+    and a label with a single event (round-2 fused K9 paths on CPU)."""
+    import numpy as np
+    import pandas as pd
+    import torch
+
+    from anovos_amd.core.frame import AnovosFrame, Column
+    from anovos_amd.data_analyzer import association_evaluator as ae
+
+    n = 5000
+    rng = np.random.default_rng(9)
+    pdf = pd.DataFrame(
+        {
+            "x": rng.normal(0, 1, n),
+            "const": np.ones(n),
+            "allnull": np.full(n, np.nan),
+            "cat": rng.choice(["a", "b"], n),
+            "label": rng.choice(["0", "1"], n, p=[0.6, 0.4]),
+        }
+    )
+    idf = AnovosFrame.from_pandas(pdf, device="cpu")
+    iv = ae.IV_calculation(ctx, idf, label_col="label", event_label="1")
+    assert set(iv["attribute"]) == {"x", "const", "allnull", "cat"}
+    assert np.isfinite(iv["iv"]).all()
+    ig = ae.IG_calculation(ctx, idf, label_col="label", event_label="1")
+    assert np.isfinite(ig["ig"]).all()
+    corr = ae.correlation_matrix(ctx, idf, ["x", "const", "allnull"])
+    m = corr[sorted(["x", "const", "allnull"])].to_numpy()
+    assert np.diag(m).tolist() == [1.0, 1.0, 1.0]  # diag forced to 1 even for degenerate cols
+
+    # single-event label: IV still computes (0.5-smoothed WOE fallback)
+    pdf2 = pdf.copy()
+    lab = np.array(["0"] * n, dtype=object)
+    lab[0] = "1"
+    pdf2["label"] = lab
+    idf2 = AnovosFrame.from_pandas(pdf2, device="cpu")
+    iv2 = ae.IV_calculation(ctx, idf2, label_col="label", event_label="1")
+    assert np.isfinite(iv2["iv"]).all()
+
+
+def test_drift_with_missing_target_column(ctx):
+    """Source has a column absent in the target: excluded with a warning,
+    remaining metrics intact (workflow parity)."""
+    import warnings as _w
+
+    import numpy as np
+    import pandas as pd
+
+    from anovos_amd.core.frame import AnovosFrame
+    from anovos_amd.drift_stability import drift_detector as dd
+
+    rng = np.random.default_rng(5)
+    n = 20000
+    tgt = pd.DataFrame({"x": rng.normal(0, 1, n), "y": rng.normal(3, 1, n)})
+    src = pd.DataFrame({"x": rng.normal(0.3, 1, n), "y": rng.normal(3, 1, n), "extra": rng.normal(0, 1, n)})
+    with _w.catch_warnings():
+        _w.simplefilter("ignore")
+        stats = dd.statistics(ctx, AnovosFrame.from_pandas(tgt), AnovosFrame.from_pandas(src),
+                              list_of_cols="all", method_type="all", use_sampling=False,
+                              model_directory="/tmp/_drift_edge")
+    assert set(stats["attribute"]) <= {"x", "y"}
+    row = stats[stats["attribute"] == "x"].iloc[0]
+    assert float(row["PSI"]) > 0
+
+
+def test_stability_two_snapshots(ctx):
+    import numpy as np
+    import pandas as pd
+
+    from anovos_amd.core.frame import AnovosFrame
+    from anovos_amd.drift_stability import stability as st
+
+    rng = np.random.default_rng(3)
+    idfs = [AnovosFrame.from_pandas(pd.DataFrame({"v": rng.normal(10, 1 + 0.2 * k, 5000)})) for k in range(2)]
+    out = st.stability_index_computation(ctx, *idfs)
+    assert "stability_index" in out.columns and len(out) == 1
+    assert 0 <= float(out["stability_index"].iloc[0]) <= 4
