@@ -66,6 +66,8 @@ def _read_value(buf: io.BytesIO, schema) -> Any:
         return _read_value(buf, schema[idx])
     if isinstance(schema, dict):
         t = schema["type"]
+        if isinstance(t, str) and t in ("int", "long", "float", "double", "boolean", "string", "bytes", "null") and "logicalType" in schema:
+            return _read_value(buf, t)
         if t == "record":
             return {f["name"]: _read_value(buf, f["type"]) for f in schema["fields"]}
         if t in ("enum",):
@@ -114,6 +116,9 @@ def _write_value(out: io.BytesIO, schema, v: Any):
     if isinstance(schema, dict) and schema["type"] == "record":
         for f in schema["fields"]:
             _write_value(out, f["type"], v[f["name"]])
+        return
+    if isinstance(schema, dict):  # logical type wrapper (timestamp-micros)
+        _write_value(out, schema["type"], v)
         return
     if schema == "null":
         return
@@ -172,14 +177,26 @@ def read_avro(path: str) -> pd.DataFrame:
             records.append(_read_value(bb, schema))
         if buf.read(16) != sync:
             raise ValueError("avro sync marker mismatch")
-    return pd.DataFrame.from_records(records)
+    pdf = pd.DataFrame.from_records(records)
+    # decode logical timestamp-micros columns
+    for f in schema.get("fields", []):
+        ftype = f["type"]
+        parts = ftype if isinstance(ftype, list) else [ftype]
+        for pt in parts:
+            if isinstance(pt, dict) and pt.get("logicalType") == "timestamp-micros":
+                name = f["name"]
+                if name in pdf.columns:
+                    pdf[name] = pd.to_datetime(pdf[name], unit="us")
+    return pdf
 
 
 def _schema_for(pdf: pd.DataFrame) -> dict:
     fields = []
     for name in pdf.columns:
         s = pdf[name]
-        if pd.api.types.is_float_dtype(s):
+        if pd.api.types.is_datetime64_any_dtype(s):
+            t = {"type": "long", "logicalType": "timestamp-micros"}
+        elif pd.api.types.is_float_dtype(s):
             t = "double"
         elif pd.api.types.is_integer_dtype(s):
             t = "long"
@@ -206,7 +223,12 @@ def write_avro(pdf: pd.DataFrame, path: str, codec: str = "deflate"):
     body = io.BytesIO()
     cols = list(pdf.columns)
     n = len(pdf)
-    arrs = {c: pdf[c].to_numpy() for c in cols}
+    arrs = {}
+    for c in cols:
+        if pd.api.types.is_datetime64_any_dtype(pdf[c]):
+            arrs[c] = pdf[c].astype("datetime64[us]").astype("int64").to_numpy()
+        else:
+            arrs[c] = pdf[c].to_numpy()
     nulls = {c: pdf[c].isna().to_numpy() for c in cols}
     for i in range(n):
         rec = {c: (None if nulls[c][i] else arrs[c][i]) for c in cols}

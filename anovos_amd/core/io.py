@@ -228,7 +228,34 @@ def _read_json(path, cfg, device):
     tables = [pajson.read_json(p) for p in parts]
     if not tables:
         raise FileNotFoundError(f"no json files under {path}")
-    return _to_frame(_concat_tables(tables), device)
+    return _to_frame(_parse_iso_timestamp_strings(_concat_tables(tables)), device)
+
+
+def _parse_iso_timestamp_strings(table: pa.Table) -> pa.Table:
+    """Arrow's JSON reader leaves ISO-8601 timestamps as strings (the
+    writer emits date_format='iso' for Spark parity); detect and parse
+    string columns whose values are ISO timestamps."""
+    import re
+
+    import pyarrow.compute as pc
+
+    for i, name in enumerate(table.column_names):
+        col = table.column(name)
+        arr = col.combine_chunks() if isinstance(col, pa.ChunkedArray) else col
+        if not pa.types.is_string(arr.type) or len(arr) == 0:
+            continue
+        sample = next((v for v in arr.to_pylist()[:50] if v), None)
+        if not (sample and re.match(r"^\d{4}-\d{2}-\d{2}T\d{2}:\d{2}:\d{2}", str(sample))):
+            continue
+        try:
+            # arrow's ISO parser handles fractional seconds (strptime's
+            # %f does not)
+            ts = pc.cast(arr, pa.timestamp("us"))
+            if pc.sum(pc.is_null(ts)).as_py() == pc.sum(pc.is_null(arr)).as_py():
+                table = table.set_column(i, name, ts)
+        except Exception:
+            pass
+    return table
 
 
 def _read_avro(path, cfg, device):
@@ -283,7 +310,9 @@ def write_dataset(idf: AnovosFrame, file_path: str, file_type: str, file_configs
             compression = None
         papq.write_table(pa.Table.from_pandas(pdf, preserve_index=False), part + ".parquet", compression=compression)
     elif ft == "json":
-        pdf.to_json(part + ".json", orient="records", lines=True)
+        # ISO timestamps (Spark JSON parity; epoch-ms ints would come
+        # back as plain numerics)
+        pdf.to_json(part + ".json", orient="records", lines=True, date_format="iso")
     elif ft == "avro":
         from anovos_amd.core import avro_codec
 
