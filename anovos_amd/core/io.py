@@ -44,7 +44,9 @@ def _expand_parts(file_path: str, exts) -> List[str]:
     return [file_path]
 
 
-def _my_parts(parts: List[str]) -> List[str]:
+def _my_parts(parts: List[str], sharded: bool = True) -> List[str]:
+    if not sharded:
+        return parts
     w, r = dist.world_size(), dist.rank()
     if w <= 1:
         return parts
@@ -52,17 +54,23 @@ def _my_parts(parts: List[str]) -> List[str]:
     return mine
 
 
-def read_dataset(file_path: str, file_type: str, file_configs: Dict = None, device="cpu") -> AnovosFrame:
+def read_dataset(file_path: str, file_type: str, file_configs: Dict = None, device="cpu",
+                 sharded: bool = True) -> AnovosFrame:
+    """sharded=True (default): part files round-robin across ranks (row
+    partitioning). sharded=False: EVERY rank reads all parts — required
+    for small stats/model artifacts consumed via the stats-reuse
+    contract (a sharded read would hand rank>0 zero parts of a
+    single-file stats CSV)."""
     file_configs = dict(file_configs or {})
     ft = file_type.lower()
     if ft == "csv":
-        idf = _read_csv(file_path, file_configs, device)
+        idf = _read_csv(file_path, file_configs, device, sharded)
     elif ft == "parquet":
-        idf = _read_parquet(file_path, file_configs, device)
+        idf = _read_parquet(file_path, file_configs, device, sharded)
     elif ft == "json":
-        idf = _read_json(file_path, file_configs, device)
+        idf = _read_json(file_path, file_configs, device, sharded)
     elif ft == "avro":
-        idf = _read_avro(file_path, file_configs, device)
+        idf = _read_avro(file_path, file_configs, device, sharded)
     else:
         raise ValueError(f"unsupported file_type: {file_type}")
     return _unify_frame_dictionaries(idf)
@@ -196,10 +204,10 @@ def _sorted_dict_column(name, codes_i64, dictionary, null_mask, dev):
     return Column(name, "string", torch.from_numpy(np.ascontiguousarray(out)).to(dev), sorted_dict)
 
 
-def _read_csv(path, cfg, device):
+def _read_csv(path, cfg, device, sharded=True):
     header = str(cfg.get("header", True)).lower() in ("true", "1")
     delim = cfg.get("delimiter", cfg.get("sep", ","))
-    parts = _my_parts(_expand_parts(path, [".csv"]))
+    parts = _my_parts(_expand_parts(path, [".csv"]), sharded)
     tables = []
     for p in parts:
         ro = pacsv.ReadOptions(autogenerate_column_names=not header)
@@ -213,18 +221,18 @@ def _read_csv(path, cfg, device):
     return _to_frame(_concat_tables(tables), device)
 
 
-def _read_parquet(path, cfg, device):
-    parts = _my_parts(_expand_parts(path, [".parquet", ".pq"]))
+def _read_parquet(path, cfg, device, sharded=True):
+    parts = _my_parts(_expand_parts(path, [".parquet", ".pq"]), sharded)
     if not parts:
         raise FileNotFoundError(f"no parquet files under {path}")
     tables = [papq.read_table(p) for p in parts]
     return _to_frame(_concat_tables(tables), device)
 
 
-def _read_json(path, cfg, device):
+def _read_json(path, cfg, device, sharded=True):
     import pyarrow.json as pajson
 
-    parts = _my_parts(_expand_parts(path, [".json", ".jsonl"]))
+    parts = _my_parts(_expand_parts(path, [".json", ".jsonl"]), sharded)
     tables = [pajson.read_json(p) for p in parts]
     if not tables:
         raise FileNotFoundError(f"no json files under {path}")
@@ -258,10 +266,10 @@ def _parse_iso_timestamp_strings(table: pa.Table) -> pa.Table:
     return table
 
 
-def _read_avro(path, cfg, device):
+def _read_avro(path, cfg, device, sharded=True):
     from anovos_amd.core import avro_codec
 
-    parts = _my_parts(_expand_parts(path, [".avro"]))
+    parts = _my_parts(_expand_parts(path, [".avro"]), sharded)
     if not parts:
         raise FileNotFoundError(f"no avro files under {path}")
     pdfs = [avro_codec.read_avro(p) for p in parts]

@@ -170,7 +170,7 @@ def nullColumns_detection(
     else:
         from anovos_amd.data_ingest.data_ingest import read_dataset
 
-        odf_print = read_dataset(ctx, **stats_missing).to_pandas()[["attribute", "missing_count", "missing_pct"]]
+        odf_print = read_dataset(ctx, **stats_missing, sharded=False).to_pandas()[["attribute", "missing_count", "missing_pct"]]
     missing_cols = odf_print[odf_print["missing_count"] > 0]["attribute"].tolist()
 
     num_cols, cat_cols, _ = attributeType_segregation(idf)
@@ -218,7 +218,7 @@ def nullColumns_detection(
             else:
                 from anovos_amd.data_ingest.data_ingest import read_dataset
 
-                uc = read_dataset(ctx, **stats_unique).to_pandas()
+                uc = read_dataset(ctx, **stats_unique, sharded=False).to_pandas()
             remove_cols = uc[uc["unique_values"] < 2]["attribute"].tolist()
             cols_ = [e for e in list_of_cols if e not in remove_cols]
             if treatment_threshold:
@@ -530,7 +530,7 @@ def IDness_detection(ctx, idf, list_of_cols="all", drop_cols=[], treatment=False
     else:
         from anovos_amd.data_ingest.data_ingest import read_dataset
 
-        odf_print = read_dataset(ctx, **stats_unique).to_pandas()
+        odf_print = read_dataset(ctx, **stats_unique, sharded=False).to_pandas()
         odf_print = odf_print[odf_print["attribute"].isin(list_of_cols)]
     odf_print = odf_print.copy()
     odf_print["flagged"] = (odf_print["IDness"] >= treatment_threshold).astype(int)
@@ -578,7 +578,7 @@ def biasedness_detection(ctx, idf, list_of_cols="all", drop_cols=[], treatment=F
     else:
         from anovos_amd.data_ingest.data_ingest import read_dataset
 
-        odf_print = read_dataset(ctx, **stats_mode).to_pandas()[["attribute", "mode", "mode_rows", "mode_pct"]]
+        odf_print = read_dataset(ctx, **stats_mode, sharded=False).to_pandas()[["attribute", "mode", "mode_rows", "mode_pct"]]
         odf_print = odf_print[odf_print["attribute"].isin(list_of_cols)]
     odf_print = odf_print.copy()
     odf_print["flagged"] = [(1 if (p is None or p != p or p >= treatment_threshold) else 0) for p in odf_print["mode_pct"]]

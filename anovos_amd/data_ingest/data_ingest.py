@@ -19,10 +19,11 @@ from anovos_amd.core.frame import AnovosFrame, Column
 from anovos_amd.shared.utils import attributeType_segregation, pairwise_reduce
 
 
-def read_dataset(ctx, file_path: str, file_type: str, file_configs: Dict = {}) -> AnovosFrame:
-    """Reference data_ingest.py:23-51 — csv/parquet/avro/json."""
+def read_dataset(ctx, file_path: str, file_type: str, file_configs: Dict = {}, sharded: bool = True) -> AnovosFrame:
+    """Reference data_ingest.py:23-51 — csv/parquet/avro/json.
+    sharded=False replicates all parts to every rank (stats artifacts)."""
     device = getattr(ctx, "device", "cpu")
-    return core_io.read_dataset(file_path, file_type, file_configs, device=device)
+    return core_io.read_dataset(file_path, file_type, file_configs, device=device, sharded=sharded)
 
 
 def write_dataset(idf: AnovosFrame, file_path: str, file_type: str, file_configs: Dict = {}, column_order: List[str] = []):

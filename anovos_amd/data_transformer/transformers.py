@@ -594,7 +594,7 @@ def imputation_MMM(
     if stats_missing:
         from anovos_amd.data_ingest.data_ingest import read_dataset as _rd
 
-        miss_df = _rd(ctx, **stats_missing).to_pandas()
+        miss_df = _rd(ctx, **stats_missing, sharded=False).to_pandas()
         missing = dict(zip(miss_df["attribute"], miss_df["missing_count"]))
     else:
         missing, _ = stats_ops.null_counts(idf, candidates)
@@ -633,7 +633,7 @@ def imputation_MMM(
             if stats_mode:
                 from anovos_amd.data_ingest.data_ingest import read_dataset as _rd
 
-                mdf = _rd(ctx, **stats_mode).to_pandas()
+                mdf = _rd(ctx, **stats_mode, sharded=False).to_pandas()
                 pre = dict(zip(mdf["attribute"], mdf["mode"]))
             else:
                 pre = {}

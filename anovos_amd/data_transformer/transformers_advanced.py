@@ -37,7 +37,7 @@ def _resolve_missing_cols(ctx, idf, list_of_cols, drop_cols, stats_missing):
     if stats_missing:
         from anovos_amd.data_ingest.data_ingest import read_dataset
 
-        miss = read_dataset(ctx, **stats_missing).to_pandas()
+        miss = read_dataset(ctx, **stats_missing, sharded=False).to_pandas()
         missing_cols = miss[miss["missing_count"] > 0]["attribute"].tolist()
     else:
         nulls, _ = stats_ops.null_counts(idf, num_cols)

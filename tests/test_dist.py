@@ -504,3 +504,96 @@ def test_dist_exact_continuous_mode_partitioned():
     res = json.load(open(out))
     assert res["mode"] == pytest.approx(0.123456789, abs=1e-12)
     assert res["count"] == 500
+
+
+def _workflow_worker(rank, port, workdir, out):
+    os.environ.update({"RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": "2",
+                       "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port)})
+    os.chdir(workdir)
+    import torch.distributed as td
+
+    from anovos_amd.core import dist
+    from anovos_amd import workflow
+
+    dist.init_from_env(timeout_s=180)
+    from anovos_amd.shared.context import init_context
+
+    init_context("cpu")
+    df = workflow.run(os.path.join(workdir, "cfg.yaml"))
+    rows = df.count()  # collective: must run on BOTH ranks
+    if rank == 0:
+        import pandas as _pd
+
+        counts = _pd.read_csv("report_stats/measures_of_counts.csv")
+        json.dump({"rows": rows,
+                   "fill_age": float(counts[counts["attribute"] == "age"]["fill_count"].iloc[0]),
+                   "report": os.path.exists("report_stats/ml_anovos_report.html")}, open(out, "w"))
+    td.barrier()
+    td.destroy_process_group()
+
+
+def test_dist_full_workflow_two_ranks(tmp_path):
+    """The entire YAML workflow (ETL -> stats -> QC -> associations ->
+    transformers -> report) as TWO ranks over gloo: part files shard by
+    rank, stats merge globally, rank 0 writes the report. Mirrors how
+    the 8-GPU run executes."""
+    import sys as _sys
+
+    import pandas as pd
+    import yaml as _yaml
+
+    _sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "tools"))
+    import make_income_data as mid
+
+    pdf = mid.make(4000)
+    d = tmp_path / "data" / "income_dataset" / "csv"
+    d.mkdir(parents=True)
+    pdf.iloc[:2000].to_csv(d / "part-00000.csv", index=False)
+    pdf.iloc[2000:].to_csv(d / "part-00001.csv", index=False)
+    cfg = {
+        "input_dataset": {
+            "read_dataset": {"file_path": "data/income_dataset/csv", "file_type": "csv",
+                             "file_configs": {"header": True, "inferSchema": True}},
+            "delete_column": ["logfnl"],
+        },
+        "stats_generator": {
+            "metric": ["global_summary", "measures_of_counts", "measures_of_dispersion"],
+            "metric_args": {"list_of_cols": "all", "drop_cols": ["ifa"]},
+        },
+        "quality_checker": {
+            "nullColumns_detection": {"list_of_cols": "all", "drop_cols": ["ifa", "income"],
+                                       "treatment": True, "treatment_method": "MMM"},
+        },
+        "association_evaluator": {
+            "IV_calculation": {"list_of_cols": "all", "drop_cols": ["ifa"],
+                               "label_col": "income", "event_label": ">50K"},
+        },
+        "transformers": {
+            "numerical_rescaling": {"z_standardization": {"list_of_cols": ["age"], "output_mode": "append"}},
+        },
+        "report_preprocessing": {
+            "master_path": "report_stats",
+            "charts_to_objects": {"list_of_cols": "all", "drop_cols": ["ifa"],
+                                   "label_col": "income", "event_label": ">50K",
+                                   "bin_method": "equal_frequency", "bin_size": 10,
+                                   "source_path": "inter"},
+        },
+        "report_generation": {"master_path": "report_stats", "final_report_path": "report_stats",
+                              "label_col": "income", "event_label": ">50K"},
+    }
+    with open(tmp_path / "cfg.yaml", "w") as f:
+        _yaml.safe_dump(cfg, f, sort_keys=False)
+    port = _free_port()
+    out = str(tmp_path / "res.json")
+    mp_ctx = mp.get_context("spawn")
+    procs = [mp_ctx.Process(target=_workflow_worker, args=(r, port, str(tmp_path), out)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=300)
+        assert p.exitcode == 0, f"workflow worker failed: exit {p.exitcode}"
+    res = json.load(open(out))
+    assert res["rows"] == 4000  # both shards merged in the global count
+    truth_fill = int(pdf["age"].notna().sum())
+    assert res["fill_age"] == truth_fill
+    assert res["report"] is True
