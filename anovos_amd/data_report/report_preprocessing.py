@@ -266,6 +266,17 @@ def plot_comparative_drift(ctx, idf, source_pdf, col, cutoffs_path, is_numcol=Fa
     return fig
 
 
+
+def _write_fig_json(fig, path):
+    """All ranks COMPUTE each chart (the underlying stats are
+    collectives), but only rank 0 WRITES — concurrent identical writes
+    to one path can tear the file."""
+    from anovos_amd.core import dist as _dist
+
+    if _dist.rank() == 0:
+        fig.write_json(path)
+
+
 def charts_to_objects(ctx, idf, list_of_cols="all", drop_cols=[], label_col=None,
                       event_label=1, bin_method="equal_range", bin_size=10, coverage=1.0,
                       drift_detector=False, outlier_charts=False, source_path="NA",
@@ -339,38 +350,42 @@ def charts_to_objects(ctx, idf, list_of_cols="all", drop_cols=[], label_col=None
         if col in cat_cols:
             view = idf_encoded
             f = plot_frequency(ctx, view, col, cutoffs_path, is_numcol=False)
-            f.write_json(ends_with(local_path) + "freqDist_" + col)
+            _write_fig_json(f, ends_with(local_path) + "freqDist_" + col)
             if label_col and col != label_col:
                 f = plot_eventRate(ctx, view, col, label_col, event_label, cutoffs_path, is_numcol=False)
-                f.write_json(ends_with(local_path) + "eventDist_" + col)
+                _write_fig_json(f, ends_with(local_path) + "eventDist_" + col)
             if drift_detector:
                 try:
                     src = pd.read_csv(os.path.join(source_path, "drift_statistics", "frequency_counts", col, "part-00000.csv"))
                     f = plot_comparative_drift(ctx, view, src, col, cutoffs_path, is_numcol=False)
-                    f.write_json(ends_with(local_path) + "drift_" + col)
+                    _write_fig_json(f, ends_with(local_path) + "drift_" + col)
                 except Exception:
                     pass
         if col in num_cols:
             if outlier_charts:
                 f = plot_outlier(ctx, idf, col)
-                f.write_json(ends_with(local_path) + "outlier_" + col)
+                _write_fig_json(f, ends_with(local_path) + "outlier_" + col)
             view = idf_encoded.drop([col]).rename({col + "_binned": col})
             f = plot_frequency(ctx, view, col, cutoffs_path, is_numcol=True)
-            f.write_json(ends_with(local_path) + "freqDist_" + col)
+            _write_fig_json(f, ends_with(local_path) + "freqDist_" + col)
             if label_col and col != label_col:
                 f = plot_eventRate(ctx, view, col, label_col, event_label, cutoffs_path, is_numcol=True)
-                f.write_json(ends_with(local_path) + "eventDist_" + col)
+                _write_fig_json(f, ends_with(local_path) + "eventDist_" + col)
             if drift_detector:
                 try:
                     src = pd.read_csv(os.path.join(source_path, "drift_statistics", "frequency_counts", col, "part-00000.csv"))
                     f = plot_comparative_drift(ctx, view, src, col, cutoffs_path, is_numcol=True)
-                    f.write_json(ends_with(local_path) + "drift_" + col)
+                    _write_fig_json(f, ends_with(local_path) + "drift_" + col)
                 except Exception:
                     pass
 
-    pd.DataFrame(idf.dtypes, columns=["attribute", "data_type"]).to_csv(
-        ends_with(local_path) + "data_type.csv", index=False)
-    if run_type in ("emr", "ak8s"):
+    from anovos_amd.core import dist as _dist
+
+    if _dist.rank() == 0:
+        pd.DataFrame(idf.dtypes, columns=["attribute", "data_type"]).to_csv(
+            ends_with(local_path) + "data_type.csv", index=False)
+    _dist.barrier()
+    if run_type in ("emr", "ak8s") and _dist.rank() == 0:
         # push the whole chart-object dir (reference report_preprocessing
         # .py:97-119 recursive aws s3 cp / azcopy side channel)
         _su.cloud_sync(local_path, master_path, run_type, auth_key, recursive=True)
