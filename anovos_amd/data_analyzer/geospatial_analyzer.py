@@ -32,6 +32,16 @@ global_paper_bg_color = "rgba(0,0,0,0)"
 _MAPBOX_STYLES = ["open-street-map", "carto-positron", "carto-darkmatter", "stamen-terrain", "stamen-toner", "stamen-watercolor"]
 
 
+
+def _rank0_write(fn):
+    """Only rank 0 persists report artifacts (stat computation above is
+    collective on all ranks; duplicate concurrent writes can tear files)."""
+    from anovos_amd.core import dist as _d
+
+    if _d.rank() == 0:
+        fn()
+
+
 def _num(df: AnovosFrame, col: str) -> torch.Tensor:
     from anovos_amd.data_transformer.geospatial import _num as _n
 
@@ -102,7 +112,7 @@ def descriptive_stats_gen(df, lat_col, long_col, geohash_col, id_col, master_pat
         )
         names = ["Overall_Summary", "Top_" + str(max_val) + "_Lat_Long"]
         for nm, tbl in zip(names, [gen_stats, top_pairs]):
-            tbl.to_csv(ends_with(master_path) + nm + "_1_" + lat_col + "_" + long_col + ".csv", index=False)
+            _rank0_write(lambda: tbl.to_csv(ends_with(master_path) + nm + "_1_" + lat_col + "_" + long_col + ".csv", index=False))
     if geohash_col is not None:
         from anovos_amd.core import dist as _dist3
 
@@ -138,8 +148,8 @@ def descriptive_stats_gen(df, lat_col, long_col, geohash_col, id_col, master_pat
             }
         )
         names = ["Overall_Summary", "Top_" + str(max_val) + "_Geohash_Distribution"]
-        gen_stats.to_csv(ends_with(master_path) + names[0] + "_2_" + geohash_col + ".csv", index=False)
-        top_gh.to_csv(ends_with(master_path) + names[1] + "_2_" + geohash_col + ".csv", index=False)
+        _rank0_write(lambda: gen_stats.to_csv(ends_with(master_path) + names[0] + "_2_" + geohash_col + ".csv", index=False))
+        _rank0_write(lambda: top_gh.to_csv(ends_with(master_path) + names[1] + "_2_" + geohash_col + ".csv", index=False))
 
 
 def lat_long_col_stats_gen(df, lat_col, long_col, id_col, master_path, max_val):
@@ -225,23 +235,23 @@ def geo_cluster_analysis(df, lat_col, long_col, max_cluster, eps, min_samples, m
     f1.update_layout(title_text=f"Elbow Curve Showing the Optimal Number of Clusters [K : {k}] <br><sup>Algorithm Used : KMeans</sup>")
     f1.layout.plot_bgcolor = global_plot_bg_color
     f1.layout.paper_bgcolor = global_paper_bg_color
-    f1.write_json(ends_with(master_path) + "cluster_plot_1_elbow_" + col_name)
+    _rank0_write(lambda: f1.write_json(ends_with(master_path) + "cluster_plot_1_elbow_" + col_name))
 
     labels, _ = _kmeans_torch(x, max(k, 2), seed=0)
     pdf["cluster"] = labels.cpu().numpy()
-    pdf.to_csv(ends_with(master_path) + "cluster_output_kmeans_" + col_name + ".csv", index=False)
+    _rank0_write(lambda: pdf.to_csv(ends_with(master_path) + "cluster_output_kmeans_" + col_name + ".csv", index=False))
 
     cluster_dtls = pdf.groupby("cluster").size().reset_index(name="counts")
     f2 = go.Figure(go.Pie(labels=cluster_dtls["cluster"], values=cluster_dtls["counts"], hole=0.3,
                           marker_colors=px.colors.sequential.Peach))
     f2.update_layout(title_text="Distribution of Clusters <br><sup>Algorithm Used : KMeans</sup>")
-    f2.write_json(ends_with(master_path) + "cluster_plot_2_kmeans_" + col_name)
+    _rank0_write(lambda: f2.write_json(ends_with(master_path) + "cluster_plot_2_kmeans_" + col_name))
 
     sample = pdf.sample(min(len(pdf), 10000), random_state=0) if len(pdf) else pdf
     f3 = px.scatter_mapbox(sample, lat=lat_col, lon=long_col, color=sample["cluster"].astype(str),
                            color_discrete_sequence=px.colors.qualitative.Safe, zoom=1)
     f3.update_layout(mapbox_style=style, title_text="Cluster-wise Geospatial Datapoints <br><sup>Algorithm Used : KMeans</sup>")
-    f3.write_json(ends_with(master_path) + "cluster_plot_3_kmeans_" + col_name)
+    _rank0_write(lambda: f3.write_json(ends_with(master_path) + "cluster_plot_3_kmeans_" + col_name))
 
     # ---- DBSCAN on a bounded driver sample (reference drove sklearn on pandas)
     from sklearn.cluster import DBSCAN
@@ -268,24 +278,24 @@ def geo_cluster_analysis(df, lat_col, long_col, max_cluster, eps, min_samples, m
     f1_ = go.Figure(go.Heatmap(z=sil, x=[str(m) for m in ms_grid], y=[f"{e:.2f}" for e in eps_grid], colorscale="Peach"))
     f1_.update_layout(title_text="Silhouette Scores across (eps x min_samples) <br><sup>Algorithm Used : DBSCAN</sup>",
                       xaxis_title="min_samples", yaxis_title="eps")
-    f1_.write_json(ends_with(master_path) + "cluster_plot_1_silhoutte_" + col_name)
+    _rank0_write(lambda: f1_.write_json(ends_with(master_path) + "cluster_plot_1_silhoutte_" + col_name))
 
     e_best, ms_best = best[0] if best[0] else (float(eps_grid[0]), int(ms_grid[0]))
     db_lab = DBSCAN(eps=e_best, min_samples=ms_best).fit_predict(samp)
     db = pd.DataFrame({lat_col: samp[:, 0], long_col: samp[:, 1], "cluster": db_lab})
-    db.to_csv(ends_with(master_path) + "cluster_output_dbscan_" + col_name + ".csv", index=False)
+    _rank0_write(lambda: db.to_csv(ends_with(master_path) + "cluster_output_dbscan_" + col_name + ".csv", index=False))
 
     db_dtls = db.groupby("cluster").size().reset_index(name="counts")
     f2_ = go.Figure(go.Pie(labels=db_dtls["cluster"], values=db_dtls["counts"], hole=0.3,
                            marker_colors=px.colors.sequential.Peach))
     f2_.update_layout(title_text="Distribution of Clusters <br><sup>Algorithm Used : DBSCAN</sup>")
-    f2_.write_json(ends_with(master_path) + "cluster_plot_2_dbscan_" + col_name)
+    _rank0_write(lambda: f2_.write_json(ends_with(master_path) + "cluster_plot_2_dbscan_" + col_name))
 
     f3_ = px.scatter_mapbox(db[db["cluster"] >= 0], lat=lat_col, lon=long_col,
                             color=db[db["cluster"] >= 0]["cluster"].astype(str),
                             color_discrete_sequence=px.colors.qualitative.Safe, zoom=1)
     f3_.update_layout(mapbox_style=style, title_text="Cluster-wise Geospatial Datapoints <br><sup>Algorithm Used : DBSCAN</sup>")
-    f3_.write_json(ends_with(master_path) + "cluster_plot_3_dbscan_" + col_name)
+    _rank0_write(lambda: f3_.write_json(ends_with(master_path) + "cluster_plot_3_dbscan_" + col_name))
 
     outliers = db[db["cluster"] == -1]
     f4 = go.Figure(go.Scatter(x=outliers[long_col], y=outliers[lat_col], mode="markers",
@@ -294,7 +304,7 @@ def geo_cluster_analysis(df, lat_col, long_col, max_cluster, eps, min_samples, m
                      xaxis_title=long_col, yaxis_title=lat_col)
     f4.layout.plot_bgcolor = global_plot_bg_color
     f4.layout.paper_bgcolor = global_paper_bg_color
-    f4.write_json(ends_with(master_path) + "cluster_plot_4_dbscan_1_" + col_name)
+    _rank0_write(lambda: f4.write_json(ends_with(master_path) + "cluster_plot_4_dbscan_1_" + col_name))
 
     # haversine-metric DBSCAN outliers
     db_lab_h = DBSCAN(eps=e_best / 60.0, min_samples=ms_best, metric="haversine").fit_predict(np.radians(samp))
@@ -304,7 +314,7 @@ def geo_cluster_analysis(df, lat_col, long_col, max_cluster, eps, min_samples, m
                                marker_symbol="x", marker_color=global_theme[4]))
     f4_.update_layout(title_text="Outlier Points <br><sup>Algorithm Used : DBSCAN (Haversine)</sup>",
                       xaxis_title=long_col, yaxis_title=lat_col)
-    f4_.write_json(ends_with(master_path) + "cluster_plot_4_dbscan_2_" + col_name)
+    _rank0_write(lambda: f4_.write_json(ends_with(master_path) + "cluster_plot_4_dbscan_2_" + col_name))
 
 
 def geo_cluster_generator(df, lat_col_list, long_col_list, geo_col_list, max_cluster, eps, min_samples, master_path, global_map_box_val, max_records):
@@ -346,7 +356,7 @@ def generate_loc_charts_processor(df, lat_col, long_col, geohash_col, max_val, i
             pdf = pd.DataFrame({la: lat[idx].cpu().numpy(), lo: lon[idx].cpu().numpy()})
             fig = px.scatter_mapbox(pdf, lat=la, lon=lo, color_discrete_sequence=[global_theme[2]], zoom=1)
             fig.update_layout(mapbox_style=style)
-            fig.write_json(ends_with(master_path) + "loc_charts_ll_" + la + "_" + lo)
+            _rank0_write(lambda: fig.write_json(ends_with(master_path) + "loc_charts_ll_" + la + "_" + lo))
     if geohash_col is not None:
         from anovos_amd.data_transformer.geospatial import geo_format_geohash
 
@@ -360,7 +370,7 @@ def generate_loc_charts_processor(df, lat_col, long_col, geohash_col, max_val, i
             pdf = pd.DataFrame({"lat": lat[idx].cpu().numpy(), "lon": lon[idx].cpu().numpy()})
             fig = px.scatter_mapbox(pdf, lat="lat", lon="lon", color_discrete_sequence=[global_theme[2]], zoom=1)
             fig.update_layout(mapbox_style=style)
-            fig.write_json(ends_with(master_path) + "loc_charts_gh_" + ghc)
+            _rank0_write(lambda: fig.write_json(ends_with(master_path) + "loc_charts_gh_" + ghc))
 
 
 def generate_loc_charts_controller(df, id_col, lat_col, long_col, geohash_col, max_val, global_map_box_val, master_path):
