@@ -78,7 +78,11 @@ def anovos_basic_report(ctx, idf, id_col="", label_col="", event_label="",
             stats = f(ctx, idf, drop_cols=drop)
             # persist like the reference (:167-210): later checkers reuse
             # these via the stats_args wiring instead of re-scanning
-            stats.to_csv(ends_with(local_path) + f.__name__ + ".csv", index=False)
+            from anovos_amd.core import dist as _dist
+
+            if _dist.rank() == 0:
+                stats.to_csv(ends_with(local_path) + f.__name__ + ".csv", index=False)
+            _dist.barrier()
             if print_impact:
                 print(f.__name__, "\n", stats.to_string(index=False))
             sg_parts.append(f"<h3>{_remove_u_score(f.__name__)}</h3>" + _tbl(stats))
@@ -115,6 +119,10 @@ def anovos_basic_report(ctx, idf, id_col="", label_col="", event_label="",
     ]
     out = render_report(tabs, title="Anovos Basic Report")
     path = ends_with(local_path) + "basic_report.html"
-    with open(path, "w") as f:
-        f.write(out)
+    from anovos_amd.core import dist as _dist
+
+    if _dist.rank() == 0:
+        with open(path, "w") as f:
+            f.write(out)
+    _dist.barrier()
     return path
