@@ -153,3 +153,22 @@ def test_monotonic_binning(ctx):
     odf = T.monotonic_binning(ctx, f, ["x"], label_col="label", event_label=1)
     b = odf.col("x").data
     assert int(b.max()) >= 3
+
+
+def test_cat_to_num_transformer_dispatch(ctx, income_frame):
+    """The supervised/unsupervised dispatcher (reference transformers.py:428)."""
+    from anovos_amd.data_transformer.transformers import cat_to_num_transformer
+
+    un = cat_to_num_transformer(ctx, income_frame, list_of_cols=["workclass"], drop_cols=[],
+                                method_type="unsupervised", encoding="label_encoding",
+                                label_col=None, event_label=None)
+    assert un.col("workclass").kind == "numerical"
+    sup = cat_to_num_transformer(ctx, income_frame, list_of_cols=["education"], drop_cols=[],
+                                 method_type="supervised", encoding=None,
+                                 label_col="income", event_label=">50K")
+    assert sup.col("education").kind == "numerical"
+    import torch
+
+    v = sup.col("education").data
+    ok = v[~torch.isnan(v)]
+    assert bool(((ok >= 0) & (ok <= 1)).all())  # target rates
