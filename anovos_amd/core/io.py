@@ -204,10 +204,19 @@ def _sorted_dict_column(name, codes_i64, dictionary, null_mask, dev):
     return Column(name, "string", torch.from_numpy(np.ascontiguousarray(out)).to(dev), sorted_dict)
 
 
+def _empty_like_first(parts_all, read_one):
+    """Fewer part files than ranks: this rank holds an EMPTY shard with
+    the schema of part 0 (the reference's Spark reader parallelism never
+    constrains part count; every merge protocol tolerates 0-row shards)."""
+    t = read_one(parts_all[0])
+    return t.slice(0, 0)
+
+
 def _read_csv(path, cfg, device, sharded=True):
     header = str(cfg.get("header", True)).lower() in ("true", "1")
     delim = cfg.get("delimiter", cfg.get("sep", ","))
-    parts = _my_parts(_expand_parts(path, [".csv"]), sharded)
+    parts_all = _expand_parts(path, [".csv"])
+    parts = _my_parts(parts_all, sharded)
     tables = []
     for p in parts:
         ro = pacsv.ReadOptions(autogenerate_column_names=not header)
@@ -217,14 +226,22 @@ def _read_csv(path, cfg, device, sharded=True):
             pass  # arrow always infers; parity is close enough (ints/floats/strings)
         tables.append(pacsv.read_csv(p, read_options=ro, parse_options=po, convert_options=co))
     if not tables:
-        raise FileNotFoundError(f"no csv part files under {path}")
+        if not parts_all:
+            raise FileNotFoundError(f"no csv part files under {path}")
+        ro = pacsv.ReadOptions(autogenerate_column_names=not header)
+        po = pacsv.ParseOptions(delimiter=delim)
+        co = pacsv.ConvertOptions(strings_can_be_null=True)
+        tables = [_empty_like_first(parts_all, lambda p: pacsv.read_csv(p, read_options=ro, parse_options=po, convert_options=co))]
     return _to_frame(_concat_tables(tables), device)
 
 
 def _read_parquet(path, cfg, device, sharded=True):
-    parts = _my_parts(_expand_parts(path, [".parquet", ".pq"]), sharded)
+    parts_all = _expand_parts(path, [".parquet", ".pq"])
+    parts = _my_parts(parts_all, sharded)
     if not parts:
-        raise FileNotFoundError(f"no parquet files under {path}")
+        if not parts_all:
+            raise FileNotFoundError(f"no parquet files under {path}")
+        return _to_frame(_empty_like_first(parts_all, papq.read_table), device)
     tables = [papq.read_table(p) for p in parts]
     return _to_frame(_concat_tables(tables), device)
 
@@ -232,10 +249,13 @@ def _read_parquet(path, cfg, device, sharded=True):
 def _read_json(path, cfg, device, sharded=True):
     import pyarrow.json as pajson
 
-    parts = _my_parts(_expand_parts(path, [".json", ".jsonl"]), sharded)
+    parts_all = _expand_parts(path, [".json", ".jsonl"])
+    parts = _my_parts(parts_all, sharded)
     tables = [pajson.read_json(p) for p in parts]
     if not tables:
-        raise FileNotFoundError(f"no json files under {path}")
+        if not parts_all:
+            raise FileNotFoundError(f"no json files under {path}")
+        tables = [_empty_like_first(parts_all, pajson.read_json)]
     return _to_frame(_parse_iso_timestamp_strings(_concat_tables(tables)), device)
 
 
@@ -269,9 +289,13 @@ def _parse_iso_timestamp_strings(table: pa.Table) -> pa.Table:
 def _read_avro(path, cfg, device, sharded=True):
     from anovos_amd.core import avro_codec
 
-    parts = _my_parts(_expand_parts(path, [".avro"]), sharded)
+    parts_all = _expand_parts(path, [".avro"])
+    parts = _my_parts(parts_all, sharded)
     if not parts:
-        raise FileNotFoundError(f"no avro files under {path}")
+        if not parts_all:
+            raise FileNotFoundError(f"no avro files under {path}")
+        pdf = avro_codec.read_avro(parts_all[0]).iloc[:0]
+        return AnovosFrame.from_pandas(pdf, device=device)
     pdfs = [avro_codec.read_avro(p) for p in parts]
     pdf = pd.concat(pdfs, ignore_index=True) if len(pdfs) > 1 else pdfs[0]
     return AnovosFrame.from_pandas(pdf, device=device)

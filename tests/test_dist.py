@@ -597,3 +597,65 @@ def test_dist_full_workflow_two_ranks(tmp_path):
     truth_fill = int(pdf["age"].notna().sum())
     assert res["fill_age"] == truth_fill
     assert res["report"] is True
+
+
+def test_dist_workflow_fewer_parts_than_ranks(tmp_path):
+    """A single-part dataset at world_size=2: rank 1 holds an EMPTY
+    shard and every stage (stats, quantiles, checkers, associations,
+    report) must still complete with globally correct results — this is
+    how an 8-GPU node behaves on small side tables."""
+    import sys as _sys
+
+    import pandas as pd
+    import yaml as _yaml
+
+    _sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "tools"))
+    import make_income_data as mid
+
+    pdf = mid.make(3000)
+    d = tmp_path / "data" / "income_dataset" / "csv"
+    d.mkdir(parents=True)
+    pdf.to_csv(d / "part-00000.csv", index=False)  # ONE part only
+    cfg = {
+        "input_dataset": {
+            "read_dataset": {"file_path": "data/income_dataset/csv", "file_type": "csv",
+                             "file_configs": {"header": True, "inferSchema": True}},
+            "delete_column": ["logfnl"],
+        },
+        "stats_generator": {
+            "metric": ["global_summary", "measures_of_counts", "measures_of_percentiles"],
+            "metric_args": {"list_of_cols": "all", "drop_cols": ["ifa"]},
+        },
+        "quality_checker": {
+            "outlier_detection": {"list_of_cols": "all", "drop_cols": ["ifa", "income"],
+                                   "detection_side": "both", "treatment": False},
+        },
+        "association_evaluator": {
+            "IV_calculation": {"list_of_cols": "all", "drop_cols": ["ifa"],
+                               "label_col": "income", "event_label": ">50K"},
+        },
+        "report_preprocessing": {
+            "master_path": "report_stats",
+            "charts_to_objects": {"list_of_cols": "all", "drop_cols": ["ifa"],
+                                   "label_col": "income", "event_label": ">50K",
+                                   "bin_method": "equal_frequency", "bin_size": 10,
+                                   "source_path": "inter"},
+        },
+        "report_generation": {"master_path": "report_stats", "final_report_path": "report_stats",
+                              "label_col": "income", "event_label": ">50K"},
+    }
+    with open(tmp_path / "cfg.yaml", "w") as f:
+        _yaml.safe_dump(cfg, f, sort_keys=False)
+    port = _free_port()
+    out = str(tmp_path / "res.json")
+    mp_ctx = mp.get_context("spawn")
+    procs = [mp_ctx.Process(target=_workflow_worker, args=(r, port, str(tmp_path), out)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=300)
+        assert p.exitcode == 0, f"worker failed: exit {p.exitcode}"
+    res = json.load(open(out))
+    assert res["rows"] == 3000
+    assert res["fill_age"] == int(pdf["age"].notna().sum())
+    assert res["report"] is True
