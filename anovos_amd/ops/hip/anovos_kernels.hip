@@ -1126,6 +1126,7 @@ int anovos_hll(const void *x, int64_t n, int p, int nchunks, int dtype,
 
 int anovos_row_null(const void *const *cols, int ncols, int64_t n, int dtype,
                     int32_t *out, hipStream_t stream) {
+  if (n <= 0 || ncols <= 0) return 0;  // empty shard (ranks > part files)
   const int ncolchunks = (ncols + COLS_PB - 1) / COLS_PB;
   const int64_t nrowchunks = (n + ROWS_PB - 1) / ROWS_PB;
   dim3 grid((uint32_t)(nrowchunks * ncolchunks));
