@@ -134,6 +134,12 @@ def _encoded_label_counts(ctx, idf, cols, label_col, event_label, label, encodin
     if hit is not None:
         return hit
     from anovos_amd.ops import backend as _backend
+    from anovos_amd.ops.groupby import align_dictionaries
+
+    # dictionary-indexed slot layout below must be rank-identical (both
+    # the fused-path eligibility check and the final all-reduce key off
+    # dictionary sizes)
+    align_dictionaries(idf, list(cols) + [label_col])
 
     use_fused = (
         bool(encoding_configs)

@@ -178,6 +178,8 @@ def _kmeans_torch(x: torch.Tensor, k: int, iters: int = 50, seed: int = 0):
     """Lloyd K-Means on device: k-means++ seeding + cdist/argmin loop.
     Returns (labels, inertia). x: [N,2] float64."""
     n = x.shape[0]
+    if n == 0:
+        return torch.zeros(0, dtype=torch.long, device=x.device), 0.0
     g = torch.Generator(device="cpu").manual_seed(seed)
     centers = x[torch.randint(0, n, (1,), generator=g).item()].unsqueeze(0)
     for _ in range(1, k):
@@ -317,26 +319,54 @@ def geo_cluster_analysis(df, lat_col, long_col, max_cluster, eps, min_samples, m
     _rank0_write(lambda: f4_.write_json(ends_with(master_path) + "cluster_plot_4_dbscan_2_" + col_name))
 
 
+def _gathered_pair_pdf(df: AnovosFrame, lat_col: str, long_col: str, max_records: int) -> pd.DataFrame:
+    """Bounded global (lat, long) sample as pandas: nan-filter + local
+    subsample, one varlen all-gather (≤ max_records × world_size rows),
+    then a deterministic re-subsample so every rank holds the SAME
+    ≤ max_records dataset-wide sample. Clustering itself is collective-
+    free and runs on rank 0 only."""
+    from anovos_amd.core import dist as _dist
+
+    lat, lon = _num(df, lat_col), _num(df, long_col)
+    ok = ~torch.isnan(lat) & ~torch.isnan(lon)
+    x = torch.stack([lat[ok], lon[ok]], dim=1)
+
+    def _sub(t):
+        if t.shape[0] > int(max_records):
+            g = torch.Generator(device="cpu").manual_seed(7)
+            idx = torch.randperm(t.shape[0], generator=g)[: int(max_records)].to(t.device)
+            return t[idx]
+        return t
+
+    x = _sub(x)
+    if _dist.world_size() > 1:
+        x = _sub(torch.cat(_dist.all_gather_tensor(x)))
+    return pd.DataFrame({lat_col: x[:, 0].cpu().numpy(), long_col: x[:, 1].cpu().numpy()})
+
+
 def geo_cluster_generator(df, lat_col_list, long_col_list, geo_col_list, max_cluster, eps, min_samples, master_path, global_map_box_val, max_records):
     """Reference geospatial_analyzer.py:734 — cluster analysis for every
-    lat-long pair and every geohash column (decoded first)."""
+    lat-long pair and every geohash column (decoded first). The sample
+    is gathered dataset-wide first; the clustering/plot pass (no
+    collectives, all writes) runs on rank 0 only."""
+    from anovos_amd.core import dist as _dist
+
+    def _run(sub, la, lo, tag):
+        if isinstance(sub, AnovosFrame):
+            sub = _gathered_pair_pdf(sub, la, lo, int(max_records))
+        if _dist.rank() == 0 and len(sub) >= 4:
+            geo_cluster_analysis(sub, la, lo, max_cluster, eps, min_samples, master_path, tag, global_map_box_val)
+        _dist.barrier()
+
     if lat_col_list:
         for la, lo in zip(list(lat_col_list), list(long_col_list)):
-            sub = df
-            if isinstance(df, AnovosFrame) and df.local_rows() > int(max_records):
-                idx = torch.randperm(df.local_rows())[: int(max_records)]
-                sub = df.filter_rows(idx)
-            geo_cluster_analysis(sub, la, lo, max_cluster, eps, min_samples, master_path, la + "_" + lo, global_map_box_val)
+            _run(df, la, lo, la + "_" + lo)
     if geo_col_list:
         from anovos_amd.data_transformer.geospatial import geo_format_geohash
 
         for ghc in list(geo_col_list):
-            dec = geo_format_geohash(df, [ghc], "dd", output_mode="append")
-            sub = dec
-            if sub.local_rows() > int(max_records):
-                idx = torch.randperm(sub.local_rows())[: int(max_records)]
-                sub = sub.filter_rows(idx)
-            geo_cluster_analysis(sub, f"{ghc}_lat_dd", f"{ghc}_lon_dd", max_cluster, eps, min_samples, master_path, ghc, global_map_box_val)
+            dec = geo_format_geohash(df, [ghc], "dd", output_mode="append") if isinstance(df, AnovosFrame) else df
+            _run(dec, f"{ghc}_lat_dd", f"{ghc}_lon_dd", ghc)
 
 
 # ------------------------------------------------------------- loc charts

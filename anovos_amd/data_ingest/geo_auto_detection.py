@@ -60,15 +60,24 @@ def ll_gh_cols(df: AnovosFrame, max_records: int = 100_000) -> Tuple[List[str], 
     columns of length 5-11 whose values all geohash-decode qualify as
     geohash columns."""
     lat_cols, long_cols, gh_cols = [], [], []
+    name_verdict = {}  # name-match detections (schema-only, rank-uniform)
+    # data-driven verdicts that must be reconciled across ranks: the
+    # reference's screens (max/std/mean/distinct, all-values-decode) are
+    # Spark dataset-wide; per-shard verdicts can disagree (an empty
+    # shard detects nothing) and a rank-dependent column list deadlocks
+    # the collective stats that follow. Name-based detection is
+    # schema-only and already rank-uniform.
+    num_verdict = {}  # name -> "lat" | "long" (data-driven only)
+    gh_state = {}  # name -> "valid" | "no" | "nodata"
     for name, dtype in df.dtypes:
         c = df.col(name)
         if c.kind == "numerical":
             lname = name.lower()
             if "latitude" in lname:
-                lat_cols.append(name)
+                name_verdict[name] = "lat"
                 continue
             if "longitude" in lname:
-                long_cols.append(name)
+                name_verdict[name] = "long"
                 continue
             x = c.data.to(torch.float64)
             v = x[~torch.isnan(x)]
@@ -86,20 +95,55 @@ def ll_gh_cols(df: AnovosFrame, max_records: int = 100_000) -> Tuple[List[str], 
             if distinct <= 2:
                 continue
             if mn_abs_max <= 90:
-                lat_cols.append(name)
+                num_verdict[name] = "lat"
             elif mn_abs_max <= 180:
-                long_cols.append(name)
+                num_verdict[name] = "long"
         elif c.kind == "categorical":
             d = [s for s in (c.dictionary or []) if s][:max_records]
             if not d:
+                gh_state[name] = "nodata"
                 continue
             max_len = max(len(str(s)) for s in d)
-            if not (4 < max_len < 12):
-                continue
-            if len(set(d)) <= 2:
-                continue
-            if all(gu.geohash_is_valid(str(s).lower()) for s in d):
-                gh_cols.append(name)
+            if (
+                4 < max_len < 12
+                and len(set(d)) > 2
+                and all(gu.geohash_is_valid(str(s).lower()) for s in d)
+            ):
+                gh_state[name] = "valid"
+            else:
+                gh_state[name] = "no"
+
+    from anovos_amd.core import dist as _dist
+
+    if _dist.world_size() > 1:
+        gathered = _dist.all_gather_object((num_verdict, gh_state))
+        # numeric: a column is longitude if ANY rank saw |max| in
+        # (90, 180] (the global |max| would too), else latitude if any
+        # rank said lat. geohash: every rank holding data must validate.
+        names = {n for nv, _ in gathered for n in nv}
+        num_verdict = {}
+        for n in names:
+            vs = {nv.get(n) for nv, _ in gathered} - {None}
+            num_verdict[n] = "long" if "long" in vs else "lat"
+        gh_names = {n for _, gs in gathered for n in gs}
+        gh_state = {}
+        for n in gh_names:
+            states = {gs.get(n, "nodata") for _, gs in gathered}
+            if "no" in states or "valid" not in states:
+                gh_state[n] = "no"
+            else:
+                gh_state[n] = "valid"
+
+    # fold verdicts back in schema order (rank-uniform; preserves the
+    # reference's lat↔long pairing order across both mechanisms)
+    for name, _ in df.dtypes:
+        v = name_verdict.get(name) or num_verdict.get(name)
+        if v == "lat":
+            lat_cols.append(name)
+        elif v == "long":
+            long_cols.append(name)
+        elif gh_state.get(name) == "valid":
+            gh_cols.append(name)
     if len(lat_cols) != len(long_cols):
         lat_cols, long_cols = [], []
     return lat_cols, long_cols, gh_cols

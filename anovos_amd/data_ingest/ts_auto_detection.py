@@ -93,18 +93,29 @@ def regex_date_time_parser(ctx, idf: AnovosFrame, col: str, precision: str = "s"
     dev = c.data.device
     if c.dtype in ("timestamp", "date"):
         return idf
+    from anovos_amd.core import dist as _dist
+
     if c.kind == "numerical":
         x = c.data.to(torch.float64)
         null = torch.isnan(x)
         v = x[~null]
-        if v.numel() == 0:
+        # range/integrality decisions must be GLOBAL (the reference's
+        # Spark min/max are dataset-wide): otherwise ranks with different
+        # shards pick different branches and the resulting schemas (and
+        # later collective counts) diverge. One batched all-reduce.
+        mn = float(v.min()) if v.numel() else float("inf")
+        mx = float(v.max()) if v.numel() else float("-inf")
+        allint = 1.0 if (v.numel() == 0 or bool(((v - v.trunc()) == 0).all())) else 0.0
+        if _dist.world_size() > 1:
+            neg_mn, mx, neg_allint = _dist.all_reduce_scalars([-mn, mx, -allint], op="max")
+            mn, allint = -neg_mn, -neg_allint
+        if mx < mn:  # globally empty column
             return idf
-        mn, mx = float(v.min()), float(v.max())
         if 1e9 <= mn and mx < 1e10:  # 10-digit epoch seconds
             ts = torch.where(null, torch.zeros_like(x), x).to(torch.int64) * US_PER_SEC
         elif 1e12 <= mn and mx < 1e13:  # 13-digit epoch millis
             ts = torch.where(null, torch.zeros_like(x), x).to(torch.int64) * 1000
-        elif 1000 <= mn and mx <= 9999 and bool(((v - v.trunc()) == 0).all()):  # yyyy
+        elif 1000 <= mn and mx <= 9999 and allint >= 1.0:  # yyyy
             days = []
             yrs = v.to(torch.int64)
             epoch = _dt.datetime(1970, 1, 1)
@@ -133,6 +144,11 @@ def regex_date_time_parser(ctx, idf: AnovosFrame, col: str, precision: str = "s"
         parsed = [_parse_one(s) for s in d]
         n_nonnull = sum(1 for s in d if s)
         n_ok = sum(1 for p in parsed if p is not None)
+        # the ≥80% acceptance rule is over the GLOBAL distinct values
+        # (dictionaries are per-rank): sum counts across ranks so every
+        # rank reaches the same convert/skip verdict.
+        if _dist.world_size() > 1:
+            n_nonnull, n_ok = _dist.all_reduce_scalars([float(n_nonnull), float(n_ok)], op="sum")
         if n_nonnull == 0 or n_ok / max(n_nonnull, 1) < 0.8:
             return idf
         lut = torch.tensor([p if p is not None else NULL_TS for p in parsed] + [NULL_TS],
@@ -150,36 +166,71 @@ def regex_date_time_parser(ctx, idf: AnovosFrame, col: str, precision: str = "s"
 
 def ts_loop_cols_pre(idf: AnovosFrame, id_col: str) -> Tuple[List[str], List[str], List[int]]:
     """Reference ts_auto_detection.py:554 — candidate classification by
-    dtype + fixed string/number width ∈ {4,6,8,10,13}."""
-    lc1, lc2, lc3 = [], [], []
+    dtype + fixed string/number width ∈ {4,6,8,10,13}.
+
+    The width/distinct screening stats are merged across ranks (len-set
+    union, max width, summed distinct counts, OR'd non-integrality): in
+    the reference these are Spark dataset-wide aggregates, and every
+    rank must classify identically or later per-ts-column collective
+    loops deadlock (empty or skewed shards would otherwise disagree)."""
+    from anovos_amd.core import dist as _dist
+
+    # pass 1 — local per-column screening stats (schema order is
+    # identical on every rank)
+    meta = []  # (name, dtype, kind, col_len, lens, distinct, nonint)
     for name, dtype in idf.dtypes:
         c = idf.col(name)
         if c.kind == "categorical":
             d = [s for s in (c.dictionary or []) if s]
             col_len = max((len(str(s)) for s in d), default=0)
             lens = {len(str(s)) for s in d}
+            nonint = False
         elif c.kind == "numerical":
             x = c.data
             v = x[~torch.isnan(x)] if x.is_floating_point() else x
             if v.numel() and bool(((v - v.trunc()) == 0).all()):
                 iv = v.to(torch.int64)
                 strs_len = torch.where(iv == 0, torch.ones_like(iv), torch.log10(iv.abs().clamp(min=1).to(torch.float64)).to(torch.int64) + 1)
-                col_len = int(strs_len.max()) if v.numel() else 0
+                col_len = int(strs_len.max())
                 lens = set(strs_len.unique().cpu().numpy().tolist())
+                nonint = False
             else:
-                col_len, lens = 0, set()
+                col_len, lens, nonint = 0, set(), bool(v.numel())
         else:
+            meta.append((name, dtype, "other", 0, set(), 0, False))
+            continue
+        nonnull_distinct = int(torch.unique(c.data[~c.null_mask()]).numel())
+        meta.append((name, dtype, c.kind, col_len, lens, nonnull_distinct, nonint))
+
+    # pass 2 — merge across ranks (one object gather; ingest-time cold path)
+    if _dist.world_size() > 1:
+        gathered = _dist.all_gather_object([(m[3], sorted(m[4]), m[5], m[6]) for m in meta])
+        merged = []
+        for i, m in enumerate(meta):
+            col_len = max(g[i][0] for g in gathered)
+            lens = set()
+            for g in gathered:
+                lens.update(g[i][1])
+            distinct = sum(g[i][2] for g in gathered)
+            nonint = any(g[i][3] for g in gathered)
+            if nonint:  # fractional values on some rank → not a fixed-width int candidate
+                col_len, lens = 0, set()
+            merged.append((m[0], m[1], m[2], col_len, lens, distinct, nonint))
+        meta = merged
+
+    # pass 3 — classify (pure function of merged stats: rank-uniform)
+    lc1, lc2, lc3 = [], [], []
+    for name, dtype, kind, col_len, lens, distinct, _nonint in meta:
+        if kind == "other":
             lc1.append(name)
             lc2.append("dt" if dtype in ("timestamp", "date") else "NA")
             lc3.append(0)
-            continue
-        nonnull_distinct = int(torch.unique(c.data[~c.null_mask()]).numel())
-        if nonnull_distinct == 0:
+        elif distinct == 0:
             lc1.append(name); lc2.append("NA"); lc3.append(col_len)
         elif name != id_col and len(lens) == 1 and col_len in _CAND_LENGTHS:
-            kind = "string_c" if c.kind == "categorical" else ("bigint_c" if dtype in ("bigint", "long", "double") else "int_c")
-            lc1.append(name); lc2.append(kind); lc3.append(col_len)
-        elif name != id_col and c.kind == "categorical":
+            tag = "string_c" if kind == "categorical" else ("bigint_c" if dtype in ("bigint", "long", "double") else "int_c")
+            lc1.append(name); lc2.append(tag); lc3.append(col_len)
+        elif name != id_col and kind == "categorical":
             lc1.append(name); lc2.append("string"); lc3.append(col_len)
         else:
             lc1.append(name); lc2.append("NA"); lc3.append(col_len)
@@ -205,6 +256,21 @@ def ts_preprocess(ctx, idf: AnovosFrame, id_col: str, output_path: str,
             odf = regex_date_time_parser(ctx, odf, i, output_mode="replace")
         except Exception:
             continue  # reference swallows per-column parse failures (:694-709)
+    # reconcile: conversion decisions are already global (see
+    # regex_date_time_parser / ts_loop_cols_pre), but a swallowed
+    # per-column exception on one rank could still leave the schemas
+    # diverged — and a rank-dependent ts_cols set deadlocks every later
+    # per-column collective loop (ts_analyzer). Keep the intersection,
+    # reverting any local-only conversion to the original column.
+    local_post = [n for n, d in odf.dtypes if d in ("timestamp", "date")]
+    if _dist.world_size() > 1:
+        sets = _dist.all_gather_object(sorted(local_post))
+        agreed = set(sets[0])
+        for s in sets[1:]:
+            agreed &= set(s)
+        for n in local_post:
+            if n not in agreed:
+                odf = odf.with_column(n, idf.col(n))
     ts_cols_post = [n for n, d in odf.dtypes if d in ("timestamp", "date")]
     num_cols, cat_cols, other_cols = attributeType_segregation(odf)
     num_cols = [x for x in num_cols if x not in [id_col] + ts_cols_post]

@@ -22,9 +22,54 @@ from anovos_amd.core.dtypes import NULL_CODE
 from anovos_amd.ops import backend
 
 
+def align_dictionaries(idf, cols: List[str]) -> None:
+    """Make categorical dictionaries rank-identical (in place).
+
+    Ingest produces sorted per-rank dictionaries, so low-cardinality
+    columns (every shard sees every category) align for free — but
+    high-cardinality columns built per rank (e.g. lat/long→geohash
+    strings) do not, and every dictionary-indexed collective
+    (cat_value_counts all-reduce, IV/IG label counts, drift binning)
+    requires identical code→value maps on all ranks. This detects
+    misalignment with one signature gather (stable blake2b — Python's
+    str hash is salted per process) and heals only the misaligned
+    columns: union-sort the gathered dictionaries, remap local codes by
+    LUT. Columns are shared between derived frames, so one heal fixes
+    every frame holding the column. Collective: must be called on all
+    ranks with the same cols."""
+    if not dist.is_dist():
+        return
+    cat = [c for c in cols if idf.col(c).kind == "categorical"]
+    if not cat:
+        # still rank-uniform: every rank sees the same schema/cols
+        return
+    import hashlib
+
+    sigs = []
+    for c in cat:
+        d = idf.col(c).dictionary or []
+        h = hashlib.blake2b("\x00".join(map(str, d)).encode(), digest_size=8).hexdigest()
+        sigs.append((len(d), h))
+    gathered = dist.all_gather_object(sigs)
+    bad = [i for i in range(len(cat)) if len({g[i] for g in gathered}) > 1]
+    for i in bad:
+        col = idf.col(cat[i])
+        d = list(col.dictionary or [])
+        union = sorted(set().union(*[set(x) for x in dist.all_gather_object(d)]))
+        pos = {v: k for k, v in enumerate(union)}
+        dev = col.data.device
+        lut = torch.tensor([pos[v] for v in d] + [NULL_CODE], dtype=torch.int64, device=dev)
+        codes = col.data.long()
+        codes = torch.where(codes == NULL_CODE, torch.full_like(codes, len(d)), codes)
+        col.data = lut[codes].to(col.data.dtype)
+        col.dictionary = union
+        col.cache.clear()
+
+
 def cat_value_counts(idf, cols: List[str]) -> Dict[str, torch.Tensor]:
     """Global counts per dictionary code for categorical columns.
     Returns {col: int64 tensor [dict_size]} on CPU. Cached per Column."""
+    align_dictionaries(idf, cols)
     out = {}
     cached = [c for c in cols if "cat_counts" in idf.col(c).cache]
     for c in cached:

@@ -659,3 +659,125 @@ def test_dist_workflow_fewer_parts_than_ranks(tmp_path):
     assert res["rows"] == 3000
     assert res["fill_age"] == int(pdf["age"].notna().sum())
     assert res["report"] is True
+
+
+def _cfg_worker(rank, port, workdir, out, artifacts):
+    os.environ.update({"RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": "2",
+                       "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port)})
+    os.chdir(workdir)
+    import torch.distributed as td
+
+    from anovos_amd import workflow
+    from anovos_amd.core import dist
+
+    dist.init_from_env(timeout_s=180)
+    from anovos_amd.shared.context import init_context
+
+    init_context("cpu")
+    df = workflow.run(os.path.join(workdir, "cfg.yaml"))
+    rows = df.count()  # collective: must run on BOTH ranks
+    if rank == 0:
+        json.dump({"rows": rows, "artifacts": {a: os.path.exists(a) for a in artifacts}},
+                  open(out, "w"))
+    td.barrier()
+    td.destroy_process_group()
+
+
+def _run_cfg_two_ranks(tmp_path, cfg, artifacts):
+    import yaml as _yaml
+
+    with open(tmp_path / "cfg.yaml", "w") as f:
+        _yaml.safe_dump(cfg, f, sort_keys=False)
+    port = _free_port()
+    out = str(tmp_path / "res.json")
+    mp_ctx = mp.get_context("spawn")
+    procs = [mp_ctx.Process(target=_cfg_worker, args=(r, port, str(tmp_path), out, artifacts))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=300)
+        assert p.exitcode == 0, f"worker failed: exit {p.exitcode}"
+    return json.load(open(out))
+
+
+def test_dist_workflow_timeseries_two_ranks(tmp_path):
+    """The ts-config workflow at world_size=2 with rank-DISJOINT date
+    ranges per shard (sorted before splitting): timestamp auto-detection
+    decisions and the viz top-category choices must be reconciled
+    globally or the per-column collective loops deadlock (r02 bug)."""
+    import sys as _sys
+
+    _sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "tools"))
+    import make_income_data as mid
+
+    pdf = mid.add_ts_cols(mid.make(3000))
+    pdf = pdf.sort_values("txn_date").reset_index(drop=True)  # disjoint dates per part
+    d = tmp_path / "data" / "income_dataset" / "csv"
+    d.mkdir(parents=True)
+    pdf.iloc[:1500].to_csv(d / "part-00000.csv", index=False)
+    pdf.iloc[1500:].to_csv(d / "part-00001.csv", index=False)
+    cfg = {
+        "input_dataset": {
+            "read_dataset": {"file_path": "data/income_dataset/csv", "file_type": "csv",
+                             "file_configs": {"header": True, "inferSchema": True}},
+            "delete_column": ["logfnl"],
+        },
+        "timeseries_analyzer": {"auto_detection": True, "id_col": "ifa", "tz_offset": "local",
+                                 "inspection": True, "analysis_level": "daily", "max_days": 3600},
+        "stats_generator": {"metric": ["global_summary", "measures_of_counts"],
+                            "metric_args": {"list_of_cols": "all", "drop_cols": ["ifa"]}},
+        "report_preprocessing": {"master_path": "report_stats"},
+    }
+    res = _run_cfg_two_ranks(tmp_path, cfg, [
+        "report_stats/ts_cols_stats.csv",
+        "report_stats/stats_txn_date_1.csv",
+        "report_stats/txn_date_age_daily.csv",
+    ])
+    assert res["rows"] == 3000
+    assert all(res["artifacts"].values()), res["artifacts"]
+    import pandas as _pd
+
+    ts_stats = _pd.read_csv(tmp_path / "report_stats" / "ts_cols_stats.csv")
+    assert "txn_date" in set(ts_stats["attribute"])
+
+
+def test_dist_workflow_geospatial_two_ranks(tmp_path):
+    """The geo-config workflow at world_size=2: lat/long/geohash
+    detection reconciled across ranks, clustering on the gathered
+    global sample (rank 1's shard alone once crashed k-means), and the
+    per-row-unique geohash dictionary healed by align_dictionaries
+    before dictionary-indexed collectives (r02 bugs)."""
+    import sys as _sys
+
+    _sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "tools"))
+    import make_income_data as mid
+
+    pdf = mid.add_geo_cols(mid.make(2400))
+    d = tmp_path / "data" / "income_dataset" / "csv"
+    d.mkdir(parents=True)
+    pdf.iloc[:1200].to_csv(d / "part-00000.csv", index=False)
+    pdf.iloc[1200:].to_csv(d / "part-00001.csv", index=False)
+    cfg = {
+        "input_dataset": {
+            "read_dataset": {"file_path": "data/income_dataset/csv", "file_type": "csv",
+                             "file_configs": {"header": True, "inferSchema": True}},
+            "delete_column": ["logfnl"],
+        },
+        "geospatial_controller": {
+            "geospatial_analyzer": {"auto_detection_analyzer": True, "id_col": "ifa",
+                                     "max_analysis_records": 10000, "top_geo_records": 50,
+                                     "max_cluster": 6, "eps": "0.3,0.5,0.1",
+                                     "min_samples": "40,120,40", "global_map_box_val": 0},
+        },
+        "stats_generator": {"metric": ["global_summary", "measures_of_counts"],
+                            "metric_args": {"list_of_cols": "all", "drop_cols": ["ifa"]}},
+        "report_preprocessing": {"master_path": "report_stats"},
+    }
+    res = _run_cfg_two_ranks(tmp_path, cfg, [
+        "report_stats/cluster_output_kmeans_latitude_longitude.csv",
+        "report_stats/Overall_Summary_2_gh7.csv",
+        "report_stats/measures_of_counts.csv",
+    ])
+    assert res["rows"] == 2400
+    assert all(res["artifacts"].values()), res["artifacts"]
