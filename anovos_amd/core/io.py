@@ -73,7 +73,94 @@ def read_dataset(file_path: str, file_type: str, file_configs: Dict = None, devi
         idf = _read_avro(file_path, file_configs, device, sharded)
     else:
         raise ValueError(f"unsupported file_type: {file_type}")
-    return _unify_frame_dictionaries(idf)
+    return _unify_frame_dictionaries(_reconcile_schema(idf))
+
+
+def _reconcile_schema(idf: AnovosFrame) -> AnovosFrame:
+    """Make per-rank inferred schemas identical before any
+    schema-shaped collective.
+
+    Sharded reads infer dtypes from each rank's own part files; a
+    header-only part (an empty shard written by another rank) infers
+    every column as string, and an all-null column can infer numeric on
+    one shard and string on another. A rank-dependent dtype changes the
+    categorical column set — and with it the number of collectives in
+    _unify_frame_dictionaries — which deadlocks. One tiny object gather
+    elects the schema of the first rank that holds rows; deviating
+    ranks coerce their (usually empty) columns to it."""
+    from anovos_amd.core import dist
+
+    if not dist.is_dist():
+        return idf
+    local = [(n, idf.col(n).dtype) for n in idf.columns]
+    gathered = dist.all_gather_object((idf.local_rows(), local))
+    schemas = [sch for _, sch in gathered]
+    if all(s == schemas[0] for s in schemas):
+        return idf
+    ref = dict(next((sch for r, sch in gathered if r > 0), schemas[0]))
+    out = idf
+    for name, dt in local:
+        want = ref.get(name, dt)
+        if dt != want:
+            out = out.with_column(name, _coerce_column(out.col(name), want))
+    return out
+
+
+def _coerce_column(col, want: str):
+    """Convert a Column to the elected dtype (schema reconciliation)."""
+    import torch
+
+    from anovos_amd.core.frame import Column
+    from anovos_amd.core.dtypes import NULL_CODE, NULL_TS, kind_of_dtype, is_timestamp_dtype
+
+    n = int(col.data.numel())
+    dev = col.data.device
+    want_kind = "other" if is_timestamp_dtype(want) else kind_of_dtype(want)
+    if n == 0:
+        if want_kind == "numerical":
+            return Column(col.name, want, torch.empty(0, dtype=torch.float64, device=dev))
+        if is_timestamp_dtype(want):
+            return Column(col.name, want, torch.empty(0, dtype=torch.int64, device=dev))
+        return Column(col.name, want, torch.empty(0, dtype=torch.int32, device=dev), [])
+    if col.kind == "categorical" and want_kind == "numerical":
+        d = col.dictionary or []
+        def _f(s):
+            try:
+                return float(s)
+            except (TypeError, ValueError):
+                return float("nan")
+        lut = torch.tensor([_f(s) for s in d] + [float("nan")], dtype=torch.float64, device=dev)
+        codes = col.data.to(torch.long)
+        codes = torch.where(codes == NULL_CODE, torch.full_like(codes, len(d)), codes)
+        return Column(col.name, want, lut[codes])
+    if col.kind == "numerical" and want_kind == "categorical":
+        from anovos_amd.ops.encode import _fmt_num
+
+        x = col.data.to(torch.float64)
+        null = torch.isnan(x)
+        uniq, inv = torch.unique(torch.nan_to_num(x), return_inverse=True)
+        strs = [_fmt_num(float(u)) for u in uniq.cpu()]
+        order = sorted(range(len(strs)), key=lambda i: strs[i])
+        rankpos = torch.empty(len(strs), dtype=torch.int64, device=dev)
+        rankpos[torch.tensor(order, device=dev)] = torch.arange(len(strs), device=dev)
+        codes = rankpos[inv].to(torch.int32)
+        codes = torch.where(null, torch.full_like(codes, NULL_CODE), codes)
+        return Column(col.name, want, codes, [strs[i] for i in order])
+    if col.kind == "categorical" and is_timestamp_dtype(want):
+        import pandas as _pd
+
+        d = col.dictionary or []
+        parsed = _pd.to_datetime(_pd.Series(d), errors="coerce")
+        vals = [int(v.value // 1000) if v == v else NULL_TS for v in parsed]
+        lut = torch.tensor(vals + [NULL_TS], dtype=torch.int64, device=dev)
+        codes = col.data.to(torch.long)
+        codes = torch.where(codes == NULL_CODE, torch.full_like(codes, len(d)), codes)
+        return Column(col.name, want, lut[codes])
+    if col.kind == "numerical" and want_kind == "numerical":
+        return Column(col.name, want, col.data)
+    raise ValueError(
+        f"schema reconciliation cannot coerce column {col.name!r} from {col.dtype} to {want}"
+    )
 
 
 def _unify_frame_dictionaries(idf: AnovosFrame) -> AnovosFrame:

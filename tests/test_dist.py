@@ -630,6 +630,16 @@ def test_dist_workflow_fewer_parts_than_ranks(tmp_path):
             "outlier_detection": {"list_of_cols": "all", "drop_cols": ["ifa", "income"],
                                    "detection_side": "both", "treatment": False},
         },
+        # intermediate write + sharded re-read: rank 1 writes a header-only
+        # part and re-infers every column as string — the schema must be
+        # reconciled across ranks or the dictionary-unify collective
+        # count diverges (r02 deadlock)
+        "write_intermediate": {
+            "file_path": "intermediate_data",
+            "file_type": "csv",
+            "file_configs": {"mode": "overwrite", "header": True, "delimiter": ",",
+                              "inferSchema": True},
+        },
         "association_evaluator": {
             "IV_calculation": {"list_of_cols": "all", "drop_cols": ["ifa"],
                                "label_col": "income", "event_label": ">50K"},
