@@ -791,3 +791,53 @@ def test_dist_workflow_geospatial_two_ranks(tmp_path):
     ])
     assert res["rows"] == 2400
     assert all(res["artifacts"].values()), res["artifacts"]
+
+
+def _align_dict_worker(rank, port, out):
+    os.environ.update({"RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": "2",
+                       "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port)})
+    import torch.distributed as td
+
+    from anovos_amd.core import dist
+    from anovos_amd.core.frame import AnovosFrame, Column
+    from anovos_amd.ops.groupby import align_dictionaries, cat_value_counts
+
+    dist.init_from_env(timeout_s=120)
+    # columns built AFTER ingest (transformer outputs) carry per-rank
+    # dictionaries; rank 0: a,b,c / rank 1: b,d (+ a null)
+    if rank == 0:
+        codes = torch.tensor([0, 0, 1, 2, 1], dtype=torch.int32)
+        d = ["a", "b", "c"]
+    else:
+        codes = torch.tensor([0, 1, 1, -1], dtype=torch.int32)
+        d = ["b", "d"]
+    idf = AnovosFrame({"g": Column("g", "string", codes, d)}, device="cpu")
+    counts = cat_value_counts(idf, ["g"])["g"]
+    col = idf.col("g")
+    res = {v: int(c) for v, c in zip(col.dictionary, counts.tolist())}
+    align_dictionaries(idf, ["g"])  # idempotent second call
+    same = list(col.dictionary)
+    if rank == 0:
+        json.dump({"counts": res, "dict": same}, open(out, "w"))
+    td.barrier()
+    td.destroy_process_group()
+
+
+def test_dist_align_dictionaries(tmp_path):
+    """Post-ingest categorical columns with rank-divergent dictionaries
+    are healed in place before the dictionary-indexed all-reduce
+    (regression: per-row-unique geohash dictionaries sheared the
+    collective)."""
+    port = _free_port()
+    out = str(tmp_path / "res.json")
+    mp_ctx = mp.get_context("spawn")
+    procs = [mp_ctx.Process(target=_align_dict_worker, args=(r, port, out)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0, f"worker failed: exit {p.exitcode}"
+    res = json.load(open(out))
+    assert res["dict"] == ["a", "b", "c", "d"]
+    # global truth: a=2, b=2+2=... rank0 b codes: idx1 x2 -> 2; rank1 b: idx0 x1 -> 1
+    assert res["counts"] == {"a": 2, "b": 3, "c": 1, "d": 2}
