@@ -158,6 +158,10 @@ def ts_viz_data(idf: AnovosFrame, x_col: str, y_col: str, output_mode="append", 
         else:
             klabels = [float(u) for u in guniq]
     G = len(klabels)
+    if G == 0:  # globally 0-row frame; rank-uniform (keys were unified)
+        if yc.kind == "categorical":
+            return pd.DataFrame(columns=[y_col, k_col, "count"])
+        return pd.DataFrame(columns=[k_col, "min", "max", "mean", "median"])
     dev = kcodes.device
     if yc.kind == "categorical":
         # top-n_cat categories by count, then count per (bucket, cat)
