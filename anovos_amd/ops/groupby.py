@@ -39,9 +39,14 @@ def align_dictionaries(idf, cols: List[str]) -> None:
     ranks with the same cols."""
     if not dist.is_dist():
         return
-    cat = [c for c in cols if idf.col(c).kind == "categorical"]
+    # skip columns already verified/healed (flag set identically on all
+    # ranks — same call sequence — so the skip stays rank-uniform and
+    # the hot path pays no collective after the first touch)
+    cat = [
+        c for c in cols
+        if idf.col(c).kind == "categorical" and "dict_aligned" not in idf.col(c).cache
+    ]
     if not cat:
-        # still rank-uniform: every rank sees the same schema/cols
         return
     import hashlib
 
@@ -64,6 +69,8 @@ def align_dictionaries(idf, cols: List[str]) -> None:
         col.data = lut[codes].to(col.data.dtype)
         col.dictionary = union
         col.cache.clear()
+    for c in cat:
+        idf.col(c).cache["dict_aligned"] = True
 
 
 def cat_value_counts(idf, cols: List[str]) -> Dict[str, torch.Tensor]:
