@@ -29,7 +29,12 @@ def read_dataset(ctx, file_path: str, file_type: str, file_configs: Dict = {}, s
 def write_dataset(idf: AnovosFrame, file_path: str, file_type: str, file_configs: Dict = {}, column_order: List[str] = []):
     """Reference data_ingest.py:54-117. repartition/coalesce hints are
     accepted and ignored (each rank writes one part: partitioning follows
-    the GPU sharding, which is the engine's unit of parallelism)."""
+    the GPU sharding, which is the engine's unit of parallelism) — but a
+    negative repartition is still rejected like Spark's
+    IllegalArgumentException (reference test_data_ingest_unit.py:79)."""
+    rp = (file_configs or {}).get("repartition")
+    if rp is not None and int(rp) < 0:
+        raise ValueError(f"Invalid repartition value: {rp}")
     core_io.write_dataset(idf, file_path, file_type, file_configs, column_order or None)
 
 

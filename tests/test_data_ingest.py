@@ -153,3 +153,36 @@ def test_io_roundtrip_all_formats_and_dtypes(tmp_path):
                         assert y is None or y == "", (fmt, c, x, y)
                     else:
                         assert x == y, (fmt, c, x, y)
+
+
+def test_write_dataset_negative_repartition_raises(tmp_path, small_pdf):
+    """Reference test_data_ingest_unit.py:79 — Spark raised
+    IllegalArgumentException; the engine rejects the same input."""
+    from anovos_amd.core.frame import AnovosFrame
+    from anovos_amd.data_ingest.data_ingest import write_dataset
+
+    idf = AnovosFrame.from_pandas(small_pdf, device="cpu")
+    with pytest.raises(ValueError):
+        write_dataset(idf, str(tmp_path / "out"), "csv",
+                      {"header": True, "delimiter": ",", "repartition": -2})
+
+
+def test_write_dataset_column_order(tmp_path, small_pdf):
+    """Reference test_data_ingest_unit.py:97/:119 — column_order is
+    honored in the written file; a wrong-length order raises."""
+    import pandas as pd
+
+    from anovos_amd.core.frame import AnovosFrame
+    from anovos_amd.data_ingest.data_ingest import write_dataset
+
+    idf = AnovosFrame.from_pandas(small_pdf, device="cpu")
+    order = list(reversed(idf.columns))
+    write_dataset(idf, str(tmp_path / "ordered"), "csv", {"header": True, "mode": "overwrite"},
+                  column_order=order)
+    import glob
+
+    part = sorted(glob.glob(str(tmp_path / "ordered" / "part-*")))[0]
+    assert list(pd.read_csv(part).columns) == order
+    with pytest.raises(ValueError):
+        write_dataset(idf, str(tmp_path / "bad"), "csv", {"header": True, "mode": "overwrite"},
+                      column_order=order[:-1])

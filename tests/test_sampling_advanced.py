@@ -172,3 +172,28 @@ def test_matrix_factorization_batched_solver():
         err_num += float(np.sum((got[sel] - truth.iloc[:, j].to_numpy()[sel]) ** 2))
         err_den += float(np.sum(truth.iloc[:, j].to_numpy()[sel] ** 2))
     assert err_num / max(err_den, 1e-9) < 0.05  # <5% relative MSE on held-out entries
+
+
+def test_imputation_matrixFactorization_empty_cols_noop(ctx, frame):
+    """Reference test_transformers.py:383 — empty list_of_cols returns
+    the input unchanged."""
+    odf = TA.imputation_matrixFactorization(ctx, frame, list_of_cols=[], id_col="")
+    assert odf.columns == frame.columns
+    assert torch.equal(torch.nan_to_num(odf.col("x").data), torch.nan_to_num(frame.col("x").data))
+
+
+def test_imputation_matrixFactorization_append(ctx, frame):
+    """Reference test_transformers.py:397 — append mode adds *_imputed
+    columns and keeps the originals untouched."""
+    odf = TA.imputation_matrixFactorization(ctx, frame, list_of_cols=["x", "y"], id_col="",
+                                             output_mode="append")
+    assert "x_imputed" in odf.columns and "y_imputed" in odf.columns
+    assert torch.equal(torch.nan_to_num(odf.col("x").data), torch.nan_to_num(frame.col("x").data))
+    assert not bool(torch.isnan(odf.col("x_imputed").data).any())
+
+
+def test_auto_imputation_empty_cols_noop(ctx, frame):
+    """Reference test_transformers.py:485."""
+    odf = TA.auto_imputation(ctx, frame, list_of_cols=[], null_pct=0.1)
+    odf = odf[0] if isinstance(odf, tuple) else odf
+    assert odf.columns == frame.columns
