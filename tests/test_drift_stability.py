@@ -104,3 +104,95 @@ def test_feature_stability_estimation(ctx):
         "stability_index_upper_bound",
     }
     assert (odf["stability_index_upper_bound"] >= odf["stability_index_lower_bound"]).all()
+
+
+def test_check_list_of_columns_empty_raises():
+    """Reference test_validations.py:13/:18 — empty selection and
+    drop-everything both raise ValueError from the decorator."""
+    import pytest as _pytest
+
+    from anovos_amd.drift_stability.validations import check_list_of_columns
+
+    @check_list_of_columns
+    def fut(spark, idf_target, idf_source, list_of_cols="all", drop_cols=[]):
+        return list_of_cols
+
+    with _pytest.raises(ValueError):
+        fut(None, None, None, list_of_cols=[], drop_cols=[])
+    with _pytest.raises(ValueError):
+        fut(None, None, None, list_of_cols=["a", "b"], drop_cols=["a", "b"])
+
+
+@pytest.fixture
+def _si_snapshots():
+    from anovos_amd.core.frame import AnovosFrame
+
+    l1 = [4.34, 4.76, 4.32, 3.39, 3.67, 4.61, 4.03, 4.93, 3.84, 3.31]
+    l2 = [6.34, 4.76, 6.32, 3.39, 5.67, 4.61, 6.03, 4.93, 5.84, 3.31]
+    l3 = [8.34, 4.76, 8.32, 3.39, 7.67, 4.61, 8.03, 4.93, 3.84, 3.31]
+    return [AnovosFrame.from_pandas(pd.DataFrame({"A": l}), device="cpu") for l in (l1, l2, l3)]
+
+
+def test_stability_index_reference_constants(ctx, _si_snapshots):
+    """The reference's own unit expectations (test_stability.py:69-81):
+    the same three snapshots must yield the same CVs/scores to 3
+    decimals."""
+    from numpy.testing import assert_almost_equal
+
+    from anovos_amd.drift_stability.stability import stability_index_computation
+
+    out = stability_index_computation(ctx, *_si_snapshots)
+    out = out.to_pandas() if not isinstance(out, pd.DataFrame) else out
+    row = out.set_index("attribute").loc["A"]
+    assert_almost_equal(
+        [row[c] for c in ["mean_cv", "stddev_cv", "kurtosis_cv", "mean_si",
+                          "stddev_si", "kurtosis_si", "stability_index", "flagged"]],
+        [0.162, 0.62, 0.198, 2.0, 0.0, 2.0, 1.4, 0.0], 3)
+
+
+def test_stability_binary_reference_constants(ctx):
+    """Reference test_stability.py:83-93 — binary columns scored by the
+    SD of the snapshot means."""
+    from numpy.testing import assert_almost_equal
+
+    from anovos_amd.core.frame import AnovosFrame
+    from anovos_amd.drift_stability.stability import stability_index_computation
+
+    snaps = [[0.0] * 10 + [1.0] * 10, [0.0] * 12 + [1.0] * 8, [0.0] * 14 + [1.0] * 6]
+    idfs = [AnovosFrame.from_pandas(pd.DataFrame({"A": l}), device="cpu") for l in snaps]
+    out = stability_index_computation(ctx, *idfs, binary_cols="A")
+    out = out.to_pandas() if not isinstance(out, pd.DataFrame) else out
+    row = out.set_index("attribute").loc["A"]
+    assert_almost_equal(
+        [float(row[c]) for c in ["mean_stddev", "mean_si", "stability_index", "flagged"]],
+        [0.1, 0.0, 0.0, 1.0], 3)
+
+
+def test_feature_stability_reference_constants(ctx, _si_snapshots, tmp_path):
+    """Reference test_stability.py:107-140 — sympy propagation of A**2
+    through the appended metric history, incl. custom weightages."""
+    import os
+
+    from numpy.testing import assert_almost_equal
+
+    from anovos_amd.drift_stability.stability import (
+        feature_stability_estimation,
+        stability_index_computation,
+    )
+
+    mp = str(tmp_path / "metrics")
+    stability_index_computation(ctx, *_si_snapshots, appended_metric_path=mp)
+    f = os.path.join(mp, os.listdir(mp)[0]) if os.path.isdir(mp) else mp
+    stats = pd.read_csv(f)
+    cols = ["mean_cv", "stddev_cv", "mean_si", "stddev_si",
+            "stability_index_lower_bound", "stability_index_upper_bound",
+            "flagged_lower", "flagged_upper"]
+    r = feature_stability_estimation(ctx, stats, {"A": "A**2"})
+    r = r.to_pandas() if not isinstance(r, pd.DataFrame) else r
+    assert_almost_equal([float(v) for v in r[cols].iloc[0]],
+                        [0.298, 0.603, 1.0, 0.0, 0.5, 1.3, 1.0, 0.0], 3)
+    r2 = feature_stability_estimation(ctx, stats, {"A": "A**2"},
+                                      metric_weightages={"mean": 0.7, "stddev": 0.3})
+    r2 = r2.to_pandas() if not isinstance(r2, pd.DataFrame) else r2
+    assert_almost_equal([float(v) for v in r2[cols].iloc[0]],
+                        [0.298, 0.603, 1.0, 0.0, 0.7, 0.7, 1.0, 1.0], 3)
