@@ -98,7 +98,8 @@ def _battery(ctx, idf):
 
 
 def _chaos_worker(rank, port, split, out_path):
-    os.environ.update({"RANK": str(rank), "LOCAL_RANK": str(rank), "WORLD_SIZE": "2",
+    os.environ.update({"RANK": str(rank), "LOCAL_RANK": str(rank),
+                       "WORLD_SIZE": str(len(split)),
                        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
                        "ANOVOS_AMD_DIST_BACKEND": "gloo"})
     import torch.distributed as td
@@ -123,6 +124,7 @@ def _chaos_worker(rank, port, split, out_path):
     pytest.param([(0, 250), (250, 500)], id="even"),
     pytest.param([(0, 500), (500, 500)], id="empty_rank1"),
     pytest.param([(0, 497), (497, 500)], id="tiny_rank1"),
+    pytest.param([(0, 200), (200, 201), (201, 500)], id="three_ranks_skewed"),
 ])
 def test_dist_chaos_battery(tmp_path, split):
     from anovos_amd.core.frame import AnovosFrame
@@ -134,7 +136,7 @@ def test_dist_chaos_battery(tmp_path, split):
     port = _free_port()
     out = str(tmp_path / "res.json")
     mp_ctx = mp.get_context("spawn")
-    procs = [mp_ctx.Process(target=_chaos_worker, args=(r, port, split, out)) for r in range(2)]
+    procs = [mp_ctx.Process(target=_chaos_worker, args=(r, port, split, out)) for r in range(len(split))]
     for p in procs:
         p.start()
     for p in procs:
