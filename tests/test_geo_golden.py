@@ -65,3 +65,20 @@ def test_null_latlon_survive_conversion():
         out = geo_format_latlon(idf, ["lat1"], ["lon1"], "dd", fmt, output_mode="replace").to_pandas()
         vals = out.iloc[:, 1]
         assert int(pd.isna(vals).sum()) >= n_null, fmt
+
+
+def test_location_distance_matches_reference_assertions(geo_idf):
+    """The reference's own expected integers (test_geospatial.py:600+):
+    haversine/vincenty/euclidean in m and km on fixture row id=1 —
+    vincenty matches geopy's WGS-84 inverse to the metre."""
+    from anovos_amd.data_transformer.geospatial import location_distance
+
+    expected = {("haversine", "m"): 17394182, ("haversine", "km"): 17394,
+                ("vincenty", "m"): 17373936, ("vincenty", "km"): 17373,
+                ("euclidean", "m"): 12473414, ("euclidean", "km"): 12473}
+    for (method, unit), exp in expected.items():
+        o = location_distance(geo_idf, ["lat1", "lon1"], ["lat2", "lon2"], "dd", "",
+                              method, unit, output_mode="replace").to_pandas()
+        o = o.sort_values(o.columns[0])
+        dcol = next(c for c in o.columns if c != o.columns[0])
+        assert int(o.iloc[0][dcol]) == exp, (method, unit)
