@@ -332,29 +332,38 @@ def centroid(idf, lat_col, long_col, id_col=None):
 
 
 def weighted_centroid(idf, id_col, lat_col, long_col):
-    """Reference geospatial.py:1099 — centroid weighted by (lat, lon)
-    pair frequency within each id group."""
+    """Reference geospatial.py:1099 — ONE dataset-wide weighted centroid
+    replicated onto every id row: each id group's cartesian vector sum is
+    scaled by the group's own row count (count-squared weighting of each
+    point), the scaled sums are pooled over all groups, and the single
+    resulting (lat, long) is attached to every id. Matches the
+    reference's output exactly (its unit test asserts the global value
+    per id, e.g. [-54, -113] on its geo fixture)."""
     lat, lon = _num(idf, lat_col), _num(idf, long_col)
     ok = gu.in_range(lat, lon) & ~torch.isnan(lat) & ~torch.isnan(lon)
     inv, valid, decode = _id_codes(idf, id_col)
     m = ok & valid
-    # the reference weights distinct (lat, lon) pairs by their row
-    # multiplicity — algebraically identical to the plain per-row mean
-    # computed here (each row contributes once either way)
     x, y, z = gu.dd_to_cartesian(lat, lon, radius=1.0)
     G = int(inv.max().item()) + 1 if inv.numel() else 0
     cnt = torch.zeros(G, dtype=torch.float64, device=lat.device).scatter_reduce(0, inv[m], torch.ones_like(x[m]), reduce="sum")
     sums = []
     for t in (x, y, z):
         s = torch.zeros(G, dtype=torch.float64, device=lat.device).scatter_reduce(0, inv[m], t[m], reduce="sum")
-        sums.append(s / cnt.clamp(min=1))
-    norm = (sums[0] ** 2 + sums[1] ** 2 + sums[2] ** 2).sqrt().clamp(min=1e-300)
-    cla, clo = gu.cartesian_to_dd(sums[0], sums[1], sums[2], radius=norm)
+        sums.append(s)
+    total_w = float(cnt.sum().clamp(min=1.0))
+    xs = float((sums[0] * cnt).sum()) / total_w
+    ys = float((sums[1] * cnt).sum()) / total_w
+    zs = float((sums[2] * cnt).sum()) / total_w
+    import math
+
+    hyp = math.sqrt(xs * xs + ys * ys)
+    cla = math.atan2(zs, hyp) * 180.0 / math.pi
+    clo = math.atan2(ys, xs) * 180.0 / math.pi
     present = (cnt > 0).nonzero(as_tuple=True)[0]
     ids = decode(present.cpu().numpy())
     pdf = pd.DataFrame({id_col: ids,
-                        "lat_weighted_centroid": cla[present].cpu().numpy(),
-                        "long_weighted_centroid": clo[present].cpu().numpy()})
+                        lat_col + "_centroid": [cla] * len(ids),
+                        long_col + "_centroid": [clo] * len(ids)})
     return AnovosFrame.from_pandas(pdf, device=idf.device)
 
 

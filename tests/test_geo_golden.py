@@ -82,3 +82,37 @@ def test_location_distance_matches_reference_assertions(geo_idf):
         o = o.sort_values(o.columns[0])
         dcol = next(c for c in o.columns if c != o.columns[0])
         assert int(o.iloc[0][dcol]) == exp, (method, unit)
+
+
+def test_weighted_centroid_matches_reference_assertions():
+    """Reference test_geospatial.py:1308-1347 — the dataset-wide
+    count²-weighted centroid replicated per id, incl. the null- and
+    invalid-row filtered variants."""
+    from anovos_amd.data_transformer.geospatial import weighted_centroid
+
+    expected = {"sample_geo_data": (1000, -54, -113),
+                "null_sample_geo_data": (811, -34, -109),
+                "invalid_sample_geo_data": (549, -15, -139)}
+    for name, (rows, la, lo) in expected.items():
+        pdf = pd.read_csv(os.path.join(BASE, name + ".csv"))
+        idf = AnovosFrame.from_pandas(pdf, device="cpu")
+        w = weighted_centroid(idf, id_col="id", lat_col="latitude", long_col="longitude").to_pandas()
+        assert len(w) == rows, name
+        assert int(w["latitude_centroid"].iloc[0]) == la, name
+        assert int(w["longitude_centroid"].iloc[0]) == lo, name
+
+
+def test_centroid_matches_reference_assertions():
+    """Reference test_geospatial.py:1268-1305: per-id centroid row
+    counts after null/invalid filtering, and id=296's value."""
+    from anovos_amd.data_transformer.geospatial import centroid
+
+    for name, rows in [("sample_geo_data", 1000), ("null_sample_geo_data", 811),
+                       ("invalid_sample_geo_data", 549)]:
+        pdf = pd.read_csv(os.path.join(BASE, name + ".csv"))
+        idf = AnovosFrame.from_pandas(pdf, device="cpu")
+        c = centroid(idf, "latitude", "longitude", id_col="id").to_pandas()
+        assert len(c) == rows, name
+        row = c[c["id"].astype(float) == 296].iloc[0]
+        assert int(row["latitude_centroid"]) == -27
+        assert int(row["longitude_centroid"]) == -120

@@ -127,8 +127,10 @@ def test_centroid_and_rog(loc_frame):
 def test_weighted_centroid(loc_frame):
     wdf = geo.weighted_centroid(loc_frame, "id", "latitude", "longitude")
     pdf = wdf.to_pandas()
-    assert set(pdf.columns) == {"id", "lat_weighted_centroid", "long_weighted_centroid"}
+    # reference semantics: ONE global weighted centroid per id row
+    assert set(pdf.columns) == {"id", "latitude_centroid", "longitude_centroid"}
     assert len(pdf) == 2
+    assert pdf["latitude_centroid"].nunique() == 1
 
 
 def test_location_in_country(ctx, loc_frame):
