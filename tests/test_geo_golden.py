@@ -116,3 +116,21 @@ def test_centroid_matches_reference_assertions():
         row = c[c["id"].astype(float) == 296].iloc[0]
         assert int(row["latitude_centroid"]) == -27
         assert int(row["longitude_centroid"]) == -120
+
+
+def test_reverse_geocoding_matches_reference_row_contract():
+    """Reference test_geospatial.py:1389-1430: filtered row counts and
+    the first row's coordinates (city names come from a different
+    offline table — see PARITY; the coordinate passthrough and
+    null/invalid filtering are the asserted contract)."""
+    from anovos_amd.data_transformer.geospatial import reverse_geocoding
+
+    for name, rows in [("sample_geo_data", 1000), ("null_sample_geo_data", 811),
+                       ("invalid_sample_geo_data", 549)]:
+        pdf = pd.read_csv(os.path.join(BASE, name + ".csv"))
+        idf = AnovosFrame.from_pandas(pdf, device="cpu")
+        o = reverse_geocoding(idf, lat_col="latitude", long_col="longitude")
+        p = o.to_pandas() if not isinstance(o, pd.DataFrame) else o
+        assert len(p) == rows, name
+        assert int(p["latitude"].iloc[0]) == -82
+        assert int(p["longitude"].iloc[0]) == -126
