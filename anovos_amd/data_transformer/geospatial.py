@@ -404,6 +404,11 @@ def reverse_geocoding(idf, lat_col, long_col):
     reverse_geocoder package; no network/package here)."""
     cities = _CITY_TABLE
     lat, lon = _num(idf, lat_col), _num(idf, long_col)
+    # the reference drops null and out-of-range rows before geocoding
+    # (geospatial.py:1346-1377; its unit test asserts the filtered row
+    # counts)
+    ok = gu.in_range(lat, lon) & ~torch.isnan(lat) & ~torch.isnan(lon)
+    lat, lon = lat[ok], lon[ok]
     clat = torch.tensor([c[0] for c in cities], dtype=torch.float64, device=lat.device)
     clon = torch.tensor([c[1] for c in cities], dtype=torch.float64, device=lat.device)
     # chord distance argmin over the city table (few hundred entries)
