@@ -134,3 +134,23 @@ def test_reverse_geocoding_matches_reference_row_contract():
         assert len(p) == rows, name
         assert int(p["latitude"].iloc[0]) == -82
         assert int(p["longitude"].iloc[0]) == -126
+
+
+def test_location_in_polygon_africa_matches_reference():
+    """Reference test_geospatial.py:1064-1101 with its africa.geojson:
+    row id=1 outside for both pairs, id=5 inside for pair 1."""
+    import json
+
+    from anovos_amd.data_transformer.geospatial import location_in_polygon
+
+    africa = json.load(open(os.path.join(BASE, "africa.geojson")))
+    for name in ["sample_geo_data_two_latlon", "null_sample_geo_data_two_latlon"]:
+        pdf = pd.read_csv(os.path.join(BASE, name + ".csv"))
+        idf = AnovosFrame.from_pandas(pdf, device="cpu")
+        o = location_in_polygon(idf, ["lat1", "lat2"], ["lon1", "lon2"], africa,
+                                output_mode="replace").to_pandas()
+        o = o.sort_values(o.columns[0]).reset_index(drop=True)
+        assert int(o.loc[0, "lat1_lon1_in_polygon"]) == 0
+        assert int(o.loc[0, "lat2_lon2_in_polygon"]) == 0
+        assert int(o.loc[4, "lat1_lon1_in_polygon"]) == 1
+        assert int(o.loc[4, "lat2_lon2_in_polygon"]) == 0
