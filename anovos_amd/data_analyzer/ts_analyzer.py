@@ -30,8 +30,10 @@ from anovos_amd.shared.utils import attributeType_segregation, ends_with
 DAYPARTS = ["early_hours", "work_hours", "late_hours", "commuting_hours", "other_hours", "Missing_NA"]
 
 
-def daypart_cat(hour):
-    """Reference ts_analyzer.py:52 — scalar hour → daypart label."""
+def daypart_cat(column):
+    """Reference ts_analyzer.py:52 — scalar hour → daypart label
+    (reference arg name: column)."""
+    hour = column
     if hour is None:
         return "Missing_NA"
     if 4 <= hour < 7:

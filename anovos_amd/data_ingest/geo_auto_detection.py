@@ -18,12 +18,12 @@ from anovos_amd.core.frame import AnovosFrame
 from anovos_amd.data_transformer import geo_utils as gu
 
 
-def geo_to_latlong(ghs, option: int = 0):
+def geo_to_latlong(x, option: int = 0):
     """Decode geohash string(s) → lat/long (reference geo_auto_detection.py:101
     via pygeohash; here the native bit-interleave decoder).
     option 0 → latitude, 1 → longitude."""
-    single = isinstance(ghs, str)
-    lst = [ghs] if single else list(ghs)
+    single = isinstance(x, str)
+    lst = [x] if single else list(x)
     if any(not gu.geohash_is_valid(str(g).lower()) for g in lst):
         raise ValueError("invalid geohash")
     ints, prec = gu.geohash_str_to_int([str(g).lower() for g in lst])
@@ -161,20 +161,20 @@ def reg_lat_lon(option: str) -> str:
     raise ValueError("option must be latitude or longitude")
 
 
-def conv_str_plus(v):
+def conv_str_plus(col):
     """Prefix '+' onto non-negative values (regex normalization)."""
-    if v is None:
+    if col is None:
         return None
-    if v < 0:
-        return v
-    return "+" + str(v)
+    if col < 0:
+        return col
+    return "+" + str(col)
 
 
-def precision_lev(v) -> int:
+def precision_lev(col) -> int:
     """Number of significant decimal places (0 for integral/None)."""
-    if v is None:
+    if col is None:
         return 0
-    frac = format(float(v), ".8f").split(".")[1]
+    frac = format(float(col), ".8f").split(".")[1]
     if float(frac) > 0:
         return len(frac)
     return 0

@@ -68,9 +68,10 @@ def save_stats(ctx, idf, master_path, function_name, reread=False, run_type="loc
     return None
 
 
-def edit_binRange(s):
+def edit_binRange(col):
     """Reference report_preprocessing.py:130 — collapse 'a-a' ranges to 'a'."""
     try:
+        s = col
         if s is None:
             return s
         parts = str(s).split("-")

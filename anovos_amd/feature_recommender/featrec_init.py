@@ -137,9 +137,9 @@ def get_column_name(df: pd.DataFrame) -> Tuple[str, str, str, str]:
     return cols[0], cols[1], cols[2], cols[3]
 
 
-def camel_case_split(s: str) -> str:
+def camel_case_split(input) -> str:  # reference arg name (featrec_init.py)
     """Reference featrec_init.py:camel_case_split — CamelCase → spaced."""
-    s = re.sub(r"([a-z0-9])([A-Z])", r"\1 \2", str(s))
+    s = re.sub(r"([a-z0-9])([A-Z])", r"\1 \2", str(input))
     s = re.sub(r"([A-Z]+)([A-Z][a-z])", r"\1 \2", s)
     return s
 

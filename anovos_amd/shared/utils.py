@@ -109,19 +109,28 @@ def transpose_dataframe(pdf: pd.DataFrame, fixed_col: str) -> pd.DataFrame:
     return out
 
 
-def output_to_local(path: str) -> str:
+def output_to_local(output_path: str) -> str:
     """dbfs:/ -> /dbfs/ path munging (reference shared/utils.py:135-154);
     other schemes pass through."""
-    p = str(path)
+    p = str(output_path)
     if p.startswith("dbfs:"):
         return "/dbfs" + p[len("dbfs:"):]
     return p
 
 
-def path_ak8s_modify(path: str, auth_key: str = "NA") -> str:
-    """wasbs:// path shim parity (reference shared/utils.py:157-179) — in
-    this engine cloud URIs are not reachable; return as-is for local use."""
-    return str(path)
+def path_ak8s_modify(output_path: str, auth_key: str = "NA") -> str:
+    """wasbs://container@account.blob.core.windows.net/p →
+    https://account.blob.core.windows.net/container/p (reference
+    shared/utils.py:157-179); non-wasbs paths pass through for local
+    runs."""
+    p = str(output_path)
+    if not p.startswith("wasbs://"):
+        return p
+    container = p.split("//")[1].split("@")[0]
+    rest = p.split("//")[1].split("@")[1]
+    url = "https://" + rest.split("windows.net/")[0] + "windows.net"
+    file_path_name = rest.split("windows.net/")[1]
+    return url + "/" + container + "/" + file_path_name
 
 
 def cloud_sync(local_path: str, master_path: str, run_type: str = "local", auth_key: str = "NA", recursive: bool = False):
