@@ -83,12 +83,22 @@ def _parse_one(s: str):
         return None
 
 
-def regex_date_time_parser(ctx, idf: AnovosFrame, col: str, precision: str = "s",
-                           tz: str = "local", output_mode: str = "replace",
-                           val_unique_cat: int = 0, trans_cat: str = "string"):
+def regex_date_time_parser(ctx, idf: AnovosFrame, id_col: str = "", col: str = "",
+                           tz: str = "local", val_unique_cat: int = 0,
+                           trans_cat: str = "string", save_output=None,
+                           output_mode: str = "replace", precision: str = "s"):
     """Reference ts_auto_detection.py:51 — convert one candidate column
     to a timestamp column. ≥80% of non-null distinct values must parse,
-    else the column is returned untouched."""
+    else the column is returned untouched.
+
+    Positional layout matches the reference: (ctx, idf, id_col, col, tz,
+    val_unique_cat, trans_cat, save_output, output_mode). A two-arg call
+    ``regex_date_time_parser(ctx, idf, "colname")`` is also accepted
+    (id_col slot holding the column, reference col empty) for engine-
+    internal use. ``save_output``: path to write the converted frame
+    (reference :520-528)."""
+    if not col:  # short form: (ctx, idf, col)
+        col = id_col
     c = idf.col(col)
     dev = c.data.device
     if c.dtype in ("timestamp", "date"):
@@ -160,8 +170,13 @@ def regex_date_time_parser(ctx, idf: AnovosFrame, col: str, precision: str = "s"
         return idf
     if output_mode == "replace":
         odf = idf.with_column(col, Column(col, "timestamp", newc.data))
-        return odf
-    return idf.with_column(col + "_ts", newc)
+    else:
+        odf = idf.with_column(col + "_ts", newc)
+    if save_output:
+        from anovos_amd.core import io as _io
+
+        _io.write_dataset(odf, str(save_output), "parquet", {"mode": "overwrite"})
+    return odf
 
 
 def ts_loop_cols_pre(idf: AnovosFrame, id_col: str) -> Tuple[List[str], List[str], List[int]]:
