@@ -120,8 +120,9 @@ def ts_eligiblity_check(ctx, idf: AnovosFrame, id_col: str, opt: int = 1, tz_off
     return pd.DataFrame(rows, columns=["attribute", "min", "1%", "5%", "10%", "25%", "50%", "75%", "90%", "95%", "99%", "max"])
 
 
-def ts_viz_data(idf: AnovosFrame, x_col: str, y_col: str, output_mode="append", output_type="daily",
-                n_cat: int = 10, tz_offset="local") -> pd.DataFrame:
+def ts_viz_data(idf: AnovosFrame, x_col: str, y_col: str, id_col: str = "",
+                tz_offset="local", output_mode="append", output_type="daily",
+                n_cat: int = 10) -> pd.DataFrame:
     """Reference ts_analyzer.py:259 — per (bucket × column) aggregates:
     counts for categorical y, min/max/mean/median for numeric y.
     Multi-rank: keys are unified and count/sum/min/max all-reduced; the
