@@ -284,7 +284,7 @@ _METRIC_DICT = [
 ]
 
 
-def wiki_generator(master_path: str, dataDict_path=None, metricDict_path=None) -> str:
+def wiki_generator(master_path: str, dataDict_path=None, metricDict_path=None, print_report=False, **kwargs) -> str:
     """Reference report_generation.py:909 — data dictionary + metric
     definitions tab. Falls back to the engine's built-in metric
     definitions when no metricDict CSV is supplied."""
@@ -341,7 +341,7 @@ def _split_charts_by_kind(master_path: str, prefix: str):
     return num, cat
 
 
-def descriptive_statistics(master_path: str) -> str:
+def descriptive_statistics(master_path: str, print_report=False, **kwargs) -> str:
     """Reference report_generation.py:994-1153 — global summary
     narrative, per-metric stat tables, numerical/categorical chart
     grids."""
@@ -365,7 +365,7 @@ def descriptive_statistics(master_path: str) -> str:
 _QC_ROW_LEVEL = ["duplicate_detection", "nullRows_detection"]
 
 
-def quality_check(master_path: str) -> str:
+def quality_check(master_path: str, print_report=False, **kwargs) -> str:
     """Reference report_generation.py:1154-1290 — row-level and
     column-level sub-sections + outlier violin charts."""
     row_parts, col_parts = [], []
@@ -383,7 +383,7 @@ def quality_check(master_path: str) -> str:
     return "".join(parts) if parts else "<p class='note'>No quality-check statistics saved.</p>"
 
 
-def attribute_associations(master_path: str, label_col, event_label, corr_threshold=0.4, iv_threshold=0.02) -> str:
+def attribute_associations(master_path: str, label_col, event_label, corr_threshold=0.4, iv_threshold=0.02, print_report=False, **kwargs) -> str:
     """Reference report_generation.py:1291 — correlation heatmap, IV/IG
     bars, variable clustering table + event-rate charts."""
     parts = []
@@ -422,7 +422,7 @@ def attribute_associations(master_path: str, label_col, event_label, corr_thresh
     return "".join(parts) if parts else "<p class='note'>No association statistics saved.</p>"
 
 
-def data_drift_stability(master_path: str, drift_threshold_model=0.1) -> str:
+def data_drift_stability(master_path: str, drift_threshold_model=0.1, print_report=False, **kwargs) -> str:
     """Reference report_generation.py:1434."""
     parts = []
     dd = _read_csv(master_path, "drift_statistics")
@@ -453,7 +453,7 @@ def data_drift_stability(master_path: str, drift_threshold_model=0.1) -> str:
     return "".join(parts) if parts else "<p class='note'>No drift / stability statistics saved.</p>"
 
 
-def ts_viz_generate(master_path: str) -> str:
+def ts_viz_generate(master_path: str, print_report=False, **kwargs) -> str:
     """Reference report_generation.py:3091 — time-series tab from
     ts_analyzer CSVs (stats_<col>_{1,2}.csv + <ts>_<attr>_<type>.csv)."""
     files = [x for x in os.listdir(master_path) if x.startswith("stats_") and x.endswith(".csv")]
@@ -495,7 +495,7 @@ def ts_viz_generate(master_path: str) -> str:
     return "".join(parts)
 
 
-def loc_report_gen(master_path: str) -> str:
+def loc_report_gen(master_path: str, print_report=False, **kwargs) -> str:
     """Reference report_generation.py:3902 — geospatial tab from
     geospatial_analyzer outputs."""
     files = os.listdir(master_path)
