@@ -95,15 +95,20 @@ def union_stats(frames: List[pd.DataFrame]) -> pd.DataFrame:
     return pd.concat(frames, ignore_index=True)
 
 
-def flatten_dataframe(pdf: pd.DataFrame, list_of_cols: List[str]) -> pd.DataFrame:
-    """Melt/unpivot (reference shared/utils.py:6-25): wide -> (attribute,
-    key, value) long format over the given value columns."""
-    id_cols = [c for c in pdf.columns if c not in list_of_cols]
-    return pdf.melt(id_vars=id_cols, value_vars=list_of_cols, var_name="key", value_name="value")
+def flatten_dataframe(idf, fixed_cols: List[str]) -> pd.DataFrame:
+    """Melt/unpivot (reference shared/utils.py:6-25): every column NOT in
+    fixed_cols becomes (key, value) long-format rows. Accepts a pandas
+    frame or an AnovosFrame (stats tables are driver-side smalls)."""
+    pdf = idf.to_pandas() if hasattr(idf, "to_pandas") else idf
+    value_cols = [c for c in pdf.columns if c not in fixed_cols]
+    return pdf.melt(id_vars=list(fixed_cols), value_vars=value_cols,
+                    var_name="key", value_name="value")
 
 
-def transpose_dataframe(pdf: pd.DataFrame, fixed_col: str) -> pd.DataFrame:
-    """Transpose a stats frame about fixed_col (reference shared/utils.py:28-45)."""
+def transpose_dataframe(idf, fixed_col: str) -> pd.DataFrame:
+    """Transpose a stats frame about fixed_col (reference
+    shared/utils.py:28-45: melt then pivot by fixed_col)."""
+    pdf = idf.to_pandas() if hasattr(idf, "to_pandas") else idf
     out = pdf.set_index(fixed_col).T.reset_index().rename(columns={"index": fixed_col})
     out.columns.name = None
     return out
