@@ -194,7 +194,8 @@ def _diagnosis_matrix(master_path: str, corr_threshold, iv_threshold) -> Optiona
 
 
 def executive_summary_gen(master_path: str, label_col, event_label,
-                          corr_threshold=0.4, iv_threshold=0.02, drift_threshold_model=0.1) -> str:
+                          corr_threshold=0.4, iv_threshold=0.02, drift_threshold_model=0.1,
+                          ds_ind=None, id_col=None, print_report=False, **kwargs) -> str:
     """Reference report_generation.py:524-908 — narrative summary, label
     distribution pie, the data-diagnosis matrix, and the drift/stability
     health BigNumbers."""
