@@ -219,3 +219,30 @@ def test_workflow_inmemory_pipeline_matches_materialized(income_csv, tmp_path, m
             import torch as _t
 
             assert float(_t.nan_to_num(a.data.double() - b.data.double()).abs().max()) < 1e-6, c
+
+
+def test_bench_json_contract():
+    """The driver's bench contract: one JSON line on stdout with the
+    required keys, value == whole-job aggregate, ms_per_step > 0."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run([sys.executable, os.path.join(repo, "bench.py"),
+                        "--steps", "2", "--warmup", "1", "--rows", "200000"],
+                       capture_output=True, text=True, timeout=600, cwd=repo)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    j = json.loads(line)
+    for k in ("metric", "value", "unit", "n_gpus", "steps", "warmup", "ms_per_step",
+              "higher_is_better", "scaling", "vs_baseline", "dtype", "data", "config"):
+        assert k in j, k
+    assert j["n_gpus"] == 1 and j["steps"] == 2 and j["warmup"] == 1
+    assert j["higher_is_better"] is True and j["scaling"] == "weak"
+    assert j["data"] == "synthetic"
+    assert j["value"] > 0 and j["ms_per_step"] > 0
+    # whole-job aggregate: rows_per_gpu * n_gpus / (ms_per_step/1000)
+    expect = j["config"]["rows_per_gpu"] * j["n_gpus"] / (j["ms_per_step"] / 1000.0)
+    assert j["value"] == pytest.approx(expect, rel=1e-6)
