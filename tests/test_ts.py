@@ -114,7 +114,12 @@ def test_ts_analyzer_outputs(ctx, tmp_path):
     assert os.path.exists(tmp_path / "stats_ts_1.csv")
     assert os.path.exists(tmp_path / "stats_ts_2.csv")
     f1 = pd.read_csv(tmp_path / "stats_ts_1.csv")
-    assert f1["mean"][0] == pytest.approx(1.0, abs=0.3)  # near-daily coverage
+    # reference opt=1 contract: pair-count percentile rows
+    assert list(f1["attribute"]) == ["id_date_pair", "date_id_pair"]
+    f2 = pd.read_csv(tmp_path / "stats_ts_2.csv")
+    assert f2["mean"][0] == pytest.approx(1.0, abs=0.3)  # near-daily coverage
+    assert {"count_unique_dates", "min_date", "max_date", "modal_date",
+            "date_diff", "missing_date", "cov"} <= set(f2.columns)
     viz = pd.read_csv(tmp_path / "ts_amount_daily.csv")
     assert {"min", "max", "mean", "median"} <= set(viz.columns)
     catviz = pd.read_csv(tmp_path / "ts_cat_daily.csv")
@@ -138,3 +143,40 @@ def test_ts_viz_weekly_and_hourly(ctx):
     assert len(wk) <= 7
     hr = tsa.ts_viz_data(feats, "ts", "v", output_type="hourly")
     assert set(hr["daypart_cat"]) <= set(tsa.DAYPARTS)
+
+
+def test_ts_eligibility_reference_productivity(ctx, tmp_path):
+    """The reference's own ts_analyzer unit expectations on its shipped
+    productivity dataset (48 states × 17 yearly snapshots —
+    test_ts_analyzer.py:62-104): pair percentiles 17/48, dates
+    1970→1986, date_diff 5844, lag stats 365.25/0.2/0.447."""
+    import datetime as _dt
+    import os as _os
+
+    from anovos_amd.data_analyzer.ts_analyzer import ts_eligiblity_check, ts_processed_feats
+    from anovos_amd.data_ingest.ts_auto_detection import ts_preprocess
+
+    data = _os.path.join(_os.path.dirname(_os.path.abspath(__file__)), "data", "productivity.csv")
+    pdf = pd.read_csv(data)
+    idf = AnovosFrame.from_pandas(pdf, device="cpu")
+    odf, ts_cols, _, _ = ts_preprocess(ctx, idf, "STATE", str(tmp_path))
+    assert ts_cols == ["YR"]
+    feats = ts_processed_feats(odf, "YR", "STATE")
+    # reference test_ts_processed_feats:49-58 (row 0 = 1970-01-01)
+    fp = feats.to_pandas().sort_values("YR").iloc[0]
+    assert int(fp["YR_hour"]) == 0 and int(fp["YR_minute"]) == 0 and int(fp["YR_second"]) == 0
+    assert int(fp["YR_dayofmonth"]) == 1 and int(fp["YR_month"]) == 1 and int(fp["YR_year"]) == 1970
+    assert int(fp["YR_quarter"]) == 1 and int(fp["YR_dayofyear"]) == 1
+
+    o1 = ts_eligiblity_check(ctx, feats, id_col="STATE", opt=1).set_index("attribute")
+    assert o1.loc["id_date_pair", "min"] == 17 and o1.loc["id_date_pair", "max"] == 17
+    assert o1.loc["date_id_pair", "min"] == 48 and o1.loc["date_id_pair", "max"] == 48
+
+    o2 = ts_eligiblity_check(ctx, feats, id_col="STATE", opt=2).iloc[0]
+    assert int(o2["count_unique_dates"]) == 17
+    assert o2["min_date"] == _dt.date(1970, 1, 1)
+    assert o2["max_date"] == _dt.date(1986, 1, 1)
+    assert int(o2["date_diff"]) == 5844
+    assert o2["mean"] == pytest.approx(365.25, abs=1e-9)
+    assert o2["variance"] == pytest.approx(0.2, abs=1e-3)
+    assert o2["stdev"] == pytest.approx(0.447, abs=1e-3)
