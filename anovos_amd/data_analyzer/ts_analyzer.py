@@ -253,8 +253,13 @@ def ts_viz_data(idf: AnovosFrame, x_col: str, y_col: str, id_col: str = "",
     o = byval[bygrp]
     gs, xs = kv[o].contiguous(), xv[o]
     starts = torch.searchsorted(gs, torch.arange(G, device=dev))
-    mid = (starts + ((cnt.to(torch.long) - 1) // 2).clamp(min=0)).clamp(max=max(int(xs.shape[0]) - 1, 0))
-    med = xs[mid] if xs.numel() else torch.zeros(G, dtype=torch.float64)
+    # interpolated median (pandas convention, like the reference): mean
+    # of the two middle elements for even group sizes
+    cl = cnt.to(torch.long)
+    cap = max(int(xs.shape[0]) - 1, 0)
+    lo = (starts + ((cl - 1) // 2).clamp(min=0)).clamp(max=cap)
+    hi = (starts + (cl // 2).clamp(min=0)).clamp(max=cap)
+    med = (xs[lo] + xs[hi]) / 2 if xs.numel() else torch.zeros(G, dtype=torch.float64)
     pdf = pd.DataFrame(
         {
             k_col: klabels,
