@@ -138,10 +138,13 @@ def get_column_name(df: pd.DataFrame) -> Tuple[str, str, str, str]:
 
 
 def camel_case_split(input) -> str:  # reference arg name (featrec_init.py)
-    """Reference featrec_init.py:camel_case_split — CamelCase → spaced."""
-    s = re.sub(r"([a-z0-9])([A-Z])", r"\1 \2", str(input))
-    s = re.sub(r"([A-Z]+)([A-Z][a-z])", r"\1 \2", s)
-    return s
+    """Reference featrec_init.py:114-130 — split camelCase boundaries,
+    each segment emitted with a trailing space (exact upstream output:
+    'accountWeeks' → 'account Weeks ')."""
+    out = ""
+    for m in re.finditer(r".+?(?:(?<=[a-z])(?=[A-Z])|(?<=[A-Z])(?=[A-Z][a-z])|$)", str(input)):
+        out += str(m.group(0)) + " "
+    return out
 
 
 def _clean_text(s: str) -> str:

@@ -184,3 +184,13 @@ def test_feature_retrieval_demo(tmp_path, monkeypatch):
     assert float(df["income"].iloc[0]) == 0.0
     assert float(df["income"].iloc[1]) == 3.0
     assert pd.isna(df["income"].iloc[2])
+
+
+def test_camel_case_split_reference_outputs():
+    """Reference test_featrec_init.py:22-35 exact outputs (trailing
+    space per segment)."""
+    from anovos_amd.feature_recommender.featrec_init import camel_case_split
+
+    assert camel_case_split("accountWeeks") == "account Weeks "
+    assert camel_case_split("account Weeks") == "account Weeks "
+    assert camel_case_split("AccountWeeksLock") == "Account Weeks Lock "
