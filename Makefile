@@ -1,5 +1,5 @@
 # anovos_amd build/test entry points (reference Makefile parity).
-.PHONY: build test test-gpu demo clean
+.PHONY: build test test-gpu demo bench clean
 
 build:
 	python -c "from anovos_amd.ops.hip.build import build; build(verbose=True)"
