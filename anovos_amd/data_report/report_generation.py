@@ -320,8 +320,13 @@ def wiki_generator(master_path: str, dataDict_path=None, metricDict_path=None, p
         except Exception:
             pass
     else:
-        parts.append("<h3>Metric Dictionary</h3>"
-                     + _tbl(pd.DataFrame(_METRIC_DICT, columns=["metric", "definition"]), 100))
+        # packaged default: the reference's full 89-row metric table
+        builtin = os.path.join(os.path.dirname(os.path.abspath(__file__)), "data", "metric_dictionary.csv")
+        try:
+            parts.append("<h3>Metric Dictionary</h3>" + _tbl(pd.read_csv(builtin), 500))
+        except Exception:
+            parts.append("<h3>Metric Dictionary</h3>"
+                         + _tbl(pd.DataFrame(_METRIC_DICT, columns=["metric", "definition"]), 100))
     return "".join(parts)
 
 

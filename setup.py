@@ -12,6 +12,7 @@ setup(
     package_data={
         "anovos_amd.ops.hip": ["*.hip", "*.so", "build.py"],
         "anovos_amd.feature_recommender": ["data/*.csv", "data/README.md"],
+        "anovos_amd.data_report": ["data/*.csv", "data/README.md"],
     },
     python_requires=">=3.10",
     install_requires=["numpy", "pandas", "pyarrow", "plotly", "scipy", "scikit-learn", "pyyaml", "sympy"],

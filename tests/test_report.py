@@ -131,3 +131,14 @@ def test_wiki_and_exec_content(ctx, frame, tmp_path):
     assert "Metric Dictionary" in html and "PSI" in html
     exec_html = rg.executive_summary_gen(mp, "label", "1")
     assert "Label Distribution" in exec_html
+
+
+def test_wiki_metric_dictionary_default(tmp_path):
+    """Without a metricDict_path, the Wiki tab renders the packaged
+    reference metric table (89 rows of section/metric definitions)."""
+    from anovos_amd.data_report.report_generation import wiki_generator
+
+    html = wiki_generator(str(tmp_path))
+    assert "Metric Dictionary" in html
+    assert "fill_count" in html  # a definition row from the packaged table
+    assert "Information Value" in html or "Measures Of" in html
