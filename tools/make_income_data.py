@@ -101,6 +101,35 @@ if __name__ == "__main__":
     os.makedirs(os.path.join(a.out, "parquet"), exist_ok=True)
     df.to_csv(os.path.join(a.out, "csv", "part-00000.csv"), index=False)
     df.to_parquet(os.path.join(a.out, "parquet", "part-00000.parquet"))
+    # data dictionary for the report Wiki tab (reference ships one per
+    # example dataset: examples/data/income_dataset/data_dictionary.csv)
+    definitions = {
+        "ifa": "unique row identifier.",
+        "age": "continuous.",
+        "workclass": "employment class (Private, Self-emp, Gov, ...).",
+        "fnlwgt": "continuous sampling weight.",
+        "logfnl": "log of fnlwgt.",
+        "education": "highest education level attained.",
+        "education-num": "continuous education rank.",
+        "marital-status": "marital status category.",
+        "occupation": "occupation category.",
+        "relationship": "household relationship category.",
+        "race": "race category.",
+        "sex": "sex category.",
+        "capital-gain": "continuous.",
+        "capital-loss": "continuous.",
+        "hours-per-week": "continuous working hours.",
+        "native-country": "country of origin.",
+        "income": "label: <=50K or >50K.",
+        "latitude": "location latitude (decimal degrees).",
+        "longitude": "location longitude (decimal degrees).",
+        "gh7": "precision-7 geohash of (latitude, longitude).",
+        "txn_date": "transaction timestamp string.",
+        "signup_epoch": "signup time, 10-digit epoch seconds.",
+    }
+    pd.DataFrame({"attribute": [c for c in df.columns],
+                  "definition": [definitions.get(c, "") for c in df.columns]}
+                 ).to_csv(os.path.join(a.out, "data_dictionary.csv"), index=False)
     # drift source: mildly shifted snapshot
     src = make(a.rows, seed=12)
     src["age"] = src["age"] * 1.05
