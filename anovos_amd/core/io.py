@@ -247,7 +247,8 @@ def _to_frame(table: pa.Table, device) -> AnovosFrame:
                 vals = arr.to_numpy(zero_copy_only=False)
                 out = vals.astype(np.float64 if t == pa.float64() else np.float32, copy=False)
                 dt = "double" if out.dtype == np.float64 else "float"
-                cols[name] = Column(name, dt, torch.from_numpy(np.ascontiguousarray(out)).to(dev))
+                cols[name] = Column(name, dt, torch.from_numpy(
+                    np.require(out, requirements=["C", "W"])).to(dev))
                 continue
             if pa.types.is_boolean(t):
                 vals = arr.to_numpy(zero_copy_only=False).astype(np.float32)
@@ -288,7 +289,8 @@ def _sorted_dict_column(name, codes_i64, dictionary, null_mask, dev):
     safe = np.maximum(codes_i64, 0)
     out = np.where((codes_i64 >= 0) & ~null_mask, remap[np.minimum(safe, max(len(cats) - 1, 0))], NULL_CODE).astype(np.int32)
     sorted_dict = [str(c) for c in cats[order]]
-    return Column(name, "string", torch.from_numpy(np.ascontiguousarray(out)).to(dev), sorted_dict)
+    return Column(name, "string", torch.from_numpy(
+        np.require(out, requirements=["C", "W"])).to(dev), sorted_dict)
 
 
 def _empty_like_first(parts_all, read_one):
